@@ -1,0 +1,774 @@
+"""RAMP discrete-event cluster simulator.
+
+Reference: ``ddls/environments/ramp_cluster/ramp_cluster_environment.py:74``.
+
+Because RAMP rules forbid sharing workers/channels between jobs, a job's JCT is
+computed once at mount time by a lookahead simulation of one training step
+(``:84-93``); the outer event loop then just advances wall-clock to completion/
+arrival times.  The lookahead inner loop here is array-vectorised over the
+job's ops/deps (numpy), in the same struct-of-arrays layout the planned HIP
+batched-env kernels consume.
+"""
+from __future__ import annotations
+
+import copy
+import math
+from collections import defaultdict
+from typing import Dict, Optional, Union
+
+import numpy as np
+
+from ..devices import Processor
+from ..jobs import Job, JobQueue, JobsGenerator
+from ..topology import Ramp, build_topology
+from ..utils import Stopwatch, get_class_from_path, seed_everything
+from .actions import Action
+from .rules import (check_if_ramp_dep_placement_rules_broken,
+                    check_if_ramp_op_placement_rules_broken)
+
+
+class RampClusterEnvironment:
+    def __init__(self,
+                 topology_config: dict,
+                 node_config: dict,
+                 name: str = "ramp_cluster",
+                 path_to_save: Optional[str] = None,
+                 save_freq: int = 1,
+                 use_sqlite_database: bool = False,
+                 suppress_warnings: bool = True,
+                 machine_epsilon: float = 1e-7):
+        self.topology_config = topology_config
+        self.node_config = node_config
+        self.name = name
+        self.path_to_save = path_to_save
+        self.save_freq = save_freq
+        self.use_sqlite_database = use_sqlite_database
+        self.suppress_warnings = suppress_warnings
+        self.machine_epsilon = machine_epsilon
+
+        self.topology = build_topology(topology_config)
+        self._populate_topology(self.topology, node_config)
+
+        self.stopwatch = Stopwatch()
+        self.reset_counter = 0
+        self._logger = None
+
+    # ------------------------------------------------------------------
+    def _populate_topology(self, topology, node_config):
+        num_cfg_nodes = sum(node_config[t]["num_nodes"] for t in node_config)
+        if num_cfg_nodes != topology.num_nodes:
+            raise ValueError(
+                f"topology has {topology.num_nodes} nodes but node_config "
+                f"specifies {num_cfg_nodes}")
+        node_iter = iter(range(topology.num_nodes))
+        self.workers = []           # dense worker list
+        self.worker_id_to_index = {}
+        for node_type in node_config:
+            for _ in range(node_config[node_type]["num_nodes"]):
+                node = next(node_iter)
+                for wc in node_config[node_type]["workers_config"]:
+                    if wc["num_workers"] > 1:
+                        raise ValueError(
+                            "RAMP supports 1 worker per server; set num_workers=1")
+                    for i in range(wc["num_workers"]):
+                        Worker = wc["worker"]
+                        if isinstance(Worker, str):
+                            Worker = get_class_from_path(Worker)
+                        w: Processor = Worker(
+                            processor_id=f"node_{topology.node_names[node]}_worker_{i}")
+                        topology.node_workers[node][w.processor_id] = w
+                        topology.worker_to_node[w.processor_id] = node
+                        topology.worker_to_type[w.processor_id] = w.device_type
+                        topology.worker_types.add(w.device_type)
+                        topology.num_workers += 1
+                        self.worker_id_to_index[w.processor_id] = len(self.workers)
+                        self.workers.append(w)
+        # dense channel index
+        self.channel_ids = list(topology.channel_id_to_channel.keys())
+        self.channel_id_to_index = {cid: i for i, cid in enumerate(self.channel_ids)}
+
+    @property
+    def device_type(self) -> str:
+        # reference assumes a homogeneous cluster (ramp_cluster_environment.py:809)
+        return next(iter(self.topology.worker_types))
+
+    # ------------------------------------------------------------------
+    def reset(self,
+              jobs_config: dict,
+              max_simulation_run_time: Union[int, float] = float("inf"),
+              job_queue_capacity: int = 10,
+              seed: Optional[int] = None,
+              verbose: bool = False):
+        self.reset_counter += 1
+        self.seed = seed
+        if seed is not None:
+            seed_everything(seed)
+
+        self.stopwatch.reset()
+        self.jobs_generator = JobsGenerator(**jobs_config)
+        self.max_simulation_run_time = (max_simulation_run_time
+                                        if max_simulation_run_time is not None
+                                        else float("inf"))
+
+        self.steps_log = defaultdict(list)
+        self.sim_log = defaultdict(list)
+        self.episode_stats = self._init_episode_stats()
+
+        self.topology.reset_devices()
+        for w in self.workers:
+            w.reset()
+
+        self.job_queue = JobQueue(queue_capacity=job_queue_capacity)
+
+        self.num_jobs_arrived = 0
+        self.num_mounted_ops = 0
+        self.num_mounted_deps = 0
+        self.load_rates = []
+        self.mounted_workers = set()
+        self.mounted_channels = set()
+        self.jobs_running: Dict[int, Job] = {}
+        self.jobs_completed: Dict[int, Job] = {}
+        self.jobs_blocked: Dict[int, Job] = {}
+        self.job_op_to_worker = {}
+        self.job_dep_to_channels = defaultdict(set)
+        self.job_idx_to_job_id = {}
+        self.job_id_to_job_idx = {}
+        self.step_counter = 0
+        self.action = None
+        self.job_op_placement = {}
+        self.job_dep_placement = {}
+
+        # memo hash tables (reference :269-277; GPU analogue: device-side open
+        # addressing keyed (model_id, partition_degree) — SURVEY.md K4)
+        self.job_model_to_max_num_partitions_to_init_details = defaultdict(dict)
+        self.job_model_to_max_num_partitions_to_lookahead = defaultdict(dict)
+
+        self.time_next_job_to_arrive = 0
+        self.job_queue.add(self._get_next_job())
+        return None
+
+    def _init_episode_stats(self):
+        stats = defaultdict(list)
+        stats["num_jobs_arrived"] = 0
+        stats["num_jobs_completed"] = 0
+        stats["num_jobs_blocked"] = 0
+        stats["episode_start_time"] = copy.copy(self.stopwatch.time())
+        return stats
+
+    def _init_step_stats(self):
+        stats = defaultdict(lambda: 0)
+        stats["step_counter"] = self.step_counter
+        stats["step_start_time"] = self.stopwatch.time()
+        for k in ("mean_num_mounted_workers", "mean_num_mounted_channels",
+                  "mean_compute_overhead_frac", "mean_communication_overhead_frac",
+                  "mean_mounted_worker_utilisation_frac",
+                  "mean_cluster_worker_utilisation_frac", "mean_num_jobs_running"):
+            stats[k] = []
+        for k in ("mean_compute_throughput", "mean_dep_throughput",
+                  "mean_cluster_throughput", "mean_demand_compute_throughput",
+                  "mean_demand_dep_throughput", "mean_demand_total_throughput",
+                  "num_jobs_completed", "num_jobs_arrived", "num_jobs_blocked"):
+            stats[k] = 0
+        return stats
+
+    # ------------------------------------------------------------------
+    def _get_next_job(self) -> Job:
+        job = self.jobs_generator.sample_job()
+        job_idx = self.num_jobs_arrived
+        job.original_job.job_id = job.job_id
+        job.original_job.details["job_idx"] = job_idx
+        job.register_job_arrived(time_arrived=self.stopwatch.time(), job_idx=job_idx)
+        self.time_last_job_arrived = self.stopwatch.time()
+        self.time_next_job_to_arrive += self.jobs_generator.sample_interarrival_time()
+        self.load_rates.append(
+            (job.original_job.details["job_total_op_memory_cost"]
+             + job.original_job.details["job_total_dep_size"])
+            / (self.time_next_job_to_arrive - self.time_last_job_arrived))
+        if job_idx in self.job_idx_to_job_id:
+            raise RuntimeError(f"duplicate job idx {job_idx}")
+        if job.job_id in self.job_id_to_job_idx:
+            raise RuntimeError(f"duplicate job id {job.job_id}")
+        self.job_idx_to_job_id[job_idx] = job.job_id
+        self.job_id_to_job_idx[job.job_id] = job_idx
+        self.num_jobs_arrived += 1
+        self.last_job_arrived_job_idx = job_idx
+        self.episode_stats["num_jobs_arrived"] += 1
+        return job
+
+    # ------------------------------------------------------------------
+    # lookahead simulation of one training step (reference :379-467)
+    # ------------------------------------------------------------------
+    def _run_lookahead(self, job_id, verbose: bool = False):
+        job_idx = self.job_id_to_job_idx[job_id]
+        job = self.jobs_running[job_idx]
+        g = job.graph
+
+        op_worker = job.op_worker            # dense worker index per op
+        op_priority = job.op_priority
+        dep_is_flow = job.dep_is_flow
+        dep_channel = job.dep_channel_idx
+        dep_priority = job.dep_priority
+
+        t = 0.0
+        tick_counter = 1
+        tick_map = {}
+        while True:
+            # 1. highest-priority ready op per worker; shortest remaining time
+            ready_ops = np.flatnonzero(job.ops_ready)
+            if len(ready_ops) > 0:
+                w = op_worker[ready_ops]
+                p = op_priority[ready_ops]
+                order = np.lexsort((p, w))
+                rs, ws = ready_ops[order], w[order]
+                last_of_group = np.flatnonzero(
+                    np.r_[ws[1:] != ws[:-1], True])
+                priority_ops = rs[last_of_group]
+                shortest_op = float(job.op_remaining[priority_ops].min())
+            else:
+                priority_ops = np.empty(0, dtype=np.int64)
+                shortest_op = float("inf")
+
+            # 2. ready deps; non-flow deps tick with 0 comm time
+            ready_deps = np.flatnonzero(job.deps_ready)
+            non_flow = ready_deps[~dep_is_flow[ready_deps]]
+            if len(non_flow) == 0:
+                ready_flows = ready_deps
+                if len(ready_flows) > 0:
+                    # per-channel highest-priority flow (contention resolution
+                    # is vacuous on the full mesh: one channel per dep)
+                    ch = dep_channel[ready_flows]
+                    pr = dep_priority[ready_flows]
+                    order = np.lexsort((pr, ch))
+                    fs, cs = ready_flows[order], ch[order]
+                    last = np.flatnonzero(np.r_[cs[1:] != cs[:-1], True])
+                    prio_flows = fs[last]
+                    shortest_comm = float(job.dep_remaining[prio_flows].min())
+                else:
+                    shortest_comm = float("inf")
+            else:
+                shortest_comm = 0.0
+
+            tick = min(shortest_op, shortest_comm)
+            if math.isinf(tick):
+                raise RuntimeError("infinite lookahead tick: deadlocked job graph")
+
+            # snapshot ready deps BEFORE op ticking (reference :429)
+            deps_to_tick = non_flow if len(non_flow) > 0 else ready_deps
+
+            # 3a. tick priority ops
+            ticked_ops = len(priority_ops) > 0
+            completed_ops = []
+            if ticked_ops:
+                rem = job.op_remaining[priority_ops]
+                rem = rem - np.minimum(tick, rem)
+                job.op_remaining[priority_ops] = rem
+                completed_ops = priority_ops[rem == 0.0]
+                for o in completed_ops:
+                    job.ops_completed[o] = True
+                    job.ops_ready[o] = False
+                    job.num_ops_completed += 1
+                    job.deps_ready[g.out_edges_of(int(o))] = True
+
+            # 3b. tick deps (non-flows, or ALL ready flows in parallel —
+            # reference TEMP HACK :756-775)
+            ticked_flows = False
+            if len(deps_to_tick) > 0:
+                if len(non_flow) == 0:
+                    ticked_flows = True
+                rem = job.dep_remaining[deps_to_tick]
+                rem = rem - np.minimum(tick, rem)
+                job.dep_remaining[deps_to_tick] = rem
+                completed_deps = deps_to_tick[rem == 0.0]
+                for e in completed_deps:
+                    e = int(e)
+                    if job.deps_completed[e]:
+                        continue
+                    job.deps_completed[e] = True
+                    job.deps_ready[e] = False
+                    job.num_deps_completed += 1
+                    child = int(g.dst[e])
+                    job.parent_deps_completed[child] += 1
+                    if job.parent_deps_completed[child] == g.true_parent_count[child]:
+                        job.ops_ready[child] = True
+
+            # 4. overhead accounting (reference :777-791)
+            if ticked_ops:
+                job.details["computation_overhead_time"] += tick
+            if ticked_flows:
+                job.details["communication_overhead_time"] += tick
+
+            tick_map[tick_counter] = [int(len(priority_ops)), tick]
+            t += tick
+
+            if (job.num_ops_completed == g.n
+                    and job.num_deps_completed == g.m):
+                lookahead_jct = t * job.num_training_steps
+                comm_oh = job.details["communication_overhead_time"] * job.num_training_steps
+                comp_oh = job.details["computation_overhead_time"] * job.num_training_steps
+                return job, lookahead_jct, comm_oh, comp_oh, tick_map
+
+            tick_counter += 1
+
+    def _perform_lookahead_job_completion_time(self, action: Action,
+                                               verbose: bool = False):
+        for job_id in action.job_ids:
+            job_idx = self.job_id_to_job_idx[job_id]
+            job = self.jobs_running[job_idx]
+            degree = self.op_partition.job_id_to_max_partition_degree[job_id]
+            model = job.details["model"]
+            memo = self.job_model_to_max_num_partitions_to_lookahead[model]
+            entry = memo.get(degree)
+            if entry is None:
+                job, jct, comm_oh, comp_oh, tick_map = self._run_lookahead(job_id)
+                memo[degree] = entry = (jct, comm_oh, comp_oh, tick_map)
+            jct, comm_oh, comp_oh, tick_map = entry
+            self._register_completed_lookahead(job, jct, comp_oh, comm_oh, tick_map)
+
+    def _register_completed_lookahead(self, job, lookahead_jct,
+                                      computation_overhead_time,
+                                      communication_overhead_time,
+                                      tick_map, verbose: bool = False):
+        device_type = self.device_type
+        if lookahead_jct > job.details["max_acceptable_job_completion_time"][device_type]:
+            # blocked: exceeds the job's max-acceptable JCT contract
+            self._register_blocked_job(job.original_job)
+            self._remove_job_from_cluster(job)
+            return
+
+        n_mounted = len(job.details["mounted_workers"])
+        util = 0.0
+        for num_active, tick_size in tick_map.values():
+            util += (num_active / n_mounted) * (tick_size / lookahead_jct)
+
+        model = job.details["model"]
+        degree = self.op_partition.job_id_to_max_partition_degree[job.job_id]
+        entry = self.job_model_to_max_num_partitions_to_init_details[model].get(degree)
+        immutable = entry["immutable"] if entry else None
+
+        job.reset_job(details={
+            "lookahead_job_completion_time": lookahead_jct,
+            "communication_overhead_time": communication_overhead_time,
+            "computation_overhead_time": computation_overhead_time,
+            "mounted_workers": job.details["mounted_workers"],
+            "mounted_channels": job.details["mounted_channels"],
+            "mean_mounted_worker_utilisation_frac": util,
+        }, immutable=immutable)
+
+        # flow-size accounting (reference :881-888)
+        job.details["job_total_flow_size"] = 0.0
+        g = job.graph
+        for e in range(g.m):
+            rt = self._set_dep_init_run_time(job, e)
+            if rt != 0:
+                job.details["job_total_flow_size"] += float(g.size[e])
+
+    # ------------------------------------------------------------------
+    def _set_dep_init_run_time(self, job: Job, dep_idx: int):
+        """Cluster-side dep init time (reference :542-560): 0 for same-node or
+        zero-size deps, else the priced init run time."""
+        g = job.graph
+        u, v = int(g.src[dep_idx]), int(g.dst[dep_idx])
+        job_idx = job.details["job_idx"]
+        src_w = self.job_op_to_worker[(job_idx, u)]
+        dst_w = self.job_op_to_worker[(job_idx, v)]
+        if self.topology.worker_to_node[src_w] == self.topology.worker_to_node[dst_w]:
+            rt = 0.0
+        elif g.size[dep_idx] == 0:
+            rt = 0.0
+        else:
+            rt = job.dep_init_run_time[dep_idx]
+        job.set_dep_init_run_time(dep_idx, rt)
+        return rt
+
+    # ------------------------------------------------------------------
+    # control-plane execution (reference :1285-1415)
+    # ------------------------------------------------------------------
+    def _partition_ops(self, op_partition):
+        self.op_partition = op_partition
+        for job_id in op_partition.action:
+            self.job_queue.jobs[job_id] = op_partition.partitioned_jobs[job_id]
+
+    def _place_ops(self, op_placement):
+        for job_id, placement in op_placement.action.items():
+            job = self.job_queue.jobs[job_id]
+            g = job.graph
+            job_idx = job.details["job_idx"]
+            # dense per-op arrays used by the lookahead
+            job.op_worker = np.full(g.n, -1, dtype=np.int64)
+            job.op_priority = np.zeros(g.n, dtype=np.int64)
+            for op_name, worker_id in placement.items():
+                op_idx = g.name_to_idx[op_name]
+                node = self.topology.worker_to_node[worker_id]
+                worker = self.topology.node_workers[node][worker_id]
+                broken = check_if_ramp_op_placement_rules_broken(worker, job)
+                if broken:
+                    raise RuntimeError(
+                        f"placement for job {job_id} op {op_name} on {worker_id} "
+                        f"breaks RAMP rules: {broken}")
+                worker.mount(job=job, op_idx=op_idx)
+                job.details["mounted_workers"].add(worker_id)
+                self.num_mounted_ops += 1
+                job.reset_op_remaining_run_time(op_idx, device_type=worker.device_type)
+                self.job_op_to_worker[(job_idx, op_idx)] = worker_id
+                job.op_worker[op_idx] = self.worker_id_to_index[worker_id]
+            self._register_running_job(job)
+            self.job_op_placement[job_id] = placement
+
+    def _register_running_job(self, job: Job):
+        job.register_job_running(time_started=self.stopwatch.time())
+        self.jobs_running[job.details["job_idx"]] = job
+        self.job_queue.remove(job)
+        g = job.graph
+        # flow classification + init run times
+        job.dep_is_flow = np.zeros(g.m, dtype=bool)
+        job.dep_channel_idx = np.full(g.m, -1, dtype=np.int64)
+        job.dep_priority = np.zeros(g.m, dtype=np.int64)
+        job_idx = job.details["job_idx"]
+        for e in range(g.m):
+            rt = self._set_dep_init_run_time(job, e)
+            u, v = int(g.src[e]), int(g.dst[e])
+            src_node = self.topology.worker_to_node[self.job_op_to_worker[(job_idx, u)]]
+            dst_node = self.topology.worker_to_node[self.job_op_to_worker[(job_idx, v)]]
+            job.dep_is_flow[e] = (g.size[e] > 0) and (src_node != dst_node)
+
+    def _place_deps(self, dep_placement):
+        for job_id, deps in dep_placement.action.items():
+            job_idx = self.job_id_to_job_idx[job_id]
+            job = self.jobs_running[job_idx]
+            for dep_idx, channel_ids in deps.items():
+                for channel_id in channel_ids:
+                    if channel_id is None:
+                        continue
+                    channel = self.topology.channel_id_to_channel[channel_id]
+                    broken = check_if_ramp_dep_placement_rules_broken(channel, job)
+                    if broken:
+                        raise RuntimeError(
+                            f"dep placement for job {job_id} dep {dep_idx} on "
+                            f"{channel_id} breaks RAMP rules: {broken}")
+                    channel.mount(job, dep_idx)
+                    job.details["mounted_channels"].add(channel_id)
+                    self.num_mounted_deps += 1
+                    job.reset_dep_remaining_run_time(dep_idx)
+                    self.job_dep_to_channels[(job_idx, dep_idx)].add(channel_id)
+                    job.dep_channel_idx[dep_idx] = self.channel_id_to_index[channel_id]
+            self.job_dep_placement[job_id] = deps
+
+    def _schedule_ops(self, op_schedule):
+        for worker_id, job_to_ops in op_schedule.action.items():
+            node = self.topology.worker_to_node[worker_id]
+            worker = self.topology.node_workers[node][worker_id]
+            for job_id, op_to_priority in job_to_ops.items():
+                job_idx = self.job_id_to_job_idx[job_id]
+                job = (self.jobs_running.get(job_idx)
+                       or self.job_queue.jobs.get(job_id))
+                for op_name, priority in op_to_priority.items():
+                    op_idx = job.graph.name_to_idx[op_name]
+                    worker.mounted_job_op_to_priority[(job_idx, op_idx)] = priority
+                    job.op_priority[op_idx] = priority
+
+    def _schedule_deps(self, dep_schedule):
+        for channel_id, job_to_deps in dep_schedule.action.items():
+            if channel_id is None:
+                continue
+            channel = self.topology.channel_id_to_channel[channel_id]
+            for job_id, dep_to_priority in job_to_deps.items():
+                job_idx = self.job_id_to_job_idx[job_id]
+                job = self.jobs_running.get(job_idx)
+                if job is None:
+                    continue
+                for dep_idx, priority in dep_to_priority.items():
+                    channel.mounted_job_dep_to_priority[(job_idx, dep_idx)] = priority
+                    job.dep_priority[dep_idx] = priority
+
+    # ------------------------------------------------------------------
+    def _remove_job_from_cluster(self, job: Job):
+        if job.job_id in self.job_queue.jobs:
+            self.job_queue.remove(job)
+        self.jobs_running.pop(job.details["job_idx"], None)
+        job_idx = job.details["job_idx"]
+        g = job.graph
+        for op_idx in range(g.n):
+            key = (job_idx, op_idx)
+            if key in self.job_op_to_worker:
+                worker_id = self.job_op_to_worker[key]
+                node = self.topology.worker_to_node[worker_id]
+                self.topology.node_workers[node][worker_id].unmount(job, op_idx)
+                self.num_mounted_ops -= 1
+                del self.job_op_to_worker[key]
+        for dep_idx in range(g.m):
+            key = (job_idx, dep_idx)
+            if key in self.job_dep_to_channels:
+                for channel_id in self.job_dep_to_channels[key]:
+                    self.topology.channel_id_to_channel[channel_id].unmount(job, dep_idx)
+                    self.num_mounted_deps -= 1
+                del self.job_dep_to_channels[key]
+        self.job_op_placement.pop(job.job_id, None)
+        self.job_dep_placement.pop(job.job_id, None)
+
+    def _register_completed_job(self, job: Job):
+        job.register_job_completed(time_completed=self.stopwatch.time())
+        self.jobs_completed[job.details["job_idx"]] = job
+        self.step_stats["num_jobs_completed"] += 1
+        self.episode_stats["num_jobs_completed"] += 1
+        dt = self.device_type
+        es = self.episode_stats
+        jct = job.details["time_completed"] - job.details["time_arrived"]
+        es["job_completion_time"].append(jct)
+        es["job_completion_time_speedup"].append(
+            job.details["job_sequential_completion_time"][dt] / jct)
+        es["job_communication_overhead_time"].append(
+            job.details["communication_overhead_time"])
+        es["job_computation_overhead_time"].append(
+            job.details["computation_overhead_time"])
+        es["jobs_completed_num_nodes"].append(job.graph.n)
+        es["jobs_completed_num_edges"].append(job.graph.m)
+        es["jobs_completed_total_operation_memory_cost"].append(
+            job.job_total_operation_memory_cost)
+        es["jobs_completed_total_dependency_size"].append(
+            job.job_total_dependency_size)
+        es["jobs_completed_max_partitions_per_op"].append(
+            job.details.get("max_partitions_per_op", 1))
+        es["jobs_completed_job_sequential_completion_time"].append(
+            job.details["job_sequential_completion_time"][dt])
+        es["jobs_completed_max_acceptable_job_completion_time_frac"].append(
+            job.max_acceptable_job_completion_time_frac)
+        es["jobs_completed_max_acceptable_job_completion_time"].append(
+            job.details["max_acceptable_job_completion_time"][dt])
+        es["jobs_completed_num_mounted_workers"].append(
+            len(job.details["mounted_workers"]))
+        es["jobs_completed_num_mounted_channels"].append(
+            len(job.details["mounted_channels"]))
+        es["jobs_completed_mean_mounted_worker_utilisation_frac"].append(
+            job.details.get("mean_mounted_worker_utilisation_frac", 0))
+        es["jobs_completed_original_demand_num_nodes"].append(
+            job.original_job.graph.n)
+        es["jobs_completed_original_demand_num_edges"].append(
+            job.original_job.graph.m)
+        es["jobs_completed_original_demand_total_operation_memory_cost"].append(
+            job.original_job.job_total_operation_memory_cost)
+        es["jobs_completed_original_demand_total_dependency_size"].append(
+            job.original_job.job_total_dependency_size)
+        self._remove_job_from_cluster(job)
+
+    def _register_blocked_job(self, job: Job):
+        if job.job_id in self.job_queue.jobs:
+            self.job_queue.remove(job)
+        self.jobs_running.pop(job.details["job_idx"], None)
+        if job.details["job_idx"] in self.jobs_blocked:
+            return
+        self.jobs_blocked[job.details["job_idx"]] = job
+        self.step_stats["num_jobs_blocked"] += 1
+        dt = self.device_type
+        es = self.episode_stats
+        es["num_jobs_blocked"] += 1
+        es["jobs_blocked_num_nodes"].append(job.graph.n)
+        es["jobs_blocked_num_edges"].append(job.graph.m)
+        es["jobs_blocked_total_operation_memory_cost"].append(
+            job.job_total_operation_memory_cost)
+        es["jobs_blocked_total_dependency_size"].append(job.job_total_dependency_size)
+        es["jobs_blocked_job_sequential_completion_time"].append(
+            job.details["job_sequential_completion_time"][dt])
+        es["jobs_blocked_max_acceptable_job_completion_time_frac"].append(
+            job.max_acceptable_job_completion_time_frac)
+        es["jobs_blocked_max_acceptable_job_completion_time"].append(
+            job.details["max_acceptable_job_completion_time"][dt])
+        es["jobs_blocked_original_demand_num_nodes"].append(job.original_job.graph.n)
+        es["jobs_blocked_original_demand_num_edges"].append(job.original_job.graph.m)
+        es["jobs_blocked_original_demand_total_operation_memory_cost"].append(
+            job.original_job.job_total_operation_memory_cost)
+        es["jobs_blocked_original_demand_total_dependency_size"].append(
+            job.original_job.job_total_dependency_size)
+
+    # ------------------------------------------------------------------
+    def step(self, action: Optional[Action] = None, verbose: bool = False):
+        if action is None:
+            action = Action()
+        self.action = action
+        self.step_stats = self._init_step_stats()
+
+        # queued jobs not handled by the action are blocked (reference :914-919)
+        for job_id, job in list(self.job_queue.jobs.items()):
+            if job_id not in action.job_ids:
+                self._register_blocked_job(job)
+
+        if action.actions["op_partition"] is not None:
+            self._partition_ops(action.actions["op_partition"])
+        if action.actions["op_placement"] is not None:
+            self._place_ops(action.actions["op_placement"])
+        if action.actions["op_schedule"] is not None:
+            self._schedule_ops(action.actions["op_schedule"])
+        if action.actions["dep_placement"] is not None:
+            self._place_deps(action.actions["dep_placement"])
+        if action.actions["dep_schedule"] is not None:
+            self._schedule_deps(action.actions["dep_schedule"])
+
+        self._perform_lookahead_job_completion_time(action)
+
+        # outer event loop: advance clock to next arrival/completion/sim end
+        step_done = False
+        while not step_done:
+            tick = min(self.time_next_job_to_arrive - self.stopwatch.time(),
+                       self.max_simulation_run_time - self.stopwatch.time())
+            for job in self.jobs_running.values():
+                elapsed = self.stopwatch.time() - job.details["time_started"]
+                remaining = job.details["lookahead_job_completion_time"] - elapsed
+                tick = min(tick, remaining)
+
+            self.mounted_workers, self.mounted_channels = set(), set()
+            mounted_worker_utilisation = []
+            ss = self.step_stats
+            for job in self.jobs_running.values():
+                frac = tick / job.details["lookahead_job_completion_time"]
+                ss["compute_info_processed"] += job.details["job_total_op_memory_cost"] * frac
+                ss["dep_info_processed"] += job.details["job_total_dep_size"] * frac
+                ss["flow_info_processed"] += job.details.get("job_total_flow_size", 0) * frac
+                ss["cluster_info_processed"] += (
+                    (job.details["job_total_op_memory_cost"]
+                     + job.details["job_total_dep_size"]) * frac)
+                ss["demand_compute_info_processed"] += (
+                    job.original_job.details["job_total_op_memory_cost"] * frac)
+                ss["demand_dep_info_processed"] += (
+                    job.original_job.details["job_total_dep_size"] * frac)
+                ss["demand_total_info_processed"] += (
+                    (job.original_job.details["job_total_op_memory_cost"]
+                     + job.original_job.details["job_total_dep_size"]) * frac)
+                ss["mean_compute_overhead_frac"].append(
+                    job.details["computation_overhead_time"]
+                    / job.details["lookahead_job_completion_time"])
+                ss["mean_communication_overhead_frac"].append(
+                    job.details["communication_overhead_time"]
+                    / job.details["lookahead_job_completion_time"])
+                self.mounted_workers.update(job.details["mounted_workers"])
+                self.mounted_channels.update(job.details["mounted_channels"])
+                mounted_worker_utilisation.append(
+                    job.details.get("mean_mounted_worker_utilisation_frac", 0))
+
+            ss["mean_num_jobs_running"].append(len(self.jobs_running))
+            ss["mean_num_mounted_workers"].append(len(self.mounted_workers))
+            ss["mean_num_mounted_channels"].append(len(self.mounted_channels))
+            if mounted_worker_utilisation:
+                mu = float(np.mean(mounted_worker_utilisation))
+                ss["mean_mounted_worker_utilisation_frac"].append(mu)
+                ss["mean_cluster_worker_utilisation_frac"].append(
+                    (len(self.mounted_workers) / self.topology.num_workers) * mu)
+            else:
+                ss["mean_mounted_worker_utilisation_frac"].append(0)
+                ss["mean_cluster_worker_utilisation_frac"].append(0)
+
+            self.stopwatch.tick(tick)
+
+            # completions
+            jobs_completed = []
+            for job in self.jobs_running.values():
+                elapsed = self.stopwatch.time() - job.details["time_started"]
+                remaining = (job.details["lookahead_job_completion_time"] - elapsed
+                             ) - self.machine_epsilon
+                if remaining <= 0:
+                    jobs_completed.append(job)
+                    step_done = True
+            for job in jobs_completed:
+                self._register_completed_job(job)
+
+            # arrivals
+            if len(self.jobs_generator) > 0:
+                if (self.stopwatch.time() + self.machine_epsilon) >= self.time_next_job_to_arrive:
+                    next_job = self._get_next_job()
+                    self.step_stats["num_jobs_arrived"] += 1
+                    if self.job_queue.can_fit(next_job):
+                        self.job_queue.add(next_job)
+                    else:
+                        self._register_blocked_job(next_job)
+                    step_done = True
+            else:
+                self.time_next_job_to_arrive = float("inf")
+
+            if self.is_done():
+                step_done = True
+
+        # ---- step-level stats ----
+        ss = self.step_stats
+        ss["step_end_time"] = self.stopwatch.time()
+        ss["step_time"] = ss["step_end_time"] - ss["step_start_time"]
+        for metric in ("mean_num_jobs_running", "mean_num_mounted_workers",
+                       "mean_num_mounted_channels", "mean_compute_overhead_frac",
+                       "mean_communication_overhead_frac",
+                       "mean_mounted_worker_utilisation_frac",
+                       "mean_cluster_worker_utilisation_frac"):
+            ss[metric] = float(np.mean(ss[metric])) if len(ss[metric]) > 0 else 0
+        for tp, info in (("mean_compute_throughput", "compute_info_processed"),
+                         ("mean_dep_throughput", "dep_info_processed"),
+                         ("mean_flow_throughput", "flow_info_processed"),
+                         ("mean_cluster_throughput", "cluster_info_processed"),
+                         ("mean_demand_compute_throughput", "demand_compute_info_processed"),
+                         ("mean_demand_dep_throughput", "demand_dep_info_processed"),
+                         ("mean_demand_total_throughput", "demand_total_info_processed")):
+            if ss[info] != 0 and ss["step_time"] != 0:
+                ss[tp] = ss[info] / ss["step_time"]
+            else:
+                ss[tp] = 0
+        ss["job_queue_length"] = len(self.job_queue)
+        for key, val in ss.items():
+            self.steps_log[key].append(val)
+
+        for metric in ("compute_info_processed", "dep_info_processed",
+                       "flow_info_processed", "cluster_info_processed",
+                       "demand_compute_info_processed", "demand_dep_info_processed",
+                       "demand_total_info_processed", "mean_compute_overhead_frac",
+                       "mean_communication_overhead_frac", "mean_num_jobs_running",
+                       "mean_num_mounted_workers",
+                       "mean_mounted_worker_utilisation_frac",
+                       "mean_cluster_worker_utilisation_frac"):
+            self.episode_stats[metric].append(ss[metric])
+
+        self.step_counter += 1
+
+        if self.is_done():
+            self._finalise_episode()
+
+        obs, action_set, reward, done, info = None, None, None, self.is_done(), None
+        return obs, action_set, reward, done, info
+
+    def _finalise_episode(self):
+        # block still-running jobs (reference :1111-1121)
+        for job in list(self.jobs_running.values()):
+            self._register_blocked_job(job.original_job)
+            self._remove_job_from_cluster(job)
+        es = self.episode_stats
+        es["episode_end_time"] = self.stopwatch.time()
+        es["episode_time"] = es["episode_end_time"] - es["episode_start_time"]
+        es["mean_load_rate"] = float(np.mean(self.load_rates)) if self.load_rates else 0
+        es["blocking_rate"] = (es["num_jobs_blocked"] / es["num_jobs_arrived"]
+                               if es["num_jobs_arrived"] else 0)
+        es["acceptance_rate"] = (es["num_jobs_completed"] / es["num_jobs_arrived"]
+                                 if es["num_jobs_arrived"] else 0)
+        for tp, info in (("mean_compute_throughput", "compute_info_processed"),
+                         ("mean_dep_throughput", "dep_info_processed"),
+                         ("mean_flow_throughput", "flow_info_processed"),
+                         ("mean_cluster_throughput", "cluster_info_processed"),
+                         ("mean_demand_compute_throughput", "demand_compute_info_processed"),
+                         ("mean_demand_dep_throughput", "demand_dep_info_processed"),
+                         ("mean_demand_total_throughput", "demand_total_info_processed")):
+            es[info] = float(np.sum(es[info]))
+            if es[info] != 0 and es["episode_time"] != 0:
+                es[tp] = es[info] / es["episode_time"]
+            else:
+                es[tp] = 0
+        for metric in ("mean_compute_overhead_frac", "mean_communication_overhead_frac",
+                       "mean_num_jobs_running", "mean_num_mounted_workers",
+                       "mean_mounted_worker_utilisation_frac",
+                       "mean_cluster_worker_utilisation_frac"):
+            if isinstance(es[metric], list):
+                es[metric] = float(np.mean(es[metric])) if len(es[metric]) > 0 else 0
+
+    def is_done(self, verbose: bool = False) -> bool:
+        if self.max_simulation_run_time is not None:
+            if self.stopwatch.time() >= self.max_simulation_run_time:
+                return True
+        if (len(self.jobs_generator) == 0 and len(self.jobs_running) == 0
+                and len(self.job_queue) == 0):
+            return True
+        return False
+
+    def __str__(self):
+        return (f"RampClusterEnvironment(topology={self.topology_config}, "
+                f"workers={self.topology.num_workers})")

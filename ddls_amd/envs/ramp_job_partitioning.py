@@ -1,0 +1,190 @@
+"""The PAC-ML gym environment: choose max partition degree per job.
+
+Reference: ``ddls/environments/ramp_job_partitioning/
+ramp_job_partitioning_environment.py:42``.  Discrete(max_partitions_per_op+1)
+actions, 0 = do not place.  Action a>0 expands to per-op partition degrees by
+the SiP-ML quantum rule, then the heuristic pipeline (first-fit block placer ->
+SRPT op scheduler -> first-fit dep placer -> SRPT dep scheduler) produces the
+full cluster Action.
+"""
+from __future__ import annotations
+
+from collections import defaultdict
+from typing import Optional, Union
+
+from ..agents.partitioners import sip_ml_num_partitions
+from ..agents.placers import FirstFitDepPlacer, RampFirstFitOpPlacer
+from ..agents.schedulers import SRPTDepScheduler, SRPTOpScheduler
+from ..cluster.actions import Action, OpPartition
+from ..cluster.environment import RampClusterEnvironment
+from ..graphs import FWD
+from . import spaces
+from .observation import RampJobPartitioningObservation
+from .rewards import REWARD_FUNCTIONS
+
+
+class RampJobPartitioningEnvironment:
+    def __init__(self,
+                 topology_config: dict,
+                 node_config: dict,
+                 jobs_config: dict,
+                 max_partitions_per_op: Optional[int] = None,
+                 min_op_run_time_quantum: Union[float, int] = 0.000006,
+                 op_placer: str = "ramp_first_fit_op_placer",
+                 op_scheduler: str = "srpt_op_scheduler",
+                 dep_placer: str = "first_fit_dep_placer",
+                 dep_scheduler: str = "srpt_dep_scheduler",
+                 observation_function: str = "ramp_job_partitioning_observation",
+                 pad_obs_kwargs: Optional[dict] = None,
+                 reward_function: str = "lookahead_job_completion_time",
+                 reward_function_kwargs: Optional[dict] = None,
+                 max_simulation_run_time: Union[int, float, None] = None,
+                 job_queue_capacity: int = 10,
+                 suppress_warnings: bool = True,
+                 name: str = "ramp_job_partitioning",
+                 path_to_save: Optional[str] = None,
+                 save_cluster_data: bool = False,
+                 save_freq: int = 1,
+                 use_sqlite_database: bool = False,
+                 apply_action_mask: bool = True):
+        self.topology_config = topology_config
+        self.node_config = node_config
+        self.jobs_config = jobs_config
+        self.apply_action_mask = apply_action_mask
+        self.max_simulation_run_time = (float("inf") if max_simulation_run_time is None
+                                        else max_simulation_run_time)
+        self.job_queue_capacity = job_queue_capacity
+        self.name = name
+        self.pad_obs_kwargs = pad_obs_kwargs
+
+        self.cluster = RampClusterEnvironment(
+            topology_config=topology_config,
+            node_config=node_config,
+            path_to_save=path_to_save if save_cluster_data else None,
+            save_freq=save_freq,
+            use_sqlite_database=use_sqlite_database,
+            suppress_warnings=suppress_warnings)
+
+        if max_partitions_per_op is None:
+            self.max_partitions_per_op = self.cluster.topology.num_workers
+        else:
+            self.max_partitions_per_op = max_partitions_per_op
+        self.min_op_run_time_quantum = min_op_run_time_quantum
+
+        if observation_function != "ramp_job_partitioning_observation":
+            raise ValueError(f"Unrecognised observation_function {observation_function}")
+        self.observation_function = RampJobPartitioningObservation(
+            self.max_partitions_per_op, pad_obs_kwargs=pad_obs_kwargs)
+
+        self.action_set = list(range(self.max_partitions_per_op + 1))
+        self.action_space = spaces.Discrete(len(self.action_set))
+        self.observation_space = spaces.Dict({})
+
+        if reward_function not in REWARD_FUNCTIONS:
+            raise ValueError(f"Unrecognised reward_function {reward_function}")
+        self.reward_function = REWARD_FUNCTIONS[reward_function](
+            **(reward_function_kwargs or {}))
+
+        if op_placer != "ramp_first_fit_op_placer":
+            raise ValueError(f"Unrecognised op_placer {op_placer}")
+        if op_scheduler != "srpt_op_scheduler":
+            raise ValueError(f"Unrecognised op_scheduler {op_scheduler}")
+        if dep_placer != "first_fit_dep_placer":
+            raise ValueError(f"Unrecognised dep_placer {dep_placer}")
+        if dep_scheduler != "srpt_dep_scheduler":
+            raise ValueError(f"Unrecognised dep_scheduler {dep_scheduler}")
+        self.op_placer = RampFirstFitOpPlacer()
+        self.op_scheduler = SRPTOpScheduler()
+        self.dep_placer = FirstFitDepPlacer()
+        self.dep_scheduler = SRPTDepScheduler()
+
+    # ------------------------------------------------------------------
+    def reset(self, seed: Optional[int] = None, verbose: bool = False):
+        self.step_counter = 1
+        self.cluster.reset(jobs_config=self.jobs_config,
+                           max_simulation_run_time=self.max_simulation_run_time,
+                           job_queue_capacity=self.job_queue_capacity,
+                           seed=seed,
+                           verbose=verbose)
+        self.obs = self.observation_function.reset(self)
+        self.observation_space = self.observation_function.observation_space
+        self.reward_function.reset(env=self)
+        return self.obs
+
+    def _is_done(self):
+        return self.cluster.is_done()
+
+    def _step_cluster(self, action):
+        self.cluster.step(action=action)
+        self.cluster_step_stats[self.cluster.step_counter] = self.cluster.step_stats
+
+    # ------------------------------------------------------------------
+    def step(self, action: int, verbose: bool = False):
+        self.cluster_step_stats = {}
+
+        if action not in self.action_set:
+            raise ValueError(f"action {action} not in action set {self.action_set}")
+        if not self.obs["action_mask"][action]:
+            if self.apply_action_mask:
+                raise ValueError(
+                    f"action {action} invalid given mask {self.obs['action_mask']}; "
+                    "set apply_action_mask=False to coerce to 0 instead")
+            action = 0
+
+        if action != 0:
+            job_id, job = next(iter(self.cluster.job_queue.jobs.items()))
+            g = job.graph
+            device_type = self.cluster.device_type
+            cc = g.compute_cost[device_type]
+            partition_action = defaultdict(dict)
+            for i in range(g.n):
+                if g.pass_type[i] != FWD:
+                    continue
+                num = sip_ml_num_partitions(float(cc[i]),
+                                            self.min_op_run_time_quantum,
+                                            max_partitions_per_op=action)
+                partition_action[job_id][g.names[i]] = num
+                partition_action[job_id][g.names[int(g.counterpart[i])]] = num
+            self.op_partition = OpPartition(dict(partition_action), cluster=self.cluster)
+        else:
+            self.op_partition = OpPartition({}, cluster=self.cluster)
+
+        self.op_placement = self.op_placer.get(op_partition=self.op_partition,
+                                               cluster=self.cluster)
+        self.op_schedule = self.op_scheduler.get(op_partition=self.op_partition,
+                                                 op_placement=self.op_placement,
+                                                 cluster=self.cluster)
+        self.dep_placement = self.dep_placer.get(op_partition=self.op_partition,
+                                                 op_placement=self.op_placement,
+                                                 cluster=self.cluster)
+        self.dep_schedule = self.dep_scheduler.get(op_partition=self.op_partition,
+                                                   dep_placement=self.dep_placement,
+                                                   cluster=self.cluster)
+        self.action = Action(op_partition=self.op_partition,
+                             op_placement=self.op_placement,
+                             op_schedule=self.op_schedule,
+                             dep_placement=self.dep_placement,
+                             dep_schedule=self.dep_schedule)
+
+        self.last_job_arrived_job_idx = self.cluster.last_job_arrived_job_idx
+
+        self._step_cluster(self.action)
+
+        # which jobs truly got placed (not blocked by the lookahead JCT check)
+        self.placed_job_idxs = set(self.action.job_idxs)
+        for job_idx in list(self.placed_job_idxs):
+            if job_idx in self.cluster.jobs_blocked:
+                self.placed_job_idxs.remove(job_idx)
+
+        self.reward = self.reward_function.extract(env=self, done=self._is_done())
+
+        # idle fast-forward until a job is queued or the sim is done
+        while len(self.cluster.job_queue) == 0 and not self.cluster.is_done():
+            self._step_cluster(Action())
+
+        self.done = self._is_done()
+        if not self.done:
+            self.obs = self.observation_function.extract(env=self, done=False)
+        self.info = {}
+        self.step_counter += 1
+        return self.obs, self.reward, self.done, self.info
