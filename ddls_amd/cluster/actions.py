@@ -153,8 +153,9 @@ class DepPlacement:
             self.job_ids.add(job_id)
             for dep_idx, channels in deps.items():
                 for channel_id in channels:
-                    if channel_id is None:
-                        continue
+                    # NB the reference keeps channel None entries for non-flow
+                    # deps (dep_placement.py:19-34); the None "channel" carries
+                    # the job through DepSchedule into the Action intersection
                     self.channel_ids.add(channel_id)
                     self.channel_to_job_to_deps[channel_id][job_id].add(dep_idx)
                     self.job_to_dep_to_channels[job_id][dep_idx].add(channel_id)
