@@ -1,0 +1,148 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: PPO training of the PAC-ML GNN partitioner on the
+32-worker RAMP cluster simulator (BASELINE.json metric: PPO env-steps/sec,
+whole node).
+
+One process per GPU (torchrun for N>1), vectorised envs per rank, fused
+gradient all-reduce with RCCL over xGMI.  A "step" is one PPO iteration:
+collect a fixed per-rank rollout (weak scaling) + the full SGD update.
+
+Synthetic workload: deterministic pipedream-format job graphs (the reference's
+"small_graphs" profile set is not distributed); random-init GNN weights.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+
+
+def build_env_fn(seed_offset: int = 0, device_type: str = "A100"):
+    from ddls_amd.envs import RampJobPartitioningEnvironment
+    from ddls_amd.workloads import ensure_default_set
+    data = ensure_default_set()
+    worker_cls = ("ddls_amd.devices.A100" if device_type == "A100"
+                  else "ddls_amd.devices.MI355X")
+
+    def make():
+        return RampJobPartitioningEnvironment(
+            topology_config={"type": "ramp", "kwargs": {
+                "num_communication_groups": 4,
+                "num_racks_per_communication_group": 4,
+                "num_servers_per_rack": 2,
+                "num_channels": 1,
+                "total_node_bandwidth": 1.6e12,
+                "intra_gpu_propagation_latency": 50e-9,
+                "worker_io_latency": 100e-9}},
+            node_config={"type_1": {"num_nodes": 32, "workers_config": [
+                {"num_workers": 1, "worker": worker_cls}]}},
+            jobs_config={"path_to_files": data,
+                         "replication_factor": 1000,
+                         "job_sampling_mode": "remove_and_repeat",
+                         "job_interarrival_time_dist": {
+                             "_target_": "ddls_amd.distributions.Fixed",
+                             "val": 1000},
+                         "max_acceptable_job_completion_time_frac_dist": {
+                             "_target_": "ddls_amd.distributions.Uniform",
+                             "min_val": 0.1, "max_val": 1, "decimals": 2},
+                         "num_training_steps": 50},
+            max_partitions_per_op=16,
+            min_op_run_time_quantum=0.01,
+            pad_obs_kwargs=None,
+            max_simulation_run_time=1e6)
+    return make
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=5)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--envs-per-rank", type=int, default=8)
+    ap.add_argument("--rollout-steps-per-env", type=int, default=32)
+    ap.add_argument("--num-sgd-iter", type=int, default=8)
+    ap.add_argument("--sgd-minibatch-size", type=int, default=128)
+    args = ap.parse_args()
+
+    from ddls_amd.models.gnn import GNNPolicy
+    from ddls_amd.parallel import (get_rank, get_world_size,
+                                   init_distributed_from_env, is_distributed)
+    from ddls_amd.rl.ppo import PPOConfig, PPOTrainer
+    from ddls_amd.rl.rollout import VectorEnv
+
+    rank = init_distributed_from_env()
+    world_size = get_world_size()
+    use_cuda = torch.cuda.is_available()
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
+
+    torch.manual_seed(0)  # identical init weights on every rank
+    policy = GNNPolicy(num_actions=17)
+
+    env_fn = build_env_fn()
+    venv = VectorEnv([env_fn for _ in range(args.envs_per_rank)],
+                     base_seed=1 + 100000 * rank)
+    per_step_env_steps = args.envs_per_rank * args.rollout_steps_per_env
+    cfg = PPOConfig(train_batch_size=per_step_env_steps,
+                    sgd_minibatch_size=args.sgd_minibatch_size,
+                    num_sgd_iter=args.num_sgd_iter)
+    trainer = PPOTrainer(venv, policy, cfg, device=device)
+
+    def barrier_sync():
+        if is_distributed():
+            torch.distributed.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        trainer.train(num_steps=args.rollout_steps_per_env)
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        trainer.train(num_steps=args.rollout_steps_per_env)
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if is_distributed():
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if use_cuda else "cpu")
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    total_env_steps = args.steps * per_step_env_steps * world_size
+    value = total_env_steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "ppo_env_steps_per_sec",
+            "value": value,
+            "unit": "env_steps/s",
+            "n_gpus": world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "pacml_gnn_msg32_hidden64_2rounds",
+                "global_batch": per_step_env_steps * world_size,
+                "seq_len": 150,
+                "parallelism": f"dp{world_size}",
+                "ramp": "4x4x2_32workers",
+                "envs_per_rank": args.envs_per_rank,
+                "num_sgd_iter": args.num_sgd_iter,
+            },
+        }))
+
+
+if __name__ == "__main__":
+    main()
