@@ -63,6 +63,32 @@ class GraphBatch:
     dst: torch.Tensor          # [E_total]
     graph_of_node: torch.Tensor  # [N_total] graph id per node
     num_graphs: int
+    _csr: Optional[tuple] = None
+    _node_ptr: Optional[torch.Tensor] = None
+
+    def csr_by_dst(self):
+        """(edge_order, indptr) with edges sorted by destination node —
+        the segment layout the HIP message_reduce kernel consumes."""
+        if self._csr is None:
+            n = self.z.shape[0]
+            order = torch.argsort(self.dst, stable=True)
+            counts = torch.bincount(self.dst, minlength=n)
+            indptr = torch.zeros(n + 1, dtype=torch.int64, device=self.z.device)
+            torch.cumsum(counts, 0, out=indptr[1:])
+            self._csr = (order, indptr)
+        return self._csr
+
+    def node_ptr(self):
+        """[G+1] prefix over nodes grouped by graph id (nodes are stored
+        graph-contiguously by construction)."""
+        if self._node_ptr is None:
+            counts = torch.bincount(self.graph_of_node,
+                                    minlength=self.num_graphs)
+            ptr = torch.zeros(self.num_graphs + 1, dtype=torch.int64,
+                              device=self.z.device)
+            torch.cumsum(counts, 0, out=ptr[1:])
+            self._node_ptr = ptr
+        return self._node_ptr
 
     @staticmethod
     def from_padded(node_features: torch.Tensor,
@@ -122,6 +148,31 @@ class MeanPoolLayer(nn.Module):
                                   module_depth)
         self.out_features_msg = out_features_msg
         self.out_features_reduce = out_features_reduce
+        # the fused HIP kernels implement the tuned LN->Linear->ReLU depth-1
+        # structure; other configs take the torch path
+        self._hip_eligible = (module_depth == 1
+                              and aggregator_activation == "relu")
+
+    def forward_batch(self, z: torch.Tensor, batch: "GraphBatch") -> torch.Tensor:
+        from .. import ops as hip_ops
+        if self._hip_eligible and hip_ops.hip_ops_enabled_for(z):
+            return self._forward_hip(z, batch)
+        return self.forward(z, batch.e, batch.src, batch.dst)
+
+    def _forward_hip(self, z: torch.Tensor, batch: "GraphBatch") -> torch.Tensor:
+        from .. import ops as hip_ops
+        ext = hip_ops.get_extension(required=True)
+        ln_n, lin_n = self.node_module[0], self.node_module[1]
+        ln_e, lin_e = self.edge_module[0], self.edge_module[1]
+        ln_r, lin_r = self.reduce_module[0], self.reduce_module[1]
+        hn = ext.row_mlp(z.contiguous(), ln_n.weight, ln_n.bias,
+                         lin_n.weight.contiguous(), lin_n.bias)
+        he = ext.row_mlp(batch.e.contiguous(), ln_e.weight, ln_e.bias,
+                         lin_e.weight.contiguous(), lin_e.bias)
+        order, indptr = batch.csr_by_dst()
+        return ext.message_reduce(hn, he, batch.src, order, indptr,
+                                  ln_r.weight, ln_r.bias,
+                                  lin_r.weight.contiguous(), lin_r.bias)
 
     def forward(self, z: torch.Tensor, e: torch.Tensor, src: torch.Tensor,
                 dst: torch.Tensor) -> torch.Tensor:
@@ -177,8 +228,25 @@ class GNN(nn.Module):
     def forward(self, batch: GraphBatch) -> torch.Tensor:
         z = batch.z
         for layer in self.layers:
-            z = layer(z, batch.e, batch.src, batch.dst)
+            z = layer.forward_batch(z, batch)
         return z
+
+
+def graph_mean(node_emb: torch.Tensor, batch: GraphBatch) -> torch.Tensor:
+    """Per-graph mean of node embeddings (HIP segment kernel on GPU)."""
+    from .. import ops as hip_ops
+    if hip_ops.hip_ops_enabled_for(node_emb):
+        ext = hip_ops.get_extension(required=True)
+        return ext.segment_mean(node_emb.contiguous(), batch.node_ptr(),
+                                batch.num_graphs)
+    B = batch.num_graphs
+    sums = torch.zeros(B, node_emb.shape[-1], dtype=node_emb.dtype,
+                       device=node_emb.device)
+    sums.index_add_(0, batch.graph_of_node, node_emb)
+    counts = torch.zeros(B, dtype=node_emb.dtype, device=node_emb.device)
+    counts.index_add_(0, batch.graph_of_node,
+                      torch.ones_like(batch.graph_of_node, dtype=node_emb.dtype))
+    return sums / counts.clamp(min=1).unsqueeze(-1)
 
 
 class GNNPolicy(nn.Module):
@@ -220,31 +288,25 @@ class GNNPolicy(nn.Module):
         self.policy_branch = branch(num_actions)
         self.value_branch = branch(1)
 
+    def forward_flat(self, batch: GraphBatch, graph_features: torch.Tensor,
+                     action_mask: torch.Tensor):
+        """Forward on a pre-collated flat batch (the rollout/SGD hot path)."""
+        node_emb = self.gnn(batch)                      # [N_total, out_node]
+        graph_node_emb = graph_mean(node_emb, batch)    # [B, out_node]
+        graph_emb = self.graph_module(graph_features)
+        final_emb = torch.cat([graph_node_emb, graph_emb], dim=-1)
+        logits = self.policy_branch(final_emb)
+        value = self.value_branch(final_emb).squeeze(-1)
+        if self.config["apply_action_mask"]:
+            inf_mask = torch.clamp(torch.log(action_mask.to(logits.dtype)),
+                                   min=torch.finfo(torch.float32).min)
+            logits = logits + inf_mask
+        return logits, value
+
     def forward(self, obs: Dict[str, torch.Tensor]):
         """obs: batched padded observation dict (torch tensors, [B, ...])."""
         batch = GraphBatch.from_padded(
             obs["node_features"], obs["edge_features"], obs["edges_src"],
             obs["edges_dst"], obs["node_split"], obs["edge_split"])
-        node_emb = self.gnn(batch)                      # [N_total, out_node]
-        # per-graph mean of node embeddings
-        B = batch.num_graphs
-        sums = torch.zeros(B, node_emb.shape[-1], dtype=node_emb.dtype,
-                           device=node_emb.device)
-        sums.index_add_(0, batch.graph_of_node, node_emb)
-        counts = torch.zeros(B, dtype=node_emb.dtype, device=node_emb.device)
-        counts.index_add_(0, batch.graph_of_node,
-                          torch.ones_like(batch.graph_of_node, dtype=node_emb.dtype))
-        graph_node_emb = sums / counts.clamp(min=1).unsqueeze(-1)
-
-        graph_emb = self.graph_module(obs["graph_features"])
-        final_emb = torch.cat([graph_node_emb, graph_emb], dim=-1)
-
-        logits = self.policy_branch(final_emb)
-        value = self.value_branch(final_emb).squeeze(-1)
-
-        if self.config["apply_action_mask"]:
-            mask = obs["action_mask"].to(logits.dtype)
-            inf_mask = torch.clamp(torch.log(mask),
-                                   min=torch.finfo(torch.float32).min)
-            logits = logits + inf_mask
-        return logits, value
+        return self.forward_flat(batch, obs["graph_features"],
+                                 obs["action_mask"])

@@ -1,0 +1,231 @@
+// Fused MeanPool message-passing kernels for gfx950 (MI355X, CDNA4).
+//
+// Replaces the per-message torch op chain of ddls_amd.models.gnn.MeanPoolLayer
+// (reference semantics: ddls/ml_models/models/mean_pool.py:107-150) with two
+// fused kernels:
+//   1. row_mlp:      y = act((LN(x) @ W^T) + b) for node/edge feature MLPs
+//   2. message_reduce: per node v, mean over {self-msg} u {in-msgs} of
+//      act(LN([hn_src || he_e]) @ Wr^T + br), zero for mail-less nodes,
+//      with edges pre-sorted by destination (CSR segments, no atomics).
+//
+// Design notes (CDNA4): wave64; one wave per row/node; LN statistics via
+// wave shuffles; per-wave LDS staging of the normalised vector so every lane
+// can compute one output feature; the reduce weight matrix (msg x out,
+// 32x64 = 8 KB) is staged once per workgroup in LDS.  These are
+// latency-bound shapes (tens of thousands of rows x K<=64); the win over the
+// eager torch chain is fusion (one kernel instead of ~10) not MFMA peak.
+
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#define WAVE 64
+#define WAVES_PER_BLOCK 4
+#define BLOCK (WAVE * WAVES_PER_BLOCK)
+#define LN_EPS 1e-5f
+
+__device__ __forceinline__ float wave_reduce_sum(float v) {
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        v += __shfl_down(v, off, WAVE);
+    return __shfl(v, 0, WAVE);
+}
+
+// y[r, j] = act( (x[r,:]-mu)/sigma * g + b_ln  dot  W[j,:] + b[j] )
+// W is torch Linear layout [H, F]. F <= 64, H <= 64.
+__global__ void row_mlp_kernel(const float* __restrict__ x,
+                               const float* __restrict__ ln_g,
+                               const float* __restrict__ ln_b,
+                               const float* __restrict__ W,
+                               const float* __restrict__ b,
+                               float* __restrict__ y,
+                               int R, int F, int H) {
+    __shared__ float xs[WAVES_PER_BLOCK][64 + 1];
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int row0 = blockIdx.x * WAVES_PER_BLOCK + wave;
+    const int stride = gridDim.x * WAVES_PER_BLOCK;
+    for (int row = row0; row < R; row += stride) {
+        float v = (lane < F) ? x[(long)row * F + lane] : 0.0f;
+        const float mean = wave_reduce_sum(v) / F;
+        const float d = (lane < F) ? (v - mean) : 0.0f;
+        const float var = wave_reduce_sum(d * d) / F;
+        const float inv_sigma = rsqrtf(var + LN_EPS);
+        if (lane < F)
+            xs[wave][lane] = d * inv_sigma * ln_g[lane] + ln_b[lane];
+        __builtin_amdgcn_wave_barrier();
+        if (lane < H) {
+            float acc = b[lane];
+            const float* wrow = W + (long)lane * F;
+            for (int k = 0; k < F; ++k)
+                acc = fmaf(xs[wave][k], wrow[k], acc);
+            y[(long)row * H + lane] = fmaxf(acc, 0.0f);  // relu
+        }
+        __builtin_amdgcn_wave_barrier();
+    }
+}
+
+// out[v,:] = has_mail(v) ? mean over messages of act(LN(m) @ Wr^T + br) : 0
+// messages of v: self message [hn[v] || 0] plus, for each in-edge e in
+// [indptr[v], indptr[v+1]) (edge ids edge_order[e], sorted by dst),
+// [hn[src[e]] || he[e]].  MSG = 2*half <= 64; OUT <= 64.
+__global__ void message_reduce_kernel(const float* __restrict__ hn,
+                                      const float* __restrict__ he,
+                                      const long* __restrict__ src,
+                                      const long* __restrict__ edge_order,
+                                      const long* __restrict__ indptr,
+                                      const float* __restrict__ ln_g,
+                                      const float* __restrict__ ln_b,
+                                      const float* __restrict__ Wr,
+                                      const float* __restrict__ br,
+                                      float* __restrict__ out,
+                                      int N, int half, int OUT) {
+    const int MSG = 2 * half;
+    // Wr staged once per block: [OUT][MSG] <= 64*64*4 = 16 KB
+    __shared__ float wr_s[64 * 64];
+    __shared__ float ms[WAVES_PER_BLOCK][64 + 1];
+    for (int i = threadIdx.x; i < OUT * MSG; i += BLOCK)
+        wr_s[i] = Wr[i];
+    __syncthreads();
+
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int v0 = blockIdx.x * WAVES_PER_BLOCK + wave;
+    const int stride = gridDim.x * WAVES_PER_BLOCK;
+    for (int v = v0; v < N; v += stride) {
+        const long e_begin = indptr[v];
+        const long e_end = indptr[v + 1];
+        const int indeg = (int)(e_end - e_begin);
+        float acc = 0.0f;  // lane's output feature accumulator
+        if (indeg > 0) {
+            // self message + in messages
+            for (int mi = -1; mi < indeg; ++mi) {
+                float mv;
+                if (mi < 0) {  // self message: [hn[v] || zeros]
+                    mv = (lane < half) ? hn[(long)v * half + lane] : 0.0f;
+                } else {
+                    const long e = edge_order[e_begin + mi];
+                    const long s = src[e];
+                    if (lane < half) mv = hn[s * half + lane];
+                    else if (lane < MSG) mv = he[e * half + (lane - half)];
+                    else mv = 0.0f;
+                }
+                const float in_range = (lane < MSG) ? 1.0f : 0.0f;
+                const float mean = wave_reduce_sum(mv * in_range) / MSG;
+                const float d = (lane < MSG) ? (mv - mean) : 0.0f;
+                const float var = wave_reduce_sum(d * d) / MSG;
+                const float inv_sigma = rsqrtf(var + LN_EPS);
+                if (lane < MSG)
+                    ms[wave][lane] = d * inv_sigma * ln_g[lane] + ln_b[lane];
+                __builtin_amdgcn_wave_barrier();
+                if (lane < OUT) {
+                    float r = br[lane];
+                    const float* wrow = wr_s + lane * MSG;
+                    for (int k = 0; k < MSG; ++k)
+                        r = fmaf(ms[wave][k], wrow[k], r);
+                    acc += fmaxf(r, 0.0f);  // relu then mean
+                }
+                __builtin_amdgcn_wave_barrier();
+            }
+            if (lane < OUT)
+                out[(long)v * OUT + lane] = acc / (indeg + 1);
+        } else {
+            // DGL zero-fills nodes with no incoming messages
+            if (lane < OUT) out[(long)v * OUT + lane] = 0.0f;
+        }
+    }
+}
+
+// segment mean of node embeddings per graph: out[g,:] = mean over nodes of g.
+// node_ptr: [G+1] prefix over nodes grouped by graph (nodes already stored
+// graph-contiguously by construction of the flat batch).
+__global__ void segment_mean_kernel(const float* __restrict__ x,
+                                    const long* __restrict__ node_ptr,
+                                    float* __restrict__ out,
+                                    int G, int F) {
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int g0 = blockIdx.x * WAVES_PER_BLOCK + wave;
+    const int stride = gridDim.x * WAVES_PER_BLOCK;
+    for (int g = g0; g < G; g += stride) {
+        const long n_begin = node_ptr[g];
+        const long n_end = node_ptr[g + 1];
+        if (lane < F) {
+            float s = 0.0f;
+            for (long n = n_begin; n < n_end; ++n)
+                s += x[n * F + lane];
+            const long cnt = n_end - n_begin;
+            out[(long)g * F + lane] = (cnt > 0) ? s / cnt : 0.0f;
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+// host wrappers
+// ---------------------------------------------------------------------------
+
+static inline int grid_for(int rows) {
+    int blocks = (rows + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
+    // >> 256 workgroups wanted to fill 256 CUs across 8 XCDs; cap + stride
+    return std::min(blocks, 2048);
+}
+
+torch::Tensor row_mlp(torch::Tensor x, torch::Tensor ln_g, torch::Tensor ln_b,
+                      torch::Tensor W, torch::Tensor b) {
+    TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kFloat32);
+    TORCH_CHECK(x.is_contiguous() && W.is_contiguous());
+    const int R = x.size(0), F = x.size(1), H = W.size(0);
+    TORCH_CHECK(F <= 64 && H <= 64, "row_mlp supports F,H <= 64");
+    auto y = torch::empty({R, H}, x.options());
+    if (R == 0) return y;
+    hipStream_t stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(row_mlp_kernel, dim3(grid_for(R)), dim3(BLOCK), 0,
+                       stream, x.data_ptr<float>(), ln_g.data_ptr<float>(),
+                       ln_b.data_ptr<float>(), W.data_ptr<float>(),
+                       b.data_ptr<float>(), y.data_ptr<float>(), R, F, H);
+    return y;
+}
+
+torch::Tensor message_reduce(torch::Tensor hn, torch::Tensor he,
+                             torch::Tensor src, torch::Tensor edge_order,
+                             torch::Tensor indptr, torch::Tensor ln_g,
+                             torch::Tensor ln_b, torch::Tensor Wr,
+                             torch::Tensor br) {
+    TORCH_CHECK(hn.is_cuda() && hn.dtype() == torch::kFloat32);
+    TORCH_CHECK(src.dtype() == torch::kInt64 && indptr.dtype() == torch::kInt64);
+    const int N = hn.size(0), half = hn.size(1), OUT = Wr.size(0);
+    TORCH_CHECK(2 * half <= 64 && OUT <= 64,
+                "message_reduce supports msg,out <= 64");
+    auto out = torch::empty({N, OUT}, hn.options());
+    if (N == 0) return out;
+    hipStream_t stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(message_reduce_kernel, dim3(grid_for(N)), dim3(BLOCK),
+                       0, stream, hn.data_ptr<float>(), he.data_ptr<float>(),
+                       src.data_ptr<long>(), edge_order.data_ptr<long>(),
+                       indptr.data_ptr<long>(), ln_g.data_ptr<float>(),
+                       ln_b.data_ptr<float>(), Wr.data_ptr<float>(),
+                       br.data_ptr<float>(), out.data_ptr<float>(),
+                       N, half, OUT);
+    return out;
+}
+
+torch::Tensor segment_mean(torch::Tensor x, torch::Tensor node_ptr, int64_t G) {
+    TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kFloat32);
+    const int F = x.size(1);
+    TORCH_CHECK(F <= 64);
+    auto out = torch::empty({G, F}, x.options());
+    if (G == 0) return out;
+    hipStream_t stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(segment_mean_kernel, dim3(grid_for((int)G)),
+                       dim3(BLOCK), 0, stream, x.data_ptr<float>(),
+                       node_ptr.data_ptr<long>(), out.data_ptr<float>(),
+                       (int)G, F);
+    return out;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("row_mlp", &row_mlp, "fused LN+Linear+ReLU over rows");
+    m.def("message_reduce", &message_reduce,
+          "fused gather + LN + Linear + ReLU + segment mean message passing");
+    m.def("segment_mean", &segment_mean, "per-graph mean of node embeddings");
+}

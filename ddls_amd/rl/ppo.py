@@ -72,26 +72,9 @@ class PPOTrainer:
 
     def _forward_flat(self, inputs):
         """Policy forward on a pre-collated flat batch."""
-        batch = inputs["batch"]
-        node_emb = self.policy.gnn(batch)
-        B = batch.num_graphs
-        sums = torch.zeros(B, node_emb.shape[-1], dtype=node_emb.dtype,
-                           device=node_emb.device)
-        sums.index_add_(0, batch.graph_of_node, node_emb)
-        counts = torch.zeros(B, dtype=node_emb.dtype, device=node_emb.device)
-        counts.index_add_(0, batch.graph_of_node,
-                          torch.ones_like(batch.graph_of_node,
-                                          dtype=node_emb.dtype))
-        graph_node_emb = sums / counts.clamp(min=1).unsqueeze(-1)
-        graph_emb = self.policy.graph_module(inputs["graph_features"])
-        final_emb = torch.cat([graph_node_emb, graph_emb], dim=-1)
-        logits = self.policy.policy_branch(final_emb)
-        value = self.policy.value_branch(final_emb).squeeze(-1)
-        if self.policy.config["apply_action_mask"]:
-            inf_mask = torch.clamp(torch.log(inputs["action_mask"]),
-                                   min=torch.finfo(torch.float32).min)
-            logits = logits + inf_mask
-        return logits, value
+        return self.policy.forward_flat(inputs["batch"],
+                                        inputs["graph_features"],
+                                        inputs["action_mask"])
 
     # ------------------------------------------------------------------
     def collect_rollout(self, num_steps: Optional[int] = None) -> Dict[str, np.ndarray]:
