@@ -57,8 +57,9 @@ class RampFirstFitOpPlacer:
             ramp_topology, op_server_info = allocated
 
             placement = {}
+            op_to_split = dict(zip(sequence, splits))
             for op, servers_of_op in op_server_info.items():
-                split = splits[sequence.index(op)]
+                split = op_to_split[op]
                 bwd = backward_name(op, num_fwd)
                 for i, server in enumerate(servers_of_op):
                     node = cluster.topology.coord_to_node[server]
@@ -131,45 +132,68 @@ class FirstFitDepPlacer:
         topo = cluster.topology
         channels_used_across_jobs = set()
         for job_id, job in op_partition.partitioned_jobs.items():
-            used_for_this_job = set()
             if job_id not in new_placements:
                 continue
+            used_for_this_job = set()
             g = job.graph
             placement = new_placements[job_id]
             job_idx = job.details["job_idx"]
+            # vectorised flow classification
+            node_of = np.array([topo.worker_to_node[placement[nm]]
+                                for nm in g.names], dtype=np.int64)
+            src_nodes = node_of[g.src]
+            dst_nodes = node_of[g.dst]
+            is_flow = (src_nodes != dst_nodes) & (g.size > 0)
             dropped = False
             for e in range(g.m):
-                u, v = int(g.src[e]), int(g.dst[e])
-                parent_node = topo.worker_to_node[placement[g.names[u]]]
-                child_node = topo.worker_to_node[placement[g.names[v]]]
-                size = float(g.size[e])
-                if parent_node != child_node and size > 0:
+                if is_flow[e]:
                     found = self._find_path_channel(
-                        cluster, parent_node, child_node, job_idx,
+                        cluster, int(src_nodes[e]), int(dst_nodes[e]), job_idx,
                         channels_used_across_jobs)
                     if found is None:
                         job_to_dep_to_channels.pop(job_id, None)
                         dropped = True
                         break
                     path, channel_num = found
-                    for idx in range(len(path) - 1):
-                        s, d = path[idx], path[idx + 1]
-                        cid = gen_channel_id(topo.node_names[s], topo.node_names[d],
-                                             channel_num)
-                        job_to_dep_to_channels[job_id][e].add(cid)
+                    direct = getattr(topo, "direct_cid", None)
+                    if direct is not None and len(path) == 2:
+                        cid = direct[(path[0], path[1], channel_num)]
+                        job_to_dep_to_channels[job_id][e] = (cid,)
                         used_for_this_job.add(cid)
+                    else:
+                        cids = set()
+                        for idx in range(len(path) - 1):
+                            sN, d = path[idx], path[idx + 1]
+                            cid = gen_channel_id(topo.node_names[sN],
+                                                 topo.node_names[d], channel_num)
+                            cids.add(cid)
+                            used_for_this_job.add(cid)
+                        job_to_dep_to_channels[job_id][e] = cids
                 else:
-                    job_to_dep_to_channels[job_id][e].add(None)
-            if not dropped:
-                channels_used_across_jobs |= used_for_this_job
-            else:
-                channels_used_across_jobs |= used_for_this_job  # reference keeps them
+                    job_to_dep_to_channels[job_id][e] = (None,)
+            # reference tracks channels used even by dropped jobs
+            channels_used_across_jobs |= used_for_this_job
 
         return DepPlacement({k: dict(v) for k, v in job_to_dep_to_channels.items()})
 
     def _find_path_channel(self, cluster, src_node, dst_node, job_idx,
                            channels_used_across_jobs):
         topo = cluster.topology
+        direct = getattr(topo, "direct_cid", None)
+        if direct is not None:  # full mesh: single-hop paths
+            if topo.num_channels == 1:
+                channel_nums = (0,)
+            else:
+                channel_nums = list(range(topo.num_channels))
+                random.shuffle(channel_nums)
+            for channel_num in channel_nums:
+                cid = direct[(src_node, dst_node, channel_num)]
+                channel = topo.channel_id_to_channel[cid]
+                if (job_idx in channel.mounted_job_idx_to_deps
+                        or (len(channel.mounted_job_idx_to_deps) == 0
+                            and cid not in channels_used_across_jobs)):
+                    return [src_node, dst_node], channel_num
+            return None
         paths = topo.shortest_paths(src_node, dst_node)
         channel_nums = list(range(topo.num_channels))
         random.shuffle(channel_nums)

@@ -115,7 +115,8 @@ class OpPlacement:
                 split_fwd_names=set(op_partition.job_id_to_mp_split_forward_op_ids[job_id]),
                 op_name_to_worker=action[job_id],
                 fwd_name_to_splits=op_partition.job_id_to_forward_op_id_to_mp_splits[job_id],
-                topology=cluster.topology)
+                topology=cluster.topology,
+                collective_time_cache=getattr(cluster, "collective_time_cache", None))
 
 
 class OpSchedule:
@@ -145,20 +146,15 @@ class DepPlacement:
 
     def __init__(self, action: Dict):
         self.action = action
-        self.job_ids = set()
+        self.job_ids = set(action.keys())
         self.channel_ids = set()
-        self.channel_to_job_to_deps = defaultdict(lambda: defaultdict(set))
-        self.job_to_dep_to_channels = defaultdict(lambda: defaultdict(set))
+        self.job_to_dep_to_channels = action  # alias: same mapping shape
         for job_id, deps in action.items():
-            self.job_ids.add(job_id)
-            for dep_idx, channels in deps.items():
-                for channel_id in channels:
-                    # NB the reference keeps channel None entries for non-flow
-                    # deps (dep_placement.py:19-34); the None "channel" carries
-                    # the job through DepSchedule into the Action intersection
-                    self.channel_ids.add(channel_id)
-                    self.channel_to_job_to_deps[channel_id][job_id].add(dep_idx)
-                    self.job_to_dep_to_channels[job_id][dep_idx].add(channel_id)
+            # NB the reference keeps channel None entries for non-flow deps
+            # (dep_placement.py:19-34); the None "channel" carries the job
+            # through DepSchedule into the Action intersection
+            for channels in deps.values():
+                self.channel_ids.update(channels)
 
 
 class Action:

@@ -104,33 +104,42 @@ def update_dep_run_times(partitioned_job,
                          split_fwd_names: Set[str],
                          op_name_to_worker: Dict[str, str],
                          fwd_name_to_splits: Dict[str, int],
-                         topology) -> None:
+                         topology,
+                         collective_time_cache: Dict = None) -> None:
     """Set every dep's init run time on the partitioned job given placement.
 
     Reference ``actions/utils.py:13-41`` + grouping + collective/one-to-one
     pricing.  ``op_name_to_worker`` maps partitioned-op NAME -> worker id.
+    Vectorised over edges; collective times are memoised by their geometry
+    key (message size, nodes, racks, cgs) in ``collective_time_cache`` (valid
+    per-topology — all other model parameters are topology constants).
     """
     pg: CompGraph = partitioned_job.graph
     og: CompGraph = original_job.graph
     n_fwd = int((og.pass_type == FWD).sum())
 
     pg_idx = pg.name_to_idx
-    # edge lookup (u_idx, v_idx) -> edge index
-    edge_lookup = {(int(u), int(v)): e
-                   for e, (u, v) in enumerate(zip(pg.src, pg.dst))}
+    rev_edge = pg.reverse_edge  # cached on the graph (shared across jobs)
 
-    worker_of = {}  # pg node idx -> worker id
+    # dense per-node server index + coordinates
+    w2n = topology.worker_to_node
+    node_server = np.fromiter((w2n[op_name_to_worker[nm]] for nm in pg.names),
+                              dtype=np.int64, count=pg.n)
+    coords_arr = getattr(topology, "_coords_arr", None)
+    if coords_arr is None:
+        coords_arr = np.asarray(topology.coords, dtype=np.int64)
+        topology._coords_arr = coords_arr
+    node_coord = coords_arr[node_server]
 
-    def worker(node_idx: int) -> str:
-        w = worker_of.get(node_idx)
-        if w is None:
-            w = op_name_to_worker[pg.names[node_idx]]
-            worker_of[node_idx] = w
-        return w
-
-    collectives: List[List[int]] = []
-    one_to_one: Set[int] = set()
-    collective_deps: Set[int] = set()
+    collectives: List[np.ndarray] = []
+    one_to_one: List[np.ndarray] = []
+    pair_lo_all: List[np.ndarray] = []
+    pair_hi_all: List[np.ndarray] = []
+    # per-class membership masks: the reference dedups via sets (an edge can
+    # be gathered twice, e.g. the join edges appear in BOTH the fwd out-group
+    # and the bwd in-group of the last forward op)
+    coll_mask = np.zeros(pg.m, dtype=bool)
+    o2o_mask = np.zeros(pg.m, dtype=bool)
 
     for f in range(og.n):
         if og.pass_type[f] != FWD:
@@ -139,92 +148,134 @@ def update_dep_run_times(partitioned_job,
         bwd_nm = backward_name(fwd_nm, n_fwd)
         if fwd_nm in split_fwd_names:
             n_splits = fwd_name_to_splits[fwd_nm]
-            fwd_deps: List[int] = []
-            bwd_deps: List[int] = []
-            sync_deps: List[int] = []
-            sync_pairs_added = set()
-            for s in range(n_splits):
-                pf = pg_idx[partitioned_name(fwd_nm, s)]
-                fwd_deps.extend(int(e) for e in pg.out_edges_of(pf))
-                pb = pg_idx[partitioned_name(bwd_nm, s)]
-                for e in pg.in_edges_of(pb):
-                    e = int(e)
-                    u, v = int(pg.src[e]), int(pg.dst[e])
-                    if (v, u) in edge_lookup:  # bidirectional sync edge
-                        key = (min(u, v), max(u, v))
-                        if key not in sync_pairs_added:
-                            sync_pairs_added.add(key)
-                            sync_deps.append(e)
-                            sync_deps.append(edge_lookup[(v, u)])
-                    else:
-                        bwd_deps.append(e)
+            fwd_deps = np.concatenate(
+                [pg.out_edges_of(pg_idx[partitioned_name(fwd_nm, s)])
+                 for s in range(n_splits)])
+            in_e = np.concatenate(
+                [pg.in_edges_of(pg_idx[partitioned_name(bwd_nm, s)])
+                 for s in range(n_splits)])
+            rev = rev_edge[in_e]
+            sync_m = rev >= 0  # bidirectional sync edges
+            bwd_deps = in_e[~sync_m]
+            sync_e, sync_r = in_e[sync_m], rev[sync_m]
             # collective type 1: symmetric parent/child server multisets
-            for group in (fwd_deps, bwd_deps):
-                parents = sorted(worker(int(pg.src[e])) for e in group)
-                children = sorted(worker(int(pg.dst[e])) for e in group)
-                if parents == children:
-                    collectives.append(list(group))
-                    collective_deps.update(group)
+            for ga in (fwd_deps, bwd_deps):
+                parents = np.sort(node_server[pg.src[ga]])
+                children = np.sort(node_server[pg.dst[ga]])
+                if len(parents) == len(children) and np.array_equal(parents, children):
+                    collectives.append(ga)
+                    coll_mask[ga] = True
                 else:
-                    one_to_one.update(group)
+                    one_to_one.append(ga)
+                    o2o_mask[ga] = True
             # collective type 2: each bidirectional sync pair is a collective
-            for i in range(0, len(sync_deps), 2):
-                pair = [sync_deps[i], sync_deps[i + 1]]
-                collectives.append(pair)
-                collective_deps.update(pair)
+            if len(sync_e) > 0:
+                lo = np.minimum(sync_e, sync_r)
+                hi = np.maximum(sync_e, sync_r)
+                _, first = np.unique(lo * pg.m + hi, return_index=True)
+                first = np.sort(first)
+                pair_lo_all.append(sync_e[first])
+                pair_hi_all.append(sync_r[first])
+                coll_mask[sync_e[first]] = True
+                coll_mask[sync_r[first]] = True
         else:
             pf = pg_idx[str(fwd_nm)]
-            one_to_one.update(int(e) for e in pg.out_edges_of(pf))
             pb = pg_idx[str(bwd_nm)]
-            one_to_one.update(int(e) for e in pg.in_edges_of(pb))
+            grp = np.unique(np.concatenate([pg.out_edges_of(pf),
+                                            pg.in_edges_of(pb)]))
+            one_to_one.append(grp)
+            o2o_mask[grp] = True
 
-    if pg.m != len(collective_deps) + len(one_to_one):
+    # NB collective multiset-equality note: node_server (dense server index)
+    # stands in for the reference's worker-id strings — bijective with one
+    # worker per server, so sorted-equality is unchanged.
+    cache = collective_time_cache if collective_time_cache is not None else {}
+
+    n_coll_deps = int(coll_mask.sum())
+    n_o2o_deps = int(o2o_mask.sum())
+    if pg.m != n_coll_deps + n_o2o_deps or (coll_mask & o2o_mask).any():
         raise AssertionError(
             f"partitioned graph has {pg.m} edges but grouped "
-            f"{len(collective_deps)} collective + {len(one_to_one)} one-to-one")
+            f"{n_coll_deps} collective + {n_o2o_deps} one-to-one")
 
-    # ---- price collectives (set_collective_dep_run_time:126-147) ----
-    for group in collectives:
-        cgs, racks, nodes, servers = set(), set(), set(), set()
-        message_size = 0.0
-        for e in group:
-            for node_idx in (int(pg.src[e]), int(pg.dst[e])):
-                w = worker(node_idx)
-                server_node = topology.worker_to_node[w]
-                c, r, s = topology.coords[server_node]
-                cgs.add(c)
-                racks.add(r)
-                nodes.add(s)
-                servers.add(w)
-            message_size += float(pg.size[e])
-        if len(servers) == 1:
+    # ---- price sync-pair collectives, batched ----
+    if pair_lo_all:
+        lo = np.concatenate(pair_lo_all)
+        hi = np.concatenate(pair_hi_all)
+        a, b = pg.src[lo], pg.dst[lo]  # the two endpoints of each pair
+        msg = pg.size[lo] + pg.size[hi]
+        same = node_server[a] == node_server[b]
+        dn = (node_coord[a, 2] != node_coord[b, 2]).astype(np.int64)
+        dr = (node_coord[a, 1] != node_coord[b, 1]).astype(np.int64)
+        dc = (node_coord[a, 0] != node_coord[b, 0]).astype(np.int64)
+        times = np.zeros(len(lo))
+        active = ~same
+        if active.any():
+            combo = (dn + 2 * dr + 4 * dc)
+            # few distinct (msg, geometry) combos per job: price each once
+            keys = np.stack([msg, combo.astype(np.float64)], axis=1)
+            uniq, inv = np.unique(keys[active], axis=0, return_inverse=True)
+            priced = np.empty(len(uniq))
+            for k, (m_, cb_) in enumerate(uniq):
+                cb_ = int(cb_)
+                key = (float(m_), 1 + (cb_ & 1), 1 + ((cb_ >> 1) & 1),
+                       1 + ((cb_ >> 2) & 1))
+                t = cache.get(key)
+                if t is None:
+                    t = calc_ramp_all_reduce_time(
+                        message_size=key[0], node_ids=key[1], racks=key[2],
+                        cgs=key[3], cont_racks=1,
+                        x=topology.num_communication_groups,
+                        DATA_RATE=topology.channel_bandwidth,
+                        latency=topology.intra_gpu_propagation_latency,
+                        IO_latency=topology.worker_io_latency)
+                    cache[key] = t
+                priced[k] = t
+            times[active] = priced[inv]
+        partitioned_job.dep_init_run_time[lo] = times
+        partitioned_job.dep_init_run_time[hi] = times
+        partitioned_job.dep_remaining[lo] = times
+        partitioned_job.dep_remaining[hi] = times
+
+    # ---- price group collectives (set_collective_dep_run_time:126-147) ----
+    for ga in collectives:
+        message_size = float(pg.size[ga].sum())
+        endpoints = np.concatenate([pg.src[ga], pg.dst[ga]])
+        coords = node_coord[endpoints]
+        servers = node_server[endpoints]
+        if len(np.unique(servers)) == 1:
             t = 0.0
         else:
-            # cont_racks: the reference's "NEW NEW" dedup over (cg, rack,
-            # server) tuples always yields 1 because the server id determines
-            # its cg (actions/utils.py:221-232)
-            t = calc_ramp_all_reduce_time(
-                message_size=message_size,
-                node_ids=len(nodes), racks=len(racks), cgs=len(cgs),
-                cont_racks=1,
-                x=topology.num_communication_groups,
-                DATA_RATE=topology.channel_bandwidth,
-                latency=topology.intra_gpu_propagation_latency,
-                IO_latency=topology.worker_io_latency)
-        for e in group:
-            partitioned_job.set_dep_init_run_time(e, t)
+            key = (message_size, len(np.unique(coords[:, 2])),
+                   len(np.unique(coords[:, 1])), len(np.unique(coords[:, 0])))
+            t = cache.get(key)
+            if t is None:
+                # cont_racks: the reference's "NEW NEW" dedup over (cg, rack,
+                # server) tuples always yields 1 because the server id
+                # determines its cg (actions/utils.py:221-232)
+                t = calc_ramp_all_reduce_time(
+                    message_size=message_size,
+                    node_ids=key[1], racks=key[2], cgs=key[3],
+                    cont_racks=1,
+                    x=topology.num_communication_groups,
+                    DATA_RATE=topology.channel_bandwidth,
+                    latency=topology.intra_gpu_propagation_latency,
+                    IO_latency=topology.worker_io_latency)
+                cache[key] = t
+        partitioned_job.dep_init_run_time[ga] = t
+        partitioned_job.dep_remaining[ga] = t
 
     # ---- price one-to-one deps (set_one_to_one_dep_run_time:149-167) ----
-    for e in one_to_one:
-        u, v = int(pg.src[e]), int(pg.dst[e])
-        if worker(u) == worker(v):
-            t = 0.0
-        elif pg.size[e] == 0:
-            t = 0.0
-        else:
-            t = calc_one_to_one_time(
-                float(pg.size[e]),
-                DATA_RATE=topology.channel_bandwidth,
-                latency=topology.intra_gpu_propagation_latency,
-                IO_latency=topology.worker_io_latency)
-        partitioned_job.set_dep_init_run_time(e, t)
+    if one_to_one:
+        o2o = np.concatenate(one_to_one)
+        same_worker = node_server[pg.src[o2o]] == node_server[pg.dst[o2o]]
+        # NB with one worker per server, same worker <=> same server
+        sizes = pg.size[o2o]
+        base = (topology.intra_gpu_propagation_latency
+                + 2 * topology.worker_io_latency)
+        times = np.where(same_worker | (sizes == 0), 0.0,
+                         base + sizes / topology.channel_bandwidth)
+        if np.isinf(times).any():
+            raise ValueError("infinite one-to-one run time")
+        partitioned_job.dep_init_run_time[o2o] = times
+        partitioned_job.dep_remaining[o2o] = times

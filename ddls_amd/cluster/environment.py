@@ -48,6 +48,8 @@ class RampClusterEnvironment:
 
         self.topology = build_topology(topology_config)
         self._populate_topology(self.topology, node_config)
+        # (msg, nodes, racks, cgs) -> collective time; valid per-topology
+        self.collective_time_cache = {}
 
         self.stopwatch = Stopwatch()
         self.reset_counter = 0
@@ -86,6 +88,10 @@ class RampClusterEnvironment:
         # dense channel index
         self.channel_ids = list(topology.channel_id_to_channel.keys())
         self.channel_id_to_index = {cid: i for i, cid in enumerate(self.channel_ids)}
+        # dense worker index -> server node index
+        self.worker_node = np.array(
+            [topology.worker_to_node[w.processor_id] for w in self.workers],
+            dtype=np.int64)
 
     @property
     def device_type(self) -> str:
@@ -354,31 +360,27 @@ class RampClusterEnvironment:
             "mean_mounted_worker_utilisation_frac": util,
         }, immutable=immutable)
 
-        # flow-size accounting (reference :881-888)
-        job.details["job_total_flow_size"] = 0.0
+        # flow-size accounting (reference :881-888): a dep counts as a flow
+        # iff it is cross-node with size > 0 (its init time is then non-zero/
+        # unset, exactly the reference's run_time != 0 condition)
         g = job.graph
-        for e in range(g.m):
-            rt = self._set_dep_init_run_time(job, e)
-            if rt != 0:
-                job.details["job_total_flow_size"] += float(g.size[e])
+        self._set_all_dep_init_run_times(job)
+        flow_mask = job.dep_cross_node & (g.size > 0)
+        job.details["job_total_flow_size"] = float(g.size[flow_mask].sum())
 
     # ------------------------------------------------------------------
-    def _set_dep_init_run_time(self, job: Job, dep_idx: int):
-        """Cluster-side dep init time (reference :542-560): 0 for same-node or
-        zero-size deps, else the priced init run time."""
+    def _set_all_dep_init_run_times(self, job: Job):
+        """Cluster-side dep init times, vectorised (reference :542-560): 0 for
+        same-node or zero-size deps, else the priced init run time."""
         g = job.graph
-        u, v = int(g.src[dep_idx]), int(g.dst[dep_idx])
-        job_idx = job.details["job_idx"]
-        src_w = self.job_op_to_worker[(job_idx, u)]
-        dst_w = self.job_op_to_worker[(job_idx, v)]
-        if self.topology.worker_to_node[src_w] == self.topology.worker_to_node[dst_w]:
-            rt = 0.0
-        elif g.size[dep_idx] == 0:
-            rt = 0.0
-        else:
-            rt = job.dep_init_run_time[dep_idx]
-        job.set_dep_init_run_time(dep_idx, rt)
-        return rt
+        if not hasattr(job, "dep_cross_node"):
+            src_nodes = self.worker_node[job.op_worker[g.src]]
+            dst_nodes = self.worker_node[job.op_worker[g.dst]]
+            job.dep_cross_node = src_nodes != dst_nodes
+        non_flow = (~job.dep_cross_node) | (g.size == 0)
+        init = np.where(non_flow, 0.0, job.dep_init_run_time)
+        job.dep_init_run_time = init
+        job.dep_remaining = init.copy()
 
     # ------------------------------------------------------------------
     # control-plane execution (reference :1285-1415)
@@ -419,38 +421,40 @@ class RampClusterEnvironment:
         self.jobs_running[job.details["job_idx"]] = job
         self.job_queue.remove(job)
         g = job.graph
-        # flow classification + init run times
-        job.dep_is_flow = np.zeros(g.m, dtype=bool)
+        # flow classification + init run times (vectorised)
         job.dep_channel_idx = np.full(g.m, -1, dtype=np.int64)
         job.dep_priority = np.zeros(g.m, dtype=np.int64)
-        job_idx = job.details["job_idx"]
-        for e in range(g.m):
-            rt = self._set_dep_init_run_time(job, e)
-            u, v = int(g.src[e]), int(g.dst[e])
-            src_node = self.topology.worker_to_node[self.job_op_to_worker[(job_idx, u)]]
-            dst_node = self.topology.worker_to_node[self.job_op_to_worker[(job_idx, v)]]
-            job.dep_is_flow[e] = (g.size[e] > 0) and (src_node != dst_node)
+        self._set_all_dep_init_run_times(job)
+        job.dep_is_flow = job.dep_cross_node & (g.size > 0)
 
     def _place_deps(self, dep_placement):
         for job_id, deps in dep_placement.action.items():
             job_idx = self.job_id_to_job_idx[job_id]
             job = self.jobs_running[job_idx]
+            mounted = []
             for dep_idx, channel_ids in deps.items():
                 for channel_id in channel_ids:
                     if channel_id is None:
                         continue
+                    mounted.append((dep_idx, channel_id))
+                    job.dep_channel_idx[dep_idx] = self.channel_id_to_index[channel_id]
+            if mounted:
+                # group by channel: one occupancy-rule check + one counted
+                # mount per channel instead of per dep
+                per_channel = defaultdict(int)
+                for dep_idx, channel_id in mounted:
+                    per_channel[channel_id] += 1
+                    job.reset_dep_remaining_run_time(dep_idx)
+                for channel_id, count in per_channel.items():
                     channel = self.topology.channel_id_to_channel[channel_id]
                     broken = check_if_ramp_dep_placement_rules_broken(channel, job)
                     if broken:
                         raise RuntimeError(
-                            f"dep placement for job {job_id} dep {dep_idx} on "
-                            f"{channel_id} breaks RAMP rules: {broken}")
-                    channel.mount(job, dep_idx)
+                            f"dep placement for job {job_id} on {channel_id} "
+                            f"breaks RAMP rules: {broken}")
+                    channel.mount(job, -1, count=count)
                     job.details["mounted_channels"].add(channel_id)
-                    self.num_mounted_deps += 1
-                    job.reset_dep_remaining_run_time(dep_idx)
-                    self.job_dep_to_channels[(job_idx, dep_idx)].add(channel_id)
-                    job.dep_channel_idx[dep_idx] = self.channel_id_to_index[channel_id]
+                self.num_mounted_deps += len(mounted)
             self.job_dep_placement[job_id] = deps
 
     def _schedule_ops(self, op_schedule):
@@ -470,14 +474,12 @@ class RampClusterEnvironment:
         for channel_id, job_to_deps in dep_schedule.action.items():
             if channel_id is None:
                 continue
-            channel = self.topology.channel_id_to_channel[channel_id]
             for job_id, dep_to_priority in job_to_deps.items():
                 job_idx = self.job_id_to_job_idx[job_id]
                 job = self.jobs_running.get(job_idx)
                 if job is None:
                     continue
                 for dep_idx, priority in dep_to_priority.items():
-                    channel.mounted_job_dep_to_priority[(job_idx, dep_idx)] = priority
                     job.dep_priority[dep_idx] = priority
 
     # ------------------------------------------------------------------
@@ -487,21 +489,25 @@ class RampClusterEnvironment:
         self.jobs_running.pop(job.details["job_idx"], None)
         job_idx = job.details["job_idx"]
         g = job.graph
-        for op_idx in range(g.n):
-            key = (job_idx, op_idx)
-            if key in self.job_op_to_worker:
-                worker_id = self.job_op_to_worker[key]
+        mounted_ops = (np.flatnonzero(job.op_worker >= 0)
+                       if hasattr(job, "op_worker") else range(g.n))
+        for op_idx in mounted_ops:
+            key = (job_idx, int(op_idx))
+            worker_id = self.job_op_to_worker.pop(key, None)
+            if worker_id is not None:
                 node = self.topology.worker_to_node[worker_id]
-                self.topology.node_workers[node][worker_id].unmount(job, op_idx)
+                self.topology.node_workers[node][worker_id].unmount(job, int(op_idx))
                 self.num_mounted_ops -= 1
-                del self.job_op_to_worker[key]
-        for dep_idx in range(g.m):
-            key = (job_idx, dep_idx)
-            if key in self.job_dep_to_channels:
-                for channel_id in self.job_dep_to_channels[key]:
-                    self.topology.channel_id_to_channel[channel_id].unmount(job, dep_idx)
-                    self.num_mounted_deps -= 1
-                del self.job_dep_to_channels[key]
+        if hasattr(job, "dep_channel_idx"):
+            mounted_deps = job.dep_channel_idx[job.dep_channel_idx >= 0]
+            if len(mounted_deps) > 0:
+                ch_idxs, counts = np.unique(mounted_deps, return_counts=True)
+                for ch_idx, count in zip(ch_idxs, counts):
+                    cid = self.channel_ids[int(ch_idx)]
+                    self.topology.channel_id_to_channel[cid].unmount(
+                        job, -1, count=int(count))
+                self.num_mounted_deps -= int(len(mounted_deps))
+                job.dep_channel_idx[:] = -1
         self.job_op_placement.pop(job.job_id, None)
         self.job_dep_placement.pop(job.job_id, None)
 

@@ -91,15 +91,20 @@ class Channel:
         return f"Channel_{self.channel_id}"
 
     def reset(self):
-        self.mounted_job_idx_to_deps = defaultdict(set)
-        self.mounted_job_dep_to_priority = {}
+        # job_idx -> number of mounted deps (the sim only ever needs the
+        # one-job-per-channel occupancy test; per-dep scheduling priorities
+        # live on the job's dense arrays)
+        self.mounted_job_idx_to_deps = {}
 
-    def mount(self, job, dep_idx: int):
-        self.mounted_job_idx_to_deps[job.details["job_idx"]].add(dep_idx)
-
-    def unmount(self, job, dep_idx: int):
+    def mount(self, job, dep_idx: int, count: int = 1):
         job_idx = job.details["job_idx"]
-        self.mounted_job_idx_to_deps[job_idx].discard(dep_idx)
-        self.mounted_job_dep_to_priority.pop((job_idx, dep_idx), None)
-        if len(self.mounted_job_idx_to_deps[job_idx]) == 0:
-            del self.mounted_job_idx_to_deps[job_idx]
+        self.mounted_job_idx_to_deps[job_idx] = (
+            self.mounted_job_idx_to_deps.get(job_idx, 0) + count)
+
+    def unmount(self, job, dep_idx: int, count: int = 1):
+        job_idx = job.details["job_idx"]
+        remaining = self.mounted_job_idx_to_deps.get(job_idx, 0) - count
+        if remaining <= 0:
+            self.mounted_job_idx_to_deps.pop(job_idx, None)
+        else:
+            self.mounted_job_idx_to_deps[job_idx] = remaining

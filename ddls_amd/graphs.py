@@ -37,7 +37,8 @@ class CompGraph:
     __slots__ = (
         "names", "name_to_idx", "compute_cost", "memory_cost", "pass_type",
         "counterpart", "src", "dst", "size", "n", "m",
-        "_out_csr", "_in_csr", "_true_parent_count", "_bidir_edge", "model",
+        "_out_csr", "_in_csr", "_true_parent_count", "_bidir_edge",
+        "_reverse_edge", "model",
     )
 
     def __init__(self,
@@ -67,6 +68,7 @@ class CompGraph:
         self._in_csr = None
         self._true_parent_count = None
         self._bidir_edge = None
+        self._reverse_edge = None
 
     def __deepcopy__(self, memo):
         # immutable: share across deep copies (jobs of one model share structure)
@@ -109,6 +111,17 @@ class CompGraph:
                 ((v, u) in pairs for u, v in zip(self.src.tolist(), self.dst.tolist())),
                 dtype=bool, count=self.m)
         return self._bidir_edge
+
+    @property
+    def reverse_edge(self) -> np.ndarray:
+        """int64[m]: index of edge (v,u) for edge e=(u,v), or -1."""
+        if self._reverse_edge is None:
+            lookup = {(int(u), int(v)): e
+                      for e, (u, v) in enumerate(zip(self.src, self.dst))}
+            self._reverse_edge = np.array(
+                [lookup.get((int(v), int(u)), -1)
+                 for u, v in zip(self.src, self.dst)], dtype=np.int64)
+        return self._reverse_edge
 
     @property
     def true_parent_count(self) -> np.ndarray:

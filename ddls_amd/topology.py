@@ -60,6 +60,17 @@ class Ramp:
                                  channel_bandwidth=self.channel_bandwidth)
                     self.channel_id_to_channel[ch.channel_id] = ch
 
+        # precomputed direct-link channel ids (full mesh -> every path is a
+        # single hop); avoids string formatting on the per-flow hot path
+        self.direct_cid: Dict[Tuple[int, int, int], str] = {}
+        for u in range(self.num_nodes):
+            for v in range(self.num_nodes):
+                if u == v:
+                    continue
+                for k in range(num_channels):
+                    self.direct_cid[(u, v, k)] = gen_channel_id(
+                        self.node_names[u], self.node_names[v], k)
+
         # populated by the cluster environment
         self.node_workers: List[dict] = [dict() for _ in range(self.num_nodes)]
         self.worker_to_node: Dict[str, int] = {}
