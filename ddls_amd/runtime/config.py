@@ -1,0 +1,140 @@
+"""YAML config groups with merging and dotted overrides (hydra-lite).
+
+Reference config system: hydra + OmegaConf groups under
+``scripts/ramp_job_partitioning_configs/`` merged in
+``train_rllib_from_config.py:46-64``, with ``_target_`` class instantiation
+and dotted-key overrides (wandb sweeps).  This rebuild implements the same
+surface with yaml + a small merger: a top-level config names its groups
+(``defaults``), each group file lives at ``<config_dir>/<group>/<name>.yaml``,
+and CLI overrides use ``a.b.c=value``.
+"""
+from __future__ import annotations
+
+import copy
+import os
+from typing import Any, Dict, List, Optional
+
+import yaml
+
+
+def load_yaml(path: str) -> dict:
+    with open(path) as f:
+        return yaml.safe_load(f) or {}
+
+
+def deep_merge(base: dict, override: dict) -> dict:
+    out = copy.deepcopy(base)
+    for k, v in override.items():
+        if isinstance(v, dict) and isinstance(out.get(k), dict):
+            out[k] = deep_merge(out[k], v)
+        else:
+            out[k] = copy.deepcopy(v)
+    return out
+
+
+def _parse_value(s: str) -> Any:
+    try:
+        return yaml.safe_load(s)
+    except yaml.YAMLError:
+        return s
+
+
+def apply_dotted_overrides(cfg: dict, overrides: List[str]) -> dict:
+    cfg = copy.deepcopy(cfg)
+    for ov in overrides:
+        if "=" not in ov:
+            raise ValueError(f"override '{ov}' must be key.path=value")
+        key, _, raw = ov.partition("=")
+        node = cfg
+        parts = key.split(".")
+        for p in parts[:-1]:
+            node = node.setdefault(p, {})
+        node[parts[-1]] = _parse_value(raw)
+    return cfg
+
+
+def load_config(config_path: str,
+                overrides: Optional[List[str]] = None) -> dict:
+    """Load a top-level config, resolving its ``defaults`` group list
+    (each entry ``{group: name}`` pulls ``<dir>/<group>/<name>.yaml`` into
+    ``cfg[group]``), then apply dotted overrides."""
+    config_dir = os.path.dirname(os.path.abspath(config_path))
+    cfg = load_yaml(config_path)
+    defaults = cfg.pop("defaults", [])
+    resolved: Dict[str, Any] = {}
+    for entry in defaults:
+        if isinstance(entry, dict):
+            for group, name in entry.items():
+                group_cfg = load_yaml(os.path.join(config_dir, group,
+                                                   f"{name}.yaml"))
+                resolved[group] = group_cfg
+        else:
+            resolved = deep_merge(resolved, load_yaml(
+                os.path.join(config_dir, f"{entry}.yaml")))
+    cfg = deep_merge(resolved, cfg)
+    if overrides:
+        cfg = apply_dotted_overrides(cfg, overrides)
+    return cfg
+
+
+# ---------------------------------------------------------------------------
+# factories
+# ---------------------------------------------------------------------------
+
+def build_env_from_config(cfg: dict):
+    """Build a RampJobPartitioningEnvironment from cfg['env_config']."""
+    from ..envs import RampJobPartitioningEnvironment
+    env_cfg = copy.deepcopy(cfg["env_config"])
+    jobs_config = env_cfg.get("jobs_config", {})
+    if jobs_config.get("path_to_files") in (None, "default", "synthetic"):
+        from ..workloads import ensure_default_set
+        jobs_config["path_to_files"] = ensure_default_set()
+    return RampJobPartitioningEnvironment(
+        topology_config=env_cfg["topology_config"],
+        node_config=env_cfg["node_config"],
+        jobs_config=jobs_config,
+        max_partitions_per_op=env_cfg.get("max_partitions_per_op"),
+        min_op_run_time_quantum=env_cfg.get("min_op_run_time_quantum", 0.01),
+        pad_obs_kwargs=env_cfg.get("pad_obs_kwargs"),
+        reward_function=env_cfg.get("reward_function",
+                                    "lookahead_job_completion_time"),
+        reward_function_kwargs=env_cfg.get("reward_function_kwargs"),
+        max_simulation_run_time=env_cfg.get("max_simulation_run_time"),
+        job_queue_capacity=env_cfg.get("job_queue_capacity", 10))
+
+
+def build_trainer_from_config(cfg: dict, device=None):
+    import torch
+
+    from ..models.gnn import GNNPolicy
+    from ..rl.ppo import PPOConfig, PPOTrainer
+    from ..rl.rollout import VectorEnv
+
+    algo = cfg.get("algo", {})
+    model_cfg = cfg.get("model", {})
+    loop_cfg = cfg.get("epoch_loop", {})
+    n_envs = loop_cfg.get("num_envs", 8)
+    seed = cfg.get("seed", 0)
+
+    def env_fn():
+        return build_env_from_config(cfg)
+
+    venv = VectorEnv([env_fn for _ in range(n_envs)], base_seed=seed)
+    torch.manual_seed(seed)
+    env0 = venv.envs[0]
+    policy = GNNPolicy(num_actions=env0.action_space.n,
+                       config=model_cfg.get("custom_model_config"))
+    ppo_cfg = PPOConfig(
+        lr=algo.get("lr", 2.785e-4),
+        gamma=algo.get("gamma", 0.997),
+        lambda_=algo.get("lambda", 1.0),
+        clip_param=algo.get("clip_param", 0.18),
+        entropy_coeff=algo.get("entropy_coeff", 3e-3),
+        kl_coeff=algo.get("kl_coeff", 0.01),
+        kl_target=algo.get("kl_target", 1e-3),
+        vf_clip_param=algo.get("vf_clip_param", 128.8),
+        grad_clip=algo.get("grad_clip", 1.5),
+        sgd_minibatch_size=algo.get("sgd_minibatch_size", 128),
+        train_batch_size=algo.get("train_batch_size", 4000),
+        num_sgd_iter=algo.get("num_sgd_iter", 50))
+    return PPOTrainer(venv, policy, ppo_cfg, device=device)

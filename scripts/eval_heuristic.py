@@ -1,0 +1,71 @@
+#!/usr/bin/env python3
+"""Evaluate heuristic baseline actors on the RAMP partitioning env.
+
+Replaces the reference's ``scripts/test_heuristic_from_config.py:33``.
+Optional cProfile via ``--profile`` (the reference's profiling hook).
+"""
+import argparse
+import json
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from ddls_amd.envs.actors import ACTORS
+from ddls_amd.runtime.config import build_env_from_config, load_config
+from ddls_amd.runtime.loops import EvalLoop
+from ddls_amd.utils import seed_everything
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--config", default=os.path.join(
+        os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        "configs", "heuristic_config.yaml"))
+    ap.add_argument("--profile", action="store_true")
+    ap.add_argument("overrides", nargs="*")
+    args = ap.parse_args()
+
+    cfg = load_config(args.config, overrides=args.overrides)
+    seed = cfg.get("seed", 1799)
+
+    def run_all():
+        results = {}
+        for actor_name in cfg.get("actors", list(ACTORS)):
+            seed_everything(seed)
+            env = build_env_from_config(cfg)
+            actor = ACTORS[actor_name]()
+            loop = EvalLoop(actor, env)
+            r = loop.run(seed=seed)
+            results[actor_name] = {
+                "blocking_rate": r["blocking_rate"],
+                "acceptance_rate": r["acceptance_rate"],
+                "mean_job_completion_time": r["mean_job_completion_time"],
+                "mean_job_completion_time_speedup":
+                    r["mean_job_completion_time_speedup"],
+                "episode_return": r["episode_return"],
+                "num_actor_steps": r["num_actor_steps"],
+                "run_time": r["run_time"],
+            }
+            print(f"{actor_name}: {json.dumps(results[actor_name])}", flush=True)
+        return results
+
+    if args.profile:
+        import cProfile
+        import pstats
+        pr = cProfile.Profile()
+        pr.enable()
+        results = run_all()
+        pr.disable()
+        pstats.Stats(pr).sort_stats("cumulative").print_stats(30)
+    else:
+        results = run_all()
+
+    out = cfg.get("experiment", {}).get("results_path")
+    if out:
+        with open(out, "w") as f:
+            json.dump(results, f, indent=2)
+
+
+if __name__ == "__main__":
+    main()
