@@ -61,8 +61,10 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=5)
     ap.add_argument("--warmup", type=int, default=2)
-    ap.add_argument("--envs-per-rank", type=int, default=8)
-    ap.add_argument("--rollout-steps-per-env", type=int, default=32)
+    ap.add_argument("--envs-per-rank", type=int, default=16)
+    ap.add_argument("--rollout-steps-per-env", type=int, default=16)
+    ap.add_argument("--env-workers", type=int, default=0,
+                    help="env worker processes per rank (0 = auto)")
     ap.add_argument("--num-sgd-iter", type=int, default=8)
     ap.add_argument("--sgd-minibatch-size", type=int, default=128)
     args = ap.parse_args()
@@ -71,7 +73,7 @@ def main():
     from ddls_amd.parallel import (get_rank, get_world_size,
                                    init_distributed_from_env, is_distributed)
     from ddls_amd.rl.ppo import PPOConfig, PPOTrainer
-    from ddls_amd.rl.rollout import VectorEnv
+    from ddls_amd.rl.subproc_env import SubprocVectorEnv
 
     rank = init_distributed_from_env()
     world_size = get_world_size()
@@ -83,8 +85,13 @@ def main():
     policy = GNNPolicy(num_actions=17)
 
     env_fn = build_env_fn()
-    venv = VectorEnv([env_fn for _ in range(args.envs_per_rank)],
-                     base_seed=1 + 100000 * rank)
+    n_workers = args.env_workers
+    if n_workers <= 0:
+        n_workers = max(1, min(args.envs_per_rank,
+                               (os.cpu_count() or 2) // max(world_size, 1) - 1))
+    venv = SubprocVectorEnv(env_fn, num_envs=args.envs_per_rank,
+                            num_workers=n_workers,
+                            base_seed=1 + 100000 * rank)
     per_step_env_steps = args.envs_per_rank * args.rollout_steps_per_env
     cfg = PPOConfig(train_batch_size=per_step_env_steps,
                     sgd_minibatch_size=args.sgd_minibatch_size,
@@ -114,6 +121,7 @@ def main():
         torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
         elapsed = float(t.item())
 
+    venv_close = getattr(venv, "close", None)
     total_env_steps = args.steps * per_step_env_steps * world_size
     value = total_env_steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
@@ -139,9 +147,12 @@ def main():
                 "parallelism": f"dp{world_size}",
                 "ramp": "4x4x2_32workers",
                 "envs_per_rank": args.envs_per_rank,
+                "env_workers": n_workers,
                 "num_sgd_iter": args.num_sgd_iter,
             },
         }))
+    if venv_close is not None:
+        venv_close()
 
 
 if __name__ == "__main__":

@@ -1,0 +1,139 @@
+"""Multiprocess vectorised envs.
+
+Replaces the reference's Ray rollout workers (RLlib ``num_workers: 8``,
+SURVEY.md K6/K8) with fork-based worker processes over pipes: each worker owns
+a shard of envs, steps them on request and ships back compact observations
+(a few KB each) — no object store, no RPC framework.
+"""
+from __future__ import annotations
+
+import multiprocessing as mp
+import os
+from typing import List, Optional
+
+import numpy as np
+
+from .rollout import CompactObs
+
+
+def _worker_loop(conn, env_fn_count: int, env_fn, base_seed: int):
+    envs = [env_fn() for _ in range(env_fn_count)]
+    episode_counters = [0] * len(envs)
+    episode_returns = np.zeros(len(envs))
+    episode_lens = np.zeros(len(envs), dtype=np.int64)
+    try:
+        while True:
+            cmd, payload = conn.recv()
+            if cmd == "reset":
+                obs = []
+                for i, env in enumerate(envs):
+                    o = env.reset(seed=base_seed + 1000 * i)
+                    obs.append(CompactObs.from_obs(o))
+                conn.send(obs)
+            elif cmd == "step":
+                actions = payload
+                obs_out, rewards, dones, stats_out = [], [], [], []
+                for i, env in enumerate(envs):
+                    o, r, done, _ = env.step(int(actions[i]))
+                    episode_returns[i] += r
+                    episode_lens[i] += 1
+                    if done:
+                        stats = dict(env.cluster.episode_stats)
+                        stats["episode_return"] = float(episode_returns[i])
+                        stats["episode_len"] = int(episode_lens[i])
+                        stats_out.append(stats)
+                        episode_returns[i] = 0.0
+                        episode_lens[i] = 0
+                        episode_counters[i] += 1
+                        o = env.reset(seed=base_seed + 1000 * i
+                                      + episode_counters[i])
+                    obs_out.append(CompactObs.from_obs(o))
+                    rewards.append(r)
+                    dones.append(done)
+                conn.send((obs_out, np.array(rewards), np.array(dones),
+                           stats_out))
+            elif cmd == "close":
+                conn.send("closed")
+                return
+            else:
+                raise ValueError(f"unknown command {cmd}")
+    except (KeyboardInterrupt, EOFError):
+        pass
+
+
+class SubprocVectorEnv:
+    """Same interface as rollout.VectorEnv, envs sharded over processes."""
+
+    def __init__(self, env_fn, num_envs: int, num_workers: Optional[int] = None,
+                 base_seed: int = 0):
+        if num_workers is None:
+            num_workers = min(num_envs, max(1, (os.cpu_count() or 2) - 1))
+        num_workers = min(num_workers, num_envs)
+        self.num_envs = num_envs
+        # shard envs as evenly as possible
+        base = num_envs // num_workers
+        rem = num_envs % num_workers
+        self.shards = [base + (1 if w < rem else 0) for w in range(num_workers)]
+        ctx = mp.get_context("fork")
+        self.conns, self.procs = [], []
+        offset = 0
+        for w, count in enumerate(self.shards):
+            parent, child = ctx.Pipe()
+            p = ctx.Process(target=_worker_loop,
+                            args=(child, count, env_fn,
+                                  base_seed + 1000000 * w),
+                            daemon=True)
+            p.start()
+            child.close()
+            self.conns.append(parent)
+            self.procs.append(p)
+            offset += count
+        self.obs: List[CompactObs] = []
+        self.completed_episode_stats: List[dict] = []
+
+    def __len__(self):
+        return self.num_envs
+
+    def reset(self):
+        for conn in self.conns:
+            conn.send(("reset", None))
+        self.obs = []
+        for conn in self.conns:
+            self.obs.extend(conn.recv())
+        return self.obs
+
+    def step(self, actions: np.ndarray):
+        start = 0
+        for conn, count in zip(self.conns, self.shards):
+            conn.send(("step", actions[start:start + count]))
+            start += count
+        obs, rewards, dones = [], [], []
+        for conn in self.conns:
+            o, r, d, stats = conn.recv()
+            obs.extend(o)
+            rewards.append(r)
+            dones.append(d)
+            self.completed_episode_stats.extend(stats)
+        self.obs = obs
+        return obs, np.concatenate(rewards), np.concatenate(dones)
+
+    def drain_episode_stats(self) -> List[dict]:
+        out = self.completed_episode_stats
+        self.completed_episode_stats = []
+        return out
+
+    def close(self):
+        for conn in self.conns:
+            try:
+                conn.send(("close", None))
+                conn.recv()
+            except (BrokenPipeError, EOFError):
+                pass
+        for p in self.procs:
+            p.join(timeout=5)
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass

@@ -105,3 +105,22 @@ def test_dp_all_reduce_gloo(tmp_path):
     for rank, (p, out) in enumerate(zip(procs, outs)):
         assert p.returncode == 0, out
         assert f"RANK{rank}_OK" in out
+
+
+def test_subproc_vector_env(tiny_model_files):
+    from ddls_amd.rl.subproc_env import SubprocVectorEnv
+
+    def fn():
+        return make_env(tiny_model_files, replication=50)
+
+    venv = SubprocVectorEnv(fn, num_envs=4, num_workers=2, base_seed=0)
+    try:
+        obs = venv.reset()
+        assert len(obs) == 4
+        for _ in range(3):
+            actions = np.array([int(o.action_mask.argmax()) for o in obs])
+            obs, rewards, dones = venv.step(actions)
+            assert len(obs) == 4 and rewards.shape == (4,)
+            assert np.isfinite(rewards).all()
+    finally:
+        venv.close()
