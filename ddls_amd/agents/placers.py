@@ -25,7 +25,9 @@ class RampFirstFitOpPlacer:
     def __init__(self, **kwargs):
         pass
 
-    def get(self, op_partition: OpPartition, cluster, verbose: bool = False) -> OpPlacement:
+    def get(self, op_partition: OpPartition, cluster, verbose: bool = False,
+            job_placement_shape=None) -> OpPlacement:
+        from .placement_utils import find_meta_block
         ramp_shape = cluster.topology.shape
         ramp_topology = dummy_ramp(ramp_shape, cluster)
 
@@ -44,10 +46,21 @@ class RampFirstFitOpPlacer:
             sequence, splits, op_server_info, parents = get_allocation_preamble(
                 og, mp_split_names, mp_splits)
 
-            # whole cluster as meta-block (reference :78-82)
-            servers = [cluster.topology.coords[i]
-                       for i in range(cluster.topology.num_nodes)]
-            meta_block_info = (servers, ramp_shape, (0, 0, 0))
+            if job_placement_shape is not None:
+                # shaping env: pack the job into a meta-block of the chosen
+                # shape (reference shaping-variant placer)
+                shape = job_placement_shape.action.get(job_id)
+                if shape is None:
+                    continue
+                meta_block_info = find_meta_block(ramp_topology, ramp_shape,
+                                                  tuple(shape))
+                if meta_block_info is None:
+                    continue
+            else:
+                # PAC-ML env: whole cluster as meta-block (reference :78-82)
+                servers = [cluster.topology.coords[i]
+                           for i in range(cluster.topology.num_nodes)]
+                meta_block_info = (servers, ramp_shape, (0, 0, 0))
 
             allocated = allocate(ramp_topology, ramp_shape, fwd_mem, num_fwd,
                                  sequence, splits, meta_block_info, parents,

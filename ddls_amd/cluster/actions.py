@@ -157,9 +157,22 @@ class DepPlacement:
                 self.channel_ids.update(channels)
 
 
+class JobPlacementShape:
+    """job_id -> (c, r, s) meta-block shape for the shaping env
+    (reference ``actions/job_placement_shape.py:1-21``)."""
+
+    def __init__(self, action: Dict):
+        self.action = dict(action)
+        self.job_ids = set(self.action.keys())
+
+    def __len__(self):
+        return len(self.action)
+
+
 class Action:
     """Bundle of the five sub-actions; a job survives only if EVERY sub-action
-    handled it (set intersection, reference ``action.py:36-52``)."""
+    handled it (set intersection, reference ``action.py:36-52``).  An optional
+    ``job_placement_shape`` joins the intersection (used by the shaping env)."""
 
     KEYS = ("op_partition", "op_placement", "op_schedule", "dep_placement",
             "dep_schedule")
@@ -169,7 +182,8 @@ class Action:
                  op_placement: Optional[OpPlacement] = None,
                  op_schedule: Optional[OpSchedule] = None,
                  dep_placement: Optional[DepPlacement] = None,
-                 dep_schedule: Optional[DepSchedule] = None):
+                 dep_schedule: Optional[DepSchedule] = None,
+                 job_placement_shape: Optional[JobPlacementShape] = None):
         self.actions = defaultdict(lambda: None)
         for key, act in zip(self.KEYS, (op_partition, op_placement, op_schedule,
                                         dep_placement, dep_schedule)):
@@ -180,6 +194,8 @@ class Action:
         if len(self.actions) > 0:
             self.job_ids = set.intersection(
                 *[set(a.job_ids) for a in self.actions.values()])
+            if job_placement_shape is not None:
+                self.job_ids &= set(job_placement_shape.job_ids)
             self.job_idxs = set(
                 op_partition.partitioned_jobs[jid].details["job_idx"]
                 for jid in self.job_ids)

@@ -173,3 +173,92 @@ def test_action_mask_shape_and_validity(tiny_model_files):
     # 1 and even actions <= 16 valid on the empty 32-worker 4x4x2 ramp
     for a in (1, 2, 4, 8, 16):
         assert mask[a]
+
+
+def test_placement_shaping_env(tiny_model_files):
+    """The shaping env variant: agent picks a (c,r,s) meta-block shape."""
+    from ddls_amd.envs.ramp_job_placement_shaping import (
+        RampJobPlacementShapingEnvironment)
+    import numpy as np
+    env = RampJobPlacementShapingEnvironment(
+        topology_config={"type": "ramp", "kwargs": {
+            "num_communication_groups": 4,
+            "num_racks_per_communication_group": 4,
+            "num_servers_per_rack": 2, "num_channels": 1,
+            "total_node_bandwidth": 1.6e12,
+            "intra_gpu_propagation_latency": 50e-9,
+            "worker_io_latency": 100e-9}},
+        node_config={"type_1": {"num_nodes": 32, "workers_config": [
+            {"num_workers": 1, "worker": "ddls_amd.devices.A100"}]}},
+        jobs_config={"path_to_files": tiny_model_files,
+                     "replication_factor": 3,
+                     "job_sampling_mode": "remove",
+                     "job_interarrival_time_dist": {
+                         "_target_": "ddls_amd.distributions.Fixed",
+                         "val": 1000},
+                     "max_acceptable_job_completion_time_frac_dist": {
+                         "_target_": "ddls_amd.distributions.Fixed", "val": 1.0},
+                     "num_training_steps": 5,
+                     "max_partitions_per_op_in_observation": 2},
+        op_partitioner_kwargs={"min_op_run_time_quantum": 0.01},
+        max_simulation_run_time=1e6)
+    obs = env.reset(seed=0)
+    assert len(obs["action_set"]) == 4 * 4 * 2 + 1
+    assert obs["action_mask"][0]
+    done, steps, completed = False, 0, 0
+    while not done and steps < 10:
+        valid = np.flatnonzero(obs["action_mask"])
+        a = int(valid[-1])
+        obs, r, done, _ = env.step(a)
+        steps += 1
+    stats = env.cluster.episode_stats
+    assert stats["num_jobs_completed"] + stats["num_jobs_blocked"] == 3
+
+
+def test_shaper_agents(tiny_model_files):
+    from ddls_amd.agents.shapers import (RampFirstFitJobPlacementShaper,
+                                         RampRandomJobPlacementShaper)
+    from ddls_amd.agents.partitioners import SipMlOpPartitioner
+    env = make_env(tiny_model_files, replication=1)
+    env.reset(seed=0)
+    op = SipMlOpPartitioner(min_op_run_time_quantum=0.01).get(
+        env.cluster, max_partitions_per_op=2)
+    for shaper in (RampFirstFitJobPlacementShaper(), RampRandomJobPlacementShaper()):
+        shape = shaper.get(op, env.cluster)
+        assert len(shape.action) == 1
+        c, r, s = next(iter(shape.action.values()))
+        assert c * r * s >= 2
+
+
+def test_legacy_cluster_env(tiny_model_files):
+    """Legacy dynamic per-tick sim: jobs run tick-by-tick, no comm model."""
+    from ddls_amd.agents.job_managers import RandomJobPlacer, SRPTJobScheduler
+    from ddls_amd.cluster.legacy_environment import ClusterEnvironment
+    from ddls_amd.utils import seed_everything
+    seed_everything(0)
+    env = ClusterEnvironment(
+        topology_config={"type": "torus", "kwargs": {
+            "x_dims": 4, "y_dims": 1, "z_dims": 1, "num_channels": 1}},
+        node_config={"type_1": {"num_nodes": 4, "workers_config": [
+            {"num_workers": 1, "worker": "ddls_amd.devices.A100"}]}})
+    env.reset(jobs_config={
+        "path_to_files": tiny_model_files,
+        "replication_factor": 2,
+        "job_sampling_mode": "remove",
+        "job_interarrival_time_dist": {
+            "_target_": "ddls_amd.distributions.Fixed", "val": 0.1},
+        "num_training_steps": 3}, seed=0)
+    placer, scheduler = RandomJobPlacer(), SRPTJobScheduler()
+    done, guard = False, 0
+    while not done and guard < 50:
+        placement = placer.get(env)
+        schedule = scheduler.get(placement, env)
+        _, _, _, done, _ = env.step({"job_placement": placement,
+                                     "job_schedule": schedule})
+        guard += 1
+    assert env.episode_stats["num_jobs_completed"] == 2
+    # per-training-step sequential lower bound: each job needs >= 3 steps of
+    # its critical path; with no comm overhead and enough workers, JCT >=
+    # num_training_steps * critical path
+    jcts = env.episode_stats["job_completion_time"]
+    assert all(j >= 3 * (0.02 + 0.03 + 0.06 + 0.04) - 1e-9 for j in jcts)
