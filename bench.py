@@ -20,7 +20,8 @@ import numpy as np
 import torch
 
 
-def build_env_fn(seed_offset: int = 0, device_type: str = "A100"):
+def build_env_fn(seed_offset: int = 0, device_type: str = "A100",
+                 lookahead_memo=None, init_details_memo=None):
     from ddls_amd.envs import RampJobPartitioningEnvironment
     from ddls_amd.workloads import ensure_default_set
     data = ensure_default_set()
@@ -29,6 +30,8 @@ def build_env_fn(seed_offset: int = 0, device_type: str = "A100"):
 
     def make():
         return RampJobPartitioningEnvironment(
+            lookahead_memo_preload=lookahead_memo,
+            init_details_memo_preload=init_details_memo,
             topology_config={"type": "ramp", "kwargs": {
                 "num_communication_groups": 4,
                 "num_racks_per_communication_group": 4,
@@ -67,6 +70,8 @@ def main():
                     help="env worker processes per rank (0 = auto)")
     ap.add_argument("--num-sgd-iter", type=int, default=8)
     ap.add_argument("--sgd-minibatch-size", type=int, default=128)
+    ap.add_argument("--no-precompute", action="store_true",
+                    help="skip the batched GPU lookahead memo precompute")
     args = ap.parse_args()
 
     from ddls_amd.models.gnn import GNNPolicy
@@ -85,6 +90,16 @@ def main():
     policy = GNNPolicy(num_actions=17)
 
     env_fn = build_env_fn()
+    if not args.no_precompute:
+        # precompute the whole (model x degree) lookahead memo table in one
+        # batched HIP kernel launch; forked env workers inherit the tables
+        from ddls_amd.cluster.batched_lookahead import precompute_lookahead_memos
+        scratch = env_fn()
+        scratch.reset(seed=0)
+        memo_l, memo_i = precompute_lookahead_memos(
+            scratch, device=device if use_cuda else "cpu")
+        env_fn = build_env_fn(lookahead_memo=memo_l, init_details_memo=memo_i)
+        del scratch
     n_workers = args.env_workers
     if n_workers <= 0:
         n_workers = max(1, min(args.envs_per_rank,

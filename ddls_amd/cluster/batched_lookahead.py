@@ -1,0 +1,267 @@
+"""Batched lookahead engine: pack many mounted jobs into SoA tensors and run
+their discrete-event lookaheads in ONE HIP kernel launch (one workgroup per
+env) — the GPU-vectorised env-step core of this rebuild.
+
+Primary use: precompute the whole (model x partition-degree) lookahead memo
+table at fleet start, so rollouts never pay a cold lookahead; the resulting
+dict feeds ``RampClusterEnvironment.reset(lookahead_memo_preload=...)`` and is
+inherited by forked env workers.
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+
+from ..graphs import FWD
+from .lookahead import run_lookahead_ticks
+
+
+def mount_job_for_lookahead(cluster, job, placement: Dict[str, str],
+                            op_schedule, dep_placement, dep_schedule) -> bool:
+    """Set the job's lookahead arrays from pipeline outputs WITHOUT mutating
+    cluster devices (no worker/channel mounts) — exactly the array state that
+    ``_place_ops``/``_register_running_job``/``_schedule_*``/``_place_deps``
+    would produce."""
+    g = job.graph
+    job.op_worker = np.full(g.n, -1, dtype=np.int64)
+    job.op_priority = np.zeros(g.n, dtype=np.int64)
+    for op_name, worker_id in placement.items():
+        op_idx = g.name_to_idx[op_name]
+        worker = cluster.workers[cluster.worker_id_to_index[worker_id]]
+        job.reset_op_remaining_run_time(op_idx, device_type=worker.device_type)
+        job.op_worker[op_idx] = cluster.worker_id_to_index[worker_id]
+    if (job.op_worker < 0).any():
+        return False
+
+    job_id = job.job_id
+    for worker_id, job_to_ops in op_schedule.action.items():
+        for op_name, priority in job_to_ops.get(job_id, {}).items():
+            job.op_priority[g.name_to_idx[op_name]] = priority
+
+    job.dep_channel_idx = np.full(g.m, -1, dtype=np.int64)
+    job.dep_priority = np.zeros(g.m, dtype=np.int64)
+    deps = dep_placement.action.get(job_id, {})
+    if len(deps) != g.m:
+        return False  # dep placement dropped the job
+    for dep_idx, channels in deps.items():
+        for cid in channels:
+            if cid is not None:
+                job.dep_channel_idx[dep_idx] = cluster.channel_id_to_index[cid]
+    for channel_id, job_to_deps in dep_schedule.action.items():
+        if channel_id is None:
+            continue
+        for dep_idx, priority in job_to_deps.get(job_id, {}).items():
+            job.dep_priority[dep_idx] = priority
+
+    cluster._set_all_dep_init_run_times(job)
+    job.dep_is_flow = job.dep_cross_node & (g.size > 0)
+    return True
+
+
+def _pack(jobs: List, device) -> Tuple:
+    import torch
+    descs = []
+    op_rem, op_w, op_p, tpc, indptrs, edges_all = [], [], [], [], [], []
+    dep_rem, dep_flow, dep_ch, dep_p, dep_dst = [], [], [], [], []
+    op_ready, op_done, dep_ready, dep_done, pdone = [], [], [], [], []
+    op_off = dep_off = csr_off = indptr_off = worker_off = channel_off = 0
+    for job in jobs:
+        g = job.graph
+        # local worker / channel remap
+        workers = np.unique(job.op_worker)
+        w_remap = {int(w): i for i, w in enumerate(workers)}
+        chans = np.unique(job.dep_channel_idx[job.dep_channel_idx >= 0])
+        c_remap = {int(c): i for i, c in enumerate(chans)}
+        n_workers = len(workers)
+        n_channels = max(1, len(chans))
+
+        indptr, order = g.out_csr
+        indptrs.append(indptr.astype(np.int64))
+        edges_all.append(order.astype(np.int32))
+
+        op_rem.append(job.op_remaining.astype(np.float64))
+        op_w.append(np.array([w_remap[int(w)] for w in job.op_worker],
+                             dtype=np.int32))
+        op_p.append(job.op_priority.astype(np.int32))
+        tpc.append(g.true_parent_count.astype(np.int32))
+        pdone.append(job.parent_deps_completed.astype(np.int32))
+        op_ready.append(job.ops_ready.astype(np.uint8))
+        op_done.append(job.ops_completed.astype(np.uint8))
+
+        dep_rem.append(job.dep_remaining.astype(np.float64))
+        dep_flow.append(job.dep_is_flow.astype(np.uint8))
+        dep_ch.append(np.array([c_remap.get(int(c), 0)
+                                for c in job.dep_channel_idx], dtype=np.int32))
+        dep_p.append(job.dep_priority.astype(np.int32))
+        dep_dst.append(g.dst.astype(np.int32))
+        dep_ready.append(job.deps_ready.astype(np.uint8))
+        dep_done.append(job.deps_completed.astype(np.uint8))
+
+        max_ticks = 8 * (g.n + g.m) + 64
+        descs.append([op_off, dep_off, csr_off, indptr_off, worker_off,
+                      channel_off, g.n, g.m, n_workers, n_channels, max_ticks])
+        op_off += g.n
+        dep_off += g.m
+        csr_off += g.m
+        indptr_off += g.n + 1
+        worker_off += n_workers
+        channel_off += n_channels
+
+    t = lambda arrs, dt: torch.as_tensor(np.concatenate(arrs), device=device,
+                                         dtype=dt)
+    return (
+        torch.as_tensor(np.array(descs, dtype=np.int64), device=device),
+        t(op_rem, torch.float64), t(op_w, torch.int32), t(op_p, torch.int32),
+        t(indptrs, torch.int64), t(edges_all, torch.int32),
+        t(tpc, torch.int32), t(pdone, torch.int32),
+        t(op_ready, torch.uint8), t(op_done, torch.uint8),
+        t(dep_rem, torch.float64), t(dep_flow, torch.uint8),
+        t(dep_ch, torch.int32), t(dep_p, torch.int32), t(dep_dst, torch.int32),
+        t(dep_ready, torch.uint8), t(dep_done, torch.uint8),
+        torch.zeros(worker_off, dtype=torch.int64, device=device),
+        torch.zeros(channel_off, dtype=torch.int64, device=device),
+    )
+
+
+def run_lookahead_batch(jobs: List, device=None,
+                        force_cpu: bool = False) -> List[Tuple]:
+    """Run one-training-step lookaheads for all jobs; returns per job
+    (t, comp_oh, comm_oh, active_time_sum).  GPU path = one HIP kernel launch;
+    CPU fallback = the reference tick loop per job."""
+    import torch
+    use_gpu = (not force_cpu) and device is not None and \
+        torch.cuda.is_available() and str(device) != "cpu"
+    if use_gpu:
+        from .. import ops as hip_ops
+        ext = hip_ops.get_extension(required=True)
+        packed = _pack(jobs, device)
+        outs = ext.lookahead_batch(*packed)
+        t_t, comp_t, comm_t, active_t, ticks_t, status_t = [o.cpu() for o in outs]
+        results = []
+        for b, job in enumerate(jobs):
+            if int(status_t[b]) != 0:
+                # fall back to CPU for this env (tick budget / anomaly)
+                t, comp, comm, tick_map = run_lookahead_ticks(job)
+                from .lookahead import active_time_sum
+                results.append((t, comp, comm, active_time_sum(tick_map)))
+            else:
+                results.append((float(t_t[b]), float(comp_t[b]),
+                                float(comm_t[b]), float(active_t[b])))
+        return results
+    results = []
+    for job in jobs:
+        t, comp, comm, tick_map = run_lookahead_ticks(job)
+        from .lookahead import active_time_sum
+        results.append((t, comp, comm, active_time_sum(tick_map)))
+    return results
+
+
+# ---------------------------------------------------------------------------
+# memo precomputation
+# ---------------------------------------------------------------------------
+
+def precompute_lookahead_memos(env, device=None, degrees: Optional[List[int]] = None,
+                               verbose: bool = False):
+    """For every (model, partition-degree) pair, run partition -> placement ->
+    pricing on the (empty) cluster and batch all lookaheads into one kernel
+    launch.  Returns (lookahead_memo, init_details_memo) dicts for
+    ``RampClusterEnvironment.reset`` preloading.
+
+    Valid under the reference's own memo assumption
+    (ramp_cluster_environment.py:271,277): lookahead results depend only on
+    (model, max partition degree).
+    """
+    from ..agents.partitioners import sip_ml_num_partitions
+    from .actions import OpPartition
+
+    cluster = env.cluster
+    if degrees is None:
+        degrees = [1] + [d for d in range(2, env.max_partitions_per_op + 1, 2)]
+
+    # one representative job per model from the generator pool
+    model_to_job = {}
+    for j in cluster.jobs_generator.job_sampler.sample_pool:
+        model_to_job.setdefault(j.details["model"], j)
+
+    lookahead_memo: Dict = {}
+    init_memo: Dict = {}
+    prepared: List = []
+    keys: List[Tuple] = []
+    device_type = cluster.device_type
+
+    saved_queue = dict(cluster.job_queue.jobs)
+    saved_idx_maps = (dict(cluster.job_id_to_job_idx),
+                      dict(cluster.job_idx_to_job_id))
+    fake_idx = 10_000_000
+    try:
+        for model, proto in model_to_job.items():
+            import copy as _copy
+            job = _copy.deepcopy(proto)
+            job.details["job_idx"] = fake_idx
+            job.original_job.details["job_idx"] = fake_idx
+            cluster.job_queue.jobs.clear()
+            cluster.job_queue.jobs[job.job_id] = job
+            cluster.job_id_to_job_idx[job.job_id] = fake_idx
+            cluster.job_idx_to_job_id[fake_idx] = job.job_id
+            fake_idx += 1
+            g = job.graph
+            cc = g.compute_cost[device_type]
+            seen_degrees = set()
+            for action in degrees:
+                partition_action = {}
+                for i in range(g.n):
+                    if g.pass_type[i] != FWD:
+                        continue
+                    num = sip_ml_num_partitions(
+                        float(cc[i]), env.min_op_run_time_quantum,
+                        max_partitions_per_op=max(action, 1))
+                    partition_action[g.names[i]] = num
+                    partition_action[g.names[int(g.counterpart[i])]] = num
+                op_partition = OpPartition({job.job_id: partition_action},
+                                           cluster=cluster)
+                degree = op_partition.job_id_to_max_partition_degree[job.job_id]
+                if degree in seen_degrees:
+                    continue
+                seen_degrees.add(degree)
+                op_placement = env.op_placer.get(op_partition=op_partition,
+                                                 cluster=cluster)
+                if job.job_id not in op_placement.action:
+                    continue  # no valid placement for this degree
+                op_schedule = env.op_scheduler.get(
+                    op_partition=op_partition, op_placement=op_placement,
+                    cluster=cluster)
+                dep_placement = env.dep_placer.get(
+                    op_partition=op_partition, op_placement=op_placement,
+                    cluster=cluster)
+                dep_schedule = env.dep_scheduler.get(
+                    op_partition=op_partition, dep_placement=dep_placement,
+                    cluster=cluster)
+                pjob = op_partition.partitioned_jobs[job.job_id]
+                ok = mount_job_for_lookahead(
+                    cluster, pjob, op_placement.action[job.job_id],
+                    op_schedule, dep_placement, dep_schedule)
+                if not ok:
+                    continue
+                prepared.append(pjob)
+                keys.append((model, degree))
+                entry = cluster.job_model_to_max_num_partitions_to_init_details[
+                    model][degree]
+                init_memo[(model, degree)] = {"graph": entry["graph"],
+                                              "immutable": entry["immutable"]}
+    finally:
+        cluster.job_queue.jobs.clear()
+        cluster.job_queue.jobs.update(saved_queue)
+        cluster.job_id_to_job_idx, cluster.job_idx_to_job_id = saved_idx_maps
+
+    results = run_lookahead_batch(prepared, device=device)
+    for (model, degree), pjob, (t, comp, comm, active_sum) in zip(
+            keys, prepared, results):
+        steps = pjob.num_training_steps
+        lookahead_memo[(model, degree)] = (
+            t * steps, comm * steps, comp * steps,
+            {"active_time_sum": active_sum})
+        if verbose:
+            print(f"lookahead memo {model} degree {degree}: "
+                  f"jct={t * steps:.4f}")
+    return lookahead_memo, init_memo

@@ -104,6 +104,8 @@ class RampClusterEnvironment:
               max_simulation_run_time: Union[int, float] = float("inf"),
               job_queue_capacity: int = 10,
               seed: Optional[int] = None,
+              lookahead_memo_preload: Optional[dict] = None,
+              init_details_memo_preload: Optional[dict] = None,
               verbose: bool = False):
         self.reset_counter += 1
         self.seed = seed
@@ -148,6 +150,14 @@ class RampClusterEnvironment:
         # addressing keyed (model_id, partition_degree) — SURVEY.md K4)
         self.job_model_to_max_num_partitions_to_init_details = defaultdict(dict)
         self.job_model_to_max_num_partitions_to_lookahead = defaultdict(dict)
+        # optionally preload memo tables (e.g. precomputed in one batched HIP
+        # lookahead launch at fleet start — ddls_amd.cluster.batched_lookahead)
+        if init_details_memo_preload:
+            for (model, degree), entry in init_details_memo_preload.items():
+                self.job_model_to_max_num_partitions_to_init_details[model][degree] = dict(entry)
+        if lookahead_memo_preload:
+            for (model, degree), entry in lookahead_memo_preload.items():
+                self.job_model_to_max_num_partitions_to_lookahead[model][degree] = entry
 
         self.time_next_job_to_arrive = 0
         self.job_queue.add(self._get_next_job())
@@ -205,115 +215,17 @@ class RampClusterEnvironment:
     # lookahead simulation of one training step (reference :379-467)
     # ------------------------------------------------------------------
     def _run_lookahead(self, job_id, verbose: bool = False):
+        from .lookahead import run_lookahead_ticks
         job_idx = self.job_id_to_job_idx[job_id]
         job = self.jobs_running[job_idx]
-        g = job.graph
-
-        op_worker = job.op_worker            # dense worker index per op
-        op_priority = job.op_priority
-        dep_is_flow = job.dep_is_flow
-        dep_channel = job.dep_channel_idx
-        dep_priority = job.dep_priority
-
-        t = 0.0
-        tick_counter = 1
-        tick_map = {}
-        while True:
-            # 1. highest-priority ready op per worker; shortest remaining time
-            ready_ops = np.flatnonzero(job.ops_ready)
-            if len(ready_ops) > 0:
-                w = op_worker[ready_ops]
-                p = op_priority[ready_ops]
-                order = np.lexsort((p, w))
-                rs, ws = ready_ops[order], w[order]
-                last_of_group = np.flatnonzero(
-                    np.r_[ws[1:] != ws[:-1], True])
-                priority_ops = rs[last_of_group]
-                shortest_op = float(job.op_remaining[priority_ops].min())
-            else:
-                priority_ops = np.empty(0, dtype=np.int64)
-                shortest_op = float("inf")
-
-            # 2. ready deps; non-flow deps tick with 0 comm time
-            ready_deps = np.flatnonzero(job.deps_ready)
-            non_flow = ready_deps[~dep_is_flow[ready_deps]]
-            if len(non_flow) == 0:
-                ready_flows = ready_deps
-                if len(ready_flows) > 0:
-                    # per-channel highest-priority flow (contention resolution
-                    # is vacuous on the full mesh: one channel per dep)
-                    ch = dep_channel[ready_flows]
-                    pr = dep_priority[ready_flows]
-                    order = np.lexsort((pr, ch))
-                    fs, cs = ready_flows[order], ch[order]
-                    last = np.flatnonzero(np.r_[cs[1:] != cs[:-1], True])
-                    prio_flows = fs[last]
-                    shortest_comm = float(job.dep_remaining[prio_flows].min())
-                else:
-                    shortest_comm = float("inf")
-            else:
-                shortest_comm = 0.0
-
-            tick = min(shortest_op, shortest_comm)
-            if math.isinf(tick):
-                raise RuntimeError("infinite lookahead tick: deadlocked job graph")
-
-            # snapshot ready deps BEFORE op ticking (reference :429)
-            deps_to_tick = non_flow if len(non_flow) > 0 else ready_deps
-
-            # 3a. tick priority ops
-            ticked_ops = len(priority_ops) > 0
-            completed_ops = []
-            if ticked_ops:
-                rem = job.op_remaining[priority_ops]
-                rem = rem - np.minimum(tick, rem)
-                job.op_remaining[priority_ops] = rem
-                completed_ops = priority_ops[rem == 0.0]
-                for o in completed_ops:
-                    job.ops_completed[o] = True
-                    job.ops_ready[o] = False
-                    job.num_ops_completed += 1
-                    job.deps_ready[g.out_edges_of(int(o))] = True
-
-            # 3b. tick deps (non-flows, or ALL ready flows in parallel —
-            # reference TEMP HACK :756-775)
-            ticked_flows = False
-            if len(deps_to_tick) > 0:
-                if len(non_flow) == 0:
-                    ticked_flows = True
-                rem = job.dep_remaining[deps_to_tick]
-                rem = rem - np.minimum(tick, rem)
-                job.dep_remaining[deps_to_tick] = rem
-                completed_deps = deps_to_tick[rem == 0.0]
-                for e in completed_deps:
-                    e = int(e)
-                    if job.deps_completed[e]:
-                        continue
-                    job.deps_completed[e] = True
-                    job.deps_ready[e] = False
-                    job.num_deps_completed += 1
-                    child = int(g.dst[e])
-                    job.parent_deps_completed[child] += 1
-                    if job.parent_deps_completed[child] == g.true_parent_count[child]:
-                        job.ops_ready[child] = True
-
-            # 4. overhead accounting (reference :777-791)
-            if ticked_ops:
-                job.details["computation_overhead_time"] += tick
-            if ticked_flows:
-                job.details["communication_overhead_time"] += tick
-
-            tick_map[tick_counter] = [int(len(priority_ops)), tick]
-            t += tick
-
-            if (job.num_ops_completed == g.n
-                    and job.num_deps_completed == g.m):
-                lookahead_jct = t * job.num_training_steps
-                comm_oh = job.details["communication_overhead_time"] * job.num_training_steps
-                comp_oh = job.details["computation_overhead_time"] * job.num_training_steps
-                return job, lookahead_jct, comm_oh, comp_oh, tick_map
-
-            tick_counter += 1
+        t, comp_oh, comm_oh, tick_map = run_lookahead_ticks(job)
+        job.details["computation_overhead_time"] += comp_oh
+        job.details["communication_overhead_time"] += comm_oh
+        lookahead_jct = t * job.num_training_steps
+        return (job, lookahead_jct,
+                comm_oh * job.num_training_steps,
+                comp_oh * job.num_training_steps,
+                tick_map)
 
     def _perform_lookahead_job_completion_time(self, action: Action,
                                                verbose: bool = False):
@@ -341,10 +253,9 @@ class RampClusterEnvironment:
             self._remove_job_from_cluster(job)
             return
 
+        from .lookahead import active_time_sum
         n_mounted = len(job.details["mounted_workers"])
-        util = 0.0
-        for num_active, tick_size in tick_map.values():
-            util += (num_active / n_mounted) * (tick_size / lookahead_jct)
+        util = active_time_sum(tick_map) / (n_mounted * lookahead_jct)
 
         model = job.details["model"]
         degree = self.op_partition.job_id_to_max_partition_degree[job.job_id]

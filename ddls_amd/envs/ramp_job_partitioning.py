@@ -46,7 +46,11 @@ class RampJobPartitioningEnvironment:
                  save_cluster_data: bool = False,
                  save_freq: int = 1,
                  use_sqlite_database: bool = False,
-                 apply_action_mask: bool = True):
+                 apply_action_mask: bool = True,
+                 lookahead_memo_preload: Optional[dict] = None,
+                 init_details_memo_preload: Optional[dict] = None):
+        self.lookahead_memo_preload = lookahead_memo_preload
+        self.init_details_memo_preload = init_details_memo_preload
         self.topology_config = topology_config
         self.node_config = node_config
         self.jobs_config = jobs_config
@@ -105,6 +109,8 @@ class RampJobPartitioningEnvironment:
                            max_simulation_run_time=self.max_simulation_run_time,
                            job_queue_capacity=self.job_queue_capacity,
                            seed=seed,
+                           lookahead_memo_preload=self.lookahead_memo_preload,
+                           init_details_memo_preload=self.init_details_memo_preload,
                            verbose=verbose)
         self.obs = self.observation_function.reset(self)
         self.observation_space = self.observation_function.observation_space

@@ -19,7 +19,9 @@ _EXT_ERR: Optional[Exception] = None
 
 _HERE = os.path.dirname(os.path.abspath(__file__))
 _BUILD_DIR = os.path.join(_HERE, "_build")
-_SOURCES = [os.path.join(_HERE, "hip", "meanpool.hip")]
+_SOURCES = [os.path.join(_HERE, "hip", "meanpool.hip"),
+            os.path.join(_HERE, "hip", "lookahead.hip"),
+            os.path.join(_HERE, "hip", "bindings.hip")]
 
 
 def build_extensions(verbose: bool = False):

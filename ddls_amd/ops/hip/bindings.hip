@@ -1,0 +1,30 @@
+// Single pybind11 module for all ddls_amd HIP kernels.
+#include <torch/extension.h>
+#include <vector>
+
+torch::Tensor row_mlp(torch::Tensor x, torch::Tensor ln_g, torch::Tensor ln_b,
+                      torch::Tensor W, torch::Tensor b);
+torch::Tensor message_reduce(torch::Tensor hn, torch::Tensor he,
+                             torch::Tensor src, torch::Tensor edge_order,
+                             torch::Tensor indptr, torch::Tensor ln_g,
+                             torch::Tensor ln_b, torch::Tensor Wr,
+                             torch::Tensor br);
+torch::Tensor segment_mean(torch::Tensor x, torch::Tensor node_ptr, int64_t G);
+std::vector<torch::Tensor> lookahead_batch(
+    torch::Tensor descs, torch::Tensor op_remaining, torch::Tensor op_worker,
+    torch::Tensor op_priority, torch::Tensor out_indptr, torch::Tensor out_edges,
+    torch::Tensor true_parent_count, torch::Tensor parent_done,
+    torch::Tensor op_ready, torch::Tensor op_completed,
+    torch::Tensor dep_remaining, torch::Tensor dep_is_flow,
+    torch::Tensor dep_channel, torch::Tensor dep_priority, torch::Tensor dep_dst,
+    torch::Tensor dep_ready, torch::Tensor dep_completed,
+    torch::Tensor worker_best, torch::Tensor channel_best);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+    m.def("row_mlp", &row_mlp, "fused LN+Linear+ReLU over rows");
+    m.def("message_reduce", &message_reduce,
+          "fused gather + LN + GEMV + ReLU + segment-mean message passing");
+    m.def("segment_mean", &segment_mean, "per-graph mean of node embeddings");
+    m.def("lookahead_batch", &lookahead_batch,
+          "batched RAMP lookahead discrete-event simulation");
+}

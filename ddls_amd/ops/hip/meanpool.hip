@@ -223,9 +223,4 @@ torch::Tensor segment_mean(torch::Tensor x, torch::Tensor node_ptr, int64_t G) {
     return out;
 }
 
-PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-    m.def("row_mlp", &row_mlp, "fused LN+Linear+ReLU over rows");
-    m.def("message_reduce", &message_reduce,
-          "fused gather + LN + Linear + ReLU + segment mean message passing");
-    m.def("segment_mean", &segment_mean, "per-graph mean of node embeddings");
-}
+// bindings live in bindings.hip
