@@ -677,6 +677,68 @@ class RampClusterEnvironment:
             if isinstance(es[metric], list):
                 es[metric] = float(np.mean(es[metric])) if len(es[metric]) > 0 else 0
 
+    # ---- metric vocabulary (reference :1181-1280) ----
+    @staticmethod
+    def episode_metrics():
+        return {
+            "episode_start_time", "episode_end_time", "episode_time",
+            "num_jobs_arrived", "num_jobs_completed", "num_jobs_blocked",
+            "compute_info_processed", "dep_info_processed",
+            "flow_info_processed", "cluster_info_processed",
+            "demand_compute_info_processed", "demand_dep_info_processed",
+            "demand_total_info_processed",
+            "mean_compute_throughput", "mean_dep_throughput",
+            "mean_cluster_throughput", "mean_load_rate", "blocking_rate",
+            "acceptance_rate", "mean_flow_throughput",
+            "mean_demand_compute_throughput", "mean_demand_dep_throughput",
+            "mean_demand_total_throughput", "mean_compute_overhead_frac",
+            "mean_communication_overhead_frac", "mean_num_jobs_running",
+            "mean_num_mounted_workers", "mean_mounted_worker_utilisation_frac",
+            "mean_cluster_worker_utilisation_frac",
+            "return", "episode_reward", "run_time", "epoch_counter",
+            "episode_counter", "actor_step_counter",
+        }
+
+    @staticmethod
+    def step_metrics():
+        return {"mean_num_mounted_workers", "mean_num_mounted_channels"}
+
+    @staticmethod
+    def episode_completion_metrics():
+        return {
+            "job_completion_time", "job_communication_overhead_time",
+            "job_computation_overhead_time", "jobs_completed_num_nodes",
+            "jobs_completed_num_edges",
+            "jobs_completed_total_operation_memory_cost",
+            "jobs_completed_total_dependency_size",
+            "job_completion_time_speedup", "jobs_completed_max_partitions_per_op",
+            "jobs_completed_job_sequential_completion_time",
+            "jobs_completed_max_acceptable_job_completion_time_frac",
+            "jobs_completed_max_acceptable_job_completion_time",
+            "jobs_completed_num_mounted_workers",
+            "jobs_completed_num_mounted_channels",
+            "jobs_completed_mean_mounted_worker_utilisation_frac",
+            "jobs_completed_original_demand_num_nodes",
+            "jobs_completed_original_demand_num_edges",
+            "jobs_completed_original_demand_total_operation_memory_cost",
+            "jobs_completed_original_demand_total_dependency_size",
+        }
+
+    @staticmethod
+    def episode_blocked_metrics():
+        return {
+            "jobs_blocked_num_nodes", "jobs_blocked_num_edges",
+            "jobs_blocked_total_operation_memory_cost",
+            "jobs_blocked_total_dependency_size",
+            "jobs_blocked_job_sequential_completion_time",
+            "jobs_blocked_max_acceptable_job_completion_time_frac",
+            "jobs_blocked_max_acceptable_job_completion_time",
+            "jobs_blocked_original_demand_num_nodes",
+            "jobs_blocked_original_demand_num_edges",
+            "jobs_blocked_original_demand_total_operation_memory_cost",
+            "jobs_blocked_original_demand_total_dependency_size",
+        }
+
     def is_done(self, verbose: bool = False) -> bool:
         if self.max_simulation_run_time is not None:
             if self.stopwatch.time() >= self.max_simulation_run_time:

@@ -306,6 +306,88 @@ def load_pipedream_graph(path: str, processor_type_profiled: str,
     return mirrored_graph_from_pipedream(nodes, edges, processor_type_profiled, model=model)
 
 
+# ---------------------------------------------------------------------------
+# REGAL CostGraphDef .pbtxt reader (reference ddls/utils.py:110-268)
+# ---------------------------------------------------------------------------
+
+def parse_pbtxt_nodes(path: str) -> List[dict]:
+    """Parse a REGAL-style CostGraphDef text proto into node dicts with
+    id / input_info (preceding node ids) / output_info (sizes) /
+    control_input / compute_cost."""
+    nodes: List[dict] = []
+    node = None
+    with open(path) as f:
+        for raw in f:
+            line = raw.replace(" ", "").replace("\n", "")
+            if line == "node{":
+                if node is not None:
+                    nodes.append(node)
+                node = {"input_info": [], "output_info": [],
+                        "control_input": [], "compute_cost": 0}
+            elif line == "}" or line == "":
+                continue
+            elif line.startswith("preceding_node"):
+                node["input_info"].append(int(line.split(":", 1)[1]))
+            elif line.startswith("preceding_port") or line.startswith("alias_input_port"):
+                continue
+            elif line.startswith("input_info") or line.startswith("output_info"):
+                continue
+            elif line.startswith("control_input"):
+                node["control_input"].append(int(line.split(":", 1)[1]))
+            elif line.startswith("compute_cost"):
+                node["compute_cost"] = int(line.split(":", 1)[1])
+            elif line.startswith("size"):
+                node["output_info"].append(int(line.split(":", 1)[1]))
+            elif line.startswith("id"):
+                node["id"] = int(line.split(":", 1)[1])
+            elif line.startswith("name"):
+                if "_SOURCE" in line:
+                    node["id"] = 0
+            else:
+                raise ValueError(f"unrecognised pbtxt line {line!r}")
+    if node is not None:
+        nodes.append(node)
+    return nodes
+
+
+def load_pbtxt_graph(path: str, processor_type_profiled: str,
+                     model: Optional[str] = None) -> CompGraph:
+    """Build a flat CompGraph from a REGAL CostGraphDef .pbtxt.  Dependency
+    sizes are randomly sampled from the parent's output_info sizes (the
+    reference's documented hack for DeepMind's ambiguous port mapping,
+    ``utils.py:171-183``).  Nodes carry no fwd/bwd mirroring (legacy generic
+    cluster workloads)."""
+    import random as _random
+    nodes = parse_pbtxt_nodes(path)
+    ids = [n.get("id", i) for i, n in enumerate(nodes)]
+    id_to_row = {nid: i for i, nid in enumerate(ids)}
+    names = [str(nid) for nid in ids]
+    comp = np.array([float(n.get("compute_cost", 0)) for n in nodes])
+    mem = np.zeros(len(nodes))
+    src, dst, size = [], [], []
+    for n, nid in zip(nodes, ids):
+        for parent in n["input_info"]:
+            out_sizes = nodes[id_to_row[parent]]["output_info"]
+            s = float(_random.choice(out_sizes)) if out_sizes else 0.0
+            src.append(id_to_row[parent])
+            dst.append(id_to_row[nid])
+            size.append(s)
+        for parent in n["control_input"]:
+            src.append(id_to_row[parent])
+            dst.append(id_to_row[nid])
+            size.append(0.0)
+    if model is None:
+        model = path.rstrip("/").split("/")[-1].replace(".pbtxt", "")
+    return CompGraph(names=names,
+                     compute_cost={processor_type_profiled: comp},
+                     memory_cost=mem,
+                     pass_type=np.zeros(len(nodes), dtype=np.int8),
+                     counterpart=np.full(len(nodes), -1, dtype=np.int64),
+                     src=np.array(src, dtype=np.int64),
+                     dst=np.array(dst, dtype=np.int64),
+                     size=np.array(size), model=model)
+
+
 def backward_name(forward_name: str, num_fwd_nodes: int) -> str:
     """Reference ``placers/utils.py:316-322``."""
     return str((2 * num_fwd_nodes) - (int(forward_name) - 1))

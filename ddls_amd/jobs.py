@@ -301,7 +301,11 @@ class JobsGenerator:
                 files = files[:max_files]
             if not files:
                 raise FileNotFoundError(f"no job profile files in {path_to_files}")
-            graphs = [load_pipedream_graph(f, processor_type_profiled) for f in files]
+            from .graphs import load_pbtxt_graph
+            graphs = [load_pbtxt_graph(f, processor_type_profiled)
+                      if f.endswith(".pbtxt")
+                      else load_pipedream_graph(f, processor_type_profiled)
+                      for f in files]
 
         self.job_interarrival_time_dist = distribution_from_config(
             job_interarrival_time_dist if job_interarrival_time_dist is not None

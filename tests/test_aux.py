@@ -1,0 +1,134 @@
+"""Aux subsystem tests: logger, checkpointer, config system, plotting,
+pbtxt reader, metrics utilities."""
+import os
+
+import numpy as np
+import pytest
+
+
+def test_logger_pkl_append(tmp_path):
+    from ddls_amd.runtime.logger import Logger
+    lg = Logger(str(tmp_path))
+    lg.write({"log": {"a": [1, 2], "b": 5}}, block=True)
+    lg.write({"log": {"a": [3], "b": 6}}, block=True)
+    out = lg.load("log")
+    assert out["a"] == [1, 2, 3]
+    assert out["b"] == 6
+
+
+def test_logger_sqlite_append(tmp_path):
+    from ddls_amd.runtime.logger import Logger
+    lg = Logger(str(tmp_path), use_sqlite_database=True)
+    lg.write({"log": {"a": [1], "c": "x"}}, block=True)
+    lg.write({"log": {"a": [2, 3]}}, block=True)
+    out = lg.load("log")
+    assert out["a"] == [1, 2, 3]
+    assert out["c"] == "x"
+
+
+def test_checkpointer_rllib_layout(tmp_path):
+    from ddls_amd.runtime.checkpointer import Checkpointer
+    ck = Checkpointer(str(tmp_path))
+    p1 = ck.write({"x": 1}, index=1)
+    p5 = ck.write({"x": 5}, index=5)
+    assert p1.endswith("checkpoints/checkpoint_000001/checkpoint-1")
+    assert ck.latest() == p5
+    assert Checkpointer.read(p5)["x"] == 5
+
+
+def test_config_overrides(tmp_path):
+    from ddls_amd.runtime.config import load_config
+    (tmp_path / "group").mkdir()
+    (tmp_path / "group" / "a.yaml").write_text("x: 1\ny:\n  z: 2\n")
+    (tmp_path / "top.yaml").write_text(
+        "defaults:\n  - group: a\nextra: 3\n")
+    cfg = load_config(str(tmp_path / "top.yaml"), overrides=["group.y.z=9",
+                                                             "new.key=hi"])
+    assert cfg["group"]["x"] == 1
+    assert cfg["group"]["y"]["z"] == 9
+    assert cfg["extra"] == 3
+    assert cfg["new"]["key"] == "hi"
+
+
+def test_pbtxt_reader(tmp_path):
+    from ddls_amd.graphs import load_pbtxt_graph
+    pb = """node {
+  name: "_SOURCE"
+  output_info {
+    size: 4
+  }
+  compute_cost: 1
+}
+node {
+  id: 1
+  input_info {
+    preceding_node: 0
+    preceding_port: 1
+  }
+  output_info {
+    size: 16
+  }
+  compute_cost: 7
+}
+node {
+  id: 2
+  input_info {
+    preceding_node: 1
+    preceding_port: 1
+  }
+  control_input: 0
+  compute_cost: 3
+}
+"""
+    p = tmp_path / "g.pbtxt"
+    p.write_text(pb)
+    g = load_pbtxt_graph(str(p), "MI355X")
+    assert g.n == 3
+    assert g.m == 3
+    cc = g.compute_cost["MI355X"]
+    assert cc[g.name_to_idx["1"]] == 7
+    sizes = {(g.names[int(u)], g.names[int(v)]): s
+             for u, v, s in zip(g.src, g.dst, g.size)}
+    assert sizes[("0", "1")] == 4       # data dep: parent output size
+    assert sizes[("1", "2")] == 16
+    assert sizes[("0", "2")] == 0       # control dep
+
+
+def test_plotting(tmp_path, tiny_model_files):
+    pytest.importorskip("matplotlib")
+    from ddls_amd.graphs import load_pipedream_graph
+    from ddls_amd.plotting import (plot_computation_graph,
+                                   plot_episode_stats_comparison,
+                                   plot_training_curves)
+    g = load_pipedream_graph(tiny_model_files + "/tiny.txt", "A100")
+    fig = plot_computation_graph(g)
+    fig.savefig(str(tmp_path / "graph.png"))
+    fig2 = plot_training_curves({"mean_reward": [1, 2, 3], "kl": [0.1, 0.2]})
+    fig2.savefig(str(tmp_path / "curves.png"))
+    fig3 = plot_episode_stats_comparison(
+        {"random": {"blocking_rate": 0.5, "acceptance_rate": 0.5,
+                    "mean_job_completion_time": 10}})
+    fig3.savefig(str(tmp_path / "cmp.png"))
+    assert os.path.getsize(str(tmp_path / "graph.png")) > 0
+
+
+def test_metric_vocabulary():
+    from ddls_amd.cluster.environment import RampClusterEnvironment as R
+    assert "blocking_rate" in R.episode_metrics()
+    assert "job_completion_time_speedup" in R.episode_completion_metrics()
+    assert "jobs_blocked_num_nodes" in R.episode_blocked_metrics()
+    assert "mean_num_mounted_workers" in R.step_metrics()
+
+
+def test_harvest_and_tables(tmp_path, tiny_model_files):
+    from tests.conftest import make_env
+    from ddls_amd.runtime.metrics import harvest_episode_stats, write_metrics_table
+    env = make_env(tiny_model_files, replication=1)
+    obs = env.reset(seed=0)
+    env.step(1)
+    out = harvest_episode_stats(env)
+    assert out["num_jobs_completed"] == 1
+    assert "mean_job_completion_time" in out
+    write_metrics_table([out], str(tmp_path / "table"))
+    assert os.path.exists(str(tmp_path / "table.csv"))
+    assert os.path.exists(str(tmp_path / "table.json"))
