@@ -64,7 +64,7 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=5)
     ap.add_argument("--warmup", type=int, default=2)
-    ap.add_argument("--envs-per-rank", type=int, default=16)
+    ap.add_argument("--envs-per-rank", type=int, default=64)
     ap.add_argument("--rollout-steps-per-env", type=int, default=16)
     ap.add_argument("--env-workers", type=int, default=0,
                     help="env worker processes per rank (0 = auto)")
