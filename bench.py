@@ -125,7 +125,11 @@ def main():
     barrier_sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        trainer.train(num_steps=args.rollout_steps_per_env)
+        st = trainer.train(num_steps=args.rollout_steps_per_env)
+        if rank == 0:
+            print(f"[bench] iter {st['iteration']}: rollout "
+                  f"{st['rollout_time_s']:.3f}s update {st['update_time_s']:.3f}s",
+                  file=sys.stderr, flush=True)
     barrier_sync()
     elapsed = time.perf_counter() - t0
 

@@ -47,7 +47,7 @@ def mount_job_for_lookahead(cluster, job, placement: Dict[str, str],
     for dep_idx, channels in deps.items():
         for cid in channels:
             if cid is not None:
-                job.dep_channel_idx[dep_idx] = cluster.channel_id_to_index[cid]
+                job.dep_channel_idx[dep_idx] = cluster.channel_index(cid)
     for channel_id, job_to_deps in dep_schedule.action.items():
         if channel_id is None:
             continue

@@ -85,13 +85,22 @@ class RampClusterEnvironment:
                         topology.num_workers += 1
                         self.worker_id_to_index[w.processor_id] = len(self.workers)
                         self.workers.append(w)
-        # dense channel index
-        self.channel_ids = list(topology.channel_id_to_channel.keys())
-        self.channel_id_to_index = {cid: i for i, cid in enumerate(self.channel_ids)}
+        # dense channel index, grown lazily (big RAMP meshes instantiate
+        # channels on demand)
+        self.channel_ids = []
+        self.channel_id_to_index = {}
         # dense worker index -> server node index
         self.worker_node = np.array(
             [topology.worker_to_node[w.processor_id] for w in self.workers],
             dtype=np.int64)
+
+    def channel_index(self, channel_id: str) -> int:
+        idx = self.channel_id_to_index.get(channel_id)
+        if idx is None:
+            idx = len(self.channel_ids)
+            self.channel_id_to_index[channel_id] = idx
+            self.channel_ids.append(channel_id)
+        return idx
 
     @property
     def device_type(self) -> str:
@@ -348,7 +357,7 @@ class RampClusterEnvironment:
                     if channel_id is None:
                         continue
                     mounted.append((dep_idx, channel_id))
-                    job.dep_channel_idx[dep_idx] = self.channel_id_to_index[channel_id]
+                    job.dep_channel_idx[dep_idx] = self.channel_index(channel_id)
             if mounted:
                 # group by channel: one occupancy-rule check + one counted
                 # mount per channel instead of per dep

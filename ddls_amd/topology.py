@@ -14,6 +14,60 @@ from typing import Dict, List, Tuple, Union
 from .devices import Channel, gen_channel_id
 
 
+class _LazyChannelMap:
+    """dict-like channel_id -> Channel that instantiates on first access."""
+
+    def __init__(self, topo: "Ramp"):
+        self._topo = topo
+        self._channels: Dict[str, Channel] = {}
+
+    def __getitem__(self, cid: str) -> Channel:
+        ch = self._channels.get(cid)
+        if ch is None:
+            src, dst, num = cid.split("|")
+            if (src not in self._topo.name_to_node
+                    or dst not in self._topo.name_to_node
+                    or not (0 <= int(num) < self._topo.num_channels)):
+                raise KeyError(cid)
+            ch = Channel(src, dst, int(num),
+                         channel_bandwidth=self._topo.channel_bandwidth)
+            self._channels[cid] = ch
+        return ch
+
+    def __contains__(self, cid: str) -> bool:
+        try:
+            self[cid]
+            return True
+        except KeyError:
+            return False
+
+    def values(self):
+        return self._channels.values()  # instantiated channels only
+
+    def keys(self):
+        return self._channels.keys()
+
+    def instantiated(self) -> Dict[str, Channel]:
+        return self._channels
+
+
+class _DirectCidCache:
+    """(src_node, dst_node, channel_num) -> channel id string, grown lazily."""
+
+    def __init__(self, topo: "Ramp"):
+        self._topo = topo
+        self._cache: Dict[Tuple[int, int, int], str] = {}
+
+    def __getitem__(self, key: Tuple[int, int, int]) -> str:
+        cid = self._cache.get(key)
+        if cid is None:
+            u, v, k = key
+            cid = gen_channel_id(self._topo.node_names[u],
+                                 self._topo.node_names[v], k)
+            self._cache[key] = cid
+        return cid
+
+
 class Ramp:
     """RAMP: (communication_groups x racks x servers) full mesh."""
 
@@ -49,27 +103,15 @@ class Ramp:
         self.coord_to_node = {co: i for i, co in enumerate(self.coords)}
         self.num_nodes = len(self.coords)
 
-        # per-direction channels on every (u, v) pair of the full mesh
-        self.channel_id_to_channel: Dict[str, Channel] = {}
-        for u in range(self.num_nodes):
-            for v in range(self.num_nodes):
-                if u == v:
-                    continue
-                for k in range(num_channels):
-                    ch = Channel(self.node_names[u], self.node_names[v], k,
-                                 channel_bandwidth=self.channel_bandwidth)
-                    self.channel_id_to_channel[ch.channel_id] = ch
+        # per-direction channels on every (u, v) pair of the full mesh.
+        # Instantiated LAZILY: a 1024-server mesh has ~1M directed channels
+        # and RAMP jobs only ever touch the ones their flows use.
+        self.channel_id_to_channel = _LazyChannelMap(self)
 
-        # precomputed direct-link channel ids (full mesh -> every path is a
-        # single hop); avoids string formatting on the per-flow hot path
-        self.direct_cid: Dict[Tuple[int, int, int], str] = {}
-        for u in range(self.num_nodes):
-            for v in range(self.num_nodes):
-                if u == v:
-                    continue
-                for k in range(num_channels):
-                    self.direct_cid[(u, v, k)] = gen_channel_id(
-                        self.node_names[u], self.node_names[v], k)
+        # cached direct-link channel ids (full mesh -> every path is a single
+        # hop); avoids string formatting on the per-flow hot path.  Grown on
+        # demand (a dense precompute is O(nodes^2)).
+        self.direct_cid = _DirectCidCache(self)
 
         # populated by the cluster environment
         self.node_workers: List[dict] = [dict() for _ in range(self.num_nodes)]
