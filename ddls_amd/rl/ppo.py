@@ -83,6 +83,36 @@ class PPOTrainer:
         n_envs = len(self.env)
         steps = num_steps if num_steps is not None else max(
             1, cfg.train_batch_size // n_envs)
+
+        if hasattr(self.env, "rollout"):
+            # RLlib-style worker-side rollouts: policy copies live in the env
+            # workers; one IPC round trip per iteration
+            data = self.env.rollout(self.policy, steps)
+            self.total_env_steps += steps * n_envs
+            rewards, dones = data["rewards"], data["dones"]
+            values, logps, actions = data["values"], data["logp"], data["actions"]
+            bootstrap_values = data["bootstrap_values"]
+            T = rewards.shape[0]
+            adv = np.zeros_like(rewards)
+            lastgaelam = np.zeros(n_envs)
+            for t in reversed(range(T)):
+                next_values = bootstrap_values if t == T - 1 else values[t + 1]
+                nonterminal = 1.0 - dones[t].astype(np.float64)
+                delta = (rewards[t] + cfg.gamma * next_values * nonterminal
+                         - values[t])
+                lastgaelam = (delta + cfg.gamma * cfg.lambda_ * nonterminal
+                              * lastgaelam)
+                adv[t] = lastgaelam
+            return {
+                "obs": data["obs"],
+                "actions": actions.reshape(-1),
+                "logp": logps.reshape(-1),
+                "advantages": adv.reshape(-1),
+                "value_targets": (adv + values).reshape(-1),
+                "values": values.reshape(-1),
+                "rewards": rewards.reshape(-1),
+            }
+
         obs_buf: List[CompactObs] = []
         act_buf, logp_buf = [], []
         rew_buf, done_buf, val_buf = [], [], []

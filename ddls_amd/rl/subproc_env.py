@@ -21,14 +21,83 @@ def _worker_loop(conn, env_fn_count: int, env_fn, base_seed: int):
     episode_counters = [0] * len(envs)
     episode_returns = np.zeros(len(envs))
     episode_lens = np.zeros(len(envs), dtype=np.int64)
+    policy = None          # lazy CPU policy copy for worker-side rollouts
+    obs_cache = None
     try:
         while True:
             cmd, payload = conn.recv()
+            if cmd == "rollout":
+                # RLlib-style worker-side rollout: the worker holds a CPU
+                # policy copy and runs T steps locally — one IPC round trip
+                # per iteration instead of per env step.
+                import torch
+                from ..models.gnn import GNNPolicy
+                from .rollout import collate
+                state_dict, steps, policy_cfg = payload
+                torch.set_num_threads(1)
+                if policy is None:
+                    torch.manual_seed(base_seed * 7919 + 13)  # decorrelate
+                    num_actions = len(envs[0].action_set)
+                    policy = GNNPolicy(num_actions=num_actions,
+                                       config=policy_cfg)
+                    policy.eval()
+                policy.load_state_dict({k: torch.as_tensor(v)
+                                        for k, v in state_dict.items()})
+                if obs_cache is None:
+                    obs_cache = [CompactObs.from_obs(
+                        env.reset(seed=base_seed + 1000 * i))
+                        for i, env in enumerate(envs)]
+                traj = {"obs": [], "actions": [], "logp": [], "values": [],
+                        "rewards": [], "dones": []}
+                stats_out = []
+                with torch.no_grad():
+                    for _t in range(steps):
+                        inputs = collate(obs_cache, torch.device("cpu"))
+                        logits, values = policy.forward_flat(
+                            inputs["batch"], inputs["graph_features"],
+                            inputs["action_mask"])
+                        dist = torch.distributions.Categorical(logits=logits)
+                        actions = dist.sample()
+                        logp = dist.log_prob(actions)
+                        traj["obs"].append(list(obs_cache))
+                        traj["actions"].append(actions.numpy())
+                        traj["logp"].append(logp.numpy())
+                        traj["values"].append(values.numpy())
+                        rewards = np.zeros(len(envs))
+                        dones = np.zeros(len(envs), dtype=bool)
+                        for i, env in enumerate(envs):
+                            o, r, done, _ = env.step(int(actions[i]))
+                            rewards[i] = r
+                            dones[i] = done
+                            episode_returns[i] += r
+                            episode_lens[i] += 1
+                            if done:
+                                st = dict(env.cluster.episode_stats)
+                                st["episode_return"] = float(episode_returns[i])
+                                st["episode_len"] = int(episode_lens[i])
+                                stats_out.append(st)
+                                episode_returns[i] = 0.0
+                                episode_lens[i] = 0
+                                episode_counters[i] += 1
+                                o = env.reset(seed=base_seed + 1000 * i
+                                              + episode_counters[i])
+                            obs_cache[i] = CompactObs.from_obs(o)
+                        traj["rewards"].append(rewards)
+                        traj["dones"].append(dones)
+                    # bootstrap values of the final obs
+                    inputs = collate(obs_cache, torch.device("cpu"))
+                    _, boot = policy.forward_flat(
+                        inputs["batch"], inputs["graph_features"],
+                        inputs["action_mask"])
+                traj["bootstrap_values"] = boot.numpy()
+                conn.send((traj, stats_out))
+                continue
             if cmd == "reset":
                 obs = []
                 for i, env in enumerate(envs):
                     o = env.reset(seed=base_seed + 1000 * i)
                     obs.append(CompactObs.from_obs(o))
+                obs_cache = list(obs)
                 conn.send(obs)
             elif cmd == "step":
                 actions = payload
@@ -50,6 +119,7 @@ def _worker_loop(conn, env_fn_count: int, env_fn, base_seed: int):
                     obs_out.append(CompactObs.from_obs(o))
                     rewards.append(r)
                     dones.append(done)
+                obs_cache = list(obs_out)
                 conn.send((obs_out, np.array(rewards), np.array(dones),
                            stats_out))
             elif cmd == "close":
@@ -116,6 +186,37 @@ class SubprocVectorEnv:
             self.completed_episode_stats.extend(stats)
         self.obs = obs
         return obs, np.concatenate(rewards), np.concatenate(dones)
+
+    def rollout(self, policy, steps: int):
+        """Worker-side rollout for all envs: ship CPU weights once, collect T
+        steps per env locally in every worker, gather trajectories.
+
+        Returns dict with [T, N] arrays (actions/logp/values/rewards/dones),
+        obs as a flat [T*N] list (t-major), and bootstrap_values [N].
+        """
+        state_dict = {k: v.detach().cpu().numpy()
+                      for k, v in policy.state_dict().items()}
+        cfg = dict(policy.config)
+        for conn in self.conns:
+            conn.send(("rollout", (state_dict, steps, cfg)))
+        per_worker = []
+        for conn in self.conns:
+            traj, stats = conn.recv()
+            per_worker.append(traj)
+            self.completed_episode_stats.extend(stats)
+        T = steps
+        obs_flat = []
+        for t in range(T):
+            for traj in per_worker:
+                obs_flat.extend(traj["obs"][t])
+        out = {"obs": obs_flat}
+        for key in ("actions", "logp", "values", "rewards", "dones"):
+            out[key] = np.stack(
+                [np.concatenate([traj[key][t] for traj in per_worker])
+                 for t in range(T)])
+        out["bootstrap_values"] = np.concatenate(
+            [traj["bootstrap_values"] for traj in per_worker])
+        return out
 
     def drain_episode_stats(self) -> List[dict]:
         out = self.completed_episode_stats

@@ -124,3 +124,34 @@ def test_subproc_vector_env(tiny_model_files):
             assert np.isfinite(rewards).all()
     finally:
         venv.close()
+
+
+def test_worker_side_rollouts(tiny_model_files):
+    """RLlib-style worker-side rollout path: trajectories have the right
+    shapes, masked actions only, finite GAE."""
+    import torch
+    from ddls_amd.models.gnn import GNNPolicy
+    from ddls_amd.rl.ppo import PPOConfig, PPOTrainer
+    from ddls_amd.rl.subproc_env import SubprocVectorEnv
+
+    def fn():
+        return make_env(tiny_model_files, replication=50)
+
+    venv = SubprocVectorEnv(fn, num_envs=4, num_workers=2, base_seed=0)
+    try:
+        torch.manual_seed(0)
+        policy = GNNPolicy(num_actions=17)
+        trainer = PPOTrainer(venv, policy,
+                             PPOConfig(train_batch_size=24,
+                                       sgd_minibatch_size=8, num_sgd_iter=2),
+                             device=torch.device("cpu"))
+        batch = trainer.collect_rollout(num_steps=6)
+        assert len(batch["actions"]) == 24
+        assert len(batch["obs"]) == 24
+        assert np.isfinite(batch["advantages"]).all()
+        stats = trainer.update(batch)
+        assert np.isfinite(stats["total_loss"])
+        full = trainer.train(num_steps=6)
+        assert full["env_steps_this_iter"] == 24
+    finally:
+        venv.close()
