@@ -1,0 +1,142 @@
+"""Numpy inference mirror of GNNPolicy for env-worker rollouts.
+
+The policy is ~2e4 params and rollout graphs are ~60 nodes; a torch forward
+on such shapes is dominated by op-dispatch overhead (~15 ms on one CPU
+thread).  This mirror runs the identical computation in numpy (~0.5 ms),
+used only for worker-side action sampling; the SGD update always recomputes
+through torch autograd (tiny fp differences only shift the importance ratio
+by rounding noise).
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+
+from ..rl.rollout import CompactObs
+
+LN_EPS = 1e-5
+
+
+def _ln(x, g, b):
+    mu = x.mean(axis=-1, keepdims=True)
+    var = x.var(axis=-1, keepdims=True)
+    return (x - mu) / np.sqrt(var + LN_EPS) * g + b
+
+
+def _relu(x):
+    return np.maximum(x, 0.0)
+
+
+class _MLP:
+    """LayerNorm -> Linear [-> act] [-> (Linear -> act) * (depth-1)]."""
+
+    def __init__(self, params: Dict[str, np.ndarray], prefix: str,
+                 input_act: bool = True):
+        self.ln_g = params[f"{prefix}.0.weight"]
+        self.ln_b = params[f"{prefix}.0.bias"]
+        self.W = params[f"{prefix}.1.weight"]
+        self.b = params[f"{prefix}.1.bias"]
+        self.input_act = input_act
+        # depth-1 extra layers
+        self.extra = []
+        i = 3 if input_act else 2
+        while f"{prefix}.{i}.weight" in params:
+            self.extra.append((params[f"{prefix}.{i}.weight"],
+                               params[f"{prefix}.{i}.bias"]))
+            i += 2
+
+    def __call__(self, x):
+        y = _ln(x, self.ln_g, self.ln_b) @ self.W.T + self.b
+        if self.input_act:
+            y = _relu(y)
+        for W, b in self.extra:
+            y = _relu(y @ W.T + b)
+        return y
+
+
+class _Branch:
+    """Linear->act stack + final Linear (policy/value branches)."""
+
+    def __init__(self, params, prefix):
+        self.layers = []
+        i = 0
+        while f"{prefix}.{i}.weight" in params:
+            self.layers.append((params[f"{prefix}.{i}.weight"],
+                                params[f"{prefix}.{i}.bias"]))
+            i += 2
+        # find trailing layer index (last Linear has no activation after it)
+
+    def __call__(self, x):
+        for k, (W, b) in enumerate(self.layers):
+            x = x @ W.T + b
+            if k < len(self.layers) - 1:
+                x = _relu(x)
+        return x
+
+
+class NumpyGNNPolicy:
+    """Inference-only mirror of models.gnn.GNNPolicy."""
+
+    def __init__(self, state_dict: Dict[str, np.ndarray], config: Dict,
+                 num_actions: int):
+        p = {k: np.asarray(v, dtype=np.float32) for k, v in state_dict.items()}
+        self.config = config
+        self.num_actions = num_actions
+        n_layers = config["num_rounds"]
+        self.layers = []
+        for li in range(n_layers):
+            self.layers.append({
+                "node": _MLP(p, f"gnn.layers.{li}.node_module"),
+                "edge": _MLP(p, f"gnn.layers.{li}.edge_module"),
+                "reduce": _MLP(p, f"gnn.layers.{li}.reduce_module"),
+            })
+        self.graph_module = _MLP(p, "graph_module", input_act=False)
+        self.policy_branch = _Branch(p, "policy_branch")
+        self.value_branch = _Branch(p, "value_branch")
+
+    def _meanpool(self, layer, z, e, src, dst):
+        hn = layer["node"](z)
+        he = layer["edge"](e)
+        msg_edge = np.concatenate([hn[src], he], axis=-1)
+        msg_self = np.concatenate([hn, np.zeros_like(hn)], axis=-1)
+        r_edge = layer["reduce"](msg_edge)
+        r_self = layer["reduce"](msg_self)
+        N = z.shape[0]
+        out = np.zeros((N, r_self.shape[-1]), dtype=np.float32)
+        np.add.at(out, dst, r_edge)
+        in_deg = np.bincount(dst, minlength=N).astype(np.float32)
+        total = out + r_self
+        mean = total / (in_deg + 1.0)[:, None]
+        mean[in_deg == 0] = 0.0
+        return mean
+
+    def forward(self, obs: CompactObs) -> Tuple[np.ndarray, float]:
+        z = obs.node_features
+        e = obs.edge_features
+        src, dst = obs.edges_src, obs.edges_dst
+        for layer in self.layers:
+            z = self._meanpool(layer, z, e, src, dst)
+        graph_node_emb = z.mean(axis=0) if len(z) else np.zeros(
+            self.config["out_features_node"], dtype=np.float32)
+        graph_emb = self.graph_module(obs.graph_features[None, :])[0]
+        final = np.concatenate([graph_node_emb, graph_emb])
+        logits = self.policy_branch(final[None, :])[0]
+        value = float(self.value_branch(final[None, :])[0, 0])
+        if self.config.get("apply_action_mask", True):
+            with np.errstate(divide="ignore"):
+                inf_mask = np.maximum(np.log(obs.action_mask),
+                                      np.finfo(np.float32).min)
+            logits = logits + inf_mask
+        return logits.astype(np.float32), value
+
+    def act(self, obs: CompactObs, rng: np.random.RandomState
+            ) -> Tuple[int, float, float]:
+        """Sample a masked action; returns (action, logp, value)."""
+        logits, value = self.forward(obs)
+        x = logits - logits.max()
+        p = np.exp(x)
+        p /= p.sum()
+        a = int(rng.choice(len(p), p=p))
+        logp = float(np.log(max(p[a], 1e-45)))
+        return a, logp, value

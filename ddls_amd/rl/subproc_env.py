@@ -23,26 +23,22 @@ def _worker_loop(conn, env_fn_count: int, env_fn, base_seed: int):
     episode_lens = np.zeros(len(envs), dtype=np.int64)
     policy = None          # lazy CPU policy copy for worker-side rollouts
     obs_cache = None
+    rng = None
     try:
         while True:
             cmd, payload = conn.recv()
             if cmd == "rollout":
-                # RLlib-style worker-side rollout: the worker holds a CPU
-                # policy copy and runs T steps locally — one IPC round trip
-                # per iteration instead of per env step.
-                import torch
-                from ..models.gnn import GNNPolicy
-                from .rollout import collate
+                # RLlib-style worker-side rollout: the worker holds a numpy
+                # inference copy of the policy (torch op-dispatch dominates at
+                # these tiny shapes) and runs T steps locally — one IPC round
+                # trip per iteration instead of per env step.
+                from ..models.numpy_policy import NumpyGNNPolicy
                 state_dict, steps, policy_cfg = payload
-                torch.set_num_threads(1)
-                if policy is None:
-                    torch.manual_seed(base_seed * 7919 + 13)  # decorrelate
-                    num_actions = len(envs[0].action_set)
-                    policy = GNNPolicy(num_actions=num_actions,
-                                       config=policy_cfg)
-                    policy.eval()
-                policy.load_state_dict({k: torch.as_tensor(v)
-                                        for k, v in state_dict.items()})
+                if rng is None:
+                    rng = np.random.RandomState((base_seed * 7919 + 13)
+                                                % (2 ** 31))
+                num_actions = len(envs[0].action_set)
+                policy = NumpyGNNPolicy(state_dict, policy_cfg, num_actions)
                 if obs_cache is None:
                     obs_cache = [CompactObs.from_obs(
                         env.reset(seed=base_seed + 1000 * i))
@@ -50,46 +46,42 @@ def _worker_loop(conn, env_fn_count: int, env_fn, base_seed: int):
                 traj = {"obs": [], "actions": [], "logp": [], "values": [],
                         "rewards": [], "dones": []}
                 stats_out = []
-                with torch.no_grad():
-                    for _t in range(steps):
-                        inputs = collate(obs_cache, torch.device("cpu"))
-                        logits, values = policy.forward_flat(
-                            inputs["batch"], inputs["graph_features"],
-                            inputs["action_mask"])
-                        dist = torch.distributions.Categorical(logits=logits)
-                        actions = dist.sample()
-                        logp = dist.log_prob(actions)
-                        traj["obs"].append(list(obs_cache))
-                        traj["actions"].append(actions.numpy())
-                        traj["logp"].append(logp.numpy())
-                        traj["values"].append(values.numpy())
-                        rewards = np.zeros(len(envs))
-                        dones = np.zeros(len(envs), dtype=bool)
-                        for i, env in enumerate(envs):
-                            o, r, done, _ = env.step(int(actions[i]))
-                            rewards[i] = r
-                            dones[i] = done
-                            episode_returns[i] += r
-                            episode_lens[i] += 1
-                            if done:
-                                st = dict(env.cluster.episode_stats)
-                                st["episode_return"] = float(episode_returns[i])
-                                st["episode_len"] = int(episode_lens[i])
-                                stats_out.append(st)
-                                episode_returns[i] = 0.0
-                                episode_lens[i] = 0
-                                episode_counters[i] += 1
-                                o = env.reset(seed=base_seed + 1000 * i
-                                              + episode_counters[i])
-                            obs_cache[i] = CompactObs.from_obs(o)
-                        traj["rewards"].append(rewards)
-                        traj["dones"].append(dones)
-                    # bootstrap values of the final obs
-                    inputs = collate(obs_cache, torch.device("cpu"))
-                    _, boot = policy.forward_flat(
-                        inputs["batch"], inputs["graph_features"],
-                        inputs["action_mask"])
-                traj["bootstrap_values"] = boot.numpy()
+                n = len(envs)
+                for _t in range(steps):
+                    actions = np.zeros(n, dtype=np.int64)
+                    logps = np.zeros(n, dtype=np.float32)
+                    values = np.zeros(n, dtype=np.float32)
+                    for i, o in enumerate(obs_cache):
+                        actions[i], logps[i], values[i] = policy.act(o, rng)
+                    traj["obs"].append(list(obs_cache))
+                    traj["actions"].append(actions)
+                    traj["logp"].append(logps)
+                    traj["values"].append(values)
+                    rewards = np.zeros(n)
+                    dones = np.zeros(n, dtype=bool)
+                    for i, env in enumerate(envs):
+                        o, r, done, _ = env.step(int(actions[i]))
+                        rewards[i] = r
+                        dones[i] = done
+                        episode_returns[i] += r
+                        episode_lens[i] += 1
+                        if done:
+                            st = dict(env.cluster.episode_stats)
+                            st["episode_return"] = float(episode_returns[i])
+                            st["episode_len"] = int(episode_lens[i])
+                            stats_out.append(st)
+                            episode_returns[i] = 0.0
+                            episode_lens[i] = 0
+                            episode_counters[i] += 1
+                            o = env.reset(seed=base_seed + 1000 * i
+                                          + episode_counters[i])
+                        obs_cache[i] = CompactObs.from_obs(o)
+                    traj["rewards"].append(rewards)
+                    traj["dones"].append(dones)
+                # bootstrap values of the final obs
+                traj["bootstrap_values"] = np.array(
+                    [policy.forward(o)[1] for o in obs_cache],
+                    dtype=np.float32)
                 conn.send((traj, stats_out))
                 continue
             if cmd == "reset":

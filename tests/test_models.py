@@ -188,3 +188,33 @@ def test_hip_policy_rollout_forward(small_graph):
             del os.environ["DDLS_AMD_DISABLE_HIP"]
     assert torch.allclose(logits_hip, logits_ref, atol=1e-4)
     assert torch.allclose(value_hip, value_ref, atol=1e-4)
+
+
+def test_numpy_policy_matches_torch(small_graph):
+    """The worker-side numpy inference mirror must match the torch policy."""
+    from ddls_amd.models.numpy_policy import NumpyGNNPolicy
+    from ddls_amd.rl.rollout import CompactObs
+    z, e, src, dst = small_graph
+    torch.manual_seed(9)
+    policy = GNNPolicy(num_actions=17)
+    policy.eval()
+    mask = np.ones(17, dtype=np.float32)
+    mask[5] = 0
+    gf = np.random.rand(34).astype(np.float32)
+    obs = CompactObs(node_features=z.numpy(), edge_features=e.numpy(),
+                     edges_src=src.numpy(), edges_dst=dst.numpy(),
+                     graph_features=gf, action_mask=mask)
+    sd = {k: v.detach().numpy() for k, v in policy.state_dict().items()}
+    np_policy = NumpyGNNPolicy(sd, policy.config, 17)
+    np_logits, np_value = np_policy.forward(obs)
+
+    batch = GraphBatch(z=z, e=e, src=src, dst=dst,
+                       graph_of_node=torch.zeros(7, dtype=torch.int64),
+                       num_graphs=1)
+    with torch.no_grad():
+        t_logits, t_value = policy.forward_flat(
+            batch, torch.as_tensor(gf)[None, :], torch.as_tensor(mask)[None, :])
+    finite = np.isfinite(np_logits)
+    assert np.allclose(np_logits[finite], t_logits[0].numpy()[finite], atol=1e-4)
+    assert t_logits[0, 5].item() < -1e30 and np_logits[5] < -1e30
+    assert np_value == pytest.approx(t_value.item(), abs=1e-4)
