@@ -32,6 +32,7 @@ def build_env_fn(seed_offset: int = 0, device_type: str = "A100",
         return RampJobPartitioningEnvironment(
             lookahead_memo_preload=lookahead_memo,
             init_details_memo_preload=init_details_memo,
+            reuse_jobs_generator=True,
             topology_config={"type": "ramp", "kwargs": {
                 "num_communication_groups": 4,
                 "num_racks_per_communication_group": 4,

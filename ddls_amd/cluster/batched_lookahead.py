@@ -181,7 +181,7 @@ def precompute_lookahead_memos(env, device=None, degrees: Optional[List[int]] = 
 
     # one representative job per model from the generator pool
     model_to_job = {}
-    for j in cluster.jobs_generator.job_sampler.sample_pool:
+    for j in cluster.jobs_generator.job_sampler.original_pool:
         model_to_job.setdefault(j.details["model"], j)
 
     lookahead_memo: Dict = {}

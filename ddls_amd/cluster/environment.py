@@ -115,6 +115,7 @@ class RampClusterEnvironment:
               seed: Optional[int] = None,
               lookahead_memo_preload: Optional[dict] = None,
               init_details_memo_preload: Optional[dict] = None,
+              reuse_jobs_generator: bool = False,
               verbose: bool = False):
         self.reset_counter += 1
         self.seed = seed
@@ -122,7 +123,12 @@ class RampClusterEnvironment:
             seed_everything(seed)
 
         self.stopwatch.reset()
-        self.jobs_generator = JobsGenerator(**jobs_config)
+        if (reuse_jobs_generator and getattr(self, "jobs_generator", None)
+                is not None and getattr(self, "_last_jobs_config", None) == jobs_config):
+            self.jobs_generator.reset()
+        else:
+            self.jobs_generator = JobsGenerator(**jobs_config)
+            self._last_jobs_config = jobs_config
         self.max_simulation_run_time = (max_simulation_run_time
                                         if max_simulation_run_time is not None
                                         else float("inf"))

@@ -48,9 +48,11 @@ class RampJobPartitioningEnvironment:
                  use_sqlite_database: bool = False,
                  apply_action_mask: bool = True,
                  lookahead_memo_preload: Optional[dict] = None,
-                 init_details_memo_preload: Optional[dict] = None):
+                 init_details_memo_preload: Optional[dict] = None,
+                 reuse_jobs_generator: bool = False):
         self.lookahead_memo_preload = lookahead_memo_preload
         self.init_details_memo_preload = init_details_memo_preload
+        self.reuse_jobs_generator = reuse_jobs_generator
         self.topology_config = topology_config
         self.node_config = node_config
         self.jobs_config = jobs_config
@@ -111,6 +113,7 @@ class RampJobPartitioningEnvironment:
                            seed=seed,
                            lookahead_memo_preload=self.lookahead_memo_preload,
                            init_details_memo_preload=self.init_details_memo_preload,
+                           reuse_jobs_generator=self.reuse_jobs_generator,
                            verbose=verbose)
         self.obs = self.observation_function.reset(self)
         self.observation_space = self.observation_function.observation_space

@@ -346,6 +346,23 @@ class JobsGenerator:
     def __len__(self):
         return len(self.job_sampler)
 
+    def reset(self, resample_fracs: bool = True):
+        """Reset for a new episode without rebuilding the pool (the reference
+        constructs a fresh JobsGenerator per reset; this is the O(1)
+        equivalent: refill the sampler and resample each prototype's
+        max-acceptable-JCT fraction)."""
+        self.job_sampler.reset()
+        if resample_fracs:
+            for job in self.job_sampler.original_pool:
+                frac = float(self.max_acceptable_job_completion_time_frac_dist.sample())
+                job.max_acceptable_job_completion_time_frac = frac
+                seq = job.details["job_sequential_completion_time"]
+                job.details["max_acceptable_job_completion_time"] = {
+                    dt: frac * v for dt, v in seq.items()}
+            self.jobs_params = self._init_jobs_params(
+                self.job_sampler.original_pool,
+                self.max_partitions_per_op_in_observation)
+
     def sample_job(self) -> Job:
         return self.job_sampler.sample()
 

@@ -26,13 +26,15 @@ class Stopwatch:
 class Sampler:
     """Pool sampler with replace/remove/remove_and_repeat modes.
 
-    On pool reset, job ids are re-based so they stay unique across refills
-    (reference ``utils.py:95-105``).
+    Copy-on-sample: the pool holds prototype jobs and ``sample()`` hands out a
+    private deep copy with its id re-based so ids stay unique across refills
+    (reference ``utils.py:95-105`` re-bases at refill; copying lazily makes
+    pool resets O(1) instead of O(pool)).
     """
 
     def __init__(self, pool: list, sampling_mode: str, shuffle: bool = False,
                  automatically_change_ids: bool = True):
-        self.original_pool = pool
+        self.original_pool = list(pool)
         self.sampling_mode = sampling_mode
         self.shuffle = shuffle
         self.automatically_change_ids = automatically_change_ids
@@ -40,34 +42,34 @@ class Sampler:
         self.reset()
 
     def sample(self):
-        idx = np.random.randint(low=0, high=len(self.sample_pool))
-        datum = self.sample_pool[idx]
+        idx = np.random.randint(low=0, high=len(self.sample_indices))
+        proto_idx = self.sample_indices[idx]
         if self.sampling_mode == "replace":
-            # hand out a private copy so cluster state never aliases the pool
-            datum = copy.deepcopy(datum)
+            pass
         elif self.sampling_mode == "remove":
-            self.sample_pool.pop(idx)
+            self.sample_indices.pop(idx)
         elif self.sampling_mode == "remove_and_repeat":
-            self.sample_pool.pop(idx)
-            if len(self.sample_pool) == 0:
+            self.sample_indices.pop(idx)
+            if len(self.sample_indices) == 0:
                 self.reset()
         else:
             raise ValueError(f"Unrecognised sampling_mode {self.sampling_mode}")
-        return datum
+        proto = self.original_pool[proto_idx]
+        job = copy.deepcopy(proto)
+        if self.automatically_change_ids:
+            job.job_id = int(self._base_ids[proto_idx] + proto.job_id)
+        return job
 
     def __len__(self):
-        return len(self.sample_pool)
+        return len(self.sample_indices)
 
     def reset(self):
-        # one deepcopy call so graphs shared between jobs stay shared
-        # (CompGraph/GraphImmutableDetails define __deepcopy__ -> self)
-        self.sample_pool = copy.deepcopy(list(self.original_pool))
-        if self.automatically_change_ids:
-            base_id = len(self.original_pool) * self.reset_counter
-            for job in self.sample_pool:
-                job.job_id = int(base_id + job.job_id)
+        n = len(self.original_pool)
+        self.sample_indices = list(range(n))
+        # re-base ids per refill generation (reference semantics)
+        self._base_ids = [n * self.reset_counter] * n
         if self.shuffle:
-            random.shuffle(self.sample_pool)
+            random.shuffle(self.sample_indices)
         self.reset_counter += 1
 
 
