@@ -21,6 +21,7 @@ Semantics parity notes:
 from __future__ import annotations
 
 import math
+import os
 from dataclasses import dataclass
 from typing import Dict, List, Optional
 
@@ -120,6 +121,48 @@ class GraphBatch:
                           graph_of_node=graph_of_node, num_graphs=B)
 
 
+class _FusedMeanPoolFn(torch.autograd.Function):
+    """Fused HIP forward+backward of one MeanPool layer (training path on
+    MI355X; kernels in ddls_amd/ops/hip/meanpool.hip + meanpool_bwd.hip)."""
+
+    @staticmethod
+    def forward(ctx, z, e, src, dst, edge_order, indptr,
+                ln_n_w, ln_n_b, W_n, b_n,
+                ln_e_w, ln_e_b, W_e, b_e,
+                ln_r_w, ln_r_b, W_r, b_r):
+        from .. import ops as hip_ops
+        ext = hip_ops.get_extension(required=True)
+        zc, ec = z.contiguous(), e.contiguous()
+        Wn, We, Wr = W_n.contiguous(), W_e.contiguous(), W_r.contiguous()
+        hn = ext.row_mlp(zc, ln_n_w, ln_n_b, Wn, b_n)
+        he = ext.row_mlp(ec, ln_e_w, ln_e_b, We, b_e)
+        out, r_edge, r_self = ext.message_reduce_train(
+            hn, he, src, edge_order, indptr, ln_r_w, ln_r_b, Wr, b_r)
+        ctx.save_for_backward(zc, ec, src, dst, indptr, hn, he, r_edge, r_self,
+                              ln_n_w, ln_n_b, Wn, ln_e_w, ln_e_b, We,
+                              ln_r_w, ln_r_b, Wr)
+        return out
+
+    @staticmethod
+    def backward(ctx, gout):
+        from .. import ops as hip_ops
+        ext = hip_ops.get_extension(required=True)
+        (z, e, src, dst, indptr, hn, he, r_edge, r_self,
+         ln_n_w, ln_n_b, W_n, ln_e_w, ln_e_b, W_e,
+         ln_r_w, ln_r_b, W_r) = ctx.saved_tensors
+        ghn, ghe, gWr, gbr, glnr_g, glnr_b = ext.message_reduce_bwd(
+            hn, he, src, dst, indptr, ln_r_w, ln_r_b, W_r, r_edge, r_self,
+            gout.contiguous())
+        gz, gWn, gbn, glnn_g, glnn_b = ext.row_mlp_bwd(
+            z, hn, ghn, ln_n_w, ln_n_b, W_n)
+        ge, gWe, gbe, glne_g, glne_b = ext.row_mlp_bwd(
+            e, he, ghe, ln_e_w, ln_e_b, W_e)
+        return (gz, ge, None, None, None, None,
+                glnn_g, glnn_b, gWn, gbn,
+                glne_g, glne_b, gWe, gbe,
+                glnr_g, glnr_b, gWr, gbr)
+
+
 def _mlp(in_dim: int, out_dim: int, act, depth: int,
          input_act: bool = True) -> nn.Sequential:
     layers: List[nn.Module] = [nn.LayerNorm(in_dim), nn.Linear(in_dim, out_dim)]
@@ -157,6 +200,19 @@ class MeanPoolLayer(nn.Module):
         from .. import ops as hip_ops
         if self._hip_eligible and hip_ops.hip_ops_enabled_for(z):
             return self._forward_hip(z, batch)
+        if (self._hip_eligible and z.is_cuda and torch.is_grad_enabled()
+                and hip_ops.get_extension() is not None
+                and os.environ.get("DDLS_AMD_DISABLE_HIP", "0") != "1"):
+            # fused HIP forward+backward training path
+            order, indptr = batch.csr_by_dst()
+            ln_n, lin_n = self.node_module[0], self.node_module[1]
+            ln_e, lin_e = self.edge_module[0], self.edge_module[1]
+            ln_r, lin_r = self.reduce_module[0], self.reduce_module[1]
+            return _FusedMeanPoolFn.apply(
+                z, batch.e, batch.src, batch.dst, order, indptr,
+                ln_n.weight, ln_n.bias, lin_n.weight, lin_n.bias,
+                ln_e.weight, ln_e.bias, lin_e.weight, lin_e.bias,
+                ln_r.weight, ln_r.bias, lin_r.weight, lin_r.bias)
         return self.forward(z, batch.e, batch.src, batch.dst)
 
     def _forward_hip(self, z: torch.Tensor, batch: "GraphBatch") -> torch.Tensor:

@@ -10,6 +10,18 @@ torch::Tensor message_reduce(torch::Tensor hn, torch::Tensor he,
                              torch::Tensor ln_b, torch::Tensor Wr,
                              torch::Tensor br);
 torch::Tensor segment_mean(torch::Tensor x, torch::Tensor node_ptr, int64_t G);
+std::vector<torch::Tensor> message_reduce_train(
+    torch::Tensor hn, torch::Tensor he, torch::Tensor src,
+    torch::Tensor edge_order, torch::Tensor indptr, torch::Tensor ln_g,
+    torch::Tensor ln_b, torch::Tensor Wr, torch::Tensor br);
+std::vector<torch::Tensor> message_reduce_bwd(
+    torch::Tensor hn, torch::Tensor he, torch::Tensor src, torch::Tensor dst,
+    torch::Tensor indptr, torch::Tensor ln_g, torch::Tensor ln_b,
+    torch::Tensor Wr, torch::Tensor r_edge, torch::Tensor r_self,
+    torch::Tensor gout);
+std::vector<torch::Tensor> row_mlp_bwd(torch::Tensor x, torch::Tensor y,
+                                       torch::Tensor gy, torch::Tensor ln_g,
+                                       torch::Tensor ln_b, torch::Tensor W);
 std::vector<torch::Tensor> lookahead_batch(
     torch::Tensor descs, torch::Tensor op_remaining, torch::Tensor op_worker,
     torch::Tensor op_priority, torch::Tensor out_indptr, torch::Tensor out_edges,
@@ -27,4 +39,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("segment_mean", &segment_mean, "per-graph mean of node embeddings");
     m.def("lookahead_batch", &lookahead_batch,
           "batched RAMP lookahead discrete-event simulation");
+    m.def("message_reduce_train", &message_reduce_train,
+          "message_reduce storing per-message activations for backward");
+    m.def("message_reduce_bwd", &message_reduce_bwd,
+          "fused MeanPool message-passing backward");
+    m.def("row_mlp_bwd", &row_mlp_bwd, "fused LN+Linear+ReLU backward");
 }

@@ -18,6 +18,7 @@
 #include <hip/hip_runtime.h>
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
+#include <vector>
 
 #define WAVE 64
 #define WAVES_PER_BLOCK 4
@@ -79,6 +80,8 @@ __global__ void message_reduce_kernel(const float* __restrict__ hn,
                                       const float* __restrict__ Wr,
                                       const float* __restrict__ br,
                                       float* __restrict__ out,
+                                      float* __restrict__ r_edge_store,  // [E][OUT] or null
+                                      float* __restrict__ r_self_store,  // [N][OUT] or null
                                       int N, int half, int OUT) {
     const int MSG = 2 * half;
     // Wr staged once per block: [OUT][MSG] <= 64*64*4 = 16 KB
@@ -123,7 +126,15 @@ __global__ void message_reduce_kernel(const float* __restrict__ hn,
                     const float* wrow = wr_s + lane * MSG;
                     for (int k = 0; k < MSG; ++k)
                         r = fmaf(ms[wave][k], wrow[k], r);
-                    acc += fmaxf(r, 0.0f);  // relu then mean
+                    r = fmaxf(r, 0.0f);  // relu then mean
+                    acc += r;
+                    if (mi < 0) {
+                        if (r_self_store)
+                            r_self_store[(long)v * OUT + lane] = r;
+                    } else if (r_edge_store) {
+                        const long e_id = edge_order[e_begin + mi];
+                        r_edge_store[e_id * OUT + lane] = r;
+                    }
                 }
                 __builtin_amdgcn_wave_barrier();
             }
@@ -205,8 +216,31 @@ torch::Tensor message_reduce(torch::Tensor hn, torch::Tensor he,
                        indptr.data_ptr<long>(), ln_g.data_ptr<float>(),
                        ln_b.data_ptr<float>(), Wr.data_ptr<float>(),
                        br.data_ptr<float>(), out.data_ptr<float>(),
-                       N, half, OUT);
+                       nullptr, nullptr, N, half, OUT);
     return out;
+}
+
+std::vector<torch::Tensor> message_reduce_train(
+        torch::Tensor hn, torch::Tensor he, torch::Tensor src,
+        torch::Tensor edge_order, torch::Tensor indptr, torch::Tensor ln_g,
+        torch::Tensor ln_b, torch::Tensor Wr, torch::Tensor br) {
+    const int N = hn.size(0), half = hn.size(1), OUT = Wr.size(0);
+    const int E = he.size(0);
+    TORCH_CHECK(2 * half <= 64 && OUT <= 64);
+    auto out = torch::empty({N, OUT}, hn.options());
+    auto r_edge = torch::empty({E, OUT}, hn.options());
+    auto r_self = torch::empty({N, OUT}, hn.options());
+    if (N == 0) return {out, r_edge, r_self};
+    hipStream_t stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(message_reduce_kernel, dim3(grid_for(N)), dim3(BLOCK),
+                       0, stream, hn.data_ptr<float>(), he.data_ptr<float>(),
+                       src.data_ptr<long>(), edge_order.data_ptr<long>(),
+                       indptr.data_ptr<long>(), ln_g.data_ptr<float>(),
+                       ln_b.data_ptr<float>(), Wr.data_ptr<float>(),
+                       br.data_ptr<float>(), out.data_ptr<float>(),
+                       r_edge.data_ptr<float>(), r_self.data_ptr<float>(),
+                       N, half, OUT);
+    return {out, r_edge, r_self};
 }
 
 torch::Tensor segment_mean(torch::Tensor x, torch::Tensor node_ptr, int64_t G) {
