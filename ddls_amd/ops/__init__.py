@@ -68,7 +68,9 @@ def hip_ops_enabled_for(tensor) -> bool:
         return False
     if os.environ.get("DDLS_AMD_DISABLE_HIP", "0") == "1":
         return False
-    if torch.is_grad_enabled() and tensor.requires_grad:
+    if torch.is_grad_enabled():
+        # training path: the fused autograd Function owns the kernels (the
+        # INPUT may be a constant while the PARAMS still need grads)
         return False
     allow_fallback = os.environ.get("DDLS_AMD_ALLOW_TORCH_FALLBACK", "0") == "1"
     ext = get_extension(required=not allow_fallback)
