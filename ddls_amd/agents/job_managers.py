@@ -106,3 +106,60 @@ class RandomJobScheduler(_BaseJobScheduler):
 JOB_PLACERS = {"random": RandomJobPlacer}
 JOB_SCHEDULERS = {"srpt": SRPTJobScheduler, "fifo": FIFOJobScheduler,
                   "random": RandomJobScheduler}
+
+
+class SRPTJobPrioritiser:
+    """Order jobs by total size ascending (reference
+    ``managers/prioritisers/srpt_job_prioritiser.py``)."""
+
+    def prioritise(self, jobs):
+        sizes = np.array([j.job_total_operation_memory_cost
+                          + j.job_total_dependency_size for j in jobs])
+        return [jobs[i] for i in np.argsort(sizes)]
+
+
+class RandomJobPartitioner:
+    """Job-level random partitioner for the legacy path (reference
+    ``managers/partitioners/random_job_partitioner.py``): picks an even
+    partition degree per job and returns the degree map (the RAMP-side
+    graph transform lives in cluster/partition.py)."""
+
+    def __init__(self, max_partitions_per_op: int = 2):
+        self.max_partitions_per_op = max_partitions_per_op
+
+    def get(self, cluster):
+        choices = [1] + list(range(2, self.max_partitions_per_op + 1, 2))
+        return {job_id: int(np.random.choice(choices))
+                for job_id in cluster.job_queue.jobs}
+
+
+class AllReduceJobCommunicator:
+    """Gradient-sync communicator for the legacy cluster.
+
+    The reference's ``managers/communicators/all_reduce_job_communicator.py``
+    raises NotImplemented; here it is functional: prices a per-training-step
+    ring all-reduce of the job's parameter bytes over the workers the job is
+    mounted on, using the analytic collective model (xGMI-profiled by
+    default: 7 p2p links, per-link bound ring -> 2(n-1)/n * S over one link).
+    """
+
+    def __init__(self, link_bandwidth: float = 153e9, num_links: int = 7,
+                 latency: float = 1e-6):
+        self.link_bandwidth = link_bandwidth
+        self.num_links = num_links
+        self.latency = latency
+
+    def all_reduce_time(self, message_size: float, num_workers: int) -> float:
+        if num_workers <= 1:
+            return 0.0
+        # ring all-reduce on point-to-point links is per-link bound
+        per_link_bytes = 2.0 * (num_workers - 1) / num_workers * message_size
+        steps = 2 * (num_workers - 1)
+        return steps * self.latency + per_link_bytes / self.link_bandwidth
+
+    def communicate(self, job, cluster) -> float:
+        workers = {cluster.job_op_to_worker.get((job.details["job_idx"], i))
+                   for i in range(job.graph.n)}
+        workers.discard(None)
+        message = float(job.graph.memory_cost.sum())
+        return self.all_reduce_time(message, len(workers))

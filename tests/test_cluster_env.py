@@ -262,3 +262,40 @@ def test_legacy_cluster_env(tiny_model_files):
     # num_training_steps * critical path
     jcts = env.episode_stats["job_completion_time"]
     assert all(j >= 3 * (0.02 + 0.03 + 0.06 + 0.04) - 1e-9 for j in jcts)
+
+
+def test_job_placing_env(tiny_model_files):
+    from ddls_amd.envs.job_placing import JobPlacingAllNodesEnvironment
+    from ddls_amd.utils import seed_everything
+    seed_everything(0)
+    env = JobPlacingAllNodesEnvironment(
+        topology_config={"type": "torus", "kwargs": {
+            "x_dims": 4, "y_dims": 1, "z_dims": 1, "num_channels": 1}},
+        node_config={"type_1": {"num_nodes": 4, "workers_config": [
+            {"num_workers": 1, "worker": "ddls_amd.devices.A100"}]}},
+        jobs_config={"path_to_files": tiny_model_files,
+                     "replication_factor": 2, "job_sampling_mode": "remove",
+                     "job_interarrival_time_dist": {
+                         "_target_": "ddls_amd.distributions.Fixed", "val": 1.0},
+                     "num_training_steps": 2})
+    obs = env.reset(seed=0)
+    assert obs.shape == (6,)
+    done, guard = False, 0
+    while not done and guard < 20:
+        obs, r, done, _ = env.step(2)
+        guard += 1
+    assert env.cluster.episode_stats["num_jobs_completed"] == 2
+
+
+def test_job_managers_extra(tiny_model_files):
+    from ddls_amd.agents.job_managers import (AllReduceJobCommunicator,
+                                              SRPTJobPrioritiser)
+    env = make_env(tiny_model_files, replication=2)
+    env.reset(seed=0)
+    jobs = list(env.cluster.job_queue.jobs.values())
+    pri = SRPTJobPrioritiser().prioritise(jobs)
+    assert len(pri) == len(jobs)
+    comm = AllReduceJobCommunicator()
+    assert comm.all_reduce_time(1e9, 1) == 0.0
+    t2, t8 = comm.all_reduce_time(1e9, 2), comm.all_reduce_time(1e9, 8)
+    assert t2 > 0 and t8 > t2 * 0.5  # per-link bound: weak n-dependence
