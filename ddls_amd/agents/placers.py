@@ -158,11 +158,21 @@ class FirstFitDepPlacer:
             dst_nodes = node_of[g.dst]
             is_flow = (src_nodes != dst_nodes) & (g.size > 0)
             dropped = False
+            # same job may reuse a (src,dst) channel freely; cache only with a
+            # single channel per link (multi-channel keeps the reference's
+            # per-flow shuffled channel choice)
+            cache_pairs = topo.num_channels == 1
+            pair_cache = {}
             for e in range(g.m):
                 if is_flow[e]:
-                    found = self._find_path_channel(
-                        cluster, int(src_nodes[e]), int(dst_nodes[e]), job_idx,
-                        channels_used_across_jobs)
+                    pair = (int(src_nodes[e]), int(dst_nodes[e]))
+                    found = pair_cache.get(pair) if cache_pairs else None
+                    if found is None:
+                        found = self._find_path_channel(
+                            cluster, pair[0], pair[1], job_idx,
+                            channels_used_across_jobs)
+                        if cache_pairs:
+                            pair_cache[pair] = found
                     if found is None:
                         job_to_dep_to_channels.pop(job_id, None)
                         dropped = True
