@@ -91,23 +91,24 @@ def main():
     policy = GNNPolicy(num_actions=17)
 
     env_fn = build_env_fn()
+    n_workers = args.env_workers
+    if n_workers <= 0:
+        n_workers = max(1, min(args.envs_per_rank,
+                               (os.cpu_count() or 2) // max(world_size, 1) - 1))
+    # fork env workers BEFORE any HIP context exists in this process
+    venv = SubprocVectorEnv(env_fn, num_envs=args.envs_per_rank,
+                            num_workers=n_workers,
+                            base_seed=1 + 100000 * rank)
     if not args.no_precompute:
         # precompute the whole (model x degree) lookahead memo table in one
-        # batched HIP kernel launch; forked env workers inherit the tables
+        # batched HIP kernel launch; ship the tables to the live workers
         from ddls_amd.cluster.batched_lookahead import precompute_lookahead_memos
         scratch = env_fn()
         scratch.reset(seed=0)
         memo_l, memo_i = precompute_lookahead_memos(
             scratch, device=device if use_cuda else "cpu")
-        env_fn = build_env_fn(lookahead_memo=memo_l, init_details_memo=memo_i)
         del scratch
-    n_workers = args.env_workers
-    if n_workers <= 0:
-        n_workers = max(1, min(args.envs_per_rank,
-                               (os.cpu_count() or 2) // max(world_size, 1) - 1))
-    venv = SubprocVectorEnv(env_fn, num_envs=args.envs_per_rank,
-                            num_workers=n_workers,
-                            base_seed=1 + 100000 * rank)
+        venv.preload_memos(memo_l, memo_i)
     per_step_env_steps = args.envs_per_rank * args.rollout_steps_per_env
     cfg = PPOConfig(train_batch_size=per_step_env_steps,
                     sgd_minibatch_size=args.sgd_minibatch_size,

@@ -114,6 +114,23 @@ def _worker_loop(conn, env_fn_count: int, env_fn, base_seed: int):
                 obs_cache = list(obs_out)
                 conn.send((obs_out, np.array(rewards), np.array(dones),
                            stats_out))
+            elif cmd == "preload_memo":
+                # inject (model, degree) memo tables computed after fork
+                # (e.g. by the batched HIP lookahead on the parent's GPU)
+                lookahead_memo, init_memo = payload
+                for env in envs:
+                    env.lookahead_memo_preload = lookahead_memo
+                    env.init_details_memo_preload = init_memo
+                    cluster = getattr(env, "cluster", None)
+                    if cluster is not None and hasattr(
+                            cluster, "job_model_to_max_num_partitions_to_lookahead"):
+                        for (model, degree), entry in (init_memo or {}).items():
+                            cluster.job_model_to_max_num_partitions_to_init_details[
+                                model][degree] = dict(entry)
+                        for (model, degree), entry in (lookahead_memo or {}).items():
+                            cluster.job_model_to_max_num_partitions_to_lookahead[
+                                model][degree] = entry
+                conn.send("ok")
             elif cmd == "close":
                 conn.send("closed")
                 return
@@ -214,6 +231,12 @@ class SubprocVectorEnv:
         out = self.completed_episode_stats
         self.completed_episode_stats = []
         return out
+
+    def preload_memos(self, lookahead_memo, init_details_memo):
+        for conn in self.conns:
+            conn.send(("preload_memo", (lookahead_memo, init_details_memo)))
+        for conn in self.conns:
+            conn.recv()
 
     def close(self):
         for conn in self.conns:

@@ -108,21 +108,36 @@ def build_trainer_from_config(cfg: dict, device=None):
 
     from ..models.gnn import GNNPolicy
     from ..rl.ppo import PPOConfig, PPOTrainer
-    from ..rl.rollout import VectorEnv
+    from ..rl.subproc_env import SubprocVectorEnv
 
     algo = cfg.get("algo", {})
     model_cfg = cfg.get("model", {})
     loop_cfg = cfg.get("epoch_loop", {})
     n_envs = loop_cfg.get("num_envs", 8)
+    n_workers = loop_cfg.get("num_env_workers") or max(
+        1, min(n_envs, (os.cpu_count() or 2) - 1))
     seed = cfg.get("seed", 0)
 
     def env_fn():
-        return build_env_from_config(cfg)
+        env = build_env_from_config(cfg)
+        env.reuse_jobs_generator = True
+        return env
 
-    venv = VectorEnv([env_fn for _ in range(n_envs)], base_seed=seed)
+    # fork env workers before any HIP context exists in this process
+    venv = SubprocVectorEnv(env_fn, num_envs=n_envs, num_workers=n_workers,
+                            base_seed=seed)
+    num_actions = cfg.get("env_config", {}).get("max_partitions_per_op", 16) + 1
+    if loop_cfg.get("precompute_lookaheads", True):
+        from ..cluster.batched_lookahead import precompute_lookahead_memos
+        scratch = build_env_from_config(cfg)
+        scratch.reset(seed=seed)
+        memo_l, memo_i = precompute_lookahead_memos(
+            scratch, device=device if (device is not None and str(device) != "cpu"
+                                       and torch.cuda.is_available()) else "cpu")
+        del scratch
+        venv.preload_memos(memo_l, memo_i)
     torch.manual_seed(seed)
-    env0 = venv.envs[0]
-    policy = GNNPolicy(num_actions=env0.action_space.n,
+    policy = GNNPolicy(num_actions=num_actions,
                        config=model_cfg.get("custom_model_config"))
     ppo_cfg = PPOConfig(
         lr=algo.get("lr", 2.785e-4),

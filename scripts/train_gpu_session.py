@@ -28,11 +28,11 @@ os.makedirs(OUT, exist_ok=True)
 device = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
 
 env_fn = build_env_fn()
+venv = SubprocVectorEnv(env_fn, num_envs=64, num_workers=64, base_seed=1)
 scratch = env_fn(); scratch.reset(seed=0)
 ml, mi = precompute_lookahead_memos(scratch, device=device)
 del scratch
-env_fn = build_env_fn(lookahead_memo=ml, init_details_memo=mi)
-venv = SubprocVectorEnv(env_fn, num_envs=64, num_workers=64, base_seed=1)
+venv.preload_memos(ml, mi)
 torch.manual_seed(0)
 policy = GNNPolicy(num_actions=17)
 trainer = PPOTrainer(venv, policy,
