@@ -1,0 +1,131 @@
+"""Observation encoding, reward-function variants, distributions."""
+import math
+
+import numpy as np
+import pytest
+
+from tests.conftest import make_env
+
+
+def test_padded_obs_shapes_and_ranges(tiny_model_files):
+    env = make_env(tiny_model_files)  # pad_obs_kwargs max_nodes=150
+    obs = env.reset(seed=0)
+    assert obs["node_features"].shape == (150, 5)
+    assert obs["edge_features"].shape == (11175, 2)
+    assert obs["graph_features"].shape == (17 + 17,)
+    assert obs["node_split"][0] == 4   # tiny graph: 4 mirrored nodes
+    assert obs["edge_split"][0] == 3
+    assert 0 <= obs["node_features"].min() and obs["node_features"].max() <= 1
+    assert 0 <= obs["graph_features"].min() and obs["graph_features"].max() <= 1
+    # padding region zeroed
+    assert np.all(obs["node_features"][4:] == 0)
+    assert np.all(obs["edge_features"][3:] == 0)
+    # observation space built from the sample obs
+    assert env.observation_space["node_features"].shape == (150, 5)
+    assert env.observation_space.contains(obs)
+
+
+def test_unpadded_obs(tiny_model_files):
+    env = make_env(tiny_model_files)
+    env.pad_obs_kwargs = None
+    env.observation_function.pad_obs_kwargs = None
+    obs = env.reset(seed=0)
+    assert obs["node_features"].shape == (4, 5)
+    assert np.isnan(obs["node_split"][0])
+
+
+def test_node_feature_semantics(tiny_model_files):
+    env = make_env(tiny_model_files)
+    obs = env.reset(seed=0)
+    nf = obs["node_features"]
+    g = next(iter(env.cluster.job_queue.jobs.values())).graph
+    # compute cost normalised by max; bwd of op2 ('3') has the max cost
+    i3 = g.name_to_idx["3"]
+    assert nf[i3, 0] == pytest.approx(1.0)
+    assert nf[i3, 1] == 1.0  # is_highest_compute flag
+    # depth normalised: last node depth 4/4
+    i4 = g.name_to_idx["4"]
+    assert nf[i4, 4] == pytest.approx(1.0)
+
+
+def test_apply_action_mask_false_coerces(tiny_model_files):
+    env = make_env(tiny_model_files, replication=1)
+    env.apply_action_mask = False
+    env.reset(seed=0)
+    obs, r, done, _ = env.step(3)  # odd => invalid => coerced to 0 (block)
+    assert env.cluster.episode_stats["num_jobs_blocked"] == 1
+
+
+def test_apply_action_mask_true_raises(tiny_model_files):
+    env = make_env(tiny_model_files, replication=1)
+    env.reset(seed=0)
+    with pytest.raises(ValueError):
+        env.step(3)
+
+
+def test_reward_variants(tiny_model_files):
+    from ddls_amd.envs.rewards import (JobAcceptance,
+                                       LookaheadJobCompletionTime)
+    seq = (0.02 + 0.03 + 0.04 + 0.06) * 10
+
+    # normalised + log-transformed JCT reward
+    env = make_env(tiny_model_files, replication=1, num_training_steps=10)
+    env.reward_function = LookaheadJobCompletionTime(
+        normaliser="job_sequential_completion_time", transform_with_log=True)
+    env.reset(seed=0)
+    obs, r, done, _ = env.step(1)
+    expected = -math.log(1 + 1.0, 10)  # JCT/seq == 1 for action 1
+    assert r == pytest.approx(expected)
+
+    # inverse
+    env = make_env(tiny_model_files, replication=1, num_training_steps=10)
+    env.reward_function = LookaheadJobCompletionTime(inverse=True, sign=1)
+    env.reset(seed=0)
+    obs, r, done, _ = env.step(1)
+    assert r == pytest.approx(1.0 / seq)
+
+    # acceptance
+    env = make_env(tiny_model_files, replication=1)
+    env.reward_function = JobAcceptance()
+    env.reset(seed=0)
+    obs, r, done, _ = env.step(1)
+    assert r == 1
+    env = make_env(tiny_model_files, replication=1)
+    env.reward_function = JobAcceptance()
+    env.reset(seed=0)
+    obs, r, done, _ = env.step(0)
+    assert r == -1
+
+
+def test_distributions():
+    from ddls_amd.distributions import (CustomSkewNorm, Fixed,
+                                        ListOfDistributions,
+                                        ProbabilityMassFunction, Uniform,
+                                        distribution_from_config)
+    np.random.seed(0)
+    assert Fixed(5).sample() == 5
+    u = Uniform(0.1, 1.0, decimals=2)
+    vals = [u.sample() for _ in range(200)]
+    assert all(0.1 <= v <= 1.0 for v in vals)
+    assert all(round(v, 2) == v for v in vals)
+    pmf = ProbabilityMassFunction({0.25: 0.5, 0.75: 0.5})
+    assert set(pmf.sample(size=50).tolist()) <= {0.25, 0.75}
+    sn = CustomSkewNorm(skewness=-5, min_val=0.1, max_val=1.0, decimals=2,
+                        num_bins=50)
+    v = sn.sample(size=100)
+    assert v.min() >= 0.0 and v.max() <= 1.0
+    lod = ListOfDistributions({"d1": {"ddls_amd.distributions.Fixed": {"val": 3}},
+                               "d2": {"ddls_amd.distributions.Fixed": {"val": 4}}})
+    assert lod.sample().sample() in (3, 4)
+    d = distribution_from_config({"_target_": "ddls_amd.distributions.Fixed",
+                                  "val": 9})
+    assert d.sample() == 9
+
+
+def test_workload_generator_deterministic(tmp_path):
+    from ddls_amd.workloads import generate_default_set
+    d1, d2 = tmp_path / "a", tmp_path / "b"
+    generate_default_set(str(d1))
+    generate_default_set(str(d2))
+    for f in sorted(d1.iterdir()):
+        assert f.read_text() == (d2 / f.name).read_text()
