@@ -299,3 +299,32 @@ def test_job_managers_extra(tiny_model_files):
     assert comm.all_reduce_time(1e9, 1) == 0.0
     t2, t8 = comm.all_reduce_time(1e9, 2), comm.all_reduce_time(1e9, 8)
     assert t2 > 0 and t8 > t2 * 0.5  # per-link bound: weak n-dependence
+
+
+def test_pipeline_cache_trace_equal(tiny_model_files):
+    """Cached empty-cluster pipeline must reproduce the uncached episode
+    exactly (rewards + episode stats)."""
+    def run(cache):
+        env = make_env(tiny_model_files, replication=6, frac_dist={
+            "_target_": "ddls_amd.distributions.Uniform",
+            "min_val": 0.1, "max_val": 1.0, "decimals": 2})
+        env.cache_pipeline = cache
+        obs = env.reset(seed=42)
+        rewards, done = [], False
+        actions = [1, 2, 4, 2, 16, 8]
+        i = 0
+        while not done:
+            a = actions[i % len(actions)]
+            if not obs["action_mask"][a]:
+                a = 0
+            obs, r, done, _ = env.step(a)
+            rewards.append(r)
+            i += 1
+        return rewards, dict(env.cluster.episode_stats)
+
+    r1, s1 = run(cache=False)
+    r2, s2 = run(cache=True)
+    assert r1 == pytest.approx(r2, rel=1e-12)
+    assert s1["num_jobs_completed"] == s2["num_jobs_completed"]
+    assert s1["num_jobs_blocked"] == s2["num_jobs_blocked"]
+    assert s1["job_completion_time"] == pytest.approx(s2["job_completion_time"])
