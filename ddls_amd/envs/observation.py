@@ -15,9 +15,10 @@ import numpy as np
 
 from ..agents.placement_utils import get_block, get_block_shapes, get_factor_pairs
 from . import spaces
+from .base import DDLSObservationFunction
 
 
-class RampJobPartitioningObservation:
+class RampJobPartitioningObservation(DDLSObservationFunction):
     def __init__(self,
                  max_partitions_per_op: int,
                  pad_obs_kwargs: Optional[dict] = None,

@@ -35,6 +35,7 @@ class RampJobPartitioningEnvironment:
                  dep_placer: str = "first_fit_dep_placer",
                  dep_scheduler: str = "srpt_dep_scheduler",
                  observation_function: str = "ramp_job_partitioning_observation",
+                 information_function: str = "default",
                  pad_obs_kwargs: Optional[dict] = None,
                  reward_function: str = "lookahead_job_completion_time",
                  reward_function_kwargs: Optional[dict] = None,
@@ -80,6 +81,8 @@ class RampJobPartitioningEnvironment:
             self.max_partitions_per_op = max_partitions_per_op
         self.min_op_run_time_quantum = min_op_run_time_quantum
 
+        if information_function != "default":
+            raise ValueError(f"Unrecognised information_function {information_function}")
         if observation_function != "ramp_job_partitioning_observation":
             raise ValueError(f"Unrecognised observation_function {observation_function}")
         self.observation_function = RampJobPartitioningObservation(

@@ -11,7 +11,10 @@ import math
 from typing import Union
 
 
-class RewardFunction:
+from .base import DDLSRewardFunction
+
+
+class RewardFunction(DDLSRewardFunction):
     def reset(self, env=None, **kwargs):
         pass
 

@@ -76,3 +76,19 @@ def all_reduce_scalar(x: float, op: str = "sum") -> float:
     if op == "mean":
         t /= get_world_size()
     return float(t.item())
+
+
+def pick_least_used_gpu() -> int:
+    """Index of the GPU with the most free memory (reference
+    ``ml_models/utils.py:134-159`` scraped nvidia-smi; rocm equivalent via
+    torch.cuda.mem_get_info).  Prefer torchrun LOCAL_RANK when present."""
+    if "LOCAL_RANK" in os.environ:
+        return int(os.environ["LOCAL_RANK"])
+    if not torch.cuda.is_available():
+        return 0
+    best, best_free = 0, -1
+    for i in range(torch.cuda.device_count()):
+        free, _total = torch.cuda.mem_get_info(i)
+        if free > best_free:
+            best, best_free = i, free
+    return best
