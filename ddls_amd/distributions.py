@@ -37,6 +37,20 @@ class Uniform(Distribution):
         return v
 
 
+class Exponential(Distribution):
+    """Exponential interarrival times -> Poisson arrival process
+    (BASELINE.json configs[4])."""
+
+    def __init__(self, mean: float):
+        self.mean = mean
+
+    def sample(self, size=None, replace=True):
+        v = np.random.exponential(scale=self.mean, size=size)
+        if size is None:
+            return float(v)
+        return v
+
+
 class ProbabilityMassFunction(Distribution):
     """Discrete PMF mapping value -> probability
     (reference ``distributions/probability_mass_function.py:7``)."""

@@ -54,6 +54,7 @@ class RampJobPartitioningEnvironment:
                  cache_pipeline: bool = True):
         self.cache_pipeline = cache_pipeline
         self._pipeline_cache = {}
+        self._obs_cache = {}
         self.lookahead_memo_preload = lookahead_memo_preload
         self.init_details_memo_preload = init_details_memo_preload
         self.reuse_jobs_generator = reuse_jobs_generator
@@ -113,6 +114,7 @@ class RampJobPartitioningEnvironment:
     # ------------------------------------------------------------------
     def reset(self, seed: Optional[int] = None, verbose: bool = False):
         self.step_counter = 1
+        self._obs_cache = {}  # per-episode: jobs_params can change on reset
         self.cluster.reset(jobs_config=self.jobs_config,
                            max_simulation_run_time=self.max_simulation_run_time,
                            job_queue_capacity=self.job_queue_capacity,
@@ -220,10 +222,27 @@ class RampJobPartitioningEnvironment:
 
         self.done = self._is_done()
         if not self.done:
-            self.obs = self.observation_function.extract(env=self, done=False)
+            self.obs = self._next_obs()
         self.info = {}
         self.step_counter += 1
         return self.obs, self.reward, self.done, self.info
+
+    def _next_obs(self):
+        """Observation of the queued job; on an EMPTY cluster the encoding is
+        fully determined by (model, max_acceptable frac) — cache per job
+        identity fields (the mask and network features are static when
+        nothing is mounted)."""
+        cluster = self.cluster
+        if (not self.cache_pipeline or len(cluster.jobs_running) > 0
+                or cluster.num_mounted_ops != 0):
+            return self.observation_function.extract(env=self, done=False)
+        job = next(iter(cluster.job_queue.jobs.values()))
+        key = (job.details["model"], job.max_acceptable_job_completion_time_frac)
+        obs = self._obs_cache.get(key)
+        if obs is None:
+            obs = self.observation_function.extract(env=self, done=False)
+            self._obs_cache[key] = obs
+        return obs
 
     # ------------------------------------------------------------------
     # empty-cluster pipeline memoisation
