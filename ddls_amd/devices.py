@@ -75,6 +75,23 @@ class A100(Processor):
     memory_capacity = int(80e9)
 
 
+class GPU(Processor):
+    """Generic parametric GPU (reference ``gpus/gpu.py:5-51``): memory /
+    compute-unit count / clock are constructor parameters, so experiments can
+    model arbitrary accelerators without subclassing.  ``device_type`` keys the
+    per-device ``compute_cost`` dicts, as for the named processors."""
+    device_type = "GPU"
+
+    def __init__(self, processor_id=None, memory_capacity: int = int(80e9),
+                 num_compute_units: int = 256, clock_frequency: float = 2.4e9,
+                 device_type: str = "GPU"):
+        self.memory_capacity = int(memory_capacity)
+        self.num_compute_units = num_compute_units
+        self.clock_frequency = clock_frequency
+        self.device_type = device_type
+        super().__init__(processor_id)
+
+
 class Channel:
     """Directed per-link channel (reference ``channel.py:7-42``)."""
 

@@ -163,3 +163,13 @@ def test_poisson_arrivals_env(tiny_model_files):
     stats = env.cluster.episode_stats
     assert max_running >= 2  # overlapping jobs exercised the occupied paths
     assert stats["num_jobs_completed"] + stats["num_jobs_blocked"] > 0
+
+
+def test_generic_gpu_device_and_scheduling_stub():
+    from ddls_amd.devices import GPU
+    from ddls_amd.envs.job_scheduling import JobSchedulingEnvironment
+    g = GPU(processor_id=7, memory_capacity=int(16e9), device_type="V100")
+    assert g.memory_capacity == int(16e9) and g.device_type == "V100"
+    assert str(g) == "V100_7"
+    with pytest.raises(NotImplementedError):
+        JobSchedulingEnvironment()
