@@ -1,0 +1,259 @@
+"""hipGraph-captured PPO SGD minibatch step.
+
+The SGD update is launch-bound on MI355X: one minibatch fwd+bwd+Adam of the
+~1e5-param GNN policy is ~120 tiny kernels, each far cheaper than its host
+dispatch.  This module captures the WHOLE minibatch step (zero-grad -> fused
+HIP GNN forward -> PPO loss -> backward -> grad clip -> capturable Adam ->
+device-side stats accumulation) in a single hipGraph (torch.cuda.CUDAGraph is
+hipGraph on ROCm) and replays it per minibatch with one launch.
+
+Static-shape strategy: graphs vary in node/edge count, so minibatches are
+packed into fixed-capacity buffers with ONE trailing dummy graph (id B) that
+absorbs every padded node and edge — padded edges point at the last padded
+node, padded nodes carry graph id B, and all loss math slices [:B], so
+garbage in the padded region can never reach a real gradient.  Capacities
+grow geometrically with re-capture on overflow.
+
+The reference has no equivalent (RLlib eager torch); this is
+MI355X-native new functionality ("capture launch-bound inner loops in
+hipGraphs").
+
+Distributed: RCCL collectives are kept OUTSIDE the graph — two captured
+segments (fwd+bwd | clip+step) with the eager fused all-reduce between them.
+"""
+from __future__ import annotations
+
+from typing import List, Optional
+
+import numpy as np
+import torch
+import torch.nn.functional as F
+
+from ..models.gnn import GraphBatch
+from ..parallel import all_reduce_gradients, is_distributed
+from .rollout import CompactObs
+
+
+class CapturedSGDStep:
+    """Replayable hipGraph of one PPO SGD minibatch step.
+
+    ``step()`` returns False (caller runs the eager path) when the minibatch
+    is partial, capture is unsupported, or a capture attempt failed.
+    """
+
+    GROWTH = 1.3
+
+    def __init__(self, policy, optimizer, config, device):
+        self.policy = policy
+        self.optimizer = optimizer
+        self.cfg = config
+        self.device = device
+        self.B = config.sgd_minibatch_size
+        self.A = policy.num_actions
+        self.Fn = policy.config["in_features_node"]
+        self.Fe = policy.config["in_features_edge"]
+        self.Fg = policy.config["in_features_graph"] + self.A
+        self.broken = not (device.type == "cuda" and torch.cuda.is_available()
+                           and hasattr(torch.cuda, "CUDAGraph"))
+        self.n_cap = 0
+        self.e_cap = 0
+        self.graph = None
+        self.graph_opt = None   # second segment when distributed
+        # [policy_loss, vf_loss, kl, entropy, total_loss] device accumulator:
+        # read ONCE per update() so replays never host-sync
+        self.stats_acc = torch.zeros(5, device=device)
+        self.kl_coeff_t = torch.zeros((), device=device)
+
+    # ------------------------------------------------------------------
+    def set_kl_coeff(self, v: float):
+        self.kl_coeff_t.fill_(float(v))
+
+    def reset_stats(self):
+        self.stats_acc.zero_()
+
+    # ------------------------------------------------------------------
+    def step(self, mb_obs: List[CompactObs], actions: np.ndarray,
+             old_logp: np.ndarray, adv: np.ndarray,
+             vtarg: np.ndarray) -> bool:
+        if self.broken or len(mb_obs) != self.B:
+            return False
+        n = sum(len(o.node_features) for o in mb_obs)
+        e = sum(len(o.edges_src) for o in mb_obs)
+        if n > self.n_cap or e > self.e_cap or self.graph is None:
+            try:
+                self._capture(max(int(n * self.GROWTH),
+                                  int(self.n_cap * self.GROWTH)),
+                              max(int(e * self.GROWTH),
+                                  int(self.e_cap * self.GROWTH)))
+            except Exception:
+                self.broken = True
+                return False
+        self._fill(mb_obs, actions, old_logp, adv, vtarg)
+        self.graph.replay()
+        if self.graph_opt is not None:
+            all_reduce_gradients(self.policy.parameters())
+            self.graph_opt.replay()
+        return True
+
+    # ------------------------------------------------------------------
+    def _alloc(self, n_cap: int, e_cap: int):
+        dev, B = self.device, self.B
+
+        def dbuf(shape, dtype):
+            return torch.zeros(shape, dtype=dtype, device=dev)
+
+        def pbuf(shape, dtype):
+            return torch.zeros(shape, dtype=dtype, pin_memory=True)
+
+        self.n_cap, self.e_cap = n_cap, e_cap
+        self.d = {
+            "z": dbuf((n_cap, self.Fn), torch.float32),
+            "e": dbuf((e_cap, self.Fe), torch.float32),
+            "src": dbuf(e_cap, torch.int64),
+            "dst": dbuf(e_cap, torch.int64),
+            "order": dbuf(e_cap, torch.int64),
+            "indptr": dbuf(n_cap + 1, torch.int64),
+            "gon": dbuf(n_cap, torch.int64),
+            "gf": dbuf((B + 1, self.Fg), torch.float32),
+            "mask": torch.ones((B + 1, self.A), dtype=torch.float32, device=dev),
+            "actions": dbuf(B, torch.int64),
+            "old_logp": dbuf(B, torch.float32),
+            "adv": dbuf(B, torch.float32),
+            "vtarg": dbuf(B, torch.float32),
+        }
+        self.p = {k: pbuf(tuple(v.shape), v.dtype) for k, v in self.d.items()
+                  if k != "mask"}
+        self.np_ = {k: v.numpy() for k, v in self.p.items()}
+        # one static flat graph: B real graphs + dummy graph B for padding
+        self.batch = GraphBatch(
+            z=self.d["z"], e=self.d["e"], src=self.d["src"], dst=self.d["dst"],
+            graph_of_node=self.d["gon"], num_graphs=B + 1,
+            _csr=(self.d["order"], self.d["indptr"]))
+
+    def _fill(self, mb_obs, actions, old_logp, adv, vtarg):
+        ns = np.array([len(o.node_features) for o in mb_obs], dtype=np.int64)
+        offsets = np.concatenate([[0], np.cumsum(ns)[:-1]])
+        n = int(ns.sum())
+        z = np.concatenate([o.node_features for o in mb_obs])
+        e = np.concatenate([o.edge_features for o in mb_obs])
+        src = np.concatenate([o.edges_src + off
+                              for o, off in zip(mb_obs, offsets)])
+        dst = np.concatenate([o.edges_dst + off
+                              for o, off in zip(mb_obs, offsets)])
+        m = len(src)
+        p = self.np_
+        p["z"][:n] = z
+        p["e"][:m] = e
+        p["src"][:m] = src
+        p["src"][m:] = self.n_cap - 1
+        p["dst"][:m] = dst
+        p["dst"][m:] = self.n_cap - 1
+        # padded edges already sit at the max node id, so only real edges need
+        # sorting; padded ones are appended in order
+        p["order"][:m] = np.argsort(dst, kind="stable")
+        p["order"][m:] = np.arange(m, self.e_cap)
+        counts = np.bincount(dst, minlength=self.n_cap)
+        counts[self.n_cap - 1] += self.e_cap - m
+        p["indptr"][0] = 0
+        np.cumsum(counts, out=p["indptr"][1:])
+        p["gon"][:n] = np.repeat(np.arange(self.B, dtype=np.int64), ns)
+        p["gon"][n:] = self.B
+        p["gf"][:self.B] = np.stack([o.graph_features for o in mb_obs])
+        # row B of gf stays zero; mask stays all-ones (log(1)=0 on device)
+        p["actions"][:] = actions
+        p["old_logp"][:] = old_logp
+        p["adv"][:] = adv
+        p["vtarg"][:] = vtarg
+        for k, dst_t in self.d.items():
+            if k != "mask":
+                dst_t.copy_(self.p[k], non_blocking=True)
+
+    # ------------------------------------------------------------------
+    def _body_fwd_bwd(self):
+        cfg, B = self.cfg, self.B
+        grads = [pa.grad for pa in self.policy.parameters()
+                 if pa.grad is not None]
+        if grads:
+            torch._foreach_zero_(grads)
+        logits, values = self.policy.forward_flat(
+            self.batch, self.d["gf"], self.d["mask"])
+        logits, values = logits[:B], values[:B]
+        # Categorical re-implemented with log_softmax: torch.distributions
+        # argument validation host-syncs, which aborts stream capture
+        logp_all = F.log_softmax(logits, dim=-1)
+        logp = logp_all.gather(1, self.d["actions"].unsqueeze(1)).squeeze(1)
+        ratio = torch.exp(logp - self.d["old_logp"])
+        adv = self.d["adv"]
+        surr = torch.min(ratio * adv,
+                         torch.clamp(ratio, 1 - cfg.clip_param,
+                                     1 + cfg.clip_param) * adv)
+        policy_loss = -surr.mean()
+        kl = (self.d["old_logp"] - logp).mean()
+        vf_err = (values - self.d["vtarg"]) ** 2
+        vf_loss = torch.clamp(vf_err, 0, cfg.vf_clip_param).mean()
+        entropy = (-(logp_all.exp() * logp_all).sum(-1)).mean()
+        loss = (policy_loss + self.kl_coeff_t * kl
+                + cfg.vf_loss_coeff * vf_loss - cfg.entropy_coeff * entropy)
+        loss.backward()
+        self.stats_acc += torch.stack([policy_loss.detach(), vf_loss.detach(),
+                                       kl.detach(), entropy.detach(),
+                                       loss.detach()])
+
+    def _body_opt(self):
+        if self.cfg.grad_clip is not None:
+            torch.nn.utils.clip_grad_norm_(self.policy.parameters(),
+                                           self.cfg.grad_clip)
+        self.optimizer.step()
+
+    def _capture(self, n_cap: int, e_cap: int):
+        self._alloc(n_cap, e_cap)
+        split = is_distributed()
+
+        def whole():
+            self._body_fwd_bwd()
+            if not split:
+                self._body_opt()
+
+        # warmup on a side stream (required before capture); lr=0 so the real
+        # optimizer steps during warmup cannot move the parameters, then the
+        # Adam state is zeroed in-place so training resumes bit-identical to
+        # an uncaptured run
+        saved_lr = [g["lr"] for g in self.optimizer.param_groups]
+        for g in self.optimizer.param_groups:
+            g["lr"] = 0.0
+            # capturable Adam keeps `step` on-device so the whole update is
+            # graph-safe; migrate any state created by earlier eager steps
+            g["capturable"] = True
+        for st in self.optimizer.state.values():
+            if "step" in st and torch.is_tensor(st["step"]):
+                st["step"] = st["step"].to(self.device)
+        saved_stats = self.stats_acc.clone()
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(3):
+                whole()
+                if split:
+                    self._body_opt()
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        for g, lr in zip(self.optimizer.param_groups, saved_lr):
+            g["lr"] = lr
+        for st in self.optimizer.state.values():
+            for v in st.values():
+                if torch.is_tensor(v) and v.is_floating_point():
+                    v.zero_()
+                elif torch.is_tensor(v):
+                    v.zero_()
+        self.stats_acc.copy_(saved_stats)
+
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            whole()
+        if split:
+            self.graph_opt = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.graph_opt):
+                self._body_opt()
+        # discard warmup/capture side effects on the accumulator
+        self.stats_acc.copy_(saved_stats)
+        torch.cuda.synchronize()

@@ -39,6 +39,9 @@ class PPOConfig:
     sgd_minibatch_size: int = 128
     train_batch_size: int = 4000
     num_sgd_iter: int = 50
+    # capture the SGD minibatch step in a hipGraph on GPU (graph_step.py);
+    # eager fallback on CPU / partial minibatches / capture failure
+    use_hip_graphs: bool = True
 
 
 class PPOTrainer:
@@ -58,6 +61,7 @@ class PPOTrainer:
         self.obs = self.env.reset()
         self.total_env_steps = 0
         self.iteration = 0
+        self._stepper = None  # lazy CapturedSGDStep (GPU only)
 
     # ------------------------------------------------------------------
     @torch.no_grad()
@@ -179,6 +183,25 @@ class PPOTrainer:
         stats = {"policy_loss": 0.0, "vf_loss": 0.0, "kl": 0.0, "entropy": 0.0,
                  "total_loss": 0.0}
         num_updates = 0
+        num_captured = 0
+
+        # hipGraph-captured minibatch step (GPU): one replay per minibatch
+        # instead of ~120 host-dispatched kernel launches
+        stepper = None
+        if (cfg.use_hip_graphs and self.device.type == "cuda"
+                and torch.cuda.is_available()):
+            if self._stepper is None:
+                from .graph_step import CapturedSGDStep
+                self._stepper = CapturedSGDStep(self.policy, self.optimizer,
+                                                cfg, self.device)
+            stepper = self._stepper
+            stepper.reset_stats()
+            stepper.set_kl_coeff(self.kl_coeff)
+        actions_np = np.asarray(batch["actions"], dtype=np.int64)
+        logp_np = np.asarray(batch["logp"], dtype=np.float32)
+        adv_np = np.asarray(adv, dtype=np.float32)
+        vtarg_np = np.asarray(batch["value_targets"], dtype=np.float32)
+
         rng = np.random.RandomState(self.iteration + 1234 * get_rank())
         for _ in range(cfg.num_sgd_iter):
             perm = rng.permutation(n)
@@ -187,6 +210,12 @@ class PPOTrainer:
                 if len(idx) < 2:
                     continue
                 mb_obs = [batch["obs"][i] for i in idx]
+                if stepper is not None and stepper.step(
+                        mb_obs, actions_np[idx], logp_np[idx], adv_np[idx],
+                        vtarg_np[idx]):
+                    num_updates += 1
+                    num_captured += 1
+                    continue
                 inputs = collate(mb_obs, self.device)
                 logits, values = self._forward_flat(inputs)
                 dist = torch.distributions.Categorical(logits=logits)
@@ -227,6 +256,11 @@ class PPOTrainer:
                 stats["total_loss"] += loss.detach().item()
                 num_updates += 1
 
+        if num_captured:
+            acc = self._stepper.stats_acc.cpu().numpy()  # one sync per update
+            for k, v in zip(("policy_loss", "vf_loss", "kl", "entropy",
+                             "total_loss"), acc):
+                stats[k] += float(v)
         for k in stats:
             stats[k] /= max(num_updates, 1)
 
@@ -281,6 +315,9 @@ class PPOTrainer:
     def load_state_dict(self, state: Dict):
         self.policy.load_state_dict(state["policy"])
         self.optimizer.load_state_dict(state["optimizer"])
+        # optimizer state tensors were replaced; any captured graph holds
+        # stale addresses, so force a re-capture
+        self._stepper = None
         self.kl_coeff = state.get("kl_coeff", self.config.kl_coeff)
         self.iteration = state.get("iteration", 0)
         self.total_env_steps = state.get("total_env_steps", 0)

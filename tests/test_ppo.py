@@ -155,3 +155,108 @@ def test_worker_side_rollouts(tiny_model_files):
         assert full["env_steps_this_iter"] == 24
     finally:
         venv.close()
+
+
+def _random_compact_obs(rng, num_actions=17):
+    from ddls_amd.rl.rollout import CompactObs
+    n = int(rng.randint(4, 20))
+    m = int(rng.randint(3, 3 * n))
+    mask = np.ones(num_actions, dtype=np.float32)
+    mask[rng.randint(1, num_actions)] = 0
+    return CompactObs(
+        node_features=rng.rand(n, 5).astype(np.float32),
+        edge_features=rng.rand(m, 2).astype(np.float32),
+        edges_src=rng.randint(0, n, size=m).astype(np.int64),
+        edges_dst=rng.randint(0, n, size=m).astype(np.int64),
+        graph_features=rng.rand(17 + num_actions).astype(np.float32),
+        action_mask=mask)
+
+
+@pytest.mark.gpu
+def test_hipgraph_captured_sgd_matches_eager():
+    """One hipGraph-captured SGD minibatch step == the eager step (same data,
+    same init) within atomics tolerance; a second step at a different
+    minibatch geometry re-captures and stays finite."""
+    import torch.nn.functional as F
+    from ddls_amd.models.gnn import GNNPolicy
+    from ddls_amd.rl.graph_step import CapturedSGDStep
+    from ddls_amd.rl.ppo import PPOConfig
+    from ddls_amd.rl.rollout import collate
+
+    device = torch.device("cuda:0")
+    B = 8
+    cfg = PPOConfig(sgd_minibatch_size=B, num_sgd_iter=1)
+    rng = np.random.RandomState(0)
+    mb_obs = [_random_compact_obs(rng) for _ in range(B)]
+    actions = rng.randint(0, 1, size=B).astype(np.int64)  # action 0 valid
+    old_logp = rng.randn(B).astype(np.float32) * 0.1 - 2.0
+    adv = rng.randn(B).astype(np.float32)
+    vtarg = rng.randn(B).astype(np.float32)
+
+    def make():
+        torch.manual_seed(3)
+        pol = GNNPolicy(num_actions=17).to(device)
+        opt = torch.optim.Adam(pol.parameters(), lr=cfg.lr)
+        return pol, opt
+
+    # eager reference step (the exact update() math)
+    pol_e, opt_e = make()
+    inputs = collate(mb_obs, device)
+    logits, values = pol_e.forward_flat(inputs["batch"],
+                                        inputs["graph_features"],
+                                        inputs["action_mask"])
+    logp_all = F.log_softmax(logits, dim=-1)
+    logp = logp_all.gather(1, torch.as_tensor(actions, device=device)
+                           .unsqueeze(1)).squeeze(1)
+    ratio = torch.exp(logp - torch.as_tensor(old_logp, device=device))
+    adv_t = torch.as_tensor(adv, device=device)
+    surr = torch.min(ratio * adv_t,
+                     torch.clamp(ratio, 1 - cfg.clip_param,
+                                 1 + cfg.clip_param) * adv_t)
+    policy_loss = -surr.mean()
+    kl = (torch.as_tensor(old_logp, device=device) - logp).mean()
+    vf_loss = torch.clamp(
+        (values - torch.as_tensor(vtarg, device=device)) ** 2,
+        0, cfg.vf_clip_param).mean()
+    entropy = (-(logp_all.exp() * logp_all).sum(-1)).mean()
+    loss = (policy_loss + cfg.kl_coeff * kl + cfg.vf_loss_coeff * vf_loss
+            - cfg.entropy_coeff * entropy)
+    opt_e.zero_grad()
+    loss.backward()
+    torch.nn.utils.clip_grad_norm_(pol_e.parameters(), cfg.grad_clip)
+    opt_e.step()
+
+    # captured step
+    pol_c, opt_c = make()
+    stepper = CapturedSGDStep(pol_c, opt_c, cfg, device)
+    stepper.set_kl_coeff(cfg.kl_coeff)
+    assert stepper.step(mb_obs, actions, old_logp, adv, vtarg), \
+        "hipGraph capture failed (fell back)"
+    torch.cuda.synchronize()
+    assert not stepper.broken
+
+    acc = stepper.stats_acc.cpu().numpy()
+    assert np.isfinite(acc).all()
+    assert acc[0] == pytest.approx(policy_loss.item(), abs=1e-3)
+    assert acc[4] == pytest.approx(loss.item(), abs=1e-3)
+    for (n1, p1), (n2, p2) in zip(pol_e.named_parameters(),
+                                  pol_c.named_parameters()):
+        assert torch.allclose(p1, p2, rtol=1e-3, atol=1e-5), n1
+
+    # different geometry -> growth + re-capture, still sane
+    mb2 = [_random_compact_obs(rng) for _ in range(B - 1)]
+    mb2.append(_random_compact_obs(np.random.RandomState(99)))
+    big = _random_compact_obs(np.random.RandomState(7))
+    big.node_features = np.random.rand(60, 5).astype(np.float32)
+    big.edge_features = np.random.rand(150, 2).astype(np.float32)
+    big.edges_src = np.random.randint(0, 60, 150).astype(np.int64)
+    big.edges_dst = np.random.randint(0, 60, 150).astype(np.int64)
+    mb2[0] = big
+    assert stepper.step(mb2, actions, old_logp, adv, vtarg)
+    torch.cuda.synchronize()
+    for _n, p in pol_c.named_parameters():
+        assert torch.isfinite(p).all()
+
+    # partial minibatch falls back
+    assert not stepper.step(mb2[:B - 1], actions[:B - 1], old_logp[:B - 1],
+                            adv[:B - 1], vtarg[:B - 1])
