@@ -130,9 +130,15 @@ def main():
         st = trainer.train(num_steps=args.rollout_steps_per_env)
         if rank == 0:
             print(f"[bench] iter {st['iteration']}: rollout "
-                  f"{st['rollout_time_s']:.3f}s update {st['update_time_s']:.3f}s",
+                  f"{st['rollout_time_s']:.3f}s update {st['update_time_s']:.3f}s "
+                  f"hipgraph_mb={st.get('hipgraph_minibatches', 0)} "
+                  f"captures={st.get('hipgraph_captures', 0)}",
                   file=sys.stderr, flush=True)
     barrier_sync()
+    if rank == 0 and getattr(trainer, "_stepper", None) is not None:
+        sp = trainer._stepper
+        print(f"[bench] hipgraph: broken={sp.broken} captures={sp.capture_count} "
+              f"err={sp.last_error}", file=sys.stderr, flush=True)
     elapsed = time.perf_counter() - t0
 
     # max over ranks

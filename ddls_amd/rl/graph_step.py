@@ -59,6 +59,8 @@ class CapturedSGDStep:
         self.e_cap = 0
         self.graph = None
         self.graph_opt = None   # second segment when distributed
+        self.capture_count = 0
+        self.last_error: Optional[str] = None
         # [policy_loss, vf_loss, kl, entropy, total_loss] device accumulator:
         # read ONCE per update() so replays never host-sync
         self.stats_acc = torch.zeros(5, device=device)
@@ -72,6 +74,29 @@ class CapturedSGDStep:
         self.stats_acc.zero_()
 
     # ------------------------------------------------------------------
+    def ensure_capacity(self, n_cap: int, e_cap: int) -> bool:
+        """(Re-)capture if the buffers cannot hold n_cap nodes / e_cap edges.
+        Call once per update with B x the rollout's max per-sample counts so
+        the minibatch loop never recaptures."""
+        if self.broken:
+            return False
+        if self.graph is not None and n_cap <= self.n_cap and e_cap <= self.e_cap:
+            return True
+        try:
+            self._capture(max(n_cap, int(self.n_cap * self.GROWTH)),
+                          max(e_cap, int(self.e_cap * self.GROWTH)))
+            return True
+        except Exception as exc:  # unsupported torch/ROCm combo -> eager
+            self.broken = True
+            self.last_error = f"{type(exc).__name__}: {exc}"
+            # capturable eager Adam is slow; revert it for the fallback path
+            for g in self.optimizer.param_groups:
+                g["capturable"] = False
+            for st in self.optimizer.state.values():
+                if "step" in st and torch.is_tensor(st["step"]):
+                    st["step"] = st["step"].cpu()
+            return False
+
     def step(self, mb_obs: List[CompactObs], actions: np.ndarray,
              old_logp: np.ndarray, adv: np.ndarray,
              vtarg: np.ndarray) -> bool:
@@ -79,15 +104,8 @@ class CapturedSGDStep:
             return False
         n = sum(len(o.node_features) for o in mb_obs)
         e = sum(len(o.edges_src) for o in mb_obs)
-        if n > self.n_cap or e > self.e_cap or self.graph is None:
-            try:
-                self._capture(max(int(n * self.GROWTH),
-                                  int(self.n_cap * self.GROWTH)),
-                              max(int(e * self.GROWTH),
-                                  int(self.e_cap * self.GROWTH)))
-            except Exception:
-                self.broken = True
-                return False
+        if not self.ensure_capacity(n, e):
+            return False
         self._fill(mb_obs, actions, old_logp, adv, vtarg)
         self.graph.replay()
         if self.graph_opt is not None:
@@ -121,8 +139,8 @@ class CapturedSGDStep:
             "adv": dbuf(B, torch.float32),
             "vtarg": dbuf(B, torch.float32),
         }
-        self.p = {k: pbuf(tuple(v.shape), v.dtype) for k, v in self.d.items()
-                  if k != "mask"}
+        self.p = {k: pbuf(tuple(v.shape), v.dtype) for k, v in self.d.items()}
+        self.p["mask"].fill_(1.0)
         self.np_ = {k: v.numpy() for k, v in self.p.items()}
         # one static flat graph: B real graphs + dummy graph B for padding
         self.batch = GraphBatch(
@@ -159,14 +177,15 @@ class CapturedSGDStep:
         p["gon"][:n] = np.repeat(np.arange(self.B, dtype=np.int64), ns)
         p["gon"][n:] = self.B
         p["gf"][:self.B] = np.stack([o.graph_features for o in mb_obs])
-        # row B of gf stays zero; mask stays all-ones (log(1)=0 on device)
+        p["mask"][:self.B] = np.stack([o.action_mask for o in mb_obs])
+        # row B of gf stays zero and of mask stays all-ones (log(1)=0), so the
+        # dummy padding graph has finite logits
         p["actions"][:] = actions
         p["old_logp"][:] = old_logp
         p["adv"][:] = adv
         p["vtarg"][:] = vtarg
         for k, dst_t in self.d.items():
-            if k != "mask":
-                dst_t.copy_(self.p[k], non_blocking=True)
+            dst_t.copy_(self.p[k], non_blocking=True)
 
     # ------------------------------------------------------------------
     def _body_fwd_bwd(self):
@@ -215,9 +234,10 @@ class CapturedSGDStep:
                 self._body_opt()
 
         # warmup on a side stream (required before capture); lr=0 so the real
-        # optimizer steps during warmup cannot move the parameters, then the
-        # Adam state is zeroed in-place so training resumes bit-identical to
-        # an uncaptured run
+        # optimizer steps during warmup cannot move the parameters, and the
+        # Adam state is snapshot/restored in-place (same tensor addresses, so
+        # the captured graph sees the live state) so training resumes exactly
+        # where an uncaptured run would be
         saved_lr = [g["lr"] for g in self.optimizer.param_groups]
         for g in self.optimizer.param_groups:
             g["lr"] = 0.0
@@ -227,23 +247,33 @@ class CapturedSGDStep:
         for st in self.optimizer.state.values():
             if "step" in st and torch.is_tensor(st["step"]):
                 st["step"] = st["step"].to(self.device)
+        snap = {p_: {k: v.clone() if torch.is_tensor(v) else v
+                     for k, v in st.items()}
+                for p_, st in self.optimizer.state.items()}
         saved_stats = self.stats_acc.clone()
-        s = torch.cuda.Stream()
-        s.wait_stream(torch.cuda.current_stream())
-        with torch.cuda.stream(s):
-            for _ in range(3):
-                whole()
-                if split:
-                    self._body_opt()
-        torch.cuda.current_stream().wait_stream(s)
-        torch.cuda.synchronize()
-        for g, lr in zip(self.optimizer.param_groups, saved_lr):
-            g["lr"] = lr
-        for st in self.optimizer.state.values():
-            for v in st.values():
-                if torch.is_tensor(v) and v.is_floating_point():
-                    v.zero_()
-                elif torch.is_tensor(v):
+        try:
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(3):
+                    whole()
+                    if split:
+                        self._body_opt()
+            torch.cuda.current_stream().wait_stream(s)
+            torch.cuda.synchronize()
+        finally:
+            for g, lr in zip(self.optimizer.param_groups, saved_lr):
+                g["lr"] = lr
+        # undo warmup's pollution of the optimizer state: restore snapshotted
+        # entries, zero entries first created by warmup (fresh-Adam semantics)
+        for p_, st in self.optimizer.state.items():
+            prev = snap.get(p_)
+            for k, v in st.items():
+                if not torch.is_tensor(v):
+                    continue
+                if prev is not None and k in prev:
+                    v.copy_(prev[k])
+                else:
                     v.zero_()
         self.stats_acc.copy_(saved_stats)
 
@@ -256,4 +286,5 @@ class CapturedSGDStep:
                 self._body_opt()
         # discard warmup/capture side effects on the accumulator
         self.stats_acc.copy_(saved_stats)
+        self.capture_count += 1
         torch.cuda.synchronize()

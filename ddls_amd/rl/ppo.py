@@ -197,6 +197,12 @@ class PPOTrainer:
             stepper = self._stepper
             stepper.reset_stats()
             stepper.set_kl_coeff(self.kl_coeff)
+            # pre-size capacity to B x the rollout's max per-sample counts so
+            # the minibatch loop never recaptures
+            max_n = max(len(o.node_features) for o in batch["obs"])
+            max_e = max(len(o.edges_src) for o in batch["obs"])
+            stepper.ensure_capacity(cfg.sgd_minibatch_size * max_n,
+                                    cfg.sgd_minibatch_size * max_e)
         actions_np = np.asarray(batch["actions"], dtype=np.int64)
         logp_np = np.asarray(batch["logp"], dtype=np.float32)
         adv_np = np.asarray(adv, dtype=np.float32)
@@ -270,6 +276,9 @@ class PPOTrainer:
         elif stats["kl"] < 0.5 * cfg.kl_target:
             self.kl_coeff *= 0.5
         stats["kl_coeff"] = self.kl_coeff
+        if stepper is not None:
+            stats["hipgraph_minibatches"] = num_captured
+            stats["hipgraph_captures"] = stepper.capture_count
         return stats
 
     # ------------------------------------------------------------------
