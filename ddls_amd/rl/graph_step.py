@@ -42,6 +42,11 @@ class CapturedSGDStep:
     """
 
     GROWTH = 1.3
+    # dummy-node slots always kept free beyond the real-node capacity: padded
+    # edges are spread across MANY dummy nodes — a single mega-segment would
+    # serialize the wave-per-node message_reduce forward and the backward's
+    # ghn atomics
+    RESERVE = 256
 
     def __init__(self, policy, optimizer, config, device):
         self.policy = policy
@@ -80,10 +85,12 @@ class CapturedSGDStep:
         the minibatch loop never recaptures."""
         if self.broken:
             return False
-        if self.graph is not None and n_cap <= self.n_cap and e_cap <= self.e_cap:
+        if (self.graph is not None and n_cap <= self.n_cap - self.RESERVE
+                and e_cap <= self.e_cap):
             return True
         try:
-            self._capture(max(n_cap, int(self.n_cap * self.GROWTH)),
+            self._capture(max(n_cap + self.RESERVE,
+                              int(self.n_cap * self.GROWTH)),
                           max(e_cap, int(self.e_cap * self.GROWTH)))
             return True
         except Exception as exc:  # unsupported torch/ROCm combo -> eager
@@ -104,7 +111,7 @@ class CapturedSGDStep:
             return False
         n = sum(len(o.node_features) for o in mb_obs)
         e = sum(len(o.edges_src) for o in mb_obs)
-        if not self.ensure_capacity(n, e):
+        if not self.ensure_capacity(n, e) or n > self.n_cap - self.RESERVE:
             return False
         self._fill(mb_obs, actions, old_logp, adv, vtarg)
         self.graph.replay()
@@ -163,15 +170,21 @@ class CapturedSGDStep:
         p["z"][:n] = z
         p["e"][:m] = e
         p["src"][:m] = src
-        p["src"][m:] = self.n_cap - 1
         p["dst"][:m] = dst
-        p["dst"][m:] = self.n_cap - 1
-        # padded edges already sit at the max node id, so only real edges need
-        # sorting; padded ones are appended in order
+        # spread padded edges monotonically over the dummy nodes [n, n_cap) as
+        # self-loops: short per-node segments, no serial mega-segment and no
+        # atomic hot spot in the backward
+        pad_e = self.e_cap - m
+        pad_nodes = self.n_cap - n
+        pad_dst = n + (np.arange(pad_e, dtype=np.int64) * pad_nodes) // max(pad_e, 1)
+        p["src"][m:] = pad_dst
+        p["dst"][m:] = pad_dst
+        # padded edges sit at node ids above every real dst and are already
+        # sorted, so only real edges need sorting
         p["order"][:m] = np.argsort(dst, kind="stable")
         p["order"][m:] = np.arange(m, self.e_cap)
         counts = np.bincount(dst, minlength=self.n_cap)
-        counts[self.n_cap - 1] += self.e_cap - m
+        counts += np.bincount(pad_dst, minlength=self.n_cap)
         p["indptr"][0] = 0
         np.cumsum(counts, out=p["indptr"][1:])
         p["gon"][:n] = np.repeat(np.arange(self.B, dtype=np.int64), ns)
