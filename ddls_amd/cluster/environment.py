@@ -317,6 +317,49 @@ class RampClusterEnvironment:
             self.job_queue.jobs[job_id] = op_partition.partitioned_jobs[job_id]
 
     def _place_ops(self, op_placement):
+        fast = getattr(op_placement, "fast_mount", None)
+        if fast is not None and len(op_placement.action) == 1:
+            # vectorised mount from the cached pipeline: dense-array state +
+            # one rule/memory check per worker instead of per op.  The
+            # per-(job,op) worker priority table is NOT filled here — the
+            # dense job.op_priority array is the source of truth for conflict
+            # resolution and Processor.unmount pops with a default.
+            (job_id, placement), = op_placement.action.items()
+            job = self.job_queue.jobs[job_id]
+            g = job.graph
+            job_idx = job.details["job_idx"]
+            op_idx_arr, widx_arr, op_groups = fast
+            job.op_worker = np.full(g.n, -1, dtype=np.int64)
+            job.op_priority = np.zeros(g.n, dtype=np.int64)
+            job.op_worker[op_idx_arr] = widx_arr
+            for worker_id, ops_w, mem_sum in op_groups:
+                node = self.topology.worker_to_node[worker_id]
+                worker = self.topology.node_workers[node][worker_id]
+                broken = check_if_ramp_op_placement_rules_broken(worker, job)
+                if broken:
+                    raise RuntimeError(
+                        f"placement for job {job_id} on {worker_id} "
+                        f"breaks RAMP rules: {broken}")
+                if worker.memory_occupied + mem_sum > worker.memory_capacity:
+                    raise MemoryError(
+                        f"allocating {mem_sum} B for job {job.job_id} but only "
+                        f"{worker.memory_capacity - worker.memory_occupied} B "
+                        f"free on {worker.processor_id}")
+                worker.mounted_job_idx_to_ops[job_idx].update(
+                    int(o) for o in ops_w)
+                worker.mounted_job_idx_to_job_id[job_idx] = job.job_id
+                worker.memory_occupied += mem_sum
+                job.details["mounted_workers"].add(worker_id)
+                dt = worker.device_type
+                job.op_remaining[ops_w] = g.compute_cost[dt][ops_w]
+                for o in ops_w:
+                    job.mounted_device_type[o] = dt
+                self.job_op_to_worker.update(
+                    ((job_idx, int(o)), worker_id) for o in ops_w)
+            self.num_mounted_ops += len(op_idx_arr)
+            self._register_running_job(job)
+            self.job_op_placement[job_id] = placement
+            return
         for job_id, placement in op_placement.action.items():
             job = self.job_queue.jobs[job_id]
             g = job.graph
@@ -354,6 +397,28 @@ class RampClusterEnvironment:
         job.dep_is_flow = job.dep_cross_node & (g.size > 0)
 
     def _place_deps(self, dep_placement):
+        fast = getattr(dep_placement, "fast_mount", None)
+        if fast is not None and len(dep_placement.action) == 1:
+            # vectorised mount from the cached pipeline (single-channel deps)
+            (job_id, deps), = dep_placement.action.items()
+            job = self.jobs_running[self.job_id_to_job_idx[job_id]]
+            dep_idxs, chan_idxs, channel_counts = fast
+            if len(dep_idxs):
+                job.dep_channel_idx[dep_idxs] = chan_idxs
+                job.dep_remaining[dep_idxs] = job.dep_init_run_time[dep_idxs]
+                for channel_id, count in channel_counts:
+                    channel = self.topology.channel_id_to_channel[channel_id]
+                    broken = check_if_ramp_dep_placement_rules_broken(channel,
+                                                                      job)
+                    if broken:
+                        raise RuntimeError(
+                            f"dep placement for job {job_id} on {channel_id} "
+                            f"breaks RAMP rules: {broken}")
+                    channel.mount(job, -1, count=count)
+                    job.details["mounted_channels"].add(channel_id)
+                self.num_mounted_deps += len(dep_idxs)
+            self.job_dep_placement[job_id] = deps
+            return
         for job_id, deps in dep_placement.action.items():
             job_idx = self.job_id_to_job_idx[job_id]
             job = self.jobs_running[job_idx]
@@ -384,6 +449,17 @@ class RampClusterEnvironment:
             self.job_dep_placement[job_id] = deps
 
     def _schedule_ops(self, op_schedule):
+        fast = getattr(op_schedule, "fast_priorities", None)
+        if fast is not None and len(op_schedule.action) > 0:
+            job_ids = {jid for jobs in op_schedule.action.values()
+                       for jid in jobs}
+            if len(job_ids) == 1:
+                job_idx = self.job_id_to_job_idx[next(iter(job_ids))]
+                job = self.jobs_running.get(job_idx)
+                if job is not None:
+                    idxs, prios = fast
+                    job.op_priority[idxs] = prios
+                    return
         for worker_id, job_to_ops in op_schedule.action.items():
             node = self.topology.worker_to_node[worker_id]
             worker = self.topology.node_workers[node][worker_id]
@@ -397,6 +473,17 @@ class RampClusterEnvironment:
                     job.op_priority[op_idx] = priority
 
     def _schedule_deps(self, dep_schedule):
+        fast = getattr(dep_schedule, "fast_priorities", None)
+        if fast is not None and len(dep_schedule.action) > 0:
+            job_ids = {jid for jobs in dep_schedule.action.values()
+                       for jid in jobs}
+            if len(job_ids) == 1:
+                job_idx = self.job_id_to_job_idx[next(iter(job_ids))]
+                job = self.jobs_running.get(job_idx)
+                if job is not None:
+                    idxs, prios = fast
+                    job.dep_priority[idxs] = prios
+                    return
         for channel_id, job_to_deps in dep_schedule.action.items():
             if channel_id is None:
                 continue
@@ -416,14 +503,34 @@ class RampClusterEnvironment:
         job_idx = job.details["job_idx"]
         g = job.graph
         mounted_ops = (np.flatnonzero(job.op_worker >= 0)
-                       if hasattr(job, "op_worker") else range(g.n))
-        for op_idx in mounted_ops:
-            key = (job_idx, int(op_idx))
-            worker_id = self.job_op_to_worker.pop(key, None)
-            if worker_id is not None:
-                node = self.topology.worker_to_node[worker_id]
-                self.topology.node_workers[node][worker_id].unmount(job, int(op_idx))
-                self.num_mounted_ops -= 1
+                       if hasattr(job, "op_worker") else None)
+        if mounted_ops is not None:
+            # vectorised unmount: one memory/bookkeeping update per worker
+            # (RAMP one-job-per-worker) instead of per op
+            widx = job.op_worker[mounted_ops]
+            for w in np.unique(widx):
+                ops_w = mounted_ops[widx == w]
+                worker = self.workers[int(w)]
+                worker.memory_occupied -= float(g.memory_cost[ops_w].sum())
+                if job_idx in worker.mounted_job_idx_to_ops:
+                    del worker.mounted_job_idx_to_ops[job_idx]
+                worker.mounted_job_idx_to_job_id.pop(job_idx, None)
+                if worker.mounted_job_op_to_priority:
+                    for o in ops_w:
+                        worker.mounted_job_op_to_priority.pop(
+                            (job_idx, int(o)), None)
+                for o in ops_w:
+                    self.job_op_to_worker.pop((job_idx, int(o)), None)
+            self.num_mounted_ops -= int(len(mounted_ops))
+        else:
+            for op_idx in range(g.n):
+                key = (job_idx, int(op_idx))
+                worker_id = self.job_op_to_worker.pop(key, None)
+                if worker_id is not None:
+                    node = self.topology.worker_to_node[worker_id]
+                    self.topology.node_workers[node][worker_id].unmount(
+                        job, int(op_idx))
+                    self.num_mounted_ops -= 1
         if hasattr(job, "dep_channel_idx"):
             mounted_deps = job.dep_channel_idx[job.dep_channel_idx >= 0]
             if len(mounted_deps) > 0:

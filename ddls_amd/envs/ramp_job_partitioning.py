@@ -259,11 +259,65 @@ class RampJobPartitioningEnvironment:
         dep_sched = {cid: dict(jobs[job_id])
                      for cid, jobs in self.dep_schedule.action.items()
                      if job_id in jobs}
+        # vectorised dep-mount arrays: (dep idxs, global channel idxs, counted
+        # per-channel mounts) — valid only when every placed dep uses exactly
+        # one channel (always true on the full-mesh RAMP: paths are one hop)
+        dpd = dict(self.dep_placement.action.get(job_id, {}))
+        fast = None
+        dep_idxs, chan_ids = [], []
+        single_channel = True
+        for dep_idx, channel_ids in dpd.items():
+            real = [c for c in channel_ids if c is not None]
+            if not real:
+                continue
+            if len(real) != 1:
+                single_channel = False
+                break
+            dep_idxs.append(dep_idx)
+            chan_ids.append(real[0])
+        if single_channel:
+            from collections import Counter as _Counter
+            fast = (_np.asarray(dep_idxs, dtype=_np.int64),
+                    _np.asarray([self.cluster.channel_index(c)
+                                 for c in chan_ids], dtype=_np.int64),
+                    list(_Counter(chan_ids).items()))
+        # vectorised op-mount arrays: dense op/worker indices + per-worker
+        # groups for the one-rule-check-per-worker mount
+        g = pj.graph
+        placement = dict(self.op_placement.action[job_id])
+        op_idx_arr = _np.asarray([g.name_to_idx[nm] for nm in placement],
+                                 dtype=_np.int64)
+        widx_arr = _np.asarray(
+            [self.cluster.worker_id_to_index[w] for w in placement.values()],
+            dtype=_np.int64)
+        per_worker = {}
+        for nm, wid in placement.items():
+            per_worker.setdefault(wid, []).append(g.name_to_idx[nm])
+        op_groups = [(wid, _np.asarray(ops, dtype=_np.int64),
+                      float(g.memory_cost[ops].sum()))
+                     for wid, ops in per_worker.items()]
+        # dense schedule arrays (op and dep priorities)
+        sch_op_idx, sch_op_prio = [], []
+        for _w, ops in sched.items():
+            for nm, prio in ops.items():
+                sch_op_idx.append(g.name_to_idx[nm])
+                sch_op_prio.append(prio)
+        sch_dep_idx, sch_dep_prio = [], []
+        for _c, dmap in dep_sched.items():
+            for didx, prio in dmap.items():
+                sch_dep_idx.append(didx)
+                sch_dep_prio.append(prio)
         self._pipeline_cache[cache_key] = {
             "placed": True,
-            "placement": dict(self.op_placement.action[job_id]),
+            "placement": placement,
             "op_schedule": sched,
-            "dep_placement": dict(self.dep_placement.action.get(job_id, {})),
+            "dep_mount_fast": fast,
+            "op_mount_fast": (op_idx_arr, widx_arr, op_groups),
+            "sched_fast": (_np.asarray(sch_op_idx, dtype=_np.int64),
+                           _np.asarray(sch_op_prio, dtype=_np.int64),
+                           _np.asarray(sch_dep_idx, dtype=_np.int64),
+                           _np.asarray(sch_dep_prio, dtype=_np.int64)),
+            "dep_placement": dpd,
             "dep_schedule": dep_sched,
             "dep_init_run_time": _np.array(pj.dep_init_run_time, copy=True),
         }
@@ -294,9 +348,17 @@ class RampJobPartitioningEnvironment:
                 {"op_id": op_name, "job_id": job_id})
         op_placement.worker_ids = set(cached["placement"].values())
         op_placement.job_id_to_worker_ids = {job_id: op_placement.worker_ids}
+        op_placement.fast_mount = cached.get("op_mount_fast")
         self.op_placement = op_placement
         self.op_schedule = OpSchedule(
             {w: {job_id: ops} for w, ops in cached["op_schedule"].items()})
+        sf = cached.get("sched_fast")
+        if sf is not None:
+            self.op_schedule.fast_priorities = (sf[0], sf[1])
         self.dep_placement = DepPlacement({job_id: cached["dep_placement"]})
+        # hand the vectorised mount arrays to RampClusterEnvironment._place_deps
+        self.dep_placement.fast_mount = cached.get("dep_mount_fast")
         self.dep_schedule = DepSchedule(
             {cid: {job_id: deps} for cid, deps in cached["dep_schedule"].items()})
+        if sf is not None:
+            self.dep_schedule.fast_priorities = (sf[2], sf[3])
