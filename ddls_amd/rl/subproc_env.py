@@ -16,7 +16,8 @@ import numpy as np
 from .rollout import CompactObs
 
 
-def _worker_loop(conn, env_fn_count: int, env_fn, base_seed: int):
+def _worker_loop(conn, env_fn_count: int, env_fn, base_seed: int,
+                 shm_buf=None):
     envs = [env_fn() for _ in range(env_fn_count)]
     episode_counters = [0] * len(envs)
     episode_returns = np.zeros(len(envs))
@@ -24,9 +25,22 @@ def _worker_loop(conn, env_fn_count: int, env_fn, base_seed: int):
     policy = None          # lazy CPU policy copy for worker-side rollouts
     obs_cache = None
     rng = None
+    shm_views = None       # zero-copy param views into the fork-shared buffer
     try:
         while True:
             cmd, payload = conn.recv()
+            if cmd == "shm_meta":
+                # parent laid out flat params in the fork-inherited RawArray;
+                # build numpy views once — every later rollout sees the
+                # freshest weights with NO weight pickling at all
+                raw = np.frombuffer(shm_buf, dtype=np.uint8)
+                shm_views = {}
+                for key, off, nbytes, shape, dtype in payload:
+                    shm_views[key] = (raw[off:off + nbytes]
+                                      .view(np.dtype(dtype)).reshape(shape))
+                policy = None
+                conn.send("ok")
+                continue
             if cmd == "rollout":
                 # RLlib-style worker-side rollout: the worker holds a numpy
                 # inference copy of the policy (torch op-dispatch dominates at
@@ -38,7 +52,15 @@ def _worker_loop(conn, env_fn_count: int, env_fn, base_seed: int):
                     rng = np.random.RandomState((base_seed * 7919 + 13)
                                                 % (2 ** 31))
                 num_actions = len(envs[0].action_set)
-                policy = NumpyGNNPolicy(state_dict, policy_cfg, num_actions)
+                if state_dict is None:
+                    # shared-memory path: the policy's arrays ARE the shm
+                    # views, so reuse it as-is
+                    if policy is None:
+                        policy = NumpyGNNPolicy(shm_views, policy_cfg,
+                                                num_actions)
+                else:
+                    policy = NumpyGNNPolicy(state_dict, policy_cfg,
+                                            num_actions)
                 if obs_cache is None:
                     obs_cache = [CompactObs.from_obs(
                         env.reset(seed=base_seed + 1000 * i))
@@ -143,6 +165,8 @@ def _worker_loop(conn, env_fn_count: int, env_fn, base_seed: int):
 class SubprocVectorEnv:
     """Same interface as rollout.VectorEnv, envs sharded over processes."""
 
+    SHM_BYTES = 16 << 20   # shared weight buffer; larger policies fall back
+
     def __init__(self, env_fn, num_envs: int, num_workers: Optional[int] = None,
                  base_seed: int = 0):
         if num_workers is None:
@@ -154,13 +178,19 @@ class SubprocVectorEnv:
         rem = num_envs % num_workers
         self.shards = [base + (1 if w < rem else 0) for w in range(num_workers)]
         ctx = mp.get_context("fork")
+        # fork-inherited shared buffer for the policy weights: one vectorised
+        # write in the parent replaces num_workers x ~400 KB pickled
+        # state_dicts per rollout
+        self._shm_buf = ctx.RawArray("b", self.SHM_BYTES)
+        self._shm_meta = None
+        self._shm_views = None
         self.conns, self.procs = [], []
         offset = 0
         for w, count in enumerate(self.shards):
             parent, child = ctx.Pipe()
             p = ctx.Process(target=_worker_loop,
                             args=(child, count, env_fn,
-                                  base_seed + 1000000 * w),
+                                  base_seed + 1000000 * w, self._shm_buf),
                             daemon=True)
             p.start()
             child.close()
@@ -203,11 +233,19 @@ class SubprocVectorEnv:
         Returns dict with [T, N] arrays (actions/logp/values/rewards/dones),
         obs as a flat [T*N] list (t-major), and bootstrap_values [N].
         """
-        state_dict = {k: v.detach().cpu().numpy()
-                      for k, v in policy.state_dict().items()}
         cfg = dict(policy.config)
+        if self._shm_meta is None and self._shm_buf is not None:
+            self._try_init_shm(policy)
+        if self._shm_views is not None:
+            for k, v in policy.state_dict().items():
+                self._shm_views[k][...] = v.detach().cpu().numpy()
+            payload = (None, steps, cfg)
+        else:
+            state_dict = {k: v.detach().cpu().numpy()
+                          for k, v in policy.state_dict().items()}
+            payload = (state_dict, steps, cfg)
         for conn in self.conns:
-            conn.send(("rollout", (state_dict, steps, cfg)))
+            conn.send(("rollout", payload))
         per_worker = []
         for conn in self.conns:
             traj, stats = conn.recv()
@@ -226,6 +264,29 @@ class SubprocVectorEnv:
         out["bootstrap_values"] = np.concatenate(
             [traj["bootstrap_values"] for traj in per_worker])
         return out
+
+    def _try_init_shm(self, policy):
+        """Lay out the policy's params in the fork-shared buffer and send the
+        layout to every worker; falls back to pickled state_dicts when the
+        policy does not fit."""
+        meta, off = [], 0
+        sd = {k: v.detach().cpu().numpy() for k, v in policy.state_dict().items()}
+        for k, v in sd.items():
+            off = (off + 63) & ~63   # 64 B alignment per tensor
+            meta.append((k, off, v.nbytes, tuple(v.shape), v.dtype.str))
+            off += v.nbytes
+        if off > len(self._shm_buf):
+            self._shm_buf = None
+            return
+        raw = np.frombuffer(self._shm_buf, dtype=np.uint8)
+        self._shm_views = {
+            k: raw[o:o + nb].view(np.dtype(dt)).reshape(shape)
+            for k, o, nb, shape, dt in meta}
+        for conn in self.conns:
+            conn.send(("shm_meta", meta))
+        for conn in self.conns:
+            conn.recv()
+        self._shm_meta = meta
 
     def drain_episode_stats(self) -> List[dict]:
         out = self.completed_episode_stats

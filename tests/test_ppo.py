@@ -260,3 +260,30 @@ def test_hipgraph_captured_sgd_matches_eager():
     # partial minibatch falls back
     assert not stepper.step(mb2[:B - 1], actions[:B - 1], old_logp[:B - 1],
                             adv[:B - 1], vtarg[:B - 1])
+
+
+def test_shared_memory_rollout_matches_pickled(tiny_model_files):
+    """The fork-shared weight buffer path must produce trajectories identical
+    to pickling the state_dict to every worker."""
+    from ddls_amd.models.gnn import GNNPolicy
+    from ddls_amd.rl.subproc_env import SubprocVectorEnv
+    from tests.conftest import make_env as _mk
+
+    def env_fn():
+        return _mk(tiny_model_files, replication=50, num_training_steps=10)
+
+    torch.manual_seed(0)
+    policy = GNNPolicy(17)
+    v1 = SubprocVectorEnv(env_fn, num_envs=4, num_workers=2, base_seed=3)
+    a1 = v1.rollout(policy, 5)
+    b1 = v1.rollout(policy, 5)
+    assert v1._shm_views is not None  # shm path actually engaged
+    v1.close()
+    v2 = SubprocVectorEnv(env_fn, num_envs=4, num_workers=2, base_seed=3)
+    v2._shm_buf = None  # force pickled fallback
+    a2 = v2.rollout(policy, 5)
+    b2 = v2.rollout(policy, 5)
+    v2.close()
+    for k in ("actions", "logp", "values", "rewards", "dones"):
+        assert np.array_equal(a1[k], a2[k]), k
+        assert np.array_equal(b1[k], b2[k]), k
