@@ -114,7 +114,6 @@ class RampJobPartitioningEnvironment:
     # ------------------------------------------------------------------
     def reset(self, seed: Optional[int] = None, verbose: bool = False):
         self.step_counter = 1
-        self._obs_cache = {}  # per-episode: jobs_params can change on reset
         self.cluster.reset(jobs_config=self.jobs_config,
                            max_simulation_run_time=self.max_simulation_run_time,
                            job_queue_capacity=self.job_queue_capacity,
@@ -123,6 +122,18 @@ class RampJobPartitioningEnvironment:
                            init_details_memo_preload=self.init_details_memo_preload,
                            reuse_jobs_generator=self.reuse_jobs_generator,
                            verbose=verbose)
+        # obs-cache entries stay valid across resets that reproduce the same
+        # normalisation stats (the only reset-dependent inputs are the
+        # resampled max-acceptable-JCT fracs) — key them by a stats salt
+        # instead of clearing, so a re-reset to the same seed keeps hot caches
+        p = self.cluster.jobs_generator.jobs_params
+        self._obs_cache_salt = (
+            p.get("min_max_acceptable_job_completion_time_fracs"),
+            p.get("max_max_acceptable_job_completion_time_fracs"),
+            p.get("min_max_acceptable_job_completion_times"),
+            p.get("max_max_acceptable_job_completion_times"))
+        if len(self._obs_cache) > 20000:
+            self._obs_cache = {}
         self.obs = self.observation_function.reset(self)
         self.observation_space = self.observation_function.observation_space
         self.reward_function.reset(env=self)
@@ -237,7 +248,8 @@ class RampJobPartitioningEnvironment:
                 or cluster.num_mounted_ops != 0):
             return self.observation_function.extract(env=self, done=False)
         job = next(iter(cluster.job_queue.jobs.values()))
-        key = (job.details["model"], job.max_acceptable_job_completion_time_frac)
+        key = (self._obs_cache_salt, job.details["model"],
+               job.max_acceptable_job_completion_time_frac)
         obs = self._obs_cache.get(key)
         if obs is None:
             obs = self.observation_function.extract(env=self, done=False)

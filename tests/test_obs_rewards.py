@@ -173,3 +173,22 @@ def test_generic_gpu_device_and_scheduling_stub():
     assert str(g) == "V100_7"
     with pytest.raises(NotImplementedError):
         JobSchedulingEnvironment()
+
+
+def test_obs_cache_across_resets(tiny_model_files):
+    """Persistent obs cache keyed by normalisation-stats salt must return the
+    exact obs the observation function would produce, across episodes with
+    resampled fracs."""
+    env_a = make_env(tiny_model_files, replication=6, num_training_steps=10)
+    env_b = make_env(tiny_model_files, replication=6, num_training_steps=10)
+    env_b.cache_pipeline = False  # disables pipeline+obs caching
+    for seed in (0, 1, 0):
+        oa, ob = env_a.reset(seed=seed), env_b.reset(seed=seed)
+        done = False
+        while not done:
+            for k in ("node_features", "graph_features", "action_mask"):
+                assert np.array_equal(oa[k], ob[k]), (seed, k)
+            valid = oa["action_set"][oa["action_mask"].astype(bool)]
+            oa, ra, done, _ = env_a.step(int(valid[-1]))
+            ob, rb, done_b, _ = env_b.step(int(valid[-1]))
+            assert ra == rb and done == done_b
