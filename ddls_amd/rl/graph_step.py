@@ -146,9 +146,19 @@ class CapturedSGDStep:
             "adv": dbuf(B, torch.float32),
             "vtarg": dbuf(B, torch.float32),
         }
-        self.p = {k: pbuf(tuple(v.shape), v.dtype) for k, v in self.d.items()}
-        self.p["mask"].fill_(1.0)
-        self.np_ = {k: v.numpy() for k, v in self.p.items()}
+        # double-buffered pinned staging: _fill for minibatch k+1 must not
+        # overwrite host memory the async H2D copies of minibatch k are still
+        # reading (no per-minibatch sync exists), so alternate two pinned sets
+        # guarded by events recorded after each copy batch
+        self.pin = []
+        self.pin_np = []
+        for _ in range(2):
+            pset = {k: pbuf(tuple(v.shape), v.dtype) for k, v in self.d.items()}
+            pset["mask"].fill_(1.0)
+            self.pin.append(pset)
+            self.pin_np.append({k: v.numpy() for k, v in pset.items()})
+        self.copy_events = [torch.cuda.Event(), torch.cuda.Event()]
+        self._buf = 0
         # one static flat graph: B real graphs + dummy graph B for padding
         self.batch = GraphBatch(
             z=self.d["z"], e=self.d["e"], src=self.d["src"], dst=self.d["dst"],
@@ -166,7 +176,10 @@ class CapturedSGDStep:
         dst = np.concatenate([o.edges_dst + off
                               for o, off in zip(mb_obs, offsets)])
         m = len(src)
-        p = self.np_
+        j = self._buf = self._buf ^ 1
+        # wait until the GPU finished the copies that last read this pinned set
+        self.copy_events[j].synchronize()
+        p = self.pin_np[j]
         p["z"][:n] = z
         p["e"][:m] = e
         p["src"][:m] = src
@@ -198,7 +211,8 @@ class CapturedSGDStep:
         p["adv"][:] = adv
         p["vtarg"][:] = vtarg
         for k, dst_t in self.d.items():
-            dst_t.copy_(self.p[k], non_blocking=True)
+            dst_t.copy_(self.pin[j][k], non_blocking=True)
+        self.copy_events[j].record()
 
     # ------------------------------------------------------------------
     def _body_fwd_bwd(self):
