@@ -113,12 +113,25 @@ class CapturedSGDStep:
         e = sum(len(o.edges_src) for o in mb_obs)
         if not self.ensure_capacity(n, e) or n > self.n_cap - self.RESERVE:
             return False
-        self._fill(mb_obs, actions, old_logp, adv, vtarg)
+        self.commit(self.prepare(mb_obs, actions, old_logp, adv, vtarg))
+        return True
+
+    def prepare(self, mb_obs, actions, old_logp, adv, vtarg) -> int:
+        """Stage one minibatch into the next pinned set (CPU work only —
+        safe to run in a prefetch thread while the GPU replays the previous
+        minibatch).  Alternates the two pinned sets; calls must be strictly
+        sequential."""
+        return self._fill(mb_obs, actions, old_logp, adv, vtarg)
+
+    def commit(self, j: int):
+        """Enqueue H2D copies from pinned set j + replay the graph."""
+        for k, dst_t in self.d.items():
+            dst_t.copy_(self.pin[j][k], non_blocking=True)
+        self.copy_events[j].record()
         self.graph.replay()
         if self.graph_opt is not None:
             all_reduce_gradients(self.policy.parameters())
             self.graph_opt.replay()
-        return True
 
     # ------------------------------------------------------------------
     def _alloc(self, n_cap: int, e_cap: int):
@@ -210,9 +223,7 @@ class CapturedSGDStep:
         p["old_logp"][:] = old_logp
         p["adv"][:] = adv
         p["vtarg"][:] = vtarg
-        for k, dst_t in self.d.items():
-            dst_t.copy_(self.pin[j][k], non_blocking=True)
-        self.copy_events[j].record()
+        return j
 
     # ------------------------------------------------------------------
     def _body_fwd_bwd(self):

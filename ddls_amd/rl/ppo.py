@@ -64,6 +64,7 @@ class PPOTrainer:
         self.total_env_steps = 0
         self.iteration = 0
         self._stepper = None  # lazy CapturedSGDStep (GPU only)
+        self._mb_executor = None  # prefetch thread for captured minibatches
 
     # ------------------------------------------------------------------
     @torch.no_grad()
@@ -211,58 +212,89 @@ class PPOTrainer:
         vtarg_np = np.asarray(batch["value_targets"], dtype=np.float32)
 
         rng = np.random.RandomState(self.iteration + 1234 * get_rank())
-        for _ in range(cfg.num_sgd_iter):
-            perm = rng.permutation(n)
-            for start in range(0, n, cfg.sgd_minibatch_size):
-                idx = perm[start:start + cfg.sgd_minibatch_size]
-                if len(idx) < 2:
-                    continue
-                mb_obs = [batch["obs"][i] for i in idx]
-                if stepper is not None and stepper.step(
-                        mb_obs, actions_np[idx], logp_np[idx], adv_np[idx],
-                        vtarg_np[idx]):
-                    num_updates += 1
-                    num_captured += 1
-                    continue
-                inputs = collate(mb_obs, self.device)
-                logits, values = self._forward_flat(inputs)
-                dist = torch.distributions.Categorical(logits=logits)
-                idx_t = torch.as_tensor(idx, device=self.device)
-                logp = dist.log_prob(actions[idx_t])
-                ratio = torch.exp(logp - old_logp[idx_t])
-                mb_adv = advantages[idx_t]
-                surr = torch.min(
-                    ratio * mb_adv,
-                    torch.clamp(ratio, 1 - cfg.clip_param,
-                                1 + cfg.clip_param) * mb_adv)
-                policy_loss = -surr.mean()
+        B = cfg.sgd_minibatch_size
 
-                # KL(old || new) sample estimate (RLlib uses action logp kl)
-                kl = (old_logp[idx_t] - logp).mean()
+        if (stepper is not None and not stepper.broken
+                and stepper.graph is not None and n >= B and n % B == 0):
+            # captured fast loop with CPU prefetch: a worker thread stages
+            # minibatch k+1 into the spare pinned set while the GPU replays
+            # minibatch k (the double-buffer events make this safe)
+            def mb_data(idx):
+                return ([batch["obs"][i] for i in idx], actions_np[idx],
+                        logp_np[idx], adv_np[idx], vtarg_np[idx])
 
-                vf_err = (values - value_targets[idx_t]) ** 2
-                vf_loss = torch.clamp(vf_err, 0, cfg.vf_clip_param).mean()
-
-                entropy = dist.entropy().mean()
-
-                loss = (policy_loss + self.kl_coeff * kl
-                        + cfg.vf_loss_coeff * vf_loss
-                        - cfg.entropy_coeff * entropy)
-
-                self.optimizer.zero_grad(set_to_none=True)
-                loss.backward()
-                all_reduce_gradients(self.policy.parameters())
-                if cfg.grad_clip is not None:
-                    torch.nn.utils.clip_grad_norm_(self.policy.parameters(),
-                                                   cfg.grad_clip)
-                self.optimizer.step()
-
-                stats["policy_loss"] += policy_loss.detach().item()
-                stats["vf_loss"] += vf_loss.detach().item()
-                stats["kl"] += kl.detach().item()
-                stats["entropy"] += entropy.detach().item()
-                stats["total_loss"] += loss.detach().item()
+            idx_lists = []
+            for _ in range(cfg.num_sgd_iter):
+                perm = rng.permutation(n)
+                for start in range(0, n, B):
+                    idx_lists.append(perm[start:start + B])
+            if self._mb_executor is None:
+                from concurrent.futures import ThreadPoolExecutor
+                self._mb_executor = ThreadPoolExecutor(max_workers=1)
+            ex = self._mb_executor
+            fut = ex.submit(stepper.prepare, *mb_data(idx_lists[0]))
+            for i in range(len(idx_lists)):
+                j = fut.result()
+                if i + 1 < len(idx_lists):
+                    fut = ex.submit(stepper.prepare, *mb_data(idx_lists[i + 1]))
+                stepper.commit(j)
                 num_updates += 1
+                num_captured += 1
+            rng = None  # permutations consumed identically to the eager loop
+
+        if rng is not None:
+            for _ in range(cfg.num_sgd_iter):
+                perm = rng.permutation(n)
+                for start in range(0, n, cfg.sgd_minibatch_size):
+                    idx = perm[start:start + cfg.sgd_minibatch_size]
+                    if len(idx) < 2:
+                        continue
+                    mb_obs = [batch["obs"][i] for i in idx]
+                    if stepper is not None and stepper.step(
+                            mb_obs, actions_np[idx], logp_np[idx], adv_np[idx],
+                            vtarg_np[idx]):
+                        num_updates += 1
+                        num_captured += 1
+                        continue
+                    inputs = collate(mb_obs, self.device)
+                    logits, values = self._forward_flat(inputs)
+                    dist = torch.distributions.Categorical(logits=logits)
+                    idx_t = torch.as_tensor(idx, device=self.device)
+                    logp = dist.log_prob(actions[idx_t])
+                    ratio = torch.exp(logp - old_logp[idx_t])
+                    mb_adv = advantages[idx_t]
+                    surr = torch.min(
+                        ratio * mb_adv,
+                        torch.clamp(ratio, 1 - cfg.clip_param,
+                                    1 + cfg.clip_param) * mb_adv)
+                    policy_loss = -surr.mean()
+
+                    # KL(old || new) sample estimate (RLlib uses action logp kl)
+                    kl = (old_logp[idx_t] - logp).mean()
+
+                    vf_err = (values - value_targets[idx_t]) ** 2
+                    vf_loss = torch.clamp(vf_err, 0, cfg.vf_clip_param).mean()
+
+                    entropy = dist.entropy().mean()
+
+                    loss = (policy_loss + self.kl_coeff * kl
+                            + cfg.vf_loss_coeff * vf_loss
+                            - cfg.entropy_coeff * entropy)
+
+                    self.optimizer.zero_grad(set_to_none=True)
+                    loss.backward()
+                    all_reduce_gradients(self.policy.parameters())
+                    if cfg.grad_clip is not None:
+                        torch.nn.utils.clip_grad_norm_(self.policy.parameters(),
+                                                       cfg.grad_clip)
+                    self.optimizer.step()
+
+                    stats["policy_loss"] += policy_loss.detach().item()
+                    stats["vf_loss"] += vf_loss.detach().item()
+                    stats["kl"] += kl.detach().item()
+                    stats["entropy"] += entropy.detach().item()
+                    stats["total_loss"] += loss.detach().item()
+                    num_updates += 1
 
         if num_captured:
             acc = self._stepper.stats_acc.cpu().numpy()  # one sync per update
