@@ -200,12 +200,6 @@ class PPOTrainer:
             stepper = self._stepper
             stepper.reset_stats()
             stepper.set_kl_coeff(self.kl_coeff)
-            # pre-size capacity to B x the rollout's max per-sample counts so
-            # the minibatch loop never recaptures
-            max_n = max(len(o.node_features) for o in batch["obs"])
-            max_e = max(len(o.edges_src) for o in batch["obs"])
-            stepper.ensure_capacity(cfg.sgd_minibatch_size * max_n,
-                                    cfg.sgd_minibatch_size * max_e)
         actions_np = np.asarray(batch["actions"], dtype=np.int64)
         logp_np = np.asarray(batch["logp"], dtype=np.float32)
         adv_np = np.asarray(adv, dtype=np.float32)
@@ -214,8 +208,29 @@ class PPOTrainer:
         rng = np.random.RandomState(self.iteration + 1234 * get_rank())
         B = cfg.sgd_minibatch_size
 
-        if (stepper is not None and not stepper.broken
-                and stepper.graph is not None and n >= B and n % B == 0):
+        captured = False
+        if stepper is not None and not stepper.broken and n >= B and n % B == 0:
+            # the permutations are fixed up front, so capacity can be sized to
+            # the LARGEST ACTUAL minibatch (sum of 128 sample sizes
+            # concentrates near B x mean — sizing by B x max-sample would
+            # nearly double the padded kernel work)
+            idx_lists = []
+            for _ in range(cfg.num_sgd_iter):
+                perm = rng.permutation(n)
+                for start in range(0, n, B):
+                    idx_lists.append(perm[start:start + B])
+            ns_arr = np.array([len(o.node_features) for o in batch["obs"]],
+                              dtype=np.int64)
+            es_arr = np.array([len(o.edges_src) for o in batch["obs"]],
+                              dtype=np.int64)
+            need_n = max(int(ns_arr[idx].sum()) for idx in idx_lists)
+            need_e = max(int(es_arr[idx].sum()) for idx in idx_lists)
+            captured = stepper.ensure_capacity(int(need_n * 1.05) + 8,
+                                               int(need_e * 1.05) + 8)
+            if not captured:  # replay the same permutation stream eagerly
+                rng = np.random.RandomState(self.iteration + 1234 * get_rank())
+
+        if captured:
             # captured fast loop with CPU prefetch: a worker thread stages
             # minibatch k+1 into the spare pinned set while the GPU replays
             # minibatch k (the double-buffer events make this safe)
@@ -223,11 +238,6 @@ class PPOTrainer:
                 return ([batch["obs"][i] for i in idx], actions_np[idx],
                         logp_np[idx], adv_np[idx], vtarg_np[idx])
 
-            idx_lists = []
-            for _ in range(cfg.num_sgd_iter):
-                perm = rng.permutation(n)
-                for start in range(0, n, B):
-                    idx_lists.append(perm[start:start + B])
             if self._mb_executor is None:
                 from concurrent.futures import ThreadPoolExecutor
                 self._mb_executor = ThreadPoolExecutor(max_workers=1)
