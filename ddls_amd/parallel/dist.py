@@ -27,7 +27,9 @@ def init_distributed_from_env(device: Optional[torch.device] = None) -> int:
         return 0
     if dist.is_initialized():
         return dist.get_rank()
-    backend = "nccl" if torch.cuda.is_available() else "gloo"
+    backend = os.environ.get(
+        "DDLS_AMD_DIST_BACKEND",
+        "nccl" if torch.cuda.is_available() else "gloo")
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29500")
     dist.init_process_group(backend=backend)
@@ -59,7 +61,13 @@ def all_reduce_gradients(params: Iterable[torch.nn.Parameter]):
     if not grads:
         return
     flat = torch._utils._flatten_dense_tensors(grads)
-    dist.all_reduce(flat, op=dist.ReduceOp.SUM)
+    if flat.is_cuda and dist.get_backend() == "gloo":
+        # gloo CUDA support varies by build; stage through host
+        host = flat.cpu()
+        dist.all_reduce(host, op=dist.ReduceOp.SUM)
+        flat = host.to(flat.device)
+    else:
+        dist.all_reduce(flat, op=dist.ReduceOp.SUM)
     flat /= get_world_size()
     for g, synced in zip(grads, torch._utils._unflatten_dense_tensors(flat, grads)):
         g.copy_(synced)
