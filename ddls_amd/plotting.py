@@ -107,3 +107,66 @@ def plot_episode_stats_comparison(actor_to_stats: Dict[str, dict],
         ax.set_title(metric)
     fig.tight_layout()
     return fig
+
+
+def computation_graph_to_dot(graph: CompGraph) -> str:
+    """Graphviz-dot export of a computation graph (usable without matplotlib;
+    reference renders via nx.draw, ``plotting/plotting.py:95-115``)."""
+    lines = [f'digraph "{graph.model}" {{', "  rankdir=LR;"]
+    for i in range(graph.n):
+        color = "indianred" if graph.pass_type[i] == BWD else "steelblue"
+        lines.append(
+            f'  "{graph.names[i]}" [style=filled, fillcolor={color}, '
+            f'label="{graph.names[i]}\\nmem={float(graph.memory_cost[i]):.3g}"];')
+    for e in range(graph.m):
+        u, v = graph.names[int(graph.src[e])], graph.names[int(graph.dst[e])]
+        lines.append(f'  "{u}" -> "{v}" [label="{float(graph.size[e]):.3g}"];')
+    lines.append("}")
+    return "\n".join(lines)
+
+
+def plot_metric_distributions(actor_to_values: Dict[str, Sequence[float]],
+                              metric_name: str = "job_completion_time",
+                              bins: int = 30, cdf: bool = True,
+                              figsize=(10, 4)):
+    """Histogram + CDF of a per-job metric across actors (paper-style
+    seaborn hist/ecdf figures, reference ``plotting/plotting.py``)."""
+    plt = _require_matplotlib()
+    fig, axes = plt.subplots(1, 2 if cdf else 1, figsize=figsize,
+                             squeeze=False)
+    for actor, vals in actor_to_values.items():
+        vals = np.asarray(list(vals), dtype=float)
+        if len(vals) == 0:
+            continue
+        axes[0][0].hist(vals, bins=bins, alpha=0.5, label=actor)
+        if cdf:
+            xs = np.sort(vals)
+            axes[0][1].plot(xs, np.arange(1, len(xs) + 1) / len(xs),
+                            label=actor)
+    axes[0][0].set_xlabel(metric_name)
+    axes[0][0].set_ylabel("count")
+    axes[0][0].legend(fontsize=8)
+    if cdf:
+        axes[0][1].set_xlabel(metric_name)
+        axes[0][1].set_ylabel("CDF")
+        axes[0][1].legend(fontsize=8)
+    fig.tight_layout()
+    return fig
+
+
+def plot_cluster_timeline(steps_log: Dict[str, Sequence[float]],
+                          metrics: Sequence[str] = (
+                              "num_jobs_running", "num_mounted_ops",
+                              "mean_num_active_workers"),
+                          figsize=(12, 4)):
+    """Per-cluster-step utilisation curves from the cluster ``steps_log``."""
+    plt = _require_matplotlib()
+    avail = [m for m in metrics if m in steps_log and len(steps_log[m]) > 0]
+    n = max(1, len(avail))
+    fig, axes = plt.subplots(1, n, figsize=figsize, squeeze=False)
+    for k, metric in enumerate(avail):
+        axes[0][k].plot(steps_log[metric])
+        axes[0][k].set_title(metric)
+        axes[0][k].set_xlabel("cluster step")
+    fig.tight_layout()
+    return fig

@@ -132,3 +132,15 @@ def test_harvest_and_tables(tmp_path, tiny_model_files):
     write_metrics_table([out], str(tmp_path / "table"))
     assert os.path.exists(str(tmp_path / "table.csv"))
     assert os.path.exists(str(tmp_path / "table.json"))
+
+
+def test_computation_graph_to_dot(tiny_model_files):
+    from ddls_amd.graphs import load_pipedream_graph
+    from ddls_amd.plotting import computation_graph_to_dot
+    import glob
+    path = glob.glob(str(tiny_model_files) + "/*")[0]
+    g = load_pipedream_graph(path, processor_type_profiled="A100")
+    dot = computation_graph_to_dot(g)
+    assert dot.startswith("digraph")
+    assert dot.count("->") == g.m
+    assert all(name in dot for name in g.names)
