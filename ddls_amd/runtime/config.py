@@ -73,7 +73,17 @@ def load_config(config_path: str,
                 os.path.join(config_dir, f"{entry}.yaml")))
     cfg = deep_merge(resolved, cfg)
     if overrides:
-        cfg = apply_dotted_overrides(cfg, overrides)
+        # hydra-style GROUP overrides first: "env_config=env_1024worker"
+        # swaps in <config_dir>/env_config/env_1024worker.yaml
+        dotted = []
+        for ov in overrides:
+            key, _, raw = ov.partition("=")
+            group_file = os.path.join(config_dir, key, f"{raw}.yaml")
+            if "." not in key and "=" in ov and os.path.isfile(group_file):
+                cfg[key] = load_yaml(group_file)
+            else:
+                dotted.append(ov)
+        cfg = apply_dotted_overrides(cfg, dotted)
     return cfg
 
 

@@ -144,3 +144,22 @@ def test_computation_graph_to_dot(tiny_model_files):
     assert dot.startswith("digraph")
     assert dot.count("->") == g.m
     assert all(name in dot for name in g.names)
+
+
+def test_config_group_override_1024_workers():
+    """hydra-style group override swaps whole env_config groups; the
+    1024-worker RAMP (BASELINE configs[3]) builds and steps via lazy
+    channels."""
+    import os
+    from ddls_amd.envs.actors import ACTORS
+    from ddls_amd.runtime.config import build_env_from_config, load_config
+    from ddls_amd.runtime.loops import EvalLoop
+    from ddls_amd.utils import seed_everything
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cfg = load_config(os.path.join(root, "configs", "heuristic_config.yaml"),
+                      overrides=["env_config=env_1024worker"])
+    seed_everything(7)
+    env = build_env_from_config(cfg)
+    assert env.cluster.topology.num_workers == 1024
+    r = EvalLoop(ACTORS["acceptable_jct"](), env, max_steps=12).run(seed=7)
+    assert r["num_actor_steps"] == 12
