@@ -134,10 +134,24 @@ class _FusedMeanPoolFn(torch.autograd.Function):
         ext = hip_ops.get_extension(required=True)
         zc, ec = z.contiguous(), e.contiguous()
         Wn, We, Wr = W_n.contiguous(), W_e.contiguous(), W_r.contiguous()
-        hn = ext.row_mlp(zc, ln_n_w, ln_n_b, Wn, b_n)
-        he = ext.row_mlp(ec, ln_e_w, ln_e_b, We, b_e)
-        out, r_edge, r_self = ext.message_reduce_train(
-            hn, he, src, edge_order, indptr, ln_r_w, ln_r_b, Wr, b_r)
+        half, out_dim = W_n.shape[0], W_r.shape[0]
+        use_mfma = (half == 16 and W_r.shape[1] == 32 and out_dim % 16 == 0
+                    and out_dim <= 64 and zc.shape[1] <= 64
+                    and ec.shape[1] <= 64
+                    and os.environ.get("DDLS_AMD_DISABLE_MFMA", "0") != "1")
+        if use_mfma:
+            # matrix-core forward: the MLP contractions on
+            # v_mfma_f32_16x16x4_f32 tiles, then a light CSR combine
+            hn = ext.row_mlp_mfma(zc, ln_n_w, ln_n_b, Wn, b_n)
+            he = ext.row_mlp_mfma(ec, ln_e_w, ln_e_b, We, b_e)
+            r_edge, r_self = ext.message_mlp_mfma(
+                hn, he, src, ln_r_w, ln_r_b, Wr, b_r)
+            out = ext.segment_combine(r_edge, r_self, edge_order, indptr)
+        else:
+            hn = ext.row_mlp(zc, ln_n_w, ln_n_b, Wn, b_n)
+            he = ext.row_mlp(ec, ln_e_w, ln_e_b, We, b_e)
+            out, r_edge, r_self = ext.message_reduce_train(
+                hn, he, src, edge_order, indptr, ln_r_w, ln_r_b, Wr, b_r)
         ctx.save_for_backward(zc, ec, src, dst, indptr, hn, he, r_edge, r_self,
                               ln_n_w, ln_n_b, Wn, ln_e_w, ln_e_b, We,
                               ln_r_w, ln_r_b, Wr)

@@ -21,6 +21,7 @@ _HERE = os.path.dirname(os.path.abspath(__file__))
 _BUILD_DIR = os.path.join(_HERE, "_build")
 _SOURCES = [os.path.join(_HERE, "hip", "meanpool.hip"),
             os.path.join(_HERE, "hip", "meanpool_bwd.hip"),
+            os.path.join(_HERE, "hip", "meanpool_mfma.hip"),
             os.path.join(_HERE, "hip", "lookahead.hip"),
             os.path.join(_HERE, "hip", "bindings.hip")]
 

@@ -22,6 +22,14 @@ std::vector<torch::Tensor> message_reduce_bwd(
 std::vector<torch::Tensor> row_mlp_bwd(torch::Tensor x, torch::Tensor y,
                                        torch::Tensor gy, torch::Tensor ln_g,
                                        torch::Tensor ln_b, torch::Tensor W);
+torch::Tensor row_mlp_mfma(torch::Tensor x, torch::Tensor ln_g,
+                           torch::Tensor ln_b, torch::Tensor W,
+                           torch::Tensor b);
+std::vector<torch::Tensor> message_mlp_mfma(
+    torch::Tensor hn, torch::Tensor he, torch::Tensor src, torch::Tensor ln_g,
+    torch::Tensor ln_b, torch::Tensor Wr, torch::Tensor br);
+torch::Tensor segment_combine(torch::Tensor r_edge, torch::Tensor r_self,
+                              torch::Tensor edge_order, torch::Tensor indptr);
 std::vector<torch::Tensor> lookahead_batch(
     torch::Tensor descs, torch::Tensor op_remaining, torch::Tensor op_worker,
     torch::Tensor op_priority, torch::Tensor out_indptr, torch::Tensor out_edges,
@@ -44,4 +52,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("message_reduce_bwd", &message_reduce_bwd,
           "fused MeanPool message-passing backward");
     m.def("row_mlp_bwd", &row_mlp_bwd, "fused LN+Linear+ReLU backward");
+    m.def("row_mlp_mfma", &row_mlp_mfma,
+          "LN+Linear+ReLU on v_mfma_f32_16x16x4_f32 tiles");
+    m.def("message_mlp_mfma", &message_mlp_mfma,
+          "per-message reduce-MLP on the matrix core");
+    m.def("segment_combine", &segment_combine,
+          "CSR segment mean of stored message activations");
 }

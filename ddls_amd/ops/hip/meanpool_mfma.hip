@@ -1,0 +1,295 @@
+// MFMA (matrix-core) forward kernels for the MeanPool GNN on gfx950.
+//
+// The row/message MLPs are GEMM-shaped (R x F @ F^T x H with R ~ 1e4..4e4,
+// F <= 64, H in {16, 64}), so the matrix units do the contraction:
+// v_mfma_f32_16x16x4_f32 tiles (f32-in/f32-acc — exact fmaf-chain numerics,
+// same data as the VALU path) with the LayerNorm fused in front via lane
+// shuffles and an LDS-staged normalised tile.  Replaces the wave-per-row GEMV
+// of meanpool.hip on the training path; the segment mean over dst-CSR is a
+// separate light kernel (segment_combine) since backward needs the
+// per-message activations materialised anyway.
+//
+// Fragment maps for 16x16x4 (cdna_hip_programming.md "FP32-input MFMA"):
+//   A[i = lane&15][k = lane>>4], B[k = lane>>4][j = lane&15],
+//   C/D: col = lane&15, row = (lane>>4)*4 + reg.
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include <vector>
+
+#define WAVE 64
+#define WAVES_PER_BLOCK 4
+#define BLOCK (WAVE * WAVES_PER_BLOCK)
+#define LN_EPS 1e-5f
+
+typedef float f32x4_t __attribute__((ext_vector_type(4)));
+
+// ---------------------------------------------------------------------------
+// y[r, :] = relu(LN(x[r, :]) @ W^T + b), W torch-layout [H, F].
+// One 16-row tile per wave; K staged through LDS zero-padded to F_pad.
+// Requires H % 16 == 0, H <= 64, F <= 64.
+__global__ void __launch_bounds__(BLOCK)
+row_mlp_mfma_kernel(const float* __restrict__ x,
+                    const float* __restrict__ ln_g,
+                    const float* __restrict__ ln_b,
+                    const float* __restrict__ W,
+                    const float* __restrict__ b,
+                    float* __restrict__ y,
+                    int R, int F, int H) {
+    const int F_pad = (F + 3) & ~3;
+    const int Q = F_pad / 4;            // elements per lane in phase A
+    __shared__ float xs[WAVES_PER_BLOCK][16][64 + 4];
+    __shared__ float ws[64][64 + 4];    // W zero-padded to [H][F_pad]
+    for (int i = threadIdx.x; i < H * F_pad; i += BLOCK) {
+        const int j = i / F_pad, k = i % F_pad;
+        ws[j][k] = (k < F) ? W[(long)j * F + k] : 0.0f;
+    }
+    __syncthreads();
+
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int mrow = lane & 15;         // M index within the tile
+    const int kpart = lane >> 4;        // 0..3
+    const int tiles = (R + 16 * WAVES_PER_BLOCK - 1) / (16 * WAVES_PER_BLOCK);
+    for (int t = blockIdx.x; t < tiles; t += gridDim.x) {
+        const int rbase = (t * WAVES_PER_BLOCK + wave) * 16;
+        const int r = rbase + mrow;
+        // phase A: LayerNorm 16 rows, 4 lanes per row
+        float vals[16];
+        float s = 0.0f;
+        for (int i = 0; i < Q; ++i) {
+            const int k = kpart * Q + i;
+            const float v = (r < R && k < F) ? x[(long)r * F + k] : 0.0f;
+            vals[i] = v;
+            s += v;
+        }
+        s += __shfl_xor(s, 16, WAVE);
+        s += __shfl_xor(s, 32, WAVE);
+        const float mean = s / F;
+        float d2 = 0.0f;
+        for (int i = 0; i < Q; ++i) {
+            const int k = kpart * Q + i;
+            const float d = (k < F) ? vals[i] - mean : 0.0f;
+            vals[i] = d;
+            d2 += d * d;
+        }
+        d2 += __shfl_xor(d2, 16, WAVE);
+        d2 += __shfl_xor(d2, 32, WAVE);
+        const float inv_sigma = rsqrtf(d2 / F + LN_EPS);
+        for (int i = 0; i < Q; ++i) {
+            const int k = kpart * Q + i;
+            if (k < F_pad)
+                xs[wave][mrow][k] =
+                    (k < F) ? vals[i] * inv_sigma * ln_g[k] + ln_b[k] : 0.0f;
+        }
+        __builtin_amdgcn_wave_barrier();
+        // phase B: 16x16 output tiles on the matrix core
+        for (int jt = 0; jt < H; jt += 16) {
+            f32x4_t acc = {0.0f, 0.0f, 0.0f, 0.0f};
+            for (int kk = 0; kk < F_pad; kk += 4) {
+                const float a = xs[wave][mrow][kk + kpart];
+                const float bb = ws[jt + mrow][kk + kpart];
+                acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bb, acc, 0, 0, 0);
+            }
+            const int col = jt + mrow;
+            const float bias = b[col];
+            for (int reg = 0; reg < 4; ++reg) {
+                const int row = rbase + kpart * 4 + reg;
+                if (row < R)
+                    y[(long)row * H + col] = fmaxf(acc[reg] + bias, 0.0f);
+            }
+        }
+        __builtin_amdgcn_wave_barrier();
+    }
+}
+
+// ---------------------------------------------------------------------------
+// Per-message reduce-MLP on the matrix core: messages 0..E-1 are edge
+// messages [hn[src[m]] || he[m]], messages E..E+N-1 are self messages
+// [hn[v] || 0].  r = relu(LN(msg) @ Wr^T + br) -> r_edge [E][OUT] (edge-id
+// indexed) and r_self [N][OUT].  Requires half == 16 (MSG = 32),
+// OUT % 16 == 0, OUT <= 64.
+__global__ void __launch_bounds__(BLOCK)
+message_mlp_mfma_kernel(const float* __restrict__ hn,
+                        const float* __restrict__ he,
+                        const long* __restrict__ src,
+                        const float* __restrict__ ln_g,
+                        const float* __restrict__ ln_b,
+                        const float* __restrict__ Wr,
+                        const float* __restrict__ br,
+                        float* __restrict__ r_edge,
+                        float* __restrict__ r_self,
+                        long E, int N, int OUT) {
+    const int MSG = 32, Q = 8;
+    __shared__ float ms[WAVES_PER_BLOCK][16][MSG + 4];
+    __shared__ float ws[64][MSG + 4];
+    for (int i = threadIdx.x; i < OUT * MSG; i += BLOCK)
+        ws[i / MSG][i % MSG] = Wr[i];
+    __syncthreads();
+
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int mrow = lane & 15;
+    const int kpart = lane >> 4;
+    const long M = E + N;
+    const long tiles = (M + 16 * WAVES_PER_BLOCK - 1) / (16 * WAVES_PER_BLOCK);
+    for (long t = blockIdx.x; t < tiles; t += gridDim.x) {
+        const long mbase = (t * WAVES_PER_BLOCK + wave) * 16;
+        const long m = mbase + mrow;
+        float vals[Q];
+        float s = 0.0f;
+        if (m < M) {
+            const bool is_edge = m < E;
+            const long node = is_edge ? src[m] : (m - E);
+            for (int i = 0; i < Q; ++i) {
+                const int k = kpart * Q + i;
+                float v;
+                if (k < 16)
+                    v = hn[node * 16 + k];
+                else if (is_edge)
+                    v = he[m * 16 + (k - 16)];
+                else
+                    v = 0.0f;  // self message upper half is zero
+                vals[i] = v;
+                s += v;
+            }
+        } else {
+            for (int i = 0; i < Q; ++i) vals[i] = 0.0f;
+        }
+        s += __shfl_xor(s, 16, WAVE);
+        s += __shfl_xor(s, 32, WAVE);
+        const float mean = s / MSG;
+        float d2 = 0.0f;
+        for (int i = 0; i < Q; ++i) {
+            const float d = (m < M) ? vals[i] - mean : 0.0f;
+            vals[i] = d;
+            d2 += d * d;
+        }
+        d2 += __shfl_xor(d2, 16, WAVE);
+        d2 += __shfl_xor(d2, 32, WAVE);
+        const float inv_sigma = rsqrtf(d2 / MSG + LN_EPS);
+        for (int i = 0; i < Q; ++i) {
+            const int k = kpart * Q + i;
+            ms[wave][mrow][k] = vals[i] * inv_sigma * ln_g[k] + ln_b[k];
+        }
+        __builtin_amdgcn_wave_barrier();
+        for (int jt = 0; jt < OUT; jt += 16) {
+            f32x4_t acc = {0.0f, 0.0f, 0.0f, 0.0f};
+            for (int kk = 0; kk < MSG; kk += 4) {
+                const float a = ms[wave][mrow][kk + kpart];
+                const float bb = ws[jt + mrow][kk + kpart];
+                acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bb, acc, 0, 0, 0);
+            }
+            const int col = jt + mrow;
+            const float bias = br[col];
+            for (int reg = 0; reg < 4; ++reg) {
+                const long row = mbase + kpart * 4 + reg;
+                if (row < M) {
+                    const float r = fmaxf(acc[reg] + bias, 0.0f);
+                    if (row < E)
+                        r_edge[row * OUT + col] = r;
+                    else
+                        r_self[(row - E) * OUT + col] = r;
+                }
+            }
+        }
+        __builtin_amdgcn_wave_barrier();
+    }
+}
+
+// ---------------------------------------------------------------------------
+// out[v] = indeg(v) > 0 ? (r_self[v] + sum_csr r_edge[edge_order[e]])
+//                         / (indeg+1)
+//                       : 0   (DGL zero-fill), summed in the same order as
+// the fused message_reduce (self first, then CSR order).
+__global__ void __launch_bounds__(BLOCK)
+segment_combine_kernel(const float* __restrict__ r_edge,
+                       const float* __restrict__ r_self,
+                       const long* __restrict__ edge_order,
+                       const long* __restrict__ indptr,
+                       float* __restrict__ out,
+                       int N, int OUT) {
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int v0 = blockIdx.x * WAVES_PER_BLOCK + wave;
+    const int stride = gridDim.x * WAVES_PER_BLOCK;
+    for (int v = v0; v < N; v += stride) {
+        const long e_begin = indptr[v];
+        const long e_end = indptr[v + 1];
+        const int indeg = (int)(e_end - e_begin);
+        if (lane < OUT) {
+            if (indeg > 0) {
+                float acc = r_self[(long)v * OUT + lane];
+                for (long e = e_begin; e < e_end; ++e)
+                    acc += r_edge[edge_order[e] * OUT + lane];
+                out[(long)v * OUT + lane] = acc / (indeg + 1);
+            } else {
+                out[(long)v * OUT + lane] = 0.0f;
+            }
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
+static int grid_for_tiles(long rows) {
+    long blocks = (rows + 16L * WAVES_PER_BLOCK - 1) / (16L * WAVES_PER_BLOCK);
+    if (blocks > 4096) blocks = 4096;
+    if (blocks < 1) blocks = 1;
+    return (int)blocks;
+}
+
+torch::Tensor row_mlp_mfma(torch::Tensor x, torch::Tensor ln_g,
+                           torch::Tensor ln_b, torch::Tensor W,
+                           torch::Tensor b) {
+    TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kFloat32);
+    const int R = (int)x.size(0), F = (int)x.size(1), H = (int)W.size(0);
+    TORCH_CHECK(F <= 64 && H <= 64 && H % 16 == 0,
+                "row_mlp_mfma needs F<=64, H multiple of 16");
+    auto y = torch::empty({R, H}, x.options());
+    if (R == 0) return y;
+    hipStream_t stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(row_mlp_mfma_kernel, dim3(grid_for_tiles(R)),
+                       dim3(BLOCK), 0, stream,
+                       x.data_ptr<float>(), ln_g.data_ptr<float>(),
+                       ln_b.data_ptr<float>(), W.data_ptr<float>(),
+                       b.data_ptr<float>(), y.data_ptr<float>(), R, F, H);
+    return y;
+}
+
+std::vector<torch::Tensor> message_mlp_mfma(
+    torch::Tensor hn, torch::Tensor he, torch::Tensor src, torch::Tensor ln_g,
+    torch::Tensor ln_b, torch::Tensor Wr, torch::Tensor br) {
+    TORCH_CHECK(hn.is_cuda() && hn.dtype() == torch::kFloat32);
+    const long E = src.size(0);
+    const int N = (int)hn.size(0), half = (int)hn.size(1);
+    const int OUT = (int)Wr.size(0);
+    TORCH_CHECK(half == 16 && (int)Wr.size(1) == 32,
+                "message_mlp_mfma needs half==16 (MSG=32)");
+    TORCH_CHECK(OUT <= 64 && OUT % 16 == 0);
+    auto r_edge = torch::empty({E, OUT}, hn.options());
+    auto r_self = torch::empty({N, OUT}, hn.options());
+    hipStream_t stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(message_mlp_mfma_kernel,
+                       dim3(grid_for_tiles(E + N)), dim3(BLOCK), 0, stream,
+                       hn.data_ptr<float>(), he.data_ptr<float>(),
+                       src.data_ptr<long>(), ln_g.data_ptr<float>(),
+                       ln_b.data_ptr<float>(), Wr.data_ptr<float>(),
+                       br.data_ptr<float>(), r_edge.data_ptr<float>(),
+                       r_self.data_ptr<float>(), E, N, OUT);
+    return {r_edge, r_self};
+}
+
+torch::Tensor segment_combine(torch::Tensor r_edge, torch::Tensor r_self,
+                              torch::Tensor edge_order, torch::Tensor indptr) {
+    const int N = (int)r_self.size(0), OUT = (int)r_self.size(1);
+    auto out = torch::empty({N, OUT}, r_self.options());
+    if (N == 0) return out;
+    hipStream_t stream = at::cuda::getCurrentCUDAStream();
+    int blocks = (N + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
+    if (blocks > 4096) blocks = 4096;
+    hipLaunchKernelGGL(segment_combine_kernel, dim3(blocks), dim3(BLOCK), 0,
+                       stream, r_edge.data_ptr<float>(),
+                       r_self.data_ptr<float>(), edge_order.data_ptr<long>(),
+                       indptr.data_ptr<long>(), out.data_ptr<float>(), N, OUT);
+    return out;
+}

@@ -293,3 +293,53 @@ def test_fused_policy_training_step_matches_torch(small_graph):
     for name in g_t:
         assert torch.allclose(g_h[name], g_t[name], atol=1e-3), \
             (name, (g_h[name] - g_t[name]).abs().max())
+
+
+@pytest.mark.gpu
+def test_mfma_kernels_match_valu():
+    """v_mfma_f32_16x16x4_f32 forward kernels vs the VALU fused kernels on
+    identical data (f32-in MFMA is an exact fmaf chain; only summation order
+    differs)."""
+    from ddls_amd import ops as hip_ops
+    ext = hip_ops.get_extension(required=True)
+    torch.manual_seed(5)
+    dev = "cuda:0"
+    N, E, F, half, OUT = 777, 2345, 5, 16, 64
+    z = torch.rand(N, F, device=dev)
+    e = torch.rand(E, 2, device=dev)
+    src = torch.randint(0, N, (E,), device=dev)
+    dst = torch.randint(0, N, (E,), device=dev)
+    order = torch.argsort(dst, stable=True)
+    counts = torch.bincount(dst, minlength=N)
+    indptr = torch.zeros(N + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(counts, 0, out=indptr[1:])
+    ln_n = (torch.rand(F, device=dev) + 0.5, torch.rand(F, device=dev) - 0.5)
+    ln_e = (torch.rand(2, device=dev) + 0.5, torch.rand(2, device=dev) - 0.5)
+    ln_r = (torch.rand(2 * half, device=dev) + 0.5,
+            torch.rand(2 * half, device=dev) - 0.5)
+    Wn, bn = torch.randn(half, F, device=dev) / 2, torch.randn(half, device=dev)
+    We, be = torch.randn(half, 2, device=dev) / 2, torch.randn(half, device=dev)
+    Wr, br = (torch.randn(OUT, 2 * half, device=dev) / 4,
+              torch.randn(OUT, device=dev))
+
+    hn_v = ext.row_mlp(z, *ln_n, Wn, bn)
+    hn_m = ext.row_mlp_mfma(z, *ln_n, Wn, bn)
+    assert torch.allclose(hn_v, hn_m, atol=1e-5), \
+        (hn_v - hn_m).abs().max().item()
+    he_v = ext.row_mlp(e, *ln_e, We, be)
+    he_m = ext.row_mlp_mfma(e, *ln_e, We, be)
+    assert torch.allclose(he_v, he_m, atol=1e-5)
+
+    out_v, r_edge_v, r_self_v = ext.message_reduce_train(
+        hn_v, he_v, src, order, indptr, *ln_r, Wr, br)
+    r_edge_m, r_self_m = ext.message_mlp_mfma(hn_v, he_v, src, *ln_r, Wr, br)
+    out_m = ext.segment_combine(r_edge_m, r_self_m, order, indptr)
+    assert torch.allclose(r_edge_v, r_edge_m, atol=1e-4), \
+        (r_edge_v - r_edge_m).abs().max().item()
+    # r_self only defined for nodes with mail
+    mail = (counts > 0)
+    assert torch.allclose(r_self_v[mail], r_self_m[mail], atol=1e-4)
+    assert torch.allclose(out_v, out_m, atol=1e-4), \
+        (out_v - out_m).abs().max().item()
+    # zero-fill semantics preserved
+    assert torch.all(out_m[~mail] == 0)
