@@ -22,6 +22,7 @@ _BUILD_DIR = os.path.join(_HERE, "_build")
 _SOURCES = [os.path.join(_HERE, "hip", "meanpool.hip"),
             os.path.join(_HERE, "hip", "meanpool_bwd.hip"),
             os.path.join(_HERE, "hip", "meanpool_mfma.hip"),
+            os.path.join(_HERE, "hip", "ppo_loss.hip"),
             os.path.join(_HERE, "hip", "lookahead.hip"),
             os.path.join(_HERE, "hip", "bindings.hip")]
 

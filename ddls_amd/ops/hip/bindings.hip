@@ -30,6 +30,15 @@ std::vector<torch::Tensor> message_mlp_mfma(
     torch::Tensor ln_b, torch::Tensor Wr, torch::Tensor br);
 torch::Tensor segment_combine(torch::Tensor r_edge, torch::Tensor r_self,
                               torch::Tensor edge_order, torch::Tensor indptr);
+std::vector<torch::Tensor> ppo_loss_fwd(
+    torch::Tensor logits, torch::Tensor values, torch::Tensor actions,
+    torch::Tensor old_logp, torch::Tensor adv, torch::Tensor vtarg,
+    torch::Tensor kl_coef, double clip, double vf_clip, double vf_coef,
+    double ent_coef);
+std::vector<torch::Tensor> ppo_loss_bwd(
+    torch::Tensor p, torch::Tensor lp, torch::Tensor coef, torch::Tensor h,
+    torch::Tensor actions, torch::Tensor values, torch::Tensor vtarg,
+    torch::Tensor gl, double vf_clip, double vf_coef, double ent_coef);
 std::vector<torch::Tensor> lookahead_batch(
     torch::Tensor descs, torch::Tensor op_remaining, torch::Tensor op_worker,
     torch::Tensor op_priority, torch::Tensor out_indptr, torch::Tensor out_edges,
@@ -58,4 +67,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "per-message reduce-MLP on the matrix core");
     m.def("segment_combine", &segment_combine,
           "CSR segment mean of stored message activations");
+    m.def("ppo_loss_fwd", &ppo_loss_fwd, "fused PPO clipped-surrogate loss");
+    m.def("ppo_loss_bwd", &ppo_loss_bwd, "analytic PPO loss backward");
 }

@@ -23,6 +23,7 @@ segments (fwd+bwd | clip+step) with the eager fused all-reduce between them.
 """
 from __future__ import annotations
 
+import os
 from typing import List, Optional
 
 import numpy as np
@@ -32,6 +33,36 @@ import torch.nn.functional as F
 from ..models.gnn import GraphBatch
 from ..parallel import all_reduce_gradients, is_distributed
 from .rollout import CompactObs
+
+
+class _PPOLossFn(torch.autograd.Function):
+    """Fused PPO loss (ops/hip/ppo_loss.hip): one kernel forward, one kernel
+    analytic backward — replaces ~100 tiny elementwise kernels per replay."""
+
+    @staticmethod
+    def forward(ctx, logits, values, actions, old_logp, adv, vtarg,
+                kl_coef_t, clip, vf_clip, vf_coef, ent_coef):
+        from .. import ops as hip_ops
+        ext = hip_ops.get_extension(required=True)
+        loss, stats, p, lp, coef, h = ext.ppo_loss_fwd(
+            logits.contiguous(), values.contiguous(), actions, old_logp,
+            adv, vtarg, kl_coef_t.reshape(1), float(clip), float(vf_clip),
+            float(vf_coef), float(ent_coef))
+        ctx.save_for_backward(p, lp, coef, h, actions, values, vtarg)
+        ctx.consts = (float(vf_clip), float(vf_coef), float(ent_coef))
+        ctx.mark_non_differentiable(stats)
+        return loss.reshape(()), stats
+
+    @staticmethod
+    def backward(ctx, gl, _gstats):
+        from .. import ops as hip_ops
+        ext = hip_ops.get_extension(required=True)
+        p, lp, coef, h, actions, values, vtarg = ctx.saved_tensors
+        vf_clip, vf_coef, ent_coef = ctx.consts
+        glogits, gvalues = ext.ppo_loss_bwd(
+            p, lp, coef, h, actions, values, vtarg,
+            gl.reshape(1).contiguous(), vf_clip, vf_coef, ent_coef)
+        return (glogits, gvalues) + (None,) * 9
 
 
 class CapturedSGDStep:
@@ -66,6 +97,15 @@ class CapturedSGDStep:
         self.graph_opt = None   # second segment when distributed
         self.capture_count = 0
         self.last_error: Optional[str] = None
+        try:
+            from .. import ops as hip_ops
+            ext = hip_ops.get_extension()
+        except Exception:
+            ext = None
+        self._fused_loss = (
+            ext is not None and hasattr(ext, "ppo_loss_fwd")
+            and self.A <= 64
+            and os.environ.get("DDLS_AMD_DISABLE_FUSED_LOSS", "0") != "1")
         # [policy_loss, vf_loss, kl, entropy, total_loss] device accumulator:
         # read ONCE per update() so replays never host-sync
         self.stats_acc = torch.zeros(5, device=device)
@@ -235,6 +275,15 @@ class CapturedSGDStep:
         logits, values = self.policy.forward_flat(
             self.batch, self.d["gf"], self.d["mask"])
         logits, values = logits[:B], values[:B]
+        if self._fused_loss:
+            loss, stats = _PPOLossFn.apply(
+                logits, values, self.d["actions"], self.d["old_logp"],
+                self.d["adv"], self.d["vtarg"], self.kl_coeff_t,
+                cfg.clip_param, cfg.vf_clip_param, cfg.vf_loss_coeff,
+                cfg.entropy_coeff)
+            loss.backward()
+            self.stats_acc += stats
+            return
         # Categorical re-implemented with log_softmax: torch.distributions
         # argument validation host-syncs, which aborts stream capture
         logp_all = F.log_softmax(logits, dim=-1)
