@@ -302,9 +302,38 @@ class GNN(nn.Module):
         return z
 
 
+class _SegmentMeanFn(torch.autograd.Function):
+    """HIP per-graph mean with a broadcast/scale backward (replaces the
+    torch index_add path, which costs ~100us per minibatch on MI355X)."""
+
+    @staticmethod
+    def forward(ctx, x, node_ptr, G):
+        from .. import ops as hip_ops
+        ext = hip_ops.get_extension(required=True)
+        out = ext.segment_mean(x.contiguous(), node_ptr, G)
+        ctx.save_for_backward(node_ptr)
+        ctx.N = x.shape[0]
+        return out
+
+    @staticmethod
+    def backward(ctx, gout):
+        from .. import ops as hip_ops
+        ext = hip_ops.get_extension(required=True)
+        (node_ptr,) = ctx.saved_tensors
+        return ext.segment_mean_bwd(gout.contiguous(), node_ptr,
+                                    ctx.N), None, None
+
+
 def graph_mean(node_emb: torch.Tensor, batch: GraphBatch) -> torch.Tensor:
     """Per-graph mean of node embeddings (HIP segment kernel on GPU)."""
     from .. import ops as hip_ops
+    if (node_emb.is_cuda and torch.is_grad_enabled()
+            and node_emb.requires_grad
+            and hip_ops.get_extension() is not None
+            and hasattr(hip_ops.get_extension(), "segment_mean_bwd")
+            and os.environ.get("DDLS_AMD_DISABLE_HIP", "0") != "1"):
+        return _SegmentMeanFn.apply(node_emb, batch.node_ptr(),
+                                    batch.num_graphs)
     if hip_ops.hip_ops_enabled_for(node_emb):
         ext = hip_ops.get_extension(required=True)
         return ext.segment_mean(node_emb.contiguous(), batch.node_ptr(),

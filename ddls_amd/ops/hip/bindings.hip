@@ -30,6 +30,11 @@ std::vector<torch::Tensor> message_mlp_mfma(
     torch::Tensor ln_b, torch::Tensor Wr, torch::Tensor br);
 torch::Tensor segment_combine(torch::Tensor r_edge, torch::Tensor r_self,
                               torch::Tensor edge_order, torch::Tensor indptr);
+torch::Tensor segment_mean_bwd(torch::Tensor gout, torch::Tensor node_ptr,
+                               int64_t N);
+void flat_adam(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+               torch::Tensor v, torch::Tensor step_t, torch::Tensor normsq,
+               double clip, double lr, double b1, double b2, double eps);
 std::vector<torch::Tensor> ppo_loss_fwd(
     torch::Tensor logits, torch::Tensor values, torch::Tensor actions,
     torch::Tensor old_logp, torch::Tensor adv, torch::Tensor vtarg,
@@ -69,4 +74,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "CSR segment mean of stored message activations");
     m.def("ppo_loss_fwd", &ppo_loss_fwd, "fused PPO clipped-surrogate loss");
     m.def("ppo_loss_bwd", &ppo_loss_bwd, "analytic PPO loss backward");
+    m.def("segment_mean_bwd", &segment_mean_bwd,
+          "per-graph mean backward (broadcast/scale)");
+    m.def("flat_adam", &flat_adam,
+          "fused grad-clip + Adam over flat param/grad/m/v buffers");
 }

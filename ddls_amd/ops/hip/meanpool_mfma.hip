@@ -293,3 +293,98 @@ torch::Tensor segment_combine(torch::Tensor r_edge, torch::Tensor r_self,
                        indptr.data_ptr<long>(), out.data_ptr<float>(), N, OUT);
     return out;
 }
+
+// ---------------------------------------------------------------------------
+// segment_mean backward: gx[n] = gout[g]/count(g) for n in graph g.
+__global__ void __launch_bounds__(BLOCK)
+segment_mean_bwd_kernel(const float* __restrict__ gout,
+                        const long* __restrict__ node_ptr,
+                        float* __restrict__ gx,
+                        int G, int F) {
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int g0 = blockIdx.x * WAVES_PER_BLOCK + wave;
+    const int stride = gridDim.x * WAVES_PER_BLOCK;
+    for (int g = g0; g < G; g += stride) {
+        const long n_begin = node_ptr[g];
+        const long n_end = node_ptr[g + 1];
+        const long cnt = n_end - n_begin;
+        if (lane < F && cnt > 0) {
+            const float val = gout[(long)g * F + lane] / cnt;
+            for (long n = n_begin; n < n_end; ++n)
+                gx[n * F + lane] = val;
+        }
+    }
+}
+
+torch::Tensor segment_mean_bwd(torch::Tensor gout, torch::Tensor node_ptr,
+                               int64_t N) {
+    const int G = (int)gout.size(0), F = (int)gout.size(1);
+    auto gx = torch::zeros({N, F}, gout.options());
+    if (G == 0) return gx;
+    hipStream_t stream = at::cuda::getCurrentCUDAStream();
+    int blocks = (G + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
+    if (blocks > 4096) blocks = 4096;
+    hipLaunchKernelGGL(segment_mean_bwd_kernel, dim3(blocks), dim3(BLOCK), 0,
+                       stream, gout.data_ptr<float>(),
+                       node_ptr.data_ptr<long>(), gx.data_ptr<float>(), G, F);
+    return gx;
+}
+
+// ---------------------------------------------------------------------------
+// Fused flat Adam + grad clip: one kernel over the flat param/grad/m/v
+// buffers (the captured step keeps every param and grad a VIEW into these).
+// scale = min(1, clip / (sqrt(normsq) + 1e-6)) matches
+// torch.nn.utils.clip_grad_norm_; the update matches torch Adam exactly
+// (step_size = lr/bias1, denom = sqrt(v)/sqrt(bias2) + eps).  Grads are
+// zeroed after consumption so the next backward accumulates from zero.
+__global__ void flat_adam_kernel(float* __restrict__ p,
+                                 float* __restrict__ g,
+                                 float* __restrict__ m,
+                                 float* __restrict__ v,
+                                 const float* __restrict__ step_t,  // [1]
+                                 const float* __restrict__ normsq,  // [1]
+                                 long numel, float clip, float lr, float b1,
+                                 float b2, float eps) {
+    const float t = step_t[0];
+    float scale = 1.0f;
+    if (clip > 0.0f) {
+        const float norm = sqrtf(normsq[0]);
+        const float c = clip / (norm + 1e-6f);
+        scale = fminf(c, 1.0f);
+    }
+    const float bias1 = 1.0f - powf(b1, t);
+    const float bias2 = 1.0f - powf(b2, t);
+    const float step_size = lr / bias1;
+    const float inv_sqrt_bias2 = rsqrtf(bias2);
+    const long i0 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+    const long stride = (long)gridDim.x * blockDim.x;
+    for (long i = i0; i < numel; i += stride) {
+        const float gi = g[i] * scale;
+        const float mi = b1 * m[i] + (1.0f - b1) * gi;
+        const float vi = b2 * v[i] + (1.0f - b2) * gi * gi;
+        m[i] = mi;
+        v[i] = vi;
+        p[i] -= step_size * mi / (sqrtf(vi) * inv_sqrt_bias2 + eps);
+        g[i] = 0.0f;
+    }
+}
+
+__global__ void step_inc_kernel(float* step_t) { step_t[0] += 1.0f; }
+
+void flat_adam(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+               torch::Tensor v, torch::Tensor step_t, torch::Tensor normsq,
+               double clip, double lr, double b1, double b2, double eps) {
+    const long numel = p.numel();
+    hipStream_t stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(step_inc_kernel, dim3(1), dim3(1), 0, stream,
+                       step_t.data_ptr<float>());
+    long blocks = (numel + BLOCK - 1) / BLOCK;
+    if (blocks > 2048) blocks = 2048;
+    hipLaunchKernelGGL(flat_adam_kernel, dim3((int)blocks), dim3(BLOCK), 0,
+                       stream, p.data_ptr<float>(), g.data_ptr<float>(),
+                       m.data_ptr<float>(), v.data_ptr<float>(),
+                       step_t.data_ptr<float>(), normsq.data_ptr<float>(),
+                       numel, (float)clip, (float)lr, (float)b1, (float)b2,
+                       (float)eps);
+}

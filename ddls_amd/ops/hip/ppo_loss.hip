@@ -108,25 +108,31 @@ ppo_loss_fwd_kernel(const float* __restrict__ logits,
         float t[4] = {0, 0, 0, 0};
         for (int w = 0; w < WAVES; ++w)
             for (int i = 0; i < 4; ++i) t[i] += acc[w][i];
-        if (gridDim.x == 1) {
-            const float inv = 1.0f / B;
-            const float pl = -t[0] * inv, kl = t[1] * inv, vf = t[2] * inv,
-                        ent = t[3] * inv;
-            const float loss = pl + kl_coef[0] * kl + vf_coef * vf
-                               - ent_coef * ent;
-            loss_out[0] = loss;
-            stats_out[0] = pl;
-            stats_out[1] = vf;
-            stats_out[2] = kl;
-            stats_out[3] = ent;
-            stats_out[4] = loss;
-        } else {
-            atomicAdd(&stats_out[0], -t[0]);  // finalised on host side
-            atomicAdd(&stats_out[1], t[2]);
-            atomicAdd(&stats_out[2], t[1]);
-            atomicAdd(&stats_out[3], t[3]);
-        }
+        // raw sums; the finalize kernel turns them into means + total loss
+        atomicAdd(&stats_out[0], -t[0]);
+        atomicAdd(&stats_out[1], t[2]);
+        atomicAdd(&stats_out[2], t[1]);
+        atomicAdd(&stats_out[3], t[3]);
     }
+}
+
+__global__ void ppo_loss_finalize_kernel(const float* __restrict__ kl_coef,
+                                         float* __restrict__ loss_out,
+                                         float* __restrict__ stats_out,
+                                         int B, float vf_coef,
+                                         float ent_coef) {
+    const float inv = 1.0f / B;
+    const float pl = stats_out[0] * inv;
+    const float vf = stats_out[1] * inv;
+    const float kl = stats_out[2] * inv;
+    const float ent = stats_out[3] * inv;
+    const float loss = pl + kl_coef[0] * kl + vf_coef * vf - ent_coef * ent;
+    stats_out[0] = pl;
+    stats_out[1] = vf;
+    stats_out[2] = kl;
+    stats_out[3] = ent;
+    stats_out[4] = loss;
+    loss_out[0] = loss;
 }
 
 __global__ void __launch_bounds__(BLOCK)
@@ -183,8 +189,10 @@ std::vector<torch::Tensor> ppo_loss_fwd(
     auto loss = torch::zeros({1}, opt);
     auto stats = torch::zeros({5}, opt);
     hipStream_t stream = at::cuda::getCurrentCUDAStream();
-    // single block so the reduction finishes on-device (B is a minibatch)
-    hipLaunchKernelGGL(ppo_loss_fwd_kernel, dim3(1), dim3(BLOCK), 0, stream,
+    int blocks = (B + WAVES - 1) / WAVES;
+    if (blocks > 128) blocks = 128;
+    hipLaunchKernelGGL(ppo_loss_fwd_kernel, dim3(blocks), dim3(BLOCK), 0,
+                       stream,
                        logits.data_ptr<float>(), values.data_ptr<float>(),
                        actions.data_ptr<long>(), old_logp.data_ptr<float>(),
                        adv.data_ptr<float>(), vtarg.data_ptr<float>(),
@@ -193,6 +201,10 @@ std::vector<torch::Tensor> ppo_loss_fwd(
                        h.data_ptr<float>(), loss.data_ptr<float>(),
                        stats.data_ptr<float>(), B, A, (float)clip,
                        (float)vf_clip, (float)vf_coef, (float)ent_coef);
+    hipLaunchKernelGGL(ppo_loss_finalize_kernel, dim3(1), dim3(1), 0, stream,
+                       kl_coef.data_ptr<float>(), loss.data_ptr<float>(),
+                       stats.data_ptr<float>(), B, (float)vf_coef,
+                       (float)ent_coef);
     return {loss, stats, p, lp, coef, h};
 }
 

@@ -106,6 +106,14 @@ class CapturedSGDStep:
             ext is not None and hasattr(ext, "ppo_loss_fwd")
             and self.A <= 64
             and os.environ.get("DDLS_AMD_DISABLE_FUSED_LOSS", "0") != "1")
+        # flat-buffer optimizer engine: every param and grad becomes a view
+        # into one flat tensor each, and clip+Adam run as ONE fused kernel
+        # (torch capturable Adam decomposes into ~2 kernels per param)
+        self._flat_engine = (
+            ext is not None and hasattr(ext, "flat_adam")
+            and os.environ.get("DDLS_AMD_DISABLE_FLAT_ADAM", "0") != "1")
+        self._ext = ext
+        self.flat_p = None
         # [policy_loss, vf_loss, kl, entropy, total_loss] device accumulator:
         # read ONCE per update() so replays never host-sync
         self.stats_acc = torch.zeros(5, device=device)
@@ -170,7 +178,10 @@ class CapturedSGDStep:
         self.copy_events[j].record()
         self.graph.replay()
         if self.graph_opt is not None:
-            all_reduce_gradients(self.policy.parameters())
+            if self.flat_p is not None:
+                self._all_reduce_flat()
+            else:
+                all_reduce_gradients(self.policy.parameters())
             self.graph_opt.replay()
 
     # ------------------------------------------------------------------
@@ -192,6 +203,7 @@ class CapturedSGDStep:
             "order": dbuf(e_cap, torch.int64),
             "indptr": dbuf(n_cap + 1, torch.int64),
             "gon": dbuf(n_cap, torch.int64),
+            "nptr": dbuf(B + 2, torch.int64),
             "gf": dbuf((B + 1, self.Fg), torch.float32),
             "mask": torch.ones((B + 1, self.A), dtype=torch.float32, device=dev),
             "actions": dbuf(B, torch.int64),
@@ -216,7 +228,8 @@ class CapturedSGDStep:
         self.batch = GraphBatch(
             z=self.d["z"], e=self.d["e"], src=self.d["src"], dst=self.d["dst"],
             graph_of_node=self.d["gon"], num_graphs=B + 1,
-            _csr=(self.d["order"], self.d["indptr"]))
+            _csr=(self.d["order"], self.d["indptr"]),
+            _node_ptr=self.d["nptr"])
 
     def _fill(self, mb_obs, actions, old_logp, adv, vtarg):
         ns = np.array([len(o.node_features) for o in mb_obs], dtype=np.int64)
@@ -255,6 +268,9 @@ class CapturedSGDStep:
         np.cumsum(counts, out=p["indptr"][1:])
         p["gon"][:n] = np.repeat(np.arange(self.B, dtype=np.int64), ns)
         p["gon"][n:] = self.B
+        p["nptr"][0] = 0
+        np.cumsum(ns, out=p["nptr"][1:self.B + 1])
+        p["nptr"][self.B + 1] = self.n_cap  # dummy graph holds all padding
         p["gf"][:self.B] = np.stack([o.graph_features for o in mb_obs])
         p["mask"][:self.B] = np.stack([o.action_mask for o in mb_obs])
         # row B of gf stays zero and of mask stays all-ones (log(1)=0), so the
@@ -268,10 +284,12 @@ class CapturedSGDStep:
     # ------------------------------------------------------------------
     def _body_fwd_bwd(self):
         cfg, B = self.cfg, self.B
-        grads = [pa.grad for pa in self.policy.parameters()
-                 if pa.grad is not None]
-        if grads:
-            torch._foreach_zero_(grads)
+        if self.flat_p is None:
+            # flat engine zeroes grads inside flat_adam instead
+            grads = [pa.grad for pa in self.policy.parameters()
+                     if pa.grad is not None]
+            if grads:
+                torch._foreach_zero_(grads)
         logits, values = self.policy.forward_flat(
             self.batch, self.d["gf"], self.d["mask"])
         logits, values = logits[:B], values[:B]
@@ -306,10 +324,81 @@ class CapturedSGDStep:
                                        loss.detach()])
 
     def _body_opt(self):
+        if self.flat_p is not None:
+            # fused grad-clip + Adam over the flat buffers: 3 kernels total,
+            # and flat_adam zeroes the grads for the next backward
+            normsq = torch.dot(self.flat_g, self.flat_g)
+            self._ext.flat_adam(self.flat_p, self.flat_g, self.flat_m,
+                                self.flat_v, self.step_t, normsq.reshape(1),
+                                float(self.cfg.grad_clip or 0.0),
+                                float(self.cfg.lr), 0.9, 0.999, 1e-8)
+            return
         if self.cfg.grad_clip is not None:
             torch.nn.utils.clip_grad_norm_(self.policy.parameters(),
                                            self.cfg.grad_clip)
         self.optimizer.step()
+
+    # ------------------------------------------------------------------
+    def _flatten_params(self):
+        """Re-materialise every param and grad as a view into one flat
+        tensor each (apex-style), enabling the fused flat_adam kernel and a
+        copy-free distributed all-reduce of flat_g."""
+        if self.flat_p is not None:
+            return
+        params = list(self.policy.parameters())
+        numel = sum(p.numel() for p in params)
+        dev = self.device
+        self.flat_p = torch.zeros(numel, device=dev)
+        self.flat_g = torch.zeros(numel, device=dev)
+        self.flat_m = torch.zeros(numel, device=dev)
+        self.flat_v = torch.zeros(numel, device=dev)
+        self.step_t = torch.zeros(1, device=dev)
+        self._params = params
+        self._offs = []
+        off = 0
+        for p in params:
+            n = p.numel()
+            self.flat_p[off:off + n].copy_(p.data.reshape(-1))
+            p.data = self.flat_p[off:off + n].view(p.shape)
+            p.grad = self.flat_g[off:off + n].view(p.shape)
+            self._offs.append((off, n))
+            off += n
+        # adopt any existing torch-Adam state (e.g. a loaded checkpoint)
+        step_val = None
+        for p, (o, n) in zip(params, self._offs):
+            st = self.optimizer.state.get(p)
+            if st and "exp_avg" in st:
+                self.flat_m[o:o + n].copy_(st["exp_avg"].reshape(-1).to(dev))
+                self.flat_v[o:o + n].copy_(
+                    st["exp_avg_sq"].reshape(-1).to(dev))
+                if "step" in st:
+                    step_val = float(st["step"])
+        if step_val is not None:
+            self.step_t.fill_(step_val)
+
+    def sync_state_to_optimizer(self):
+        """Write the flat Adam state back into the torch optimizer (for
+        checkpointing and for the eager fallback path)."""
+        if self.flat_p is None:
+            return
+        step_val = float(self.step_t.item())
+        for p, (o, n) in zip(self._params, self._offs):
+            st = self.optimizer.state.setdefault(p, {})
+            st["step"] = torch.tensor(step_val)
+            st["exp_avg"] = self.flat_m[o:o + n].detach().clone().view(p.shape)
+            st["exp_avg_sq"] = (self.flat_v[o:o + n].detach().clone()
+                                .view(p.shape))
+
+    def _all_reduce_flat(self):
+        import torch.distributed as dist
+        g = self.flat_g
+        if dist.get_backend() == "gloo" and g.is_cuda:
+            host = g.cpu()
+            dist.all_reduce(host)
+            g.copy_(host.to(g.device))
+        else:
+            dist.all_reduce(g)
+        g.div_(dist.get_world_size())
 
     def _capture(self, n_cap: int, e_cap: int):
         self._alloc(n_cap, e_cap)
@@ -320,25 +409,16 @@ class CapturedSGDStep:
             if not split:
                 self._body_opt()
 
-        # warmup on a side stream (required before capture); lr=0 so the real
-        # optimizer steps during warmup cannot move the parameters, and the
-        # Adam state is snapshot/restored in-place (same tensor addresses, so
-        # the captured graph sees the live state) so training resumes exactly
-        # where an uncaptured run would be
-        saved_lr = [g["lr"] for g in self.optimizer.param_groups]
-        for g in self.optimizer.param_groups:
-            g["lr"] = 0.0
-            # capturable Adam keeps `step` on-device so the whole update is
-            # graph-safe; migrate any state created by earlier eager steps
-            g["capturable"] = True
-        for st in self.optimizer.state.values():
-            if "step" in st and torch.is_tensor(st["step"]):
-                st["step"] = st["step"].to(self.device)
-        snap = {p_: {k: v.clone() if torch.is_tensor(v) else v
-                     for k, v in st.items()}
-                for p_, st in self.optimizer.state.items()}
         saved_stats = self.stats_acc.clone()
-        try:
+        if self._flat_engine:
+            self._flatten_params()
+            # warmup (required before capture) runs REAL steps; snapshot the
+            # flat training state and restore it in place afterwards so the
+            # captured graph sees the live tensors but training resumes
+            # exactly where an uncaptured run would be
+            snap = tuple(t.clone() for t in (self.flat_p, self.flat_g,
+                                             self.flat_m, self.flat_v,
+                                             self.step_t))
             s = torch.cuda.Stream()
             s.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(s):
@@ -348,20 +428,50 @@ class CapturedSGDStep:
                         self._body_opt()
             torch.cuda.current_stream().wait_stream(s)
             torch.cuda.synchronize()
-        finally:
-            for g, lr in zip(self.optimizer.param_groups, saved_lr):
-                g["lr"] = lr
-        # undo warmup's pollution of the optimizer state: restore snapshotted
-        # entries, zero entries first created by warmup (fresh-Adam semantics)
-        for p_, st in self.optimizer.state.items():
-            prev = snap.get(p_)
-            for k, v in st.items():
-                if not torch.is_tensor(v):
-                    continue
-                if prev is not None and k in prev:
-                    v.copy_(prev[k])
-                else:
-                    v.zero_()
+            for dst, src in zip((self.flat_p, self.flat_g, self.flat_m,
+                                 self.flat_v, self.step_t), snap):
+                dst.copy_(src)
+        else:
+            # torch-Adam path: lr=0 during warmup so the optimizer steps
+            # cannot move the parameters, and the Adam state is
+            # snapshot/restored in place (same tensor addresses)
+            saved_lr = [g["lr"] for g in self.optimizer.param_groups]
+            for g in self.optimizer.param_groups:
+                g["lr"] = 0.0
+                # capturable Adam keeps `step` on-device so the whole update
+                # is graph-safe; migrate state created by earlier eager steps
+                g["capturable"] = True
+            for st in self.optimizer.state.values():
+                if "step" in st and torch.is_tensor(st["step"]):
+                    st["step"] = st["step"].to(self.device)
+            snap = {p_: {k: v.clone() if torch.is_tensor(v) else v
+                         for k, v in st.items()}
+                    for p_, st in self.optimizer.state.items()}
+            try:
+                s = torch.cuda.Stream()
+                s.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(s):
+                    for _ in range(3):
+                        whole()
+                        if split:
+                            self._body_opt()
+                torch.cuda.current_stream().wait_stream(s)
+                torch.cuda.synchronize()
+            finally:
+                for g, lr in zip(self.optimizer.param_groups, saved_lr):
+                    g["lr"] = lr
+            # undo warmup's pollution of the optimizer state: restore
+            # snapshotted entries, zero entries first created by warmup
+            # (fresh-Adam semantics)
+            for p_, st in self.optimizer.state.items():
+                prev = snap.get(p_)
+                for k, v in st.items():
+                    if not torch.is_tensor(v):
+                        continue
+                    if prev is not None and k in prev:
+                        v.copy_(prev[k])
+                    else:
+                        v.zero_()
         self.stats_acc.copy_(saved_stats)
 
         self.graph = torch.cuda.CUDAGraph()

@@ -357,6 +357,10 @@ class PPOTrainer:
 
     # ------------------------------------------------------------------
     def state_dict(self) -> Dict:
+        if self._stepper is not None:
+            # the captured step keeps Adam state in flat buffers; mirror it
+            # back so checkpoints stay torch-Adam compatible
+            self._stepper.sync_state_to_optimizer()
         return {
             "policy": self.policy.state_dict(),
             "optimizer": self.optimizer.state_dict(),
