@@ -141,6 +141,8 @@ def main():
     if rank == 0 and getattr(trainer, "_stepper", None) is not None:
         sp = trainer._stepper
         print(f"[bench] hipgraph: broken={sp.broken} captures={sp.capture_count} "
+              f"flat_adam={getattr(sp, 'flat_p', None) is not None} "
+              f"fused_loss={getattr(sp, '_fused_loss', False)} "
               f"err={sp.last_error}", file=sys.stderr, flush=True)
     elapsed = time.perf_counter() - t0
 

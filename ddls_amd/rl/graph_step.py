@@ -242,6 +242,21 @@ class CapturedSGDStep:
         dst = np.concatenate([o.edges_dst + off
                               for o, off in zip(mb_obs, offsets)])
         m = len(src)
+        # per-sample CSR cached on the obs (reused across the 8 SGD epochs):
+        # samples are node-offset-ordered, so concatenating sample-local
+        # dst-sorted orders IS the global dst-sorted order
+        orders, cnts = [], []
+        eoff = 0
+        for o in mb_obs:
+            c = getattr(o, "_csr", None)
+            if c is None:
+                c = (np.argsort(o.edges_dst, kind="stable").astype(np.int64),
+                     np.bincount(o.edges_dst,
+                                 minlength=len(o.node_features)))
+                o._csr = c
+            orders.append(c[0] + eoff)
+            cnts.append(c[1])
+            eoff += len(c[0])
         j = self._buf = self._buf ^ 1
         # wait until the GPU finished the copies that last read this pinned set
         self.copy_events[j].synchronize()
@@ -259,10 +274,12 @@ class CapturedSGDStep:
         p["src"][m:] = pad_dst
         p["dst"][m:] = pad_dst
         # padded edges sit at node ids above every real dst and are already
-        # sorted, so only real edges need sorting
-        p["order"][:m] = np.argsort(dst, kind="stable")
+        # sorted, so only real edges need sorting (from the per-sample cache)
+        p["order"][:m] = np.concatenate(orders) if orders else []
         p["order"][m:] = np.arange(m, self.e_cap)
-        counts = np.bincount(dst, minlength=self.n_cap)
+        counts = np.zeros(self.n_cap, dtype=np.int64)
+        cat_cnts = np.concatenate(cnts) if cnts else np.zeros(0, np.int64)
+        counts[:len(cat_cnts)] = cat_cnts
         counts += np.bincount(pad_dst, minlength=self.n_cap)
         p["indptr"][0] = 0
         np.cumsum(counts, out=p["indptr"][1:])
