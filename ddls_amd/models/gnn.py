@@ -155,6 +155,7 @@ class _FusedMeanPoolFn(torch.autograd.Function):
         ctx.save_for_backward(zc, ec, src, dst, indptr, hn, he, r_edge, r_self,
                               ln_n_w, ln_n_b, Wn, ln_e_w, ln_e_b, We,
                               ln_r_w, ln_r_b, Wr)
+        ctx.use_mfma = use_mfma
         return out
 
     @staticmethod
@@ -164,9 +165,14 @@ class _FusedMeanPoolFn(torch.autograd.Function):
         (z, e, src, dst, indptr, hn, he, r_edge, r_self,
          ln_n_w, ln_n_b, W_n, ln_e_w, ln_e_b, W_e,
          ln_r_w, ln_r_b, W_r) = ctx.saved_tensors
-        ghn, ghe, gWr, gbr, glnr_g, glnr_b = ext.message_reduce_bwd(
-            hn, he, src, dst, indptr, ln_r_w, ln_r_b, W_r, r_edge, r_self,
-            gout.contiguous())
+        if ctx.use_mfma and hasattr(ext, "message_reduce_bwd_mfma"):
+            ghn, ghe, gWr, gbr, glnr_g, glnr_b = ext.message_reduce_bwd_mfma(
+                hn, he, src, dst, indptr, ln_r_w, ln_r_b, W_r, r_edge, r_self,
+                gout.contiguous())
+        else:
+            ghn, ghe, gWr, gbr, glnr_g, glnr_b = ext.message_reduce_bwd(
+                hn, he, src, dst, indptr, ln_r_w, ln_r_b, W_r, r_edge, r_self,
+                gout.contiguous())
         gz, gWn, gbn, glnn_g, glnn_b = ext.row_mlp_bwd(
             z, hn, ghn, ln_n_w, ln_n_b, W_n)
         ge, gWe, gbe, glne_g, glne_b = ext.row_mlp_bwd(

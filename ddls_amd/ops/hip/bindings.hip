@@ -32,6 +32,11 @@ torch::Tensor segment_combine(torch::Tensor r_edge, torch::Tensor r_self,
                               torch::Tensor edge_order, torch::Tensor indptr);
 torch::Tensor segment_mean_bwd(torch::Tensor gout, torch::Tensor node_ptr,
                                int64_t N);
+std::vector<torch::Tensor> message_reduce_bwd_mfma(
+    torch::Tensor hn, torch::Tensor he, torch::Tensor src, torch::Tensor dst,
+    torch::Tensor indptr, torch::Tensor ln_g, torch::Tensor ln_b,
+    torch::Tensor Wr, torch::Tensor r_edge, torch::Tensor r_self,
+    torch::Tensor gout);
 void flat_adam(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, torch::Tensor step_t, torch::Tensor normsq,
                double clip, double lr, double b1, double b2, double eps);
@@ -78,4 +83,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "per-graph mean backward (broadcast/scale)");
     m.def("flat_adam", &flat_adam,
           "fused grad-clip + Adam over flat param/grad/m/v buffers");
+    m.def("message_reduce_bwd_mfma", &message_reduce_bwd_mfma,
+          "matrix-core message-passing backward (3-stage MFMA)");
 }

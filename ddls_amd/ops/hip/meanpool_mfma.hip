@@ -390,3 +390,235 @@ void flat_adam(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                        numel, (float)clip, (float)lr, (float)b1, (float)b2,
                        (float)eps);
 }
+
+// ===========================================================================
+// MFMA backward for the message reduce-MLP (replaces the wave-per-message
+// meanpool_bwd kernel for the tuned shape half=16/MSG=32).  Three stages:
+//   K1 msg_bwd_ga:   gA[m,j] = relu'(r[m,j]) * gout[node(m),j] / (deg+1)
+//   K2 msg_bwd_data: gXln = gA @ Wr (MFMA) -> LayerNorm backward ->
+//                    scatter ghn (atomic) / ghe; stores mhat & gy for the
+//                    parameter-gradient column sums (done with torch sums)
+//   K3 msg_bwd_wr:   gWr = gA^T @ xln (MFMA over message chunks, xln
+//                    recomputed as mhat*g+b), atomically accumulated
+// Message order: 0..E-1 edge messages, E..E+N-1 self messages (as forward).
+// ===========================================================================
+
+__global__ void __launch_bounds__(BLOCK)
+msg_bwd_ga_kernel(const float* __restrict__ r_edge,
+                  const float* __restrict__ r_self,
+                  const long* __restrict__ dst,
+                  const long* __restrict__ indptr,
+                  const float* __restrict__ gout,
+                  float* __restrict__ gA,
+                  long E, int N, int OUT) {
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const long M = E + (long)N;
+    const long m0 = (long)blockIdx.x * WAVES_PER_BLOCK + wave;
+    const long stride = (long)gridDim.x * WAVES_PER_BLOCK;
+    for (long m = m0; m < M; m += stride) {
+        const long node = (m < E) ? dst[m] : (m - E);
+        const long deg = indptr[node + 1] - indptr[node];
+        if (lane < OUT) {
+            float v = 0.0f;
+            if (deg > 0) {
+                const float r = (m < E) ? r_edge[m * OUT + lane]
+                                        : r_self[(m - E) * OUT + lane];
+                if (r > 0.0f)
+                    v = gout[node * OUT + lane] / (float)(deg + 1);
+            }
+            gA[m * OUT + lane] = v;
+        }
+    }
+}
+
+__global__ void __launch_bounds__(BLOCK)
+msg_bwd_data_kernel(const float* __restrict__ hn,
+                    const float* __restrict__ he,
+                    const long* __restrict__ src,
+                    const float* __restrict__ ln_g,
+                    const float* __restrict__ Wr,
+                    const float* __restrict__ gA,
+                    float* __restrict__ ghn,     // [N,16] pre-zeroed
+                    float* __restrict__ ghe,     // [E,16]
+                    float* __restrict__ mhat_out,  // [M,32]
+                    float* __restrict__ gy_out,    // [M,32]
+                    long E, int N, int OUT) {
+    const int MSG = 32, Q = 8;
+    __shared__ float ws[64][MSG + 4];          // Wr [OUT][MSG]
+    __shared__ float gx[WAVES_PER_BLOCK][16][MSG + 4];
+    for (int i = threadIdx.x; i < OUT * MSG; i += BLOCK)
+        ws[i / MSG][i % MSG] = Wr[i];
+    __syncthreads();
+
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int mrow = lane & 15;
+    const int kpart = lane >> 4;
+    const long M = E + (long)N;
+    const long tiles = (M + 16 * WAVES_PER_BLOCK - 1) / (16 * WAVES_PER_BLOCK);
+    for (long t = blockIdx.x; t < tiles; t += gridDim.x) {
+        const long mbase = (t * WAVES_PER_BLOCK + wave) * 16;
+        // gXln tile = gA_tile @ Wr on the matrix core
+        for (int jt = 0; jt < MSG; jt += 16) {
+            f32x4_t acc = {0.0f, 0.0f, 0.0f, 0.0f};
+            for (int kk = 0; kk < OUT; kk += 4) {
+                const long m = mbase + mrow;
+                const float a = (m < M)
+                    ? gA[m * OUT + kk + kpart] : 0.0f;
+                const float bb = ws[kk + kpart][jt + mrow];
+                acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bb, acc, 0, 0, 0);
+            }
+            for (int reg = 0; reg < 4; ++reg)
+                gx[wave][kpart * 4 + reg][jt + mrow] = acc[reg];
+        }
+        __builtin_amdgcn_wave_barrier();
+
+        // per-message LayerNorm backward + scatter (fwd lane layout)
+        const long m = mbase + mrow;
+        float vals[Q];
+        float s = 0.0f;
+        const bool live = (m < M);
+        const bool is_edge = live && (m < E);
+        const long node = !live ? 0 : (is_edge ? src[m] : (m - E));
+        for (int i = 0; i < Q; ++i) {
+            const int k = kpart * Q + i;
+            float v = 0.0f;
+            if (live) {
+                if (k < 16) v = hn[node * 16 + k];
+                else if (is_edge) v = he[m * 16 + (k - 16)];
+            }
+            vals[i] = v;
+            s += v;
+        }
+        s += __shfl_xor(s, 16, WAVE);
+        s += __shfl_xor(s, 32, WAVE);
+        const float mean = s / MSG;
+        float d2 = 0.0f;
+        for (int i = 0; i < Q; ++i) {
+            const float d = live ? vals[i] - mean : 0.0f;
+            vals[i] = d;
+            d2 += d * d;
+        }
+        d2 += __shfl_xor(d2, 16, WAVE);
+        d2 += __shfl_xor(d2, 32, WAVE);
+        const float inv_sigma = rsqrtf(d2 / MSG + LN_EPS);
+        float gyg[Q], mh[Q];
+        float s1 = 0.0f, s2 = 0.0f;
+        for (int i = 0; i < Q; ++i) {
+            const int k = kpart * Q + i;
+            mh[i] = vals[i] * inv_sigma;
+            const float gy = gx[wave][mrow][k];
+            gyg[i] = gy * ln_g[k];
+            s1 += gyg[i];
+            s2 += gyg[i] * mh[i];
+        }
+        s1 += __shfl_xor(s1, 16, WAVE);
+        s1 += __shfl_xor(s1, 32, WAVE);
+        s2 += __shfl_xor(s2, 16, WAVE);
+        s2 += __shfl_xor(s2, 32, WAVE);
+        const float m1 = s1 / MSG, m2 = s2 / MSG;
+        if (live) {
+            for (int i = 0; i < Q; ++i) {
+                const int k = kpart * Q + i;
+                mhat_out[m * MSG + k] = mh[i];
+                gy_out[m * MSG + k] = gx[wave][mrow][k];
+                const float gmsg = inv_sigma * (gyg[i] - m1 - mh[i] * m2);
+                if (k < 16)
+                    atomicAdd(&ghn[node * 16 + k], gmsg);
+                else if (is_edge)
+                    ghe[m * 16 + (k - 16)] = gmsg;
+            }
+        }
+        __builtin_amdgcn_wave_barrier();
+    }
+}
+
+#define K3_CHUNK 1024
+
+__global__ void __launch_bounds__(BLOCK)
+msg_bwd_wr_kernel(const float* __restrict__ gA,
+                  const float* __restrict__ mhat,
+                  const float* __restrict__ ln_g,
+                  const float* __restrict__ ln_b,
+                  float* __restrict__ gWr,   // [OUT,32] pre-zeroed
+                  long M, int OUT) {
+    const int MSG = 32;
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    const int mrow = lane & 15;
+    const int kpart = lane >> 4;
+    const int T = (OUT / 16) * 2;          // (jt, kt) tiles
+    const long chunks = (M + K3_CHUNK - 1) / K3_CHUNK;
+    for (long c = blockIdx.x; c < chunks; c += gridDim.x) {
+        const long c0 = c * K3_CHUNK;
+        for (int t = wave; t < T; t += WAVES_PER_BLOCK) {
+            const int jt = (t >> 1) * 16;   // OUT tile base
+            const int kt = (t & 1) * 16;    // msg-dim tile base
+            f32x4_t acc = {0.0f, 0.0f, 0.0f, 0.0f};
+            for (int kk = 0; kk < K3_CHUNK; kk += 4) {
+                const long m = c0 + kk + kpart;
+                float a = 0.0f, bb = 0.0f;
+                if (m < M) {
+                    a = gA[m * OUT + jt + mrow];
+                    const int k = kt + mrow;
+                    bb = mhat[m * MSG + k] * ln_g[k] + ln_b[k];
+                }
+                acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bb, acc, 0, 0, 0);
+            }
+            for (int reg = 0; reg < 4; ++reg)
+                atomicAdd(&gWr[(long)(jt + kpart * 4 + reg) * MSG + kt + mrow],
+                          acc[reg]);
+        }
+    }
+}
+
+std::vector<torch::Tensor> message_reduce_bwd_mfma(
+    torch::Tensor hn, torch::Tensor he, torch::Tensor src, torch::Tensor dst,
+    torch::Tensor indptr, torch::Tensor ln_g, torch::Tensor ln_b,
+    torch::Tensor Wr, torch::Tensor r_edge, torch::Tensor r_self,
+    torch::Tensor gout) {
+    const long E = src.size(0);
+    const int N = (int)hn.size(0), OUT = (int)Wr.size(0);
+    TORCH_CHECK((int)hn.size(1) == 16 && (int)Wr.size(1) == 32);
+    TORCH_CHECK(OUT % 16 == 0 && OUT <= 64);
+    const long M = E + N;
+    auto opt = hn.options();
+    auto gA = torch::empty({M, (long)OUT}, opt);
+    auto mhat = torch::empty({M, 32L}, opt);
+    auto gy = torch::empty({M, 32L}, opt);
+    auto ghn = torch::zeros({(long)N, 16L}, opt);
+    auto ghe = torch::zeros({E, 16L}, opt);
+    auto gWr = torch::zeros({(long)OUT, 32L}, opt);
+    hipStream_t stream = at::cuda::getCurrentCUDAStream();
+    {
+        long blocks = (M + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
+        if (blocks > 4096) blocks = 4096;
+        hipLaunchKernelGGL(msg_bwd_ga_kernel, dim3((int)blocks), dim3(BLOCK),
+                           0, stream, r_edge.data_ptr<float>(),
+                           r_self.data_ptr<float>(), dst.data_ptr<long>(),
+                           indptr.data_ptr<long>(), gout.data_ptr<float>(),
+                           gA.data_ptr<float>(), E, N, OUT);
+    }
+    hipLaunchKernelGGL(msg_bwd_data_kernel, dim3(grid_for_tiles(M)),
+                       dim3(BLOCK), 0, stream, hn.data_ptr<float>(),
+                       he.data_ptr<float>(), src.data_ptr<long>(),
+                       ln_g.data_ptr<float>(), Wr.data_ptr<float>(),
+                       gA.data_ptr<float>(), ghn.data_ptr<float>(),
+                       ghe.data_ptr<float>(), mhat.data_ptr<float>(),
+                       gy.data_ptr<float>(), E, N, OUT);
+    {
+        long blocks = (M + K3_CHUNK - 1) / K3_CHUNK;
+        if (blocks > 2048) blocks = 2048;
+        hipLaunchKernelGGL(msg_bwd_wr_kernel, dim3((int)blocks), dim3(BLOCK),
+                           0, stream, gA.data_ptr<float>(),
+                           mhat.data_ptr<float>(), ln_g.data_ptr<float>(),
+                           ln_b.data_ptr<float>(), gWr.data_ptr<float>(),
+                           M, OUT);
+    }
+    // parameter-gradient column sums on temporaries (3 reduce kernels)
+    auto gbr = gA.sum(0);
+    auto gln_b = gy.sum(0);
+    auto gln_g = (gy * mhat).sum(0);
+    return {ghn, ghe, gWr, gbr, gln_g, gln_b};
+}

@@ -343,3 +343,37 @@ def test_mfma_kernels_match_valu():
         (out_v - out_m).abs().max().item()
     # zero-fill semantics preserved
     assert torch.all(out_m[~mail] == 0)
+
+
+@pytest.mark.gpu
+def test_mfma_message_backward_matches_valu():
+    """3-stage MFMA message backward vs the fused VALU backward kernel."""
+    from ddls_amd import ops as hip_ops
+    ext = hip_ops.get_extension(required=True)
+    torch.manual_seed(6)
+    dev = "cuda:0"
+    N, E, half, OUT = 555, 1777, 16, 64
+    hn = torch.rand(N, half, device=dev)
+    he = torch.rand(E, half, device=dev)
+    src = torch.randint(0, N, (E,), device=dev)
+    dst = torch.randint(0, N, (E,), device=dev)
+    counts = torch.bincount(dst, minlength=N)
+    indptr = torch.zeros(N + 1, dtype=torch.int64, device=dev)
+    torch.cumsum(counts, 0, out=indptr[1:])
+    ln_g = torch.rand(2 * half, device=dev) + 0.5
+    ln_b = torch.rand(2 * half, device=dev) - 0.5
+    Wr = torch.randn(OUT, 2 * half, device=dev) / 4
+    br = torch.randn(OUT, device=dev)
+    order = torch.argsort(dst, stable=True)
+    _out, r_edge, r_self = ext.message_reduce_train(
+        hn, he, src, order, indptr, ln_g, ln_b, Wr, br)
+    gout = torch.randn(N, OUT, device=dev)
+
+    ref = ext.message_reduce_bwd(hn, he, src, dst, indptr, ln_g, ln_b, Wr,
+                                 r_edge, r_self, gout)
+    got = ext.message_reduce_bwd_mfma(hn, he, src, dst, indptr, ln_g, ln_b,
+                                      Wr, r_edge, r_self, gout)
+    names = ("ghn", "ghe", "gWr", "gbr", "gln_g", "gln_b")
+    for n_, a, b in zip(names, ref, got):
+        assert torch.allclose(a, b, rtol=1e-3, atol=1e-4), \
+            (n_, (a - b).abs().max().item())
