@@ -165,7 +165,8 @@ class _FusedMeanPoolFn(torch.autograd.Function):
         (z, e, src, dst, indptr, hn, he, r_edge, r_self,
          ln_n_w, ln_n_b, W_n, ln_e_w, ln_e_b, W_e,
          ln_r_w, ln_r_b, W_r) = ctx.saved_tensors
-        if ctx.use_mfma and hasattr(ext, "message_reduce_bwd_mfma"):
+        if (ctx.use_mfma and hasattr(ext, "message_reduce_bwd_mfma")
+                and os.environ.get("DDLS_AMD_DISABLE_MFMA_BWD", "0") != "1"):
             ghn, ghe, gWr, gbr, glnr_g, glnr_b = ext.message_reduce_bwd_mfma(
                 hn, he, src, dst, indptr, ln_r_w, ln_r_b, W_r, r_edge, r_self,
                 gout.contiguous())

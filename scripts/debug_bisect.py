@@ -1,0 +1,49 @@
+"""Bisect GPU-only training-path features via their disable flags."""
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+import numpy as np, torch
+from bench import build_env_fn
+from ddls_amd.cluster.batched_lookahead import precompute_lookahead_memos
+from ddls_amd.models.gnn import GNNPolicy
+from ddls_amd.rl.ppo import PPOConfig, PPOTrainer
+from ddls_amd.rl.subproc_env import SubprocVectorEnv
+
+env_fn = build_env_fn()
+venv = SubprocVectorEnv(env_fn, num_envs=64, num_workers=64, base_seed=1)
+scratch = env_fn(); scratch.reset(seed=0)
+ml, mi = precompute_lookahead_memos(scratch, device="cuda:0")
+venv.preload_memos(ml, mi)
+
+FLAGS = ("DDLS_AMD_DISABLE_FLAT_ADAM", "DDLS_AMD_DISABLE_FUSED_LOSS",
+         "DDLS_AMD_DISABLE_MFMA", "DDLS_AMD_DISABLE_MFMA_BWD")
+
+def run(tag, on, use_graphs=True, iters=20):
+    for k in FLAGS:
+        os.environ[k] = "1" if k in on else "0"
+    torch.manual_seed(0)
+    policy = GNNPolicy(num_actions=17)
+    tr = PPOTrainer(venv, policy,
+                    PPOConfig(train_batch_size=1024, sgd_minibatch_size=128,
+                              num_sgd_iter=8, use_hip_graphs=use_graphs),
+                    device=torch.device("cuda:0"))
+    ents, dps, kls = [], [], []
+    st = {}
+    prev = torch.cat([p.detach().reshape(-1).clone()
+                      for p in policy.parameters()])
+    for i in range(iters):
+        st = tr.train(num_steps=16)
+        cur = torch.cat([p.detach().reshape(-1).clone()
+                         for p in policy.parameters()])
+        dps.append(round(float((cur - prev).norm()), 4))
+        prev = cur
+        ents.append(round(st["entropy"], 3))
+        kls.append(round(st["kl"], 5))
+    print(f"{tag:18s} ent={ents[::4]} dP={dps[::4]} kl={kls[::4]} "
+          f"reward={st['mean_reward']:.1f} "
+          f"caps={getattr(tr._stepper, 'capture_count', None)}", flush=True)
+
+run("all-on", ())
+run("no-mfma-bwd", ("DDLS_AMD_DISABLE_MFMA_BWD",))
+run("no-mfma", ("DDLS_AMD_DISABLE_MFMA",))
+run("no-flat-adam", ("DDLS_AMD_DISABLE_FLAT_ADAM",))
+venv.close()
