@@ -38,12 +38,19 @@ def run(tag, on, use_graphs=True, iters=20):
         prev = cur
         ents.append(round(st["entropy"], 3))
         kls.append(round(st["kl"], 5))
+        sp = tr._stepper
+        grad_linked = None
+        if sp is not None and sp.flat_p is not None:
+            base = sp.flat_g.data_ptr()
+            end = base + sp.flat_g.numel() * 4
+            grad_linked = all(
+                p.grad is not None and base <= p.grad.data_ptr() < end
+                for p in policy.parameters())
+        print(f"  iter {i+1}: dP={dps[-1]} caps="
+              f"{getattr(sp, 'capture_count', None)} "
+              f"grad_linked={grad_linked}", flush=True)
     print(f"{tag:18s} ent={ents[::4]} dP={dps[::4]} kl={kls[::4]} "
-          f"reward={st['mean_reward']:.1f} "
-          f"caps={getattr(tr._stepper, 'capture_count', None)}", flush=True)
+          f"reward={st['mean_reward']:.1f}", flush=True)
 
-run("all-on", ())
-run("no-mfma-bwd", ("DDLS_AMD_DISABLE_MFMA_BWD",))
-run("no-mfma", ("DDLS_AMD_DISABLE_MFMA",))
-run("no-flat-adam", ("DDLS_AMD_DISABLE_FLAT_ADAM",))
+run("all-on", (), iters=8)
 venv.close()
