@@ -10,6 +10,7 @@ one process per GPU, fused gradient all-reduce with RCCL over xGMI
 from __future__ import annotations
 
 import math
+import os
 import time
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional
@@ -229,6 +230,17 @@ class PPOTrainer:
                                                int(need_e * 1.05) + 8)
             if not captured:  # replay the same permutation stream eagerly
                 rng = np.random.RandomState(self.iteration + 1234 * get_rank())
+
+        if captured and os.environ.get("DDLS_AMD_DISABLE_PREFETCH", "0") == "1":
+            for idx in idx_lists:
+                ok = self._stepper.step([batch["obs"][i] for i in idx],
+                                        actions_np[idx], logp_np[idx],
+                                        adv_np[idx], vtarg_np[idx])
+                assert ok
+                num_updates += 1
+                num_captured += 1
+            rng = None
+            captured = False
 
         if captured:
             # captured fast loop with CPU prefetch: a worker thread stages

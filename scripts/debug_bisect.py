@@ -17,14 +17,15 @@ venv.preload_memos(ml, mi)
 FLAGS = ("DDLS_AMD_DISABLE_FLAT_ADAM", "DDLS_AMD_DISABLE_FUSED_LOSS",
          "DDLS_AMD_DISABLE_MFMA", "DDLS_AMD_DISABLE_MFMA_BWD")
 
-def run(tag, on, use_graphs=True, iters=20):
+def run(tag, on, use_graphs=True, iters=20, sgd_iters=8, noprefetch=False):
     for k in FLAGS:
         os.environ[k] = "1" if k in on else "0"
+    os.environ["DDLS_AMD_DISABLE_PREFETCH"] = "1" if noprefetch else "0"
     torch.manual_seed(0)
     policy = GNNPolicy(num_actions=17)
     tr = PPOTrainer(venv, policy,
                     PPOConfig(train_batch_size=1024, sgd_minibatch_size=128,
-                              num_sgd_iter=8, use_hip_graphs=use_graphs),
+                              num_sgd_iter=sgd_iters, use_hip_graphs=use_graphs),
                     device=torch.device("cuda:0"))
     ents, dps, kls = [], [], []
     st = {}
@@ -52,5 +53,7 @@ def run(tag, on, use_graphs=True, iters=20):
     print(f"{tag:18s} ent={ents[::4]} dP={dps[::4]} kl={kls[::4]} "
           f"reward={st['mean_reward']:.1f}", flush=True)
 
-run("all-on", (), iters=8)
+run("all-on", (), iters=6)
+run("all-on-noprefetch", (), iters=6, noprefetch=True)
+run("all-on-sgd1", (), iters=10, sgd_iters=1)
 venv.close()
