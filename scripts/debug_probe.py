@@ -56,7 +56,8 @@ for i in range(5):
     obs = last_batch["obs"][:128]
     inputs = collate(obs, dev)
     for p in policy.parameters():
-        if p.grad is not None: p.grad.detach_(); p.grad.zero_()
+        if p.grad is not None:
+            with torch.no_grad(): p.grad.zero_()
     logits, values = policy.forward_flat(inputs["batch"], inputs["graph_features"], inputs["action_mask"])
     loss = logits.sum() * 1e-3 + values.sum() * 1e-3
     loss.backward()
