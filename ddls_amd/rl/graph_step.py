@@ -307,6 +307,16 @@ class CapturedSGDStep:
                      if pa.grad is not None]
             if grads:
                 torch._foreach_zero_(grads)
+        else:
+            # pin every grad to its flat_g view at RECORD time: autograd's
+            # AccumulateGrad may otherwise accumulate OUT-OF-PLACE (rebinding
+            # p.grad to a capture-pool tensor), in which case replays never
+            # write flat_g and training silently freezes
+            base = self.flat_g.data_ptr()
+            for p_, (o, np_) in zip(self._params, self._offs):
+                if (p_.grad is None
+                        or p_.grad.data_ptr() != base + 4 * o):
+                    p_.grad = self.flat_g[o:o + np_].view(p_.shape)
         logits, values = self.policy.forward_flat(
             self.batch, self.d["gf"], self.d["mask"])
         logits, values = logits[:B], values[:B]
@@ -317,6 +327,7 @@ class CapturedSGDStep:
                 cfg.clip_param, cfg.vf_clip_param, cfg.vf_loss_coeff,
                 cfg.entropy_coeff)
             loss.backward()
+            self._repair_grad_views()
             self.stats_acc += stats
             return
         # Categorical re-implemented with log_softmax: torch.distributions
@@ -336,9 +347,23 @@ class CapturedSGDStep:
         loss = (policy_loss + self.kl_coeff_t * kl
                 + cfg.vf_loss_coeff * vf_loss - cfg.entropy_coeff * entropy)
         loss.backward()
+        self._repair_grad_views()
         self.stats_acc += torch.stack([policy_loss.detach(), vf_loss.detach(),
                                        kl.detach(), entropy.detach(),
                                        loss.detach()])
+
+    def _repair_grad_views(self):
+        """If AccumulateGrad rebound any p.grad off its flat_g view during
+        this (possibly capture-recorded) backward, record a copy from the
+        rebound tensor into flat_g — pool tensor addresses are stable across
+        replays, so the recorded copies keep flat_adam fed."""
+        if self.flat_p is None:
+            return
+        base = self.flat_g.data_ptr()
+        for p_, (o, np_) in zip(self._params, self._offs):
+            g = p_.grad
+            if g is not None and g.data_ptr() != base + 4 * o:
+                self.flat_g[o:o + np_].copy_(g.detach().reshape(-1))
 
     def _body_opt(self):
         if self.flat_p is not None:

@@ -53,7 +53,6 @@ def run(tag, on, use_graphs=True, iters=20, sgd_iters=8, noprefetch=False):
     print(f"{tag:18s} ent={ents[::4]} dP={dps[::4]} kl={kls[::4]} "
           f"reward={st['mean_reward']:.1f}", flush=True)
 
-run("all-on", (), iters=6)
-run("all-on-noprefetch", (), iters=6, noprefetch=True)
-run("all-on-sgd1", (), iters=10, sgd_iters=1)
+run("all-on", (), iters=12)
+run("no-mfma-bwd", ("DDLS_AMD_DISABLE_MFMA_BWD",), iters=12)
 venv.close()
