@@ -44,27 +44,33 @@ def collect(*a, **k):
 tr.collect_rollout = collect
 
 prev = flat()
-for i in range(5):
+for i in range(4):
     st = tr.train(num_steps=16)
     cur = flat(); dp = float((cur-prev).norm()); prev = cur
-    sp = tr._stepper
-    # probe A: captured step with synthetic data
-    p0 = flat()
-    ok = sp.step(smb, sacts, solp, sadv, svt); torch.cuda.synchronize()
-    dpa = float((flat()-p0).norm())
-    # probe B: EAGER fwd/bwd on a real minibatch from the last rollout
-    obs = last_batch["obs"][:128]
-    inputs = collate(obs, dev)
-    for p in policy.parameters():
-        if p.grad is not None:
-            with torch.no_grad(): p.grad.zero_()
-    logits, values = policy.forward_flat(inputs["batch"], inputs["graph_features"], inputs["action_mask"])
-    loss = logits.sum() * 1e-3 + values.sum() * 1e-3
-    loss.backward()
-    gnorm = float(torch.cat([p.grad.reshape(-1) for p in policy.parameters()]).norm())
-    # restore params (probe A moved them)
-    prev = flat()
-    print(f"iter {i+1}: dP={dp:.4f} caps={sp.capture_count} "
-          f"probeA_captured_synth_dP={dpa:.5f} probeB_eager_real_gnorm={gnorm:.5f} "
-          f"kl={st['kl']:.5f} ent={st['entropy']:.3f}", flush=True)
+    print(f"iter {i+1}: dP={dp:.4f} kl={st['kl']:.5f}", flush=True)
+
+sp = tr._stepper
+print("state: |m|", float(sp.flat_m.norm()), "|v|", float(sp.flat_v.norm()),
+      "step", float(sp.step_t.item()), "|g|", float(sp.flat_g.norm()), flush=True)
+
+# one captured replay with real data, stats inspected
+obs = last_batch["obs"][:128]
+acts = np.asarray(last_batch["actions"][:128], dtype=np.int64)
+olp = np.asarray(last_batch["logp"][:128], dtype=np.float32)
+a = np.asarray(last_batch["advantages"][:128], dtype=np.float32)
+a = (a - a.mean()) / max(a.std(), 1e-4)
+vt = np.asarray(last_batch["value_targets"][:128], dtype=np.float32)
+sp.reset_stats()
+p0 = flat()
+assert sp.step(obs, acts, olp, a, vt); torch.cuda.synchronize()
+print("captured replay: dP", float((flat()-p0).norm()),
+      "stats", sp.stats_acc.cpu().numpy().round(4),
+      "|g| after", float(sp.flat_g.norm()), flush=True)
+
+# eager body on the same (already-filled) buffers
+p0 = flat()
+sp._body_fwd_bwd(); torch.cuda.synchronize()
+print("eager body: |g|", float(sp.flat_g.norm()), flush=True)
+sp._body_opt(); torch.cuda.synchronize()
+print("eager body dP:", float((flat()-p0).norm()), flush=True)
 venv.close()
