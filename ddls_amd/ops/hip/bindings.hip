@@ -40,6 +40,7 @@ std::vector<torch::Tensor> message_reduce_bwd_mfma(
 void flat_adam(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, torch::Tensor step_t, torch::Tensor normsq,
                double clip, double lr, double b1, double b2, double eps);
+void flat_sumsq(torch::Tensor x, torch::Tensor out);
 std::vector<torch::Tensor> ppo_loss_fwd(
     torch::Tensor logits, torch::Tensor values, torch::Tensor actions,
     torch::Tensor old_logp, torch::Tensor adv, torch::Tensor vtarg,
@@ -85,4 +86,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
           "fused grad-clip + Adam over flat param/grad/m/v buffers");
     m.def("message_reduce_bwd_mfma", &message_reduce_bwd_mfma,
           "matrix-core message-passing backward (3-stage MFMA)");
+    m.def("flat_sumsq", &flat_sumsq, "capture-safe sum of squares");
 }

@@ -396,11 +396,14 @@ void flat_adam(torch::Tensor p, torch::Tensor g, torch::Tensor m,
 // meanpool_bwd kernel for the tuned shape half=16/MSG=32).  Three stages:
 //   K1 msg_bwd_ga:   gA[m,j] = relu'(r[m,j]) * gout[node(m),j] / (deg+1)
 //   K2 msg_bwd_data: gXln = gA @ Wr (MFMA) -> LayerNorm backward ->
-//                    scatter ghn (atomic) / ghe; stores mhat & gy for the
-//                    parameter-gradient column sums (done with torch sums)
+//                    scatter ghn (atomic) / ghe; accumulates gln_g/gln_b via
+//                    LDS block partials + atomics
 //   K3 msg_bwd_wr:   gWr = gA^T @ xln (MFMA over message chunks, xln
-//                    recomputed as mhat*g+b), atomically accumulated
+//                    recomputed as mhat*g+b) + gbr column sums, atomics
 // Message order: 0..E-1 edge messages, E..E+N-1 self messages (as forward).
+// ALL reductions live in these kernels: torch's multi-pass .sum() proved
+// unstable under hipGraph replay (outputs drift after ~100 replays), so no
+// torch reduction may run inside the captured region.
 // ===========================================================================
 
 __global__ void __launch_bounds__(BLOCK)
@@ -442,13 +445,19 @@ msg_bwd_data_kernel(const float* __restrict__ hn,
                     float* __restrict__ ghn,     // [N,16] pre-zeroed
                     float* __restrict__ ghe,     // [E,16]
                     float* __restrict__ mhat_out,  // [M,32]
-                    float* __restrict__ gy_out,    // [M,32]
+                    float* __restrict__ gln_g_out,  // [32] pre-zeroed
+                    float* __restrict__ gln_b_out,  // [32] pre-zeroed
                     long E, int N, int OUT) {
     const int MSG = 32, Q = 8;
     __shared__ float ws[64][MSG + 4];          // Wr [OUT][MSG]
     __shared__ float gx[WAVES_PER_BLOCK][16][MSG + 4];
+    __shared__ float lng_acc[MSG], lnb_acc[MSG];
     for (int i = threadIdx.x; i < OUT * MSG; i += BLOCK)
         ws[i / MSG][i % MSG] = Wr[i];
+    if (threadIdx.x < MSG) {
+        lng_acc[threadIdx.x] = 0.0f;
+        lnb_acc[threadIdx.x] = 0.0f;
+    }
     __syncthreads();
 
     const int wave = threadIdx.x / WAVE;
@@ -522,7 +531,9 @@ msg_bwd_data_kernel(const float* __restrict__ hn,
             for (int i = 0; i < Q; ++i) {
                 const int k = kpart * Q + i;
                 mhat_out[m * MSG + k] = mh[i];
-                gy_out[m * MSG + k] = gx[wave][mrow][k];
+                const float gy = gx[wave][mrow][k];
+                atomicAdd(&lnb_acc[k], gy);
+                atomicAdd(&lng_acc[k], gy * mh[i]);
                 const float gmsg = inv_sigma * (gyg[i] - m1 - mh[i] * m2);
                 if (k < 16)
                     atomicAdd(&ghn[node * 16 + k], gmsg);
@@ -531,6 +542,11 @@ msg_bwd_data_kernel(const float* __restrict__ hn,
             }
         }
         __builtin_amdgcn_wave_barrier();
+    }
+    __syncthreads();
+    if (threadIdx.x < MSG) {
+        atomicAdd(&gln_g_out[threadIdx.x], lng_acc[threadIdx.x]);
+        atomicAdd(&gln_b_out[threadIdx.x], lnb_acc[threadIdx.x]);
     }
 }
 
@@ -542,6 +558,7 @@ msg_bwd_wr_kernel(const float* __restrict__ gA,
                   const float* __restrict__ ln_g,
                   const float* __restrict__ ln_b,
                   float* __restrict__ gWr,   // [OUT,32] pre-zeroed
+                  float* __restrict__ gbr,   // [OUT] pre-zeroed
                   long M, int OUT) {
     const int MSG = 32;
     const int wave = threadIdx.x / WAVE;
@@ -570,6 +587,19 @@ msg_bwd_wr_kernel(const float* __restrict__ gA,
                 atomicAdd(&gWr[(long)(jt + kpart * 4 + reg) * MSG + kt + mrow],
                           acc[reg]);
         }
+        // gbr = column sums of gA (each j column read once: kt==0 tiles)
+        for (int t = wave; t < T; t += WAVES_PER_BLOCK) {
+            if (t & 1) continue;
+            const int jt = (t >> 1) * 16;
+            float s = 0.0f;
+            for (int kk = 0; kk < K3_CHUNK; kk += 4) {
+                const long m = c0 + kk + kpart;
+                if (m < M) s += gA[m * OUT + jt + mrow];
+            }
+            s += __shfl_xor(s, 16, WAVE);
+            s += __shfl_xor(s, 32, WAVE);
+            if (kpart == 0) atomicAdd(&gbr[jt + mrow], s);
+        }
     }
 }
 
@@ -586,10 +616,12 @@ std::vector<torch::Tensor> message_reduce_bwd_mfma(
     auto opt = hn.options();
     auto gA = torch::empty({M, (long)OUT}, opt);
     auto mhat = torch::empty({M, 32L}, opt);
-    auto gy = torch::empty({M, 32L}, opt);
     auto ghn = torch::zeros({(long)N, 16L}, opt);
     auto ghe = torch::zeros({E, 16L}, opt);
     auto gWr = torch::zeros({(long)OUT, 32L}, opt);
+    auto gbr = torch::zeros({(long)OUT}, opt);
+    auto gln_g = torch::zeros({32L}, opt);
+    auto gln_b = torch::zeros({32L}, opt);
     hipStream_t stream = at::cuda::getCurrentCUDAStream();
     {
         long blocks = (M + WAVES_PER_BLOCK - 1) / WAVES_PER_BLOCK;
@@ -600,13 +632,18 @@ std::vector<torch::Tensor> message_reduce_bwd_mfma(
                            indptr.data_ptr<long>(), gout.data_ptr<float>(),
                            gA.data_ptr<float>(), E, N, OUT);
     }
-    hipLaunchKernelGGL(msg_bwd_data_kernel, dim3(grid_for_tiles(M)),
-                       dim3(BLOCK), 0, stream, hn.data_ptr<float>(),
-                       he.data_ptr<float>(), src.data_ptr<long>(),
-                       ln_g.data_ptr<float>(), Wr.data_ptr<float>(),
-                       gA.data_ptr<float>(), ghn.data_ptr<float>(),
-                       ghe.data_ptr<float>(), mhat.data_ptr<float>(),
-                       gy.data_ptr<float>(), E, N, OUT);
+    {
+        long blocks = grid_for_tiles(M);
+        if (blocks > 1024) blocks = 1024;
+        hipLaunchKernelGGL(msg_bwd_data_kernel, dim3((int)blocks),
+                           dim3(BLOCK), 0, stream, hn.data_ptr<float>(),
+                           he.data_ptr<float>(), src.data_ptr<long>(),
+                           ln_g.data_ptr<float>(), Wr.data_ptr<float>(),
+                           gA.data_ptr<float>(), ghn.data_ptr<float>(),
+                           ghe.data_ptr<float>(), mhat.data_ptr<float>(),
+                           gln_g.data_ptr<float>(), gln_b.data_ptr<float>(),
+                           E, N, OUT);
+    }
     {
         long blocks = (M + K3_CHUNK - 1) / K3_CHUNK;
         if (blocks > 2048) blocks = 2048;
@@ -614,11 +651,44 @@ std::vector<torch::Tensor> message_reduce_bwd_mfma(
                            0, stream, gA.data_ptr<float>(),
                            mhat.data_ptr<float>(), ln_g.data_ptr<float>(),
                            ln_b.data_ptr<float>(), gWr.data_ptr<float>(),
-                           M, OUT);
+                           gbr.data_ptr<float>(), M, OUT);
     }
-    // parameter-gradient column sums on temporaries (3 reduce kernels)
-    auto gbr = gA.sum(0);
-    auto gln_b = gy.sum(0);
-    auto gln_g = (gy * mhat).sum(0);
     return {ghn, ghe, gWr, gbr, gln_g, gln_b};
+}
+
+// ---------------------------------------------------------------------------
+// capture-safe sum of squares (torch multi-pass reductions proved unstable
+// under hipGraph replay); out is zeroed by the leading 1-thread kernel.
+__global__ void zero1_kernel(float* out) { out[0] = 0.0f; }
+
+__global__ void __launch_bounds__(BLOCK)
+sumsq_kernel(const float* __restrict__ x, float* __restrict__ out, long n) {
+    __shared__ float part[WAVES_PER_BLOCK];
+    const int wave = threadIdx.x / WAVE;
+    const int lane = threadIdx.x % WAVE;
+    float s = 0.0f;
+    for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < n;
+         i += (long)gridDim.x * BLOCK)
+        s += x[i] * x[i];
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        s += __shfl_down(s, off, WAVE);
+    if (lane == 0) part[wave] = s;
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        float t = 0.0f;
+        for (int w = 0; w < WAVES_PER_BLOCK; ++w) t += part[w];
+        atomicAdd(out, t);
+    }
+}
+
+void flat_sumsq(torch::Tensor x, torch::Tensor out) {
+    const long n = x.numel();
+    hipStream_t stream = at::cuda::getCurrentCUDAStream();
+    hipLaunchKernelGGL(zero1_kernel, dim3(1), dim3(1), 0, stream,
+                       out.data_ptr<float>());
+    long blocks = (n + BLOCK - 1) / BLOCK;
+    if (blocks > 512) blocks = 512;
+    hipLaunchKernelGGL(sumsq_kernel, dim3((int)blocks), dim3(BLOCK), 0,
+                       stream, x.data_ptr<float>(), out.data_ptr<float>(), n);
 }

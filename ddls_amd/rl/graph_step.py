@@ -367,11 +367,11 @@ class CapturedSGDStep:
 
     def _body_opt(self):
         if self.flat_p is not None:
-            # fused grad-clip + Adam over the flat buffers: 3 kernels total,
-            # and flat_adam zeroes the grads for the next backward
-            normsq = torch.dot(self.flat_g, self.flat_g)
+            # fused grad-clip + Adam over the flat buffers (the norm uses the
+            # in-house sumsq kernel: torch reductions drift under replay)
+            self._ext.flat_sumsq(self.flat_g, self.normsq)
             self._ext.flat_adam(self.flat_p, self.flat_g, self.flat_m,
-                                self.flat_v, self.step_t, normsq.reshape(1),
+                                self.flat_v, self.step_t, self.normsq,
                                 float(self.cfg.grad_clip or 0.0),
                                 float(self.cfg.lr), 0.9, 0.999, 1e-8)
             return
@@ -395,6 +395,7 @@ class CapturedSGDStep:
         self.flat_m = torch.zeros(numel, device=dev)
         self.flat_v = torch.zeros(numel, device=dev)
         self.step_t = torch.zeros(1, device=dev)
+        self.normsq = torch.zeros(1, device=dev)
         self._params = params
         self._offs = []
         off = 0
