@@ -418,3 +418,46 @@ def test_fused_ppo_loss_matches_torch():
     assert torch.allclose(lf.grad, lt.grad, atol=1e-5), \
         (lf.grad - lt.grad).abs().max().item()
     assert torch.allclose(vf_.grad, vt.grad, atol=1e-6)
+
+
+@pytest.mark.gpu
+def test_flat_adam_state_checkpoint_roundtrip():
+    """Flat Adam state syncs back into the torch optimizer for checkpoints
+    and reloads into a fresh stepper."""
+    from ddls_amd.models.gnn import GNNPolicy
+    from ddls_amd.rl.graph_step import CapturedSGDStep
+    from ddls_amd.rl.ppo import PPOConfig
+
+    device = torch.device("cuda:0")
+    B = 8
+    cfg = PPOConfig(sgd_minibatch_size=B)
+    rng = np.random.RandomState(1)
+    mb = [_random_compact_obs(rng) for _ in range(B)]
+    acts = np.zeros(B, dtype=np.int64)
+    olp = (rng.randn(B) * 0.1 - 2).astype(np.float32)
+    adv = rng.randn(B).astype(np.float32)
+    vt = rng.randn(B).astype(np.float32)
+
+    torch.manual_seed(3)
+    pol = GNNPolicy(num_actions=17).to(device)
+    opt = torch.optim.Adam(pol.parameters(), lr=cfg.lr, foreach=True)
+    st = CapturedSGDStep(pol, opt, cfg, device)
+    st.set_kl_coeff(cfg.kl_coeff)
+    for _ in range(3):
+        assert st.step(mb, acts, olp, adv, vt)
+    torch.cuda.synchronize()
+    st.sync_state_to_optimizer()
+    p0 = next(iter(pol.parameters()))
+    state = opt.state[p0]
+    assert float(state["step"]) == 3.0
+    assert torch.allclose(state["exp_avg"].reshape(-1),
+                          st.flat_m[:p0.numel()].cpu()
+                          if not state["exp_avg"].is_cuda
+                          else st.flat_m[:p0.numel()].view(p0.shape))
+
+    # reload path: a fresh stepper adopts the synced torch state
+    st2 = CapturedSGDStep(pol, opt, cfg, device)
+    st2.set_kl_coeff(cfg.kl_coeff)
+    assert st2.step(mb, acts, olp, adv, vt)
+    torch.cuda.synchronize()
+    assert float(st2.step_t.item()) == 4.0  # continued from step 3
