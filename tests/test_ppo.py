@@ -450,10 +450,8 @@ def test_flat_adam_state_checkpoint_roundtrip():
     p0 = next(iter(pol.parameters()))
     state = opt.state[p0]
     assert float(state["step"]) == 3.0
-    assert torch.allclose(state["exp_avg"].reshape(-1),
-                          st.flat_m[:p0.numel()].cpu()
-                          if not state["exp_avg"].is_cuda
-                          else st.flat_m[:p0.numel()].view(p0.shape))
+    assert torch.allclose(state["exp_avg"].reshape(-1).to(st.flat_m.device),
+                          st.flat_m[:p0.numel()])
 
     # reload path: a fresh stepper adopts the synced torch state
     st2 = CapturedSGDStep(pol, opt, cfg, device)
