@@ -550,7 +550,9 @@ msg_bwd_data_kernel(const float* __restrict__ hn,
     }
 }
 
-#define K3_CHUNK 1024
+// small chunks: the chip needs >>256 workgroups to fill (8 XCDs x 32 CU);
+// 1024-message chunks left only ~21 blocks in flight
+#define K3_CHUNK 128
 
 __global__ void __launch_bounds__(BLOCK)
 msg_bwd_wr_kernel(const float* __restrict__ gA,
@@ -573,6 +575,7 @@ msg_bwd_wr_kernel(const float* __restrict__ gA,
             const int jt = (t >> 1) * 16;   // OUT tile base
             const int kt = (t & 1) * 16;    // msg-dim tile base
             f32x4_t acc = {0.0f, 0.0f, 0.0f, 0.0f};
+            float sba = 0.0f;               // gbr partial (kt==0 tiles only)
             for (int kk = 0; kk < K3_CHUNK; kk += 4) {
                 const long m = c0 + kk + kpart;
                 float a = 0.0f, bb = 0.0f;
@@ -581,24 +584,17 @@ msg_bwd_wr_kernel(const float* __restrict__ gA,
                     const int k = kt + mrow;
                     bb = mhat[m * MSG + k] * ln_g[k] + ln_b[k];
                 }
+                if (kt == 0) sba += a;
                 acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, bb, acc, 0, 0, 0);
             }
             for (int reg = 0; reg < 4; ++reg)
                 atomicAdd(&gWr[(long)(jt + kpart * 4 + reg) * MSG + kt + mrow],
                           acc[reg]);
-        }
-        // gbr = column sums of gA (each j column read once: kt==0 tiles)
-        for (int t = wave; t < T; t += WAVES_PER_BLOCK) {
-            if (t & 1) continue;
-            const int jt = (t >> 1) * 16;
-            float s = 0.0f;
-            for (int kk = 0; kk < K3_CHUNK; kk += 4) {
-                const long m = c0 + kk + kpart;
-                if (m < M) s += gA[m * OUT + jt + mrow];
+            if (kt == 0) {
+                sba += __shfl_xor(sba, 16, WAVE);
+                sba += __shfl_xor(sba, 32, WAVE);
+                if (kpart == 0) atomicAdd(&gbr[jt + mrow], sba);
             }
-            s += __shfl_xor(s, 16, WAVE);
-            s += __shfl_xor(s, 32, WAVE);
-            if (kpart == 0) atomicAdd(&gbr[jt + mrow], s);
         }
     }
 }
