@@ -309,6 +309,41 @@ class GNN(nn.Module):
         return z
 
 
+class _FusedHeadFn(torch.autograd.Function):
+    """Fused policy/value head (ops/hip/policy_head.hip): LN(graph feats) ->
+    graph MLP -> concat(pooled node emb) -> both FC branches + action mask,
+    one kernel each way — replaces ~8 rocBLAS GEMMs and ~30 elementwise
+    kernels per captured replay.  graph_features and mask need no grad."""
+
+    @staticmethod
+    def forward(ctx, gm, gf, mask, ln_w, ln_b, Wg, bg,
+                W1p, b1p, W2p, b2p, W1v, b1v, W2v, b2v):
+        from .. import ops as hip_ops
+        ext = hip_ops.get_extension(required=True)
+        logits, value, emb, h1p, h1v = ext.head_fwd(
+            gm.contiguous(), gf.contiguous(), mask.contiguous(),
+            ln_w, ln_b, Wg.contiguous(), bg, W1p.contiguous(), b1p,
+            W2p.contiguous(), b2p, W1v.contiguous(), b1v,
+            W2v.contiguous(), b2v)
+        ctx.save_for_backward(gf, emb, h1p, h1v, ln_w, ln_b, Wg,
+                              W1p, W2p, W1v, W2v)
+        return logits, value
+
+    @staticmethod
+    def backward(ctx, glogits, gvalue):
+        from .. import ops as hip_ops
+        ext = hip_ops.get_extension(required=True)
+        (gf, emb, h1p, h1v, ln_w, ln_b, Wg, W1p, W2p, W1v, W2v) = \
+            ctx.saved_tensors
+        (ggm, gln_w, gln_b, gWg, gbg, gW1p, gb1p, gW2p, gb2p,
+         gW1v, gb1v, gW2v, gb2v) = ext.head_bwd(
+            gf, emb, h1p, h1v, glogits.contiguous(), gvalue.contiguous(),
+            ln_w, ln_b, Wg.contiguous(), W1p.contiguous(), W2p.contiguous(),
+            W1v.contiguous(), W2v.contiguous())
+        return (ggm, None, None, gln_w, gln_b, gWg, gbg,
+                gW1p, gb1p, gW2p, gb2p, gW1v, gb1v, gW2v, gb2v)
+
+
 class _SegmentMeanFn(torch.autograd.Function):
     """HIP per-graph mean with a broadcast/scale backward (replaces the
     torch index_add path, which costs ~100us per minibatch on MI355X)."""
@@ -393,12 +428,34 @@ class GNNPolicy(nn.Module):
 
         self.policy_branch = branch(num_actions)
         self.value_branch = branch(1)
+        # fused-head eligibility (policy_head.hip is built for the tuned
+        # PAC-ML head shape)
+        self._fused_head_ok = (
+            cfg["module_depth"] == 1 and hiddens == [256]
+            and cfg["fcnet_activation"] == "relu"
+            and cfg["apply_action_mask"]
+            and cfg["out_features_graph"] == 8
+            and cfg["out_features_node"] == 16
+            and num_actions <= 32 and in_graph <= 64)
 
     def forward_flat(self, batch: GraphBatch, graph_features: torch.Tensor,
                      action_mask: torch.Tensor):
         """Forward on a pre-collated flat batch (the rollout/SGD hot path)."""
         node_emb = self.gnn(batch)                      # [N_total, out_node]
         graph_node_emb = graph_mean(node_emb, batch)    # [B, out_node]
+        if (self._fused_head_ok and graph_node_emb.is_cuda
+                and torch.is_grad_enabled() and graph_node_emb.requires_grad
+                and os.environ.get("DDLS_AMD_DISABLE_FUSED_HEAD", "0") != "1"):
+            from .. import ops as hip_ops
+            ext = hip_ops.get_extension()
+            if ext is not None and hasattr(ext, "head_fwd"):
+                ln, lin_g = self.graph_module[0], self.graph_module[1]
+                pb, vb = self.policy_branch, self.value_branch
+                return _FusedHeadFn.apply(
+                    graph_node_emb, graph_features, action_mask,
+                    ln.weight, ln.bias, lin_g.weight, lin_g.bias,
+                    pb[0].weight, pb[0].bias, pb[2].weight, pb[2].bias,
+                    vb[0].weight, vb[0].bias, vb[2].weight, vb[2].bias)
         graph_emb = self.graph_module(graph_features)
         final_emb = torch.cat([graph_node_emb, graph_emb], dim=-1)
         logits = self.policy_branch(final_emb)

@@ -23,6 +23,7 @@ _SOURCES = [os.path.join(_HERE, "hip", "meanpool.hip"),
             os.path.join(_HERE, "hip", "meanpool_bwd.hip"),
             os.path.join(_HERE, "hip", "meanpool_mfma.hip"),
             os.path.join(_HERE, "hip", "ppo_loss.hip"),
+            os.path.join(_HERE, "hip", "policy_head.hip"),
             os.path.join(_HERE, "hip", "lookahead.hip"),
             os.path.join(_HERE, "hip", "bindings.hip")]
 

@@ -41,6 +41,17 @@ void flat_adam(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, torch::Tensor step_t, torch::Tensor normsq,
                double clip, double lr, double b1, double b2, double eps);
 void flat_sumsq(torch::Tensor x, torch::Tensor out);
+std::vector<torch::Tensor> head_fwd(
+    torch::Tensor gm, torch::Tensor gf, torch::Tensor mask,
+    torch::Tensor ln_w, torch::Tensor ln_b, torch::Tensor Wg,
+    torch::Tensor bg, torch::Tensor W1p, torch::Tensor b1p,
+    torch::Tensor W2p, torch::Tensor b2p, torch::Tensor W1v,
+    torch::Tensor b1v, torch::Tensor W2v, torch::Tensor b2v);
+std::vector<torch::Tensor> head_bwd(
+    torch::Tensor gf, torch::Tensor emb, torch::Tensor h1p, torch::Tensor h1v,
+    torch::Tensor glogits, torch::Tensor gvalue, torch::Tensor ln_w,
+    torch::Tensor ln_b, torch::Tensor Wg, torch::Tensor W1p,
+    torch::Tensor W2p, torch::Tensor W1v, torch::Tensor W2v);
 std::vector<torch::Tensor> ppo_loss_fwd(
     torch::Tensor logits, torch::Tensor values, torch::Tensor actions,
     torch::Tensor old_logp, torch::Tensor adv, torch::Tensor vtarg,
@@ -87,4 +98,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("message_reduce_bwd_mfma", &message_reduce_bwd_mfma,
           "matrix-core message-passing backward (3-stage MFMA)");
     m.def("flat_sumsq", &flat_sumsq, "capture-safe sum of squares");
+    m.def("head_fwd", &head_fwd,
+          "fused policy/value head forward (LN+graphMLP+concat+2 branches)");
+    m.def("head_bwd", &head_bwd, "fused policy/value head backward");
 }

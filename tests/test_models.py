@@ -377,3 +377,69 @@ def test_mfma_message_backward_matches_valu():
     for n_, a, b in zip(names, ref, got):
         assert torch.allclose(a, b, rtol=1e-3, atol=1e-4), \
             (n_, (a - b).abs().max().item())
+
+
+@pytest.mark.gpu
+def test_fused_policy_head_matches_eager():
+    """policy_head.hip fwd+bwd vs the eager torch head: logits, value, and
+    every parameter gradient (+ the pooled-embedding grad)."""
+    import os as _os
+    from ddls_amd.models.gnn import GNNPolicy, GraphBatch
+    dev = torch.device("cuda:0")
+    torch.manual_seed(7)
+    policy = GNNPolicy(num_actions=17).to(dev)
+    B = 33
+    gm = torch.randn(B, 16, device=dev)
+    gf = torch.rand(B, 34, device=dev)
+    mask = torch.ones(B, 17, device=dev)
+    mask[torch.rand(B, 17, device=dev) < 0.2] = 0.0
+    mask[:, 0] = 1.0
+    gout_l = torch.randn(B, 17, device=dev)
+    gout_l[mask == 0] = 0.0
+    gout_v = torch.randn(B, device=dev)
+
+    def run(disable):
+        _os.environ["DDLS_AMD_DISABLE_FUSED_HEAD"] = disable
+        policy.zero_grad(set_to_none=True)
+        gm_in = gm.clone().requires_grad_(True)
+        # bypass the GNN: call the head path directly via forward_flat
+        # internals (graph_mean result substituted)
+        from ddls_amd.models import gnn as gnn_mod
+        if disable == "1":
+            graph_emb = policy.graph_module(gf)
+            final_emb = torch.cat([gm_in, graph_emb], dim=-1)
+            logits = policy.policy_branch(final_emb)
+            value = policy.value_branch(final_emb).squeeze(-1)
+            inf_mask = torch.clamp(torch.log(mask),
+                                   min=torch.finfo(torch.float32).min)
+            logits = logits + inf_mask
+        else:
+            ln, lin_g = policy.graph_module[0], policy.graph_module[1]
+            pb, vb = policy.policy_branch, policy.value_branch
+            logits, value = gnn_mod._FusedHeadFn.apply(
+                gm_in, gf, mask, ln.weight, ln.bias, lin_g.weight,
+                lin_g.bias, pb[0].weight, pb[0].bias, pb[2].weight,
+                pb[2].bias, vb[0].weight, vb[0].bias, vb[2].weight,
+                vb[2].bias)
+        (logits * gout_l).sum().add_((value * gout_v).sum()).backward()
+        grads = {n: p.grad.clone() for n, p in policy.named_parameters()
+                 if p.grad is not None and ("graph_module" in n
+                                            or "branch" in n)}
+        return logits.detach().clone(), value.detach().clone(), \
+            gm_in.grad.clone(), grads
+
+    try:
+        l_e, v_e, ggm_e, g_e = run("1")
+        l_f, v_f, ggm_f, g_f = run("0")
+    finally:
+        _os.environ.pop("DDLS_AMD_DISABLE_FUSED_HEAD", None)
+    finite = mask > 0
+    assert torch.allclose(l_e[finite], l_f[finite], atol=2e-4), \
+        (l_e - l_f)[finite].abs().max().item()
+    assert torch.allclose(v_e, v_f, atol=2e-4)
+    assert torch.allclose(ggm_e, ggm_f, atol=2e-4), \
+        (ggm_e - ggm_f).abs().max().item()
+    assert set(g_e) == set(g_f)
+    for n in g_e:
+        assert torch.allclose(g_e[n], g_f[n], rtol=1e-3, atol=2e-4), \
+            (n, (g_e[n] - g_f[n]).abs().max().item())
