@@ -293,6 +293,9 @@ class RampJobPartitioningEnvironment:
                     _np.asarray([self.cluster.channel_index(c)
                                  for c in chan_ids], dtype=_np.int64),
                     list(_Counter(chan_ids).items()))
+        dep_channel_ids = set()
+        for channels in dpd.values():
+            dep_channel_ids.update(channels)
         # vectorised op-mount arrays: dense op/worker indices + per-worker
         # groups for the one-rule-check-per-worker mount
         g = pj.graph
@@ -324,6 +327,8 @@ class RampJobPartitioningEnvironment:
             "placement": placement,
             "op_schedule": sched,
             "dep_mount_fast": fast,
+            "dep_channel_ids": dep_channel_ids,
+            "worker_ids_set": set(placement.values()),
             "op_mount_fast": (op_idx_arr, widx_arr, op_groups),
             "sched_fast": (_np.asarray(sch_op_idx, dtype=_np.int64),
                            _np.asarray(sch_op_prio, dtype=_np.int64),
@@ -354,11 +359,11 @@ class RampJobPartitioningEnvironment:
         op_placement = OpPlacement.__new__(OpPlacement)
         op_placement.action = {job_id: cached["placement"]}
         op_placement.job_ids = {job_id}
+        # only the (uncached) SRPT scheduler reads worker_to_ops; the cached
+        # pipeline restores schedules directly, so skip building it
         op_placement.worker_to_ops = _dd(list)
-        for op_name, worker_id in cached["placement"].items():
-            op_placement.worker_to_ops[worker_id].append(
-                {"op_id": op_name, "job_id": job_id})
-        op_placement.worker_ids = set(cached["placement"].values())
+        op_placement.worker_ids = cached.get("worker_ids_set") or set(
+            cached["placement"].values())
         op_placement.job_id_to_worker_ids = {job_id: op_placement.worker_ids}
         op_placement.fast_mount = cached.get("op_mount_fast")
         self.op_placement = op_placement
@@ -367,7 +372,14 @@ class RampJobPartitioningEnvironment:
         sf = cached.get("sched_fast")
         if sf is not None:
             self.op_schedule.fast_priorities = (sf[0], sf[1])
-        self.dep_placement = DepPlacement({job_id: cached["dep_placement"]})
+        # skip DepPlacement.__init__'s per-dep channel sweep: the channel-id
+        # set is placement-determined and cached
+        dp = DepPlacement.__new__(DepPlacement)
+        dp.action = {job_id: cached["dep_placement"]}
+        dp.job_ids = {job_id}
+        dp.channel_ids = cached.get("dep_channel_ids", set())
+        dp.job_to_dep_to_channels = dp.action
+        self.dep_placement = dp
         # hand the vectorised mount arrays to RampClusterEnvironment._place_deps
         self.dep_placement.fast_mount = cached.get("dep_mount_fast")
         self.dep_schedule = DepSchedule(

@@ -90,6 +90,36 @@ class Job:
         self.reset_job(self.details, immutable=immutable)
         self.original_job = original_job if original_job is not None else self
 
+    # runtime details rebuilt by reset_job — must NOT be shared with a clone
+    _RUNTIME_DETAIL_KEYS = frozenset((
+        "mounted_workers", "mounted_channels", "communication_overhead_time",
+        "computation_overhead_time", "max_acceptable_job_completion_time",
+        "job_sequential_completion_time", "job_total_op_memory_cost",
+        "job_total_dep_size", "max_compute_node", "max_compute_cost",
+        "max_memory_node", "max_memory_cost", "max_depth", "node_to_depth",
+        "max_node_throughput", "max_dep_size_dep", "max_dep_size"))
+
+    def clone(self) -> "Job":
+        """Fast copy of a PRISTINE (never-run) job — what ``Sampler`` needs
+        per sample.  Shares the immutable graph + GraphImmutableDetails and
+        rebuilds the tick state fresh; generic ``copy.deepcopy`` recursion
+        over the details dict was ~0.5 ms per env step."""
+        import copy as _copy
+        overlay = {}
+        for k, v in self.details.items():
+            if k in self._RUNTIME_DETAIL_KEYS:
+                continue  # rebuilt by reset_job from the immutable details
+            if isinstance(v, (set, dict, list)):
+                overlay[k] = _copy.deepcopy(v)
+            else:
+                overlay[k] = v
+        return Job(graph=self.graph,
+                   num_training_steps=self.num_training_steps,
+                   max_acceptable_job_completion_time_frac=
+                   self.max_acceptable_job_completion_time_frac,
+                   job_id=self._job_id, details=overlay,
+                   immutable=self.immutable)
+
     # ---- identity ----
     @property
     def job_id(self):

@@ -19,9 +19,11 @@ LN_EPS = 1e-5
 
 
 def _ln(x, g, b):
+    # single-pass: np.var recomputes the mean internally
     mu = x.mean(axis=-1, keepdims=True)
-    var = x.var(axis=-1, keepdims=True)
-    return (x - mu) / np.sqrt(var + LN_EPS) * g + b
+    d = x - mu
+    var = np.einsum("...k,...k->...", d, d)[..., None] / x.shape[-1]
+    return d * (g / np.sqrt(var + LN_EPS)) + b
 
 
 def _relu(x):
@@ -95,7 +97,27 @@ class NumpyGNNPolicy:
         self.policy_branch = _Branch(p, "policy_branch")
         self.value_branch = _Branch(p, "value_branch")
 
-    def _meanpool(self, layer, z, e, src, dst):
+    @staticmethod
+    def _segments(obs):
+        """Per-obs CSR segments for the reduceat-based scatter (np.add.at is
+        an order of magnitude slower); cached on the CompactObs, which the
+        worker loop memoises per underlying env obs dict."""
+        seg = getattr(obs, "_np_seg", None)
+        if seg is None:
+            dst = obs.edges_dst
+            n = len(obs.node_features)
+            order = np.argsort(dst, kind="stable")
+            dst_sorted = dst[order]
+            uniq, starts = np.unique(dst_sorted, return_index=True)
+            in_deg = np.bincount(dst, minlength=n).astype(np.float32)
+            seg = (order, starts, uniq, in_deg)
+            try:
+                obs._np_seg = seg
+            except Exception:
+                pass
+        return seg
+
+    def _meanpool(self, layer, z, e, src, dst, seg):
         hn = layer["node"](z)
         he = layer["edge"](e)
         msg_edge = np.concatenate([hn[src], he], axis=-1)
@@ -103,9 +125,10 @@ class NumpyGNNPolicy:
         r_edge = layer["reduce"](msg_edge)
         r_self = layer["reduce"](msg_self)
         N = z.shape[0]
+        order, starts, uniq, in_deg = seg
         out = np.zeros((N, r_self.shape[-1]), dtype=np.float32)
-        np.add.at(out, dst, r_edge)
-        in_deg = np.bincount(dst, minlength=N).astype(np.float32)
+        if len(order):
+            out[uniq] = np.add.reduceat(r_edge[order], starts, axis=0)
         total = out + r_self
         mean = total / (in_deg + 1.0)[:, None]
         mean[in_deg == 0] = 0.0
@@ -115,8 +138,9 @@ class NumpyGNNPolicy:
         z = obs.node_features
         e = obs.edge_features
         src, dst = obs.edges_src, obs.edges_dst
+        seg = self._segments(obs)
         for layer in self.layers:
-            z = self._meanpool(layer, z, e, src, dst)
+            z = self._meanpool(layer, z, e, src, dst, seg)
         graph_node_emb = z.mean(axis=0) if len(z) else np.zeros(
             self.config["out_features_node"], dtype=np.float32)
         graph_emb = self.graph_module(obs.graph_features[None, :])[0]

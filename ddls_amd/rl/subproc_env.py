@@ -26,6 +26,18 @@ def _worker_loop(conn, env_fn_count: int, env_fn, base_seed: int,
     obs_cache = None
     rng = None
     shm_views = None       # zero-copy param views into the fork-shared buffer
+    compact_memo = {}      # env obs dict (cached per (model,frac)) -> CompactObs
+
+    def to_compact(o):
+        key = id(o)
+        hit = compact_memo.get(key)
+        if hit is not None and hit[0] is o:
+            return hit[1]
+        if len(compact_memo) > 8192:
+            compact_memo.clear()
+        co = CompactObs.from_obs(o)
+        compact_memo[key] = (o, co)
+        return co
     try:
         while True:
             cmd, payload = conn.recv()
@@ -62,9 +74,8 @@ def _worker_loop(conn, env_fn_count: int, env_fn, base_seed: int,
                     policy = NumpyGNNPolicy(state_dict, policy_cfg,
                                             num_actions)
                 if obs_cache is None:
-                    obs_cache = [CompactObs.from_obs(
-                        env.reset(seed=base_seed + 1000 * i))
-                        for i, env in enumerate(envs)]
+                    obs_cache = [to_compact(env.reset(seed=base_seed + 1000 * i))
+                                 for i, env in enumerate(envs)]
                 traj = {"obs": [], "actions": [], "logp": [], "values": [],
                         "rewards": [], "dones": []}
                 stats_out = []
@@ -97,7 +108,7 @@ def _worker_loop(conn, env_fn_count: int, env_fn, base_seed: int,
                             episode_counters[i] += 1
                             o = env.reset(seed=base_seed + 1000 * i
                                           + episode_counters[i])
-                        obs_cache[i] = CompactObs.from_obs(o)
+                        obs_cache[i] = to_compact(o)
                     traj["rewards"].append(rewards)
                     traj["dones"].append(dones)
                 # bootstrap values of the final obs
@@ -129,7 +140,7 @@ def _worker_loop(conn, env_fn_count: int, env_fn, base_seed: int,
                 obs = []
                 for i, env in enumerate(envs):
                     o = env.reset(seed=base_seed + 1000 * i)
-                    obs.append(CompactObs.from_obs(o))
+                    obs.append(to_compact(o))
                 obs_cache = list(obs)
                 conn.send(obs)
             elif cmd == "step":
@@ -149,7 +160,7 @@ def _worker_loop(conn, env_fn_count: int, env_fn, base_seed: int,
                         episode_counters[i] += 1
                         o = env.reset(seed=base_seed + 1000 * i
                                       + episode_counters[i])
-                    obs_out.append(CompactObs.from_obs(o))
+                    obs_out.append(to_compact(o))
                     rewards.append(r)
                     dones.append(done)
                 obs_cache = list(obs_out)

@@ -55,7 +55,9 @@ class Sampler:
         else:
             raise ValueError(f"Unrecognised sampling_mode {self.sampling_mode}")
         proto = self.original_pool[proto_idx]
-        job = copy.deepcopy(proto)
+        # pool prototypes are pristine (never run), so the fast field-level
+        # clone is equivalent to deepcopy at a fraction of the cost
+        job = proto.clone() if hasattr(proto, "clone") else copy.deepcopy(proto)
         if self.automatically_change_ids:
             job.job_id = int(self._base_ids[proto_idx] + proto.job_id)
         return job
