@@ -127,10 +127,13 @@ def main():
     for _ in range(args.warmup):
         trainer.train(num_steps=args.rollout_steps_per_env)
 
+    jct_samples = []
     barrier_sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
         st = trainer.train(num_steps=args.rollout_steps_per_env)
+        if "mean_job_completion_time" in st:
+            jct_samples.append(st["mean_job_completion_time"])
         if rank == 0:
             print(f"[bench] iter {st['iteration']}: rollout "
                   f"{st['rollout_time_s']:.3f}s update {st['update_time_s']:.3f}s "
@@ -143,6 +146,7 @@ def main():
         print(f"[bench] hipgraph: broken={sp.broken} captures={sp.capture_count} "
               f"flat_adam={getattr(sp, 'flat_p', None) is not None} "
               f"fused_loss={getattr(sp, '_fused_loss', False)} "
+              f"repairs={getattr(sp, 'repair_count', None)} "
               f"err={sp.last_error}", file=sys.stderr, flush=True)
     elapsed = time.perf_counter() - t0
 
@@ -181,6 +185,12 @@ def main():
                 "envs_per_rank": args.envs_per_rank,
                 "env_workers": n_workers,
                 "num_sgd_iter": args.num_sgd_iter,
+                # second half of the BASELINE metric ("mean simulated JCT"):
+                # mean over episodes COMPLETED inside the timed window (null
+                # on short runs — episodes are 1000 jobs long); the trained-
+                # policy JCT numbers live in docs/RESULTS.md
+                "mean_simulated_jct": (float(np.mean(jct_samples))
+                                       if jct_samples else None),
             },
         }))
     if venv_close is not None:

@@ -352,6 +352,8 @@ class CapturedSGDStep:
                                        kl.detach(), entropy.detach(),
                                        loss.detach()])
 
+    repair_count = 0  # grads rebound off their views at last capture
+
     def _repair_grad_views(self):
         """If AccumulateGrad rebound any p.grad off its flat_g view during
         this (possibly capture-recorded) backward, record a copy from the
@@ -360,10 +362,12 @@ class CapturedSGDStep:
         if self.flat_p is None:
             return
         base = self.flat_g.data_ptr()
+        self.repair_count = 0
         for p_, (o, np_) in zip(self._params, self._offs):
             g = p_.grad
             if g is not None and g.data_ptr() != base + 4 * o:
                 self.flat_g[o:o + np_].copy_(g.detach().reshape(-1))
+                self.repair_count += 1
 
     def _body_opt(self):
         if self.flat_p is not None:
@@ -444,6 +448,10 @@ class CapturedSGDStep:
         g.div_(dist.get_world_size())
 
     def _capture(self, n_cap: int, e_cap: int):
+        # drop the old graph first: its recorded state holds references that
+        # push AccumulateGrad off the in-place path during re-capture
+        self.graph = None
+        self.graph_opt = None
         self._alloc(n_cap, e_cap)
         split = is_distributed()
 
