@@ -208,6 +208,8 @@ class Action:
             self.job_idxs = set()
 
         for key, act in self.actions.items():
+            if act.job_ids == self.job_ids:
+                continue  # every handled job survived: nothing to filter
             self._filter_action(key, act)
 
     def _filter_action(self, key, act):
