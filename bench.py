@@ -63,8 +63,8 @@ def build_env_fn(seed_offset: int = 0, device_type: str = "A100",
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=5)
-    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--steps", type=int, default=6)
+    ap.add_argument("--warmup", type=int, default=3)
     ap.add_argument("--envs-per-rank", type=int, default=64)
     ap.add_argument("--rollout-steps-per-env", type=int, default=16)
     ap.add_argument("--env-workers", type=int, default=0,
