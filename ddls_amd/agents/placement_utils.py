@@ -9,7 +9,7 @@ Reference: ``ddls/environments/ramp_cluster/agents/placers/utils.py`` —
 from __future__ import annotations
 
 import math
-from typing import Dict, List, Optional, Sequence, Tuple
+from typing import Dict, List, Sequence, Tuple
 
 from ..graphs import CompGraph, FWD, backward_name, partitioned_name
 

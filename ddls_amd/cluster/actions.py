@@ -14,8 +14,7 @@ import copy
 from collections import defaultdict
 from typing import Dict, Optional, Set
 
-from ..graphs import FWD
-from ..jobs import GraphImmutableDetails, Job
+from ..jobs import Job
 from .comm_model import update_dep_run_times
 from .partition import build_partitioned_graph
 

@@ -15,11 +15,11 @@ effective bandwidth term is per-link, not switched.
 from __future__ import annotations
 
 import math
-from typing import Dict, List, Sequence, Set, Tuple
+from typing import Dict, List, Set
 
 import numpy as np
 
-from ..graphs import BWD, FWD, CompGraph, backward_name, partitioned_name
+from ..graphs import FWD, CompGraph, backward_name, partitioned_name
 
 # device compute-side constants of the reference model (actions/utils.py:52-60)
 DEFAULT_MEM_FRQ = 2e12

@@ -12,7 +12,6 @@ batched-env kernels consume.
 from __future__ import annotations
 
 import copy
-import math
 from collections import defaultdict
 from typing import Dict, Optional, Union
 

@@ -16,7 +16,7 @@ import numpy as np
 
 from ..jobs import Job, JobQueue, JobsGenerator
 from ..topology import build_topology
-from ..utils import Stopwatch, get_class_from_path, seed_everything
+from ..utils import Stopwatch, seed_everything
 from .environment import RampClusterEnvironment
 
 

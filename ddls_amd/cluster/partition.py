@@ -17,7 +17,7 @@ from typing import Dict, List, Optional, Sequence, Tuple
 
 import numpy as np
 
-from ..graphs import BWD, FWD, CompGraph, backward_name, partitioned_name
+from ..graphs import BWD, FWD, CompGraph, partitioned_name
 
 
 class _MutableGraph:

@@ -7,7 +7,7 @@ Reference: ``ddls/environments/ramp_job_partitioning/agents/`` — ``random.py:3
 from __future__ import annotations
 
 import math
-from typing import Optional, Union
+from typing import Optional
 
 import numpy as np
 

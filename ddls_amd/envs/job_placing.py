@@ -15,7 +15,6 @@ from typing import Optional, Union
 import numpy as np
 
 from ..cluster.legacy_environment import ClusterEnvironment
-from ..graphs import FWD
 from . import spaces
 
 

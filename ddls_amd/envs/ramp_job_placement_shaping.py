@@ -13,7 +13,6 @@ Action treats it as a sixth handled-job constraint.
 """
 from __future__ import annotations
 
-from collections import defaultdict
 from typing import Optional, Union
 
 import numpy as np

@@ -14,9 +14,8 @@ used at API boundaries.
 from __future__ import annotations
 
 import json
-import math
 import re
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional, Sequence, Tuple
 
 import numpy as np

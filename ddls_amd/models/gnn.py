@@ -20,7 +20,6 @@ Semantics parity notes:
 """
 from __future__ import annotations
 
-import math
 import os
 from dataclasses import dataclass
 from typing import Dict, List, Optional

@@ -9,10 +9,9 @@ one process per GPU, fused gradient all-reduce with RCCL over xGMI
 """
 from __future__ import annotations
 
-import math
 import os
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Dict, List, Optional
 
 import numpy as np

@@ -7,7 +7,6 @@ checkpoint/log cadence and best-checkpoint tracking).
 """
 from __future__ import annotations
 
-import copy
 import time
 from collections import defaultdict
 from typing import Callable, Dict, List, Optional

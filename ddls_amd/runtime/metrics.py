@@ -12,8 +12,7 @@ import csv
 import glob
 import json
 import os
-from collections import defaultdict
-from typing import Dict, List, Optional, Sequence
+from typing import Dict, List, Sequence
 
 import numpy as np
 
