@@ -18,8 +18,7 @@ from typing import Optional, Union
 import numpy as np
 
 from ..agents.partitioners import RandomOpPartitioner, SipMlOpPartitioner
-from ..agents.placement_utils import (check_meta_block_valid, dummy_ramp,
-                                      find_meta_block)
+from ..agents.placement_utils import check_meta_block_valid, dummy_ramp
 from ..agents.placers import FirstFitDepPlacer, RampFirstFitOpPlacer
 from ..agents.schedulers import SRPTDepScheduler, SRPTOpScheduler
 from ..cluster.actions import Action, JobPlacementShape, OpPartition
