@@ -70,7 +70,22 @@ class OpPartition:
                 memo[model][degree] = entry = {"graph": pgraph, "immutable": None}
             pgraph = entry["graph"]
 
-            details = copy.deepcopy(job.details)
+            # structured shallow copy == deepcopy for this dict's shapes
+            # (scalars, flat dicts of scalars, sets of ids, read-only arrays);
+            # generic deepcopy recursion was ~0.2 ms per env step.  The
+            # reference's overlay quirk (original-job detail values win over
+            # the partitioned job's recomputed ones, job.py:364-378) is
+            # preserved because every key is still carried over.
+            details = {}
+            for k, v in job.details.items():
+                if isinstance(v, set):
+                    details[k] = set(v)
+                elif isinstance(v, dict):
+                    details[k] = dict(v)
+                elif isinstance(v, list):
+                    details[k] = list(v)
+                else:
+                    details[k] = v  # scalars / read-only numpy arrays
             details["max_partitions_per_op"] = degree
             self.partitioned_jobs[job_id] = Job(
                 graph=pgraph,
