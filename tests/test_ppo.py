@@ -356,12 +356,19 @@ def test_distributed_split_capture_one_gpu():
         env["MASTER_PORT"] = "29641"
         env["DDLS_AMD_DIST_BACKEND"] = "gloo"
         procs = []
-        for rank in range(2):
-            e = dict(env, RANK=str(rank), WORLD_SIZE="2", LOCAL_RANK="0")
-            procs.append(subprocess.Popen([sys.executable, script], env=e,
-                                          stdout=subprocess.PIPE,
-                                          stderr=subprocess.STDOUT))
-        outs = [p.communicate(timeout=300)[0].decode() for p in procs]
+        try:
+            for rank in range(2):
+                e = dict(env, RANK=str(rank), WORLD_SIZE="2", LOCAL_RANK="0")
+                procs.append(subprocess.Popen([sys.executable, script], env=e,
+                                              stdout=subprocess.PIPE,
+                                              stderr=subprocess.STDOUT))
+            outs = [p.communicate(timeout=240)[0].decode() for p in procs]
+        finally:
+            # never leave stragglers holding the GPU / rendezvous port
+            for p in procs:
+                if p.poll() is None:
+                    p.kill()
+                    p.wait(timeout=30)
         for rank, (p, out) in enumerate(zip(procs, outs)):
             assert p.returncode == 0, out
             assert f"RANK{rank}_CAPTURE_OK" in out
