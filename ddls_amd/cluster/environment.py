@@ -798,6 +798,45 @@ class RampClusterEnvironment:
             if isinstance(es[metric], list):
                 es[metric] = float(np.mean(es[metric])) if len(es[metric]) > 0 else 0
 
+        self._episodes_finalised = getattr(self, "_episodes_finalised", 0) + 1
+        if (self.path_to_save is not None
+                and self._episodes_finalised % max(self.save_freq, 1) == 0):
+            self.save_cluster_data()
+
+    def save_cluster_data(self, blocking: bool = False):
+        """Persist steps_log + episode_stats to ``path_to_save`` as gzip-pkl
+        or a sqlite KV table, in a background thread (reference
+        ``ramp_cluster_environment.py:1570-1598``; ``use_sqlite_database``
+        flag ``:81,103-104``)."""
+        import threading
+        data = {"steps_log": {k: list(v) for k, v in self.steps_log.items()},
+                "episode_stats": dict(self.episode_stats)}
+        ep = getattr(self, "_episodes_finalised", 0)
+
+        def _save():
+            import gzip
+            import os as _os
+            import pickle
+            _os.makedirs(self.path_to_save, exist_ok=True)
+            if self.use_sqlite_database:
+                from ..runtime.logger import SqliteKV
+                db = SqliteKV(_os.path.join(self.path_to_save,
+                                            "cluster_data.sqlite"))
+                db[f"episode_{ep:06d}"] = data
+                db.commit()
+                db.close()
+            else:
+                with gzip.open(_os.path.join(
+                        self.path_to_save,
+                        f"cluster_data_episode_{ep:06d}.pkl.gz"), "wb") as f:
+                    pickle.dump(data, f)
+
+        t = threading.Thread(target=_save, daemon=True)
+        t.start()
+        self._save_thread = t
+        if blocking:
+            t.join()
+
     # ---- metric vocabulary (reference :1181-1280) ----
     @staticmethod
     def episode_metrics():

@@ -163,3 +163,37 @@ def test_config_group_override_1024_workers():
     assert env.cluster.topology.num_workers == 1024
     r = EvalLoop(ACTORS["acceptable_jct"](), env, max_steps=12).run(seed=7)
     assert r["num_actor_steps"] == 12
+
+
+def test_cluster_data_persistence(tiny_model_files, tmp_path):
+    """path_to_save + use_sqlite_database persist steps_log/episode_stats at
+    episode end (reference ramp_cluster_environment.py:1570-1598)."""
+    import gzip
+    import pickle
+    from ddls_amd.runtime.logger import SqliteKV
+    from tests.conftest import make_env
+
+    for sqlite in (False, True):
+        out = tmp_path / ("sq" if sqlite else "gz")
+        env = make_env(tiny_model_files, replication=2)
+        env.cluster.path_to_save = str(out)
+        env.cluster.use_sqlite_database = sqlite
+        obs = env.reset(seed=0)
+        done = False
+        while not done:
+            valid = obs["action_set"][obs["action_mask"].astype(bool)]
+            obs, _r, done, _ = env.step(int(valid[-1]))
+        env.cluster._save_thread.join(timeout=30)
+        if sqlite:
+            db = SqliteKV(str(out / "cluster_data.sqlite"))
+            keys = db.keys()
+            assert keys, "no episodes saved"
+            data = db[keys[0]]
+            db.close()
+        else:
+            files = sorted(out.glob("cluster_data_episode_*.pkl.gz"))
+            assert files
+            with gzip.open(files[0], "rb") as f:
+                data = pickle.load(f)
+        assert data["episode_stats"]["num_jobs_arrived"] >= 1
+        assert "num_jobs_running" in data["steps_log"]
