@@ -196,4 +196,4 @@ def test_cluster_data_persistence(tiny_model_files, tmp_path):
             with gzip.open(files[0], "rb") as f:
                 data = pickle.load(f)
         assert data["episode_stats"]["num_jobs_arrived"] >= 1
-        assert "num_jobs_running" in data["steps_log"]
+        assert "compute_info_processed" in data["steps_log"]
