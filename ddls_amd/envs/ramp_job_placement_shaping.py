@@ -173,3 +173,47 @@ class RampJobPlacementShapingEnvironment:
             self.obs = self.observation_function.extract(env=self, done=False)
         self.step_counter += 1
         return self.obs, self.reward, self.done, {}
+
+
+class FirstFitJobPlacementShaper:
+    """Pick the FIRST valid meta-block shape (reference
+    ``ramp_job_placement_shaping/agents/ramp_first_fit_job_placement_shaper.py:10``)."""
+
+    def compute_action(self, obs, **kwargs):
+        mask = np.asarray(obs["action_mask"]).astype(bool)
+        action_set = np.asarray(obs["action_set"])
+        valid = action_set[mask]
+        nonzero = valid[valid != 0]
+        return int(nonzero[0]) if len(nonzero) else 0
+
+
+class LastFitJobPlacementShaper:
+    """Pick the LAST valid meta-block shape."""
+
+    def compute_action(self, obs, **kwargs):
+        mask = np.asarray(obs["action_mask"]).astype(bool)
+        action_set = np.asarray(obs["action_set"])
+        valid = action_set[mask]
+        nonzero = valid[valid != 0]
+        return int(nonzero[-1]) if len(nonzero) else 0
+
+
+class RandomJobPlacementShaper:
+    """Uniform over valid shapes (reference
+    ``agents/ramp_random_job_placement_shaper.py:10``)."""
+
+    def compute_action(self, obs, **kwargs):
+        mask = np.asarray(obs["action_mask"]).astype(bool)
+        action_set = np.asarray(obs["action_set"])
+        valid = action_set[mask]
+        nonzero = valid[valid != 0]
+        if len(nonzero) == 0:
+            return 0
+        return int(np.random.choice(nonzero))
+
+
+SHAPING_AGENTS = {
+    "first_fit": FirstFitJobPlacementShaper,
+    "last_fit": LastFitJobPlacementShaper,
+    "random": RandomJobPlacementShaper,
+}

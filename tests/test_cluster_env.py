@@ -328,3 +328,48 @@ def test_pipeline_cache_trace_equal(tiny_model_files):
     assert s1["num_jobs_completed"] == s2["num_jobs_completed"]
     assert s1["num_jobs_blocked"] == s2["num_jobs_blocked"]
     assert s1["job_completion_time"] == pytest.approx(s2["job_completion_time"])
+
+
+def _make_shaping_env(tiny_model_files):
+    from ddls_amd.envs.ramp_job_placement_shaping import (
+        RampJobPlacementShapingEnvironment)
+    return RampJobPlacementShapingEnvironment(
+        topology_config={"type": "ramp", "kwargs": {
+            "num_communication_groups": 4,
+            "num_racks_per_communication_group": 4,
+            "num_servers_per_rack": 2, "num_channels": 1,
+            "total_node_bandwidth": 1.6e12,
+            "intra_gpu_propagation_latency": 50e-9,
+            "worker_io_latency": 100e-9}},
+        node_config={"type_1": {"num_nodes": 32, "workers_config": [
+            {"num_workers": 1, "worker": "ddls_amd.devices.A100"}]}},
+        jobs_config={"path_to_files": tiny_model_files,
+                     "replication_factor": 3,
+                     "job_sampling_mode": "remove",
+                     "job_interarrival_time_dist": {
+                         "_target_": "ddls_amd.distributions.Fixed",
+                         "val": 1000},
+                     "max_acceptable_job_completion_time_frac_dist": {
+                         "_target_": "ddls_amd.distributions.Fixed", "val": 1.0},
+                     "num_training_steps": 5,
+                     "max_partitions_per_op_in_observation": 2},
+        op_partitioner_kwargs={"min_op_run_time_quantum": 0.01},
+        max_simulation_run_time=1e6)
+
+
+def test_shaping_env_baseline_agents(tiny_model_files):
+    """first/last/random shape pickers run episodes on the shaping env
+    (reference ramp_job_placement_shaping/agents/, 3 baseline agents)."""
+    from ddls_amd.envs.ramp_job_placement_shaping import (
+        SHAPING_AGENTS, RampJobPlacementShapingEnvironment)
+    from ddls_amd.utils import seed_everything
+    for name, cls in SHAPING_AGENTS.items():
+        seed_everything(4)
+        env = _make_shaping_env(tiny_model_files)
+        agent = cls()
+        obs = env.reset(seed=4)
+        done, steps = False, 0
+        while not done and steps < 25:
+            obs, r, done, _ = env.step(agent.compute_action(obs))
+            steps += 1
+        assert env.cluster.episode_stats["num_jobs_arrived"] >= 1, name
