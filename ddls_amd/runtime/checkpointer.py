@@ -30,8 +30,27 @@ class Checkpointer:
 
     @staticmethod
     def read(checkpoint_path: str) -> dict:
+        """Load a checkpoint; GPU-written tensors map to CPU automatically
+        when no GPU is visible (RLlib restore is likewise device-portable)."""
+        import torch
         with open(checkpoint_path, "rb") as f:
-            return pickle.load(f)
+            if torch.cuda.is_available():
+                return pickle.load(f)
+            return _CpuUnpickler(f).load()
+
+
+class _CpuUnpickler(pickle.Unpickler):
+    """Unpickler that remaps torch storages to CPU (plain-pickle checkpoints
+    carry the saving device inside ``torch.storage._load_from_bytes``)."""
+
+    def find_class(self, module, name):
+        if module == "torch.storage" and name == "_load_from_bytes":
+            import io
+
+            import torch
+            return lambda b: torch.load(io.BytesIO(b), map_location="cpu",
+                                        weights_only=False)
+        return super().find_class(module, name)
 
     def latest(self) -> Optional[str]:
         entries = sorted(os.listdir(self.checkpoint_dir))
