@@ -38,6 +38,14 @@ class Checkpointer:
                 return pickle.load(f)
             return _CpuUnpickler(f).load()
 
+    def latest(self) -> Optional[str]:
+        entries = sorted(os.listdir(self.checkpoint_dir))
+        if not entries:
+            return None
+        last = entries[-1]
+        index = int(last.split("_")[-1])
+        return os.path.join(self.checkpoint_dir, last, f"checkpoint-{index}")
+
 
 class _CpuUnpickler(pickle.Unpickler):
     """Unpickler that remaps torch storages to CPU (plain-pickle checkpoints
@@ -51,11 +59,3 @@ class _CpuUnpickler(pickle.Unpickler):
             return lambda b: torch.load(io.BytesIO(b), map_location="cpu",
                                         weights_only=False)
         return super().find_class(module, name)
-
-    def latest(self) -> Optional[str]:
-        entries = sorted(os.listdir(self.checkpoint_dir))
-        if not entries:
-            return None
-        last = entries[-1]
-        index = int(last.split("_")[-1])
-        return os.path.join(self.checkpoint_dir, last, f"checkpoint-{index}")
