@@ -217,3 +217,29 @@ def test_artifact_checkpoint_loads_and_acts(tiny_model_files):
     obs = env.reset(seed=0)
     a = actor.compute_action(obs)
     assert obs["action_mask"][list(obs["action_set"]).index(a)] == 1
+
+
+def test_artifact_eval_reproduces_gpu_result():
+    """Greedy eval of the committed GPU-trained checkpoint on CPU reproduces
+    the recorded GPU eval (docs/RESULTS.md): deterministic simulator +
+    device-portable checkpoint."""
+    import os
+    import pytest as _pytest
+    from bench import build_env_fn
+    from ddls_amd.runtime.loops import EvalLoop, PolicyActor
+    from ddls_amd.utils import seed_everything
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    ckpt = os.path.join(root, "artifacts", "round1", "final_train_500",
+                        "checkpoint_000500", "checkpoint-500")
+    if not os.path.exists(ckpt):
+        _pytest.skip("artifact checkpoint not present")
+    actor = PolicyActor.from_checkpoint(ckpt, num_actions=17, device="cpu")
+    seed_everything(1799)
+    env = build_env_fn()()
+    r = EvalLoop(actor, env, max_steps=600).run(seed=1799)
+    es = env.cluster.episode_stats
+    assert r["episode_return"] == _pytest.approx(-31233.27, abs=1.0)
+    assert es["num_jobs_blocked"] / es["num_jobs_arrived"] == \
+        _pytest.approx(0.0832, abs=0.002)
+    assert r["mean_job_completion_time_speedup"] == \
+        _pytest.approx(6.287, abs=0.01)
