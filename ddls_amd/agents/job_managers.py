@@ -158,7 +158,8 @@ class AllReduceJobCommunicator:
         return steps * self.latency + per_link_bytes / self.link_bandwidth
 
     def communicate(self, job, cluster) -> float:
-        workers = {cluster.job_op_to_worker.get((job.details["job_idx"], i))
+        _jmap = cluster.job_op_to_worker.get(job.details["job_idx"], {})
+        workers = {_jmap.get(i)
                    for i in range(job.graph.n)}
         workers.discard(None)
         message = float(job.graph.memory_cost.sum())

@@ -151,6 +151,8 @@ class RampClusterEnvironment:
         self.jobs_running: Dict[int, Job] = {}
         self.jobs_completed: Dict[int, Job] = {}
         self.jobs_blocked: Dict[int, Job] = {}
+        # job_idx -> {op_idx -> worker_id}: nested so job removal is one
+        # del instead of a per-op pop sweep
         self.job_op_to_worker = {}
         self.job_dep_to_channels = defaultdict(set)
         self.job_idx_to_job_id = {}
@@ -353,8 +355,8 @@ class RampClusterEnvironment:
                 job.op_remaining[ops_w] = g.compute_cost[dt][ops_w]
                 for o in ops_w:
                     job.mounted_device_type[o] = dt
-                self.job_op_to_worker.update(
-                    ((job_idx, int(o)), worker_id) for o in ops_w)
+                self.job_op_to_worker.setdefault(job_idx, {}).update(
+                    (int(o), worker_id) for o in ops_w)
             self.num_mounted_ops += len(op_idx_arr)
             self._register_running_job(job)
             self.job_op_placement[job_id] = placement
@@ -379,7 +381,7 @@ class RampClusterEnvironment:
                 job.details["mounted_workers"].add(worker_id)
                 self.num_mounted_ops += 1
                 job.reset_op_remaining_run_time(op_idx, device_type=worker.device_type)
-                self.job_op_to_worker[(job_idx, op_idx)] = worker_id
+                self.job_op_to_worker.setdefault(job_idx, {})[op_idx] = worker_id
                 job.op_worker[op_idx] = self.worker_id_to_index[worker_id]
             self._register_running_job(job)
             self.job_op_placement[job_id] = placement
@@ -518,13 +520,12 @@ class RampClusterEnvironment:
                     for o in ops_w:
                         worker.mounted_job_op_to_priority.pop(
                             (job_idx, int(o)), None)
-                for o in ops_w:
-                    self.job_op_to_worker.pop((job_idx, int(o)), None)
+            self.job_op_to_worker.pop(job_idx, None)
             self.num_mounted_ops -= int(len(mounted_ops))
         else:
             for op_idx in range(g.n):
-                key = (job_idx, int(op_idx))
-                worker_id = self.job_op_to_worker.pop(key, None)
+                worker_id = self.job_op_to_worker.get(job_idx, {}).pop(
+                    int(op_idx), None)
                 if worker_id is not None:
                     node = self.topology.worker_to_node[worker_id]
                     self.topology.node_workers[node][worker_id].unmount(
