@@ -243,3 +243,26 @@ def test_artifact_eval_reproduces_gpu_result():
         _pytest.approx(0.0832, abs=0.002)
     assert r["mean_job_completion_time_speedup"] == \
         _pytest.approx(6.287, abs=0.01)
+
+
+def test_mi355x_flavoured_env_config():
+    """MI355X workers (288 GB) + xGMI-derived link parameters drive the
+    simulator via the env_mi355x config group."""
+    import os
+    from ddls_amd.cluster.comm_model import xgmi_profile
+    from ddls_amd.runtime.config import build_env_from_config, load_config
+    from ddls_amd.utils import seed_everything
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cfg = load_config(os.path.join(root, "configs", "heuristic_config.yaml"),
+                      overrides=["env_config=env_mi355x"])
+    prof = xgmi_profile()
+    assert cfg["env_config"]["topology_config"]["kwargs"][
+        "total_node_bandwidth"] == 1.071e12
+    assert prof["total_node_bandwidth"] == 7 * 153e9
+    seed_everything(5)
+    env = build_env_from_config(cfg)
+    assert env.cluster.workers[0].memory_capacity == int(288e9)
+    obs = env.reset(seed=5)
+    valid = obs["action_set"][obs["action_mask"].astype(bool)]
+    _obs, r, _done, _ = env.step(int(valid[-1]))
+    assert r is not None
