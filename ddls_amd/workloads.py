@@ -94,3 +94,28 @@ if __name__ == "__main__":
     target = sys.argv[1] if len(sys.argv) > 1 else default_data_dir()
     generate_default_set(target)
     print(f"wrote synthetic pipedream profiles to {target}")
+
+
+MEDIUM_MODELS = {
+    # larger graphs near the 150-node observation cap (after fwd+bwd
+    # mirroring a 70-fwd-node model has 140 obs nodes): stresses padding,
+    # block placement and the batched lookahead at scale
+    "synth_bert": (64, 8, 1.4, 11),
+    "synth_resnet152": (75, 15, 0.9, 12),
+    "synth_wideresnet": (70, 5, 1.8, 13),
+}
+
+
+def generate_medium_set(out_dir: str):
+    os.makedirs(out_dir, exist_ok=True)
+    for name, (n, skips, scale, seed) in MEDIUM_MODELS.items():
+        nodes, edges = generate_model(name, n, skips, scale, seed)
+        write_pipedream_txt(os.path.join(out_dir, f"{name}.txt"), nodes, edges)
+
+
+def ensure_medium_set(out_dir: Optional[str] = None) -> str:
+    out_dir = out_dir or os.path.join(
+        os.path.dirname(default_data_dir()), "medium_graphs")
+    if not os.path.isdir(out_dir) or not os.listdir(out_dir):
+        generate_medium_set(out_dir)
+    return out_dir
