@@ -99,6 +99,9 @@ def build_env_from_config(cfg: dict):
     if jobs_config.get("path_to_files") in (None, "default", "synthetic"):
         from ..workloads import ensure_default_set
         jobs_config["path_to_files"] = ensure_default_set()
+    elif jobs_config.get("path_to_files") == "medium":
+        from ..workloads import ensure_medium_set
+        jobs_config["path_to_files"] = ensure_medium_set()
     return RampJobPartitioningEnvironment(
         topology_config=env_cfg["topology_config"],
         node_config=env_cfg["node_config"],

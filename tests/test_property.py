@@ -104,3 +104,18 @@ def test_sampler_cycles(n, mode):
         assert sorted(got2) == list(range(n))
     else:
         assert len(s) == 0
+
+
+@settings(max_examples=12, deadline=None)
+@given(size=st.floats(1e6, 1e11), factor=st.floats(1.5, 8.0))
+def test_comm_model_monotone(size, factor):
+    """Collective and one-to-one times are strictly increasing in message
+    size and always >= their latency floors."""
+    from ddls_amd.cluster.comm_model import (calc_one_to_one_time,
+                                             calc_ramp_all_reduce_time)
+    t1 = calc_one_to_one_time(size)
+    t2 = calc_one_to_one_time(size * factor)
+    assert t2 > t1 > 1.25e-6
+    c1 = calc_ramp_all_reduce_time(size, node_ids=8, racks=4, cgs=4)
+    c2 = calc_ramp_all_reduce_time(size * factor, node_ids=8, racks=4, cgs=4)
+    assert c2 > c1 > 0
