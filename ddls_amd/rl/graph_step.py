@@ -144,6 +144,12 @@ class CapturedSGDStep:
         except Exception as exc:  # unsupported torch/ROCm combo -> eager
             self.broken = True
             self.last_error = f"{type(exc).__name__}: {exc}"
+            # hand the flat Adam state back so eager training continues
+            # exactly where the captured steps left off
+            try:
+                self.sync_state_to_optimizer()
+            except Exception:
+                pass
             # capturable eager Adam is slow; revert it for the fallback path
             for g in self.optimizer.param_groups:
                 g["capturable"] = False
