@@ -119,3 +119,26 @@ def test_comm_model_monotone(size, factor):
     c1 = calc_ramp_all_reduce_time(size, node_ids=8, racks=4, cgs=4)
     c2 = calc_ramp_all_reduce_time(size * factor, node_ids=8, racks=4, cgs=4)
     assert c2 > c1 > 0
+
+
+@settings(max_examples=20, deadline=None)
+@given(num=st.sampled_from([2, 4, 6, 8, 10, 12, 14, 16]))
+def test_block_shapes_fit_ramp(num):
+    """Every shape from get_block_shapes multiplies to the requested degree
+    and fits inside the RAMP shape; every emitted block has `num` distinct
+    in-bounds workers (reference placers/utils.py:445-531)."""
+    from ddls_amd.agents.placement_utils import (get_block, get_block_shapes,
+                                                 get_factor_pairs)
+    ramp_shape = (4, 4, 2)
+    pairs = get_factor_pairs(num)
+    shapes = get_block_shapes(pairs, ramp_shape)
+    for (c, r, s) in shapes:
+        assert c * r * s == num
+        assert c <= ramp_shape[0] and r <= ramp_shape[1] and s <= ramp_shape[2]
+        for block in get_block(c, r, s, ramp_shape):
+            assert len(block) == num
+            assert len(set(map(tuple, block))) == num
+            for (ci, ri, si) in block:
+                assert 0 <= ci < ramp_shape[0]
+                assert 0 <= ri < ramp_shape[1]
+                assert 0 <= si < ramp_shape[2]
