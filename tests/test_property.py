@@ -135,10 +135,10 @@ def test_block_shapes_fit_ramp(num):
     for (c, r, s) in shapes:
         assert c * r * s == num
         assert c <= ramp_shape[0] and r <= ramp_shape[1] and s <= ramp_shape[2]
-        for block in get_block(c, r, s, ramp_shape):
-            assert len(block) == num
-            assert len(set(map(tuple, block))) == num
-            for (ci, ri, si) in block:
-                assert 0 <= ci < ramp_shape[0]
-                assert 0 <= ri < ramp_shape[1]
-                assert 0 <= si < ramp_shape[2]
+        block = get_block(c, r, s, ramp_shape)
+        assert len(block) == num
+        assert len(set(map(tuple, block))) == num
+        for (ci, ri, si) in block:
+            assert 0 <= ci < ramp_shape[0]
+            assert 0 <= ri < ramp_shape[1]
+            assert 0 <= si < ramp_shape[2]
