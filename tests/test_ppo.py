@@ -466,3 +466,33 @@ def test_flat_adam_state_checkpoint_roundtrip():
     assert st2.step(mb, acts, olp, adv, vt)
     torch.cuda.synchronize()
     assert float(st2.step_t.item()) == 4.0  # continued from step 3
+
+
+def test_launcher_checkpoints_and_best_tracking(tiny_model_files, tmp_path):
+    """Launcher: checkpoint at epoch 0 + every evaluation_interval, best
+    checkpoint tracked by eval episode return, results log populated
+    (reference launcher.py:97-184, rllib_epoch_loop.py:144-230)."""
+    from ddls_amd.models.gnn import GNNPolicy
+    from ddls_amd.rl.ppo import PPOConfig, PPOTrainer
+    from ddls_amd.rl.rollout import VectorEnv
+    from ddls_amd.runtime.loops import EpochLoop, Launcher
+
+    torch.manual_seed(0)
+    venv = VectorEnv([lambda: make_env(tiny_model_files, replication=50)
+                      for _ in range(2)], base_seed=3)
+    trainer = PPOTrainer(venv, GNNPolicy(num_actions=17),
+                         PPOConfig(train_batch_size=8, sgd_minibatch_size=8,
+                                   num_sgd_iter=1),
+                         device=torch.device("cpu"))
+    evals = iter([{"episode_return": -100.0}, {"episode_return": -50.0}])
+    launcher = Launcher(EpochLoop(trainer), num_epochs=4,
+                        evaluation_interval=2, eval_fn=lambda: next(evals),
+                        path_to_save=str(tmp_path / "run"), verbose=False)
+    launcher.run()
+    ckpts = sorted((tmp_path / "run" / "checkpoints").iterdir())
+    assert [c.name for c in ckpts] == ["checkpoint_000000", "checkpoint_000002",
+                                       "checkpoint_000004"]
+    assert launcher.best_eval_return == -50.0
+    assert launcher.best_checkpoint.endswith("checkpoint_000004/checkpoint-4")
+    assert len(launcher.results_log["epoch_counter"]) == 4
+    assert launcher.results_log["eval_episode_return"] == [-100.0, -50.0]
