@@ -266,3 +266,23 @@ def test_mi355x_flavoured_env_config():
     valid = obs["action_set"][obs["action_mask"].astype(bool)]
     _obs, r, _done, _ = env.step(int(valid[-1]))
     assert r is not None
+
+
+def test_training_run_loaders(tmp_path):
+    """load_training_run / summarise_runs read what the Launcher's Logger
+    writes (reference results loaders, ramp_cluster/utils.py:129-473)."""
+    from ddls_amd.runtime.logger import Logger
+    from ddls_amd.runtime.metrics import load_training_run, summarise_runs
+    for name, rewards in (("a", [1.0, 2.0]), ("b", [3.0])):
+        d = tmp_path / name
+        lg = Logger(str(d))
+        lg.write({"train_log": {"mean_reward": rewards,
+                                "epoch_counter": list(range(1, len(rewards) + 1))}},
+                 block=True)
+        log = load_training_run(str(d))
+        assert log["mean_reward"] == rewards
+    rows = summarise_runs(str(tmp_path))
+    assert {r["run"] for r in rows} == {"a", "b"}
+    by = {r["run"]: r for r in rows}
+    assert by["a"]["final_reward"] == 2.0
+    assert by["a"]["epochs"] == 2
