@@ -10,5 +10,20 @@ __version__ = "0.1.0"
 from .graphs import CompGraph, load_pipedream_graph
 from .jobs import Job, JobQueue, JobsGenerator
 
+
+def __getattr__(name):  # lazy: keep `import ddls_amd` light
+    if name == "RampClusterEnvironment":
+        from .cluster.environment import RampClusterEnvironment
+        return RampClusterEnvironment
+    if name == "RampJobPartitioningEnvironment":
+        from .envs import RampJobPartitioningEnvironment
+        return RampJobPartitioningEnvironment
+    if name == "Action":
+        from .cluster.actions import Action
+        return Action
+    raise AttributeError(name)
+
+
 __all__ = ["CompGraph", "load_pipedream_graph", "Job", "JobQueue",
-           "JobsGenerator", "__version__"]
+           "JobsGenerator", "RampClusterEnvironment",
+           "RampJobPartitioningEnvironment", "Action", "__version__"]
