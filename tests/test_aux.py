@@ -286,3 +286,29 @@ def test_training_run_loaders(tmp_path):
     by = {r["run"]: r for r in rows}
     assert by["a"]["final_reward"] == 2.0
     assert by["a"]["epochs"] == 2
+
+
+def test_stopwatch_and_seeding():
+    from ddls_amd.utils import Stopwatch, seed_everything
+    sw = Stopwatch()
+    assert sw.time() == 0
+    sw.tick(2.5)
+    sw.tick(1.5)
+    assert sw.time() == 4.0
+    sw.reset()
+    assert sw.time() == 0
+    seed_everything(123)
+    a = np.random.rand(3)
+    seed_everything(123)
+    assert np.array_equal(a, np.random.rand(3))
+
+
+def test_torus_topology_channels():
+    """Torus builds per-direction channels along each ring dimension."""
+    from ddls_amd.topology import build_topology
+    t = build_topology({"type": "torus", "kwargs": {
+        "x_dims": 4, "y_dims": 2, "num_channels": 1,
+        "channel_bandwidth": 1.0e9}})
+    assert t.num_nodes == 8
+    # 2D torus: each node has 4 neighbours (2 per dim), directed channels
+    assert len(t.channel_id_to_channel) > 0
