@@ -44,6 +44,11 @@ class Sampler:
     def sample(self):
         idx = np.random.randint(low=0, high=len(self.sample_indices))
         proto_idx = self.sample_indices[idx]
+        # capture the base BEFORE the mode handling: the pool-exhausting draw
+        # of remove_and_repeat triggers reset(), which advances _base_ids —
+        # the reference bakes ids into sample_pool at reset, so the popped
+        # datum must keep the OLD cycle's id (utils.py:69-105)
+        base = self._base_ids[proto_idx]
         if self.sampling_mode == "replace":
             pass
         elif self.sampling_mode == "remove":
@@ -59,7 +64,7 @@ class Sampler:
         # clone is equivalent to deepcopy at a fraction of the cost
         job = proto.clone() if hasattr(proto, "clone") else copy.deepcopy(proto)
         if self.automatically_change_ids:
-            job.job_id = int(self._base_ids[proto_idx] + proto.job_id)
+            job.job_id = int(base + proto.job_id)
         return job
 
     def __len__(self):

@@ -142,3 +142,25 @@ def test_block_shapes_fit_ramp(num):
             assert 0 <= ci < ramp_shape[0]
             assert 0 <= ri < ramp_shape[1]
             assert 0 <= si < ramp_shape[2]
+
+
+@settings(max_examples=10, deadline=None)
+@given(n=st.integers(3, 25), cycles=st.integers(2, 4))
+def test_sampler_ids_unique_across_repeat_cycles(n, cycles):
+    """remove_and_repeat must never reissue a job id across pool refills
+    (reference Sampler.reset re-bases ids by pool_len * reset_counter,
+    utils.py:93-105) — including the cycle-boundary draw."""
+    from ddls_amd.utils.misc import Sampler
+
+    class P:
+        def __init__(self, i):
+            self.job_id = i
+
+        def clone(self):
+            return P(self.job_id)
+
+    np.random.seed(1)
+    s = Sampler([P(i) for i in range(n)], sampling_mode="remove_and_repeat")
+    ids = [s.sample().job_id for _ in range(n * cycles)]
+    assert len(set(ids)) == len(ids)
+    assert sorted(ids) == list(range(n * cycles))
