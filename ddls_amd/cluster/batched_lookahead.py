@@ -171,6 +171,14 @@ def precompute_lookahead_memos(env, device=None, degrees: Optional[List[int]] = 
     Valid under the reference's own memo assumption
     (ramp_cluster_environment.py:271,277): lookahead results depend only on
     (model, max partition degree).
+
+    Caveat (inherited reference quirk): the true JCT also depends on the
+    placement GEOMETRY (rack/cg-spanning collectives), which the memo key
+    ignores.  Without preloading, the memo stores whatever geometry the
+    FIRST (model, degree) encounter got — possibly on an occupied cluster;
+    preloading pins the canonical empty-cluster geometry instead.  The two
+    streams can therefore differ on busy clusters, each self-consistent
+    under the reference's memoisation rule.
     """
     from ..agents.partitioners import sip_ml_num_partitions
     from .actions import OpPartition
