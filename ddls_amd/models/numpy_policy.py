@@ -89,7 +89,14 @@ class NumpyGNNPolicy:
 
     def __init__(self, state_dict: Dict[str, np.ndarray], config: Dict,
                  num_actions: int):
+        # NB: np.asarray must be a VIEW for the shared-memory weight path
+        # (subproc_env fork-shared buffer) to see live weight updates; a
+        # non-fp32 or non-contiguous param would silently copy,
+        # so assert aliasing and let subproc_env fall back to pickling.
         p = {k: np.asarray(v, dtype=np.float32) for k, v in state_dict.items()}
+        self.aliases_inputs = all(
+            isinstance(v, np.ndarray) and (p[k] is v or p[k].base is v)
+            for k, v in state_dict.items())
         self.config = config
         self.num_actions = num_actions
         n_layers = config["num_rounds"]

@@ -70,6 +70,11 @@ def _worker_loop(conn, env_fn_count: int, env_fn, base_seed: int,
                     if policy is None:
                         policy = NumpyGNNPolicy(shm_views, policy_cfg,
                                                 num_actions)
+                        # the zero-copy contract must hold or rollout weights
+                        # would freeze at first-build values
+                        assert policy.aliases_inputs, (
+                            "shared-memory policy views were copied, not "
+                            "aliased — non-fp32/non-contiguous param?")
                 else:
                     policy = NumpyGNNPolicy(state_dict, policy_cfg,
                                             num_actions)
@@ -301,6 +306,12 @@ class SubprocVectorEnv:
         policy does not fit."""
         meta, off = [], 0
         sd = {k: v.detach().cpu().numpy() for k, v in policy.state_dict().items()}
+        if any(v.dtype != np.float32 for v in sd.values()):
+            # a non-fp32 param would make NumpyGNNPolicy copy (not alias) the
+            # shm view, silently freezing worker weights — fall back to
+            # pickled state_dicts (ADVICE r01)
+            self._shm_buf = None
+            return
         for k, v in sd.items():
             off = (off + 63) & ~63   # 64 B alignment per tensor
             meta.append((k, off, v.nbytes, tuple(v.shape), v.dtype.str))

@@ -18,9 +18,21 @@ class Checkpointer:
         self.checkpoint_dir = os.path.join(path_to_save, "checkpoints")
         os.makedirs(self.checkpoint_dir, exist_ok=True)
 
+    def _indices(self):
+        """Existing checkpoint indices, ignoring stray files."""
+        out = []
+        for e in os.listdir(self.checkpoint_dir):
+            if e.startswith("checkpoint_"):
+                try:
+                    out.append(int(e.split("_")[-1]))
+                except ValueError:
+                    pass
+        return sorted(out)
+
     def write(self, state: dict, index: Optional[int] = None) -> str:
         if index is None:
-            index = len(os.listdir(self.checkpoint_dir)) + 1
+            existing = self._indices()
+            index = (existing[-1] + 1) if existing else 1
         d = os.path.join(self.checkpoint_dir, f"checkpoint_{index:06d}")
         os.makedirs(d, exist_ok=True)
         path = os.path.join(d, f"checkpoint-{index}")
@@ -39,12 +51,12 @@ class Checkpointer:
             return _CpuUnpickler(f).load()
 
     def latest(self) -> Optional[str]:
-        entries = sorted(os.listdir(self.checkpoint_dir))
-        if not entries:
+        existing = self._indices()
+        if not existing:
             return None
-        last = entries[-1]
-        index = int(last.split("_")[-1])
-        return os.path.join(self.checkpoint_dir, last, f"checkpoint-{index}")
+        index = existing[-1]
+        return os.path.join(self.checkpoint_dir, f"checkpoint_{index:06d}",
+                            f"checkpoint-{index}")
 
 
 class _CpuUnpickler(pickle.Unpickler):
