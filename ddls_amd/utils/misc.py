@@ -41,7 +41,11 @@ class Sampler:
         self.reset_counter = 0
         self.reset()
 
-    def sample(self):
+    def _draw(self):
+        """One RNG draw + index bookkeeping; returns (proto_idx, base_id).
+        Shared by ``sample`` (clones the datum) and ``sample_ref`` (no clone:
+        the vectorised env engine drains whole-episode job schedules without
+        paying per-draw Job construction)."""
         idx = np.random.randint(low=0, high=len(self.sample_indices))
         proto_idx = self.sample_indices[idx]
         # capture the base BEFORE the mode handling: the pool-exhausting draw
@@ -59,6 +63,16 @@ class Sampler:
                 self.reset()
         else:
             raise ValueError(f"Unrecognised sampling_mode {self.sampling_mode}")
+        return proto_idx, base
+
+    def sample_ref(self):
+        """RNG-identical to ``sample`` but returns the pool PROTOTYPE without
+        cloning (read-only use)."""
+        proto_idx, _base = self._draw()
+        return self.original_pool[proto_idx]
+
+    def sample(self):
+        proto_idx, base = self._draw()
         proto = self.original_pool[proto_idx]
         # pool prototypes are pristine (never run), so the fast field-level
         # clone is equivalent to deepcopy at a fraction of the cost
