@@ -624,30 +624,41 @@ class EngineState:
         k = self.queued[b]
         if k < 0:
             return
-        mid = int(sched.model_id[k])
-        ms = spec.models[mid]
-        gf = ms.gf_static.copy()
-        acc = sched.max_acceptable[k]
-        frac = sched.frac[k]
-        lo, hi = sched.p_min_acc, sched.p_max_acc
-        gf[3] = (acc - lo) / (hi - lo) if hi - lo != 0 else 1.0
-        lo, hi = sched.p_min_frac, sched.p_max_frac
-        gf[4] = (frac - lo) / (hi - lo) if hi - lo != 0 else 1.0
-        gf[5] = frac
-        gf[15] = self.snapshot[b] / spec.W
-        gf[16] = self.n_running[b] / spec.W
-        f32 = gf.astype(np.float32)
-        f32[f32 < 0] += spec.eps   # numpy f32 += f64 scalar (obs fn order)
-        self.obs_gf[b] = f32
-        num_avail = spec.W - int(self.snapshot[b])
-        mask = np.zeros(spec.A, dtype=np.float32)
-        mask[0] = 1.0
-        for a in range(1, spec.A):
-            if spec.static_shape_ok[a] and a <= num_avail:
-                mask[a] = 1.0
+        gf, mask, mid = compute_obs(spec, sched, int(k),
+                                    int(self.snapshot[b]),
+                                    int(self.n_running[b]))
+        self.obs_gf[b] = gf
         self.obs_mask[b] = mask
         self.obs_model[b] = mid
         self.obs_sched[b] = k
+
+
+def compute_obs(spec: EngineSpec, sched: EpisodeSchedule, k: int,
+                snapshot: int, n_running: int):
+    """Observation encode for queued job k (mirror of
+    ``RampJobPartitioningObservation._graph_features`` dynamic entries +
+    ``get_action_set_and_action_mask``)."""
+    mid = int(sched.model_id[k])
+    ms = spec.models[mid]
+    gf = ms.gf_static.copy()
+    acc = sched.max_acceptable[k]
+    frac = sched.frac[k]
+    lo, hi = sched.p_min_acc, sched.p_max_acc
+    gf[3] = (acc - lo) / (hi - lo) if hi - lo != 0 else 1.0
+    lo, hi = sched.p_min_frac, sched.p_max_frac
+    gf[4] = (frac - lo) / (hi - lo) if hi - lo != 0 else 1.0
+    gf[5] = frac
+    gf[15] = snapshot / spec.W
+    gf[16] = n_running / spec.W
+    f32 = gf.astype(np.float32)
+    f32[f32 < 0] += spec.eps   # numpy f32 += f64 scalar (obs fn order)
+    num_avail = spec.W - snapshot
+    mask = np.zeros(spec.A, dtype=np.float32)
+    mask[0] = 1.0
+    for a in range(1, spec.A):
+        if spec.static_shape_ok[a] and a <= num_avail:
+            mask[a] = 1.0
+    return f32, mask, mid
 
 
 def _block_servers(spec: EngineSpec, Cs: int, Rs: int, Ss: int,

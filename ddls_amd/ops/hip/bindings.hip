@@ -61,6 +61,8 @@ std::vector<torch::Tensor> ppo_loss_bwd(
     torch::Tensor p, torch::Tensor lp, torch::Tensor coef, torch::Tensor h,
     torch::Tensor actions, torch::Tensor values, torch::Tensor vtarg,
     torch::Tensor gl, double vf_clip, double vf_coef, double ent_coef);
+void env_step_batch(std::vector<torch::Tensor> T, std::vector<double> fscal,
+                    std::vector<int64_t> iscal);
 std::vector<torch::Tensor> lookahead_batch(
     torch::Tensor descs, torch::Tensor op_remaining, torch::Tensor op_worker,
     torch::Tensor op_priority, torch::Tensor out_indptr, torch::Tensor out_edges,
@@ -101,4 +103,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("head_fwd", &head_fwd,
           "fused policy/value head forward (LN+graphMLP+concat+2 branches)");
     m.def("head_bwd", &head_bwd, "fused policy/value head backward");
+    m.def("env_step_batch", &env_step_batch,
+          "batched RAMP env step over vectorised envs (placement search + "
+          "memo hash probe + event loop + obs encode)");
 }

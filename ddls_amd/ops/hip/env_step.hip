@@ -1,0 +1,572 @@
+// Batched RAMP env step for gfx950: one workgroup per env, the whole
+// steady-state PAC-ML env step in ONE kernel launch over B vectorised envs
+// (reference hot loop ramp_cluster_environment.py:894-1044
+// + the first-fit block placement search agents/placers/utils.py:532).
+//
+// This kernel is a LINE-FOR-LINE translation of the CPU mirror
+// (ddls_amd/cluster/vec_engine.py::cpu_step_env) and must stay in lockstep
+// with it: the GPU parity test asserts BITWISE f64 equality of state and
+// f32 equality of observations over whole episodes.
+//
+// Memo probes go through an open-addressing hash table in HBM keyed
+// (model_id << 20 | degree) — the device-side analogue of the reference's
+// job_model_to_max_num_partitions_to_lookahead dict
+// (ramp_cluster_environment.py:269-277).  A probe miss aborts the env's
+// step WITHOUT mutating state (status = ST_MISS); the host services the
+// miss (runs the lookahead, inserts) and relaunches for the missed envs.
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include <vector>
+
+// ---- status / log codes (mirror vec_engine.py) ----
+#define LOG_PENDING 0
+#define LOG_RUNNING 1
+#define LOG_COMPLETED 2
+#define LOG_BLOCKED 3
+#define ST_IDLE 0
+#define ST_STEP 1
+#define ST_OK 2
+#define ST_MISS 3
+#define ST_ERR 4
+
+// ---- tensor-list indices (mirror gpu_engine.py TENSOR_LAYOUT) ----
+enum {
+  T_STATIC_OK = 0, T_SHAPE_PTR, T_SHAPES,
+  T_MODEL_GF, T_MODEL_SEQ, T_MODEL_A2MD, T_MODEL_SEQ_LEN,
+  T_OP_MEM, T_PAR_PTR, T_PAR_IDX,
+  T_MD_SPLITS, T_MD_PJSEQ, T_MD_DEGREE, T_MD_MODEL,
+  T_HASH_KEYS, T_HASH_VALS,
+  T_SCH_MODEL, T_SCH_FRAC, T_SCH_ACC, T_SCH_NOMINAL, T_SCH_N, T_SCH_PARAMS,
+  T_T, T_NEXT_ARRIVE, T_ARR_PTR, T_QUEUED, T_N_RUNNING,
+  T_SLOT_MD, T_SLOT_SCHED, T_SLOT_START, T_SLOT_JCT, T_SLOT_OCC, T_OCC,
+  T_SNAPSHOT, T_EP_RETURN, T_EP_LEN, T_DONE, T_STATUS,
+  T_LOG_STATUS, T_LOG_MD, T_LOG_T_ARR, T_LOG_T_END, T_LOG_ORDER,
+  T_ORDER_COUNTER,
+  T_OBS_MODEL, T_OBS_SCHED, T_OBS_GF, T_OBS_MASK,
+  T_REWARD, T_STEP_DONE, T_ACTIONS,
+  T_COUNT
+};
+
+enum {  // iscal indices
+  I_B = 0, I_C, I_R, I_S, I_W, I_WW, I_A, I_K, I_SEQ_CAP, I_PAR_CAP,
+  I_SCH, I_NJOBS, I_HS, I_INFINITE, I_R_INVERSE, I_R_LOG, I_R_NORMALISER,
+  I_R_FAIL_IS_SEQ, I_COUNT
+};
+enum {  // fscal indices
+  F_EPS = 0, F_MAX_SIM, F_MEM_CAP, F_R_SIGN, F_R_FAIL_FACTOR, F_R_FAIL_CONST,
+  F_R_ACC_SUCCESS, F_R_ACC_FAIL, F_R_JCT_W, F_R_BLK_W, F_COUNT
+};
+
+struct EnvPtrs {
+  // spec
+  const uint8_t* static_ok;       // [A]
+  const int* shape_ptr;           // [Amax+1]
+  const int* shapes;              // [NS,3]
+  const double* model_gf;         // [M,17]
+  const double* model_seq;        // [M]
+  const int* model_a2md;          // [M,A]
+  const int* model_seq_len;       // [M]
+  const double* op_mem;           // [M,SEQ_CAP]
+  const int* par_ptr;             // [M,SEQ_CAP+1]
+  const int* par_idx;             // [M,PAR_CAP]
+  const int* md_splits;           // [D,SEQ_CAP]
+  const double* md_pjseq;         // [D]
+  const int* md_degree;           // [D]
+  const int* md_model;            // [D]
+  const long* hash_keys;          // [HS]
+  const double* hash_vals;        // [HS,4]
+  // schedules
+  const int* sch_model;           // [B,SCH]
+  const double* sch_frac;         // [B,SCH]
+  const double* sch_acc;          // [B,SCH]
+  const double* sch_nominal;      // [B,SCH+1]
+  const int* sch_n;               // [B]
+  const double* sch_params;       // [B,4]
+  // state
+  double* t; double* next_arrive; int* arr_ptr; int* queued; int* n_running;
+  int* slot_md; int* slot_sched; double* slot_start; double* slot_jct;
+  unsigned long long* slot_occ; unsigned long long* occ;
+  int* snapshot; double* ep_return; int* ep_len; uint8_t* done; int* status;
+  uint8_t* log_status; int* log_md; double* log_t_arr; double* log_t_end;
+  int* log_order; int* order_counter;
+  int* obs_model; int* obs_sched; float* obs_gf; float* obs_mask;
+  double* reward; uint8_t* step_done;
+  const int* actions;
+};
+
+struct EnvDims {
+  int B, C, R, S, W, WW, A, K, SEQ_CAP, PAR_CAP, SCH, NJOBS, HS;
+  int infinite_pool, r_inverse, r_log, r_normaliser, r_fail_is_seq;
+  double eps, max_sim, mem_cap, r_sign, r_fail_factor, r_fail_const,
+         r_acc_success, r_acc_fail, r_jct_w, r_blk_w;
+};
+
+// ---- placement search (mirror of _search_placement) ----
+// Runs on thread 0; LDS scratch: free_mem[W] + op_servers[SEQ_CAP][MAXS]
+// + op_count[SEQ_CAP].  MAXS = A-1 (max split degree).
+
+__device__ bool occ_get(const unsigned long long* occ_row, int s) {
+  return (occ_row[s >> 6] >> (unsigned)(s & 63)) & 1ull;
+}
+
+// get_block mirror: writes block server ids; returns length or -1 (invalid
+// diagonal wrap, the reference's %(shape+1) dict-miss quirk)
+__device__ int block_servers(const EnvDims& D, int Cs, int Rs, int Ss,
+                             int oi, int oj, int ok, short* out) {
+  int n_out = 0;
+  if (Ss == -1) {
+    for (int n = 0; n < Cs; ++n) {
+      int c = (oi + n) % (D.C + 1);
+      int r = (oj + n) % (D.R + 1);
+      int s = ok % D.S;
+      if (c >= D.C || r >= D.R) return -1;
+      out[n_out++] = (short)((c * D.R + r) * D.S + s);
+    }
+  } else {
+    for (int c = 0; c < Cs; ++c)
+      for (int r = 0; r < Rs; ++r)
+        for (int s = 0; s < Ss; ++s) {
+          int cc = (oi + c) % D.C;
+          int rr = (oj + r) % D.R;
+          int ss = (ok + s) % D.S;
+          out[n_out++] = (short)((cc * D.R + rr) * D.S + ss);
+        }
+  }
+  return n_out;
+}
+
+__device__ bool check_block(const EnvDims& D,
+                            const unsigned long long* occ_row,
+                            const double* free_mem, const short* block,
+                            int blen, double op_size) {
+  if (blen <= 0) return false;
+  for (int i = 0; i < blen; ++i) {
+    int s = block[i];
+    if (occ_get(occ_row, s)) return false;
+    if (free_mem[s] < op_size) return false;
+  }
+  return true;
+}
+
+// find_sub_block / ff_block mirror: first fit over shapes x origins
+__device__ int find_sub_block(const EnvPtrs& P, const EnvDims& D,
+                              const unsigned long long* occ_row,
+                              const double* free_mem, int split,
+                              double op_size, short* out) {
+  for (int si = P.shape_ptr[split]; si < P.shape_ptr[split + 1]; ++si) {
+    int Cs = P.shapes[si * 3 + 0];
+    int Rs = P.shapes[si * 3 + 1];
+    int Ss = P.shapes[si * 3 + 2];
+    int I = D.C - Cs + 1;
+    int J = D.R - Rs + 1;
+    int K = D.S - Ss + 1;
+    if (I <= 0 || J <= 0 || K <= 0) continue;
+    for (int i = 0; i < I; ++i)
+      for (int j = 0; j < J; ++j)
+        for (int k = 0; k < K; ++k) {
+          int blen = block_servers(D, Cs, Rs, Ss, i, j, k, out);
+          if (blen > 0 && check_block(D, occ_row, free_mem, out, blen,
+                                      op_size))
+            return blen;
+        }
+  }
+  return -1;
+}
+
+// allocate() mirror; returns number of union servers written to out_union
+// (sorted ascending), or -1 on failure.
+__device__ int search_placement(const EnvPtrs& P, const EnvDims& D,
+                                int mid, int mdi,
+                                const unsigned long long* occ_row,
+                                double* free_mem, short* op_servers,
+                                short* op_count, short* scratch,
+                                short* out_union) {
+  const int seq_len = P.model_seq_len[mid];
+  const int MAXS = D.A - 1;
+  for (int s = 0; s < D.W; ++s) free_mem[s] = D.mem_cap;
+  for (int k = 0; k < seq_len; ++k) op_count[k] = 0;
+
+  for (int k = 0; k < seq_len; ++k) {
+    int split = P.md_splits[(long)mdi * D.SEQ_CAP + k];
+    double req = P.op_mem[(long)mid * D.SEQ_CAP + k];
+    bool placed = false;
+    // parent-collective placement (placement_utils.py:207-238)
+    for (int pi = P.par_ptr[(long)mid * (D.SEQ_CAP + 1) + k];
+         pi < P.par_ptr[(long)mid * (D.SEQ_CAP + 1) + k + 1]; ++pi) {
+      int p = P.par_idx[(long)mid * D.PAR_CAP + pi];
+      if (split != op_count[p]) continue;
+      double avail = 0.0;
+      for (int q = 0; q < op_count[p]; ++q)
+        avail += free_mem[op_servers[p * MAXS + q]];
+      if (avail >= req) {
+        for (int q = 0; q < op_count[p]; ++q) {
+          short s = op_servers[p * MAXS + q];
+          free_mem[s] -= req / split;
+          op_servers[k * MAXS + op_count[k]++] = s;
+        }
+        placed = true;
+        break;
+      }
+    }
+    if (placed) continue;
+    if (split > D.W) return -1;
+    double op_size = req / split;
+    int blen = find_sub_block(P, D, occ_row, free_mem, split, op_size,
+                              scratch);
+    if (blen < 0) return -1;
+    for (int i = 0; i < blen; ++i) {
+      short s = scratch[i];
+      free_mem[s] -= op_size;
+      op_servers[k * MAXS + op_count[k]++] = s;
+    }
+  }
+  // sorted union of all op servers (membership via a W-bit set in free_mem's
+  // space is overkill; W <= 1024 so scan a bitmask on the stack)
+  unsigned long long seen[16];  // up to 1024 servers
+  for (int w = 0; w < D.WW; ++w) seen[w] = 0ull;
+  for (int k = 0; k < seq_len; ++k)
+    for (int q = 0; q < op_count[k]; ++q) {
+      int s = op_servers[k * MAXS + q];
+      seen[s >> 6] |= 1ull << (unsigned)(s & 63);
+    }
+  int n_union = 0;
+  for (int s = 0; s < D.W; ++s)
+    if ((seen[s >> 6] >> (unsigned)(s & 63)) & 1ull)
+      out_union[n_union++] = (short)s;
+  return n_union;
+}
+
+// ---- reward (mirror of _jct_reward) ----
+__device__ double jct_reward(const EnvDims& D, double value, double seq_jct,
+                             bool blocked, double norm_seq) {
+  double reward;
+  if (blocked) {
+    double base = D.r_fail_is_seq ? seq_jct : D.r_fail_const;
+    reward = base * D.r_fail_factor;
+  } else {
+    reward = value;
+  }
+  if (D.r_normaliser != 0 && reward != 0.0) {
+    double den = (D.r_normaliser == 1) ? norm_seq
+                                       : norm_seq * D.r_fail_factor;
+    reward = reward / den;
+  }
+  if (D.r_inverse && reward != 0.0) reward = 1.0 / reward;
+  reward *= D.r_sign;
+  if (D.r_log) {
+    double sgn = copysign(1.0, reward);
+    reward = sgn * (log(1.0 + fabs(reward)) / log(10.0));
+  }
+  return reward;
+}
+
+// ---- observation (mirror of compute_obs) ----
+__device__ void write_obs(const EnvPtrs& P, const EnvDims& D, int b) {
+  int k = P.queued[b];
+  if (k < 0) return;
+  long sb = (long)b * D.SCH;
+  int mid = P.sch_model[sb + k];
+  double gf[17];
+  for (int i = 0; i < 17; ++i) gf[i] = P.model_gf[(long)mid * 17 + i];
+  double acc = P.sch_acc[sb + k];
+  double frac = P.sch_frac[sb + k];
+  double lo = P.sch_params[(long)b * 4 + 0];
+  double hi = P.sch_params[(long)b * 4 + 1];
+  gf[3] = (hi - lo != 0.0) ? (acc - lo) / (hi - lo) : 1.0;
+  lo = P.sch_params[(long)b * 4 + 2];
+  hi = P.sch_params[(long)b * 4 + 3];
+  gf[4] = (hi - lo != 0.0) ? (frac - lo) / (hi - lo) : 1.0;
+  gf[5] = frac;
+  gf[15] = (double)P.snapshot[b] / (double)D.W;
+  gf[16] = (double)P.n_running[b] / (double)D.W;
+  for (int i = 0; i < 17; ++i) {
+    float v = (float)gf[i];
+    if (v < 0.0f) v += (float)D.eps;
+    P.obs_gf[(long)b * 17 + i] = v;
+  }
+  int num_avail = D.W - P.snapshot[b];
+  for (int a = 0; a < D.A; ++a) {
+    float m = 0.0f;
+    if (a == 0) m = 1.0f;
+    else if (P.static_ok[a] && a <= num_avail) m = 1.0f;
+    P.obs_mask[(long)b * D.A + a] = m;
+  }
+  P.obs_model[b] = mid;
+  P.obs_sched[b] = k;
+}
+
+// ---- the step kernel (mirror of cpu_step_env) ----
+__global__ void __launch_bounds__(64)
+env_step_kernel(EnvPtrs P, EnvDims D) {
+  const int b = blockIdx.x;
+  if (b >= D.B) return;
+  if (P.status[b] != ST_STEP) return;
+  if (threadIdx.x != 0) return;   // sequential control flow per env (v1)
+
+  extern __shared__ char smem[];
+  const int MAXS = D.A - 1;
+  double* free_mem = (double*)smem;                       // [W]
+  short* op_servers = (short*)(free_mem + D.W);           // [SEQ_CAP][MAXS]
+  short* op_count = op_servers + (long)D.SEQ_CAP * MAXS;  // [SEQ_CAP]
+  short* scratch = op_count + D.SEQ_CAP;                  // [MAXS]
+  short* out_union = scratch + MAXS;                      // [W]
+
+  if (P.done[b] || P.queued[b] < 0) { P.status[b] = ST_ERR; return; }
+
+  const long sb = (long)b * D.SCH;
+  const long nb = (long)b * D.NJOBS;
+  const int k = P.queued[b];
+  const int mid = P.sch_model[sb + k];
+  const int action = P.actions[b];
+  const double seq_jct = P.model_seq[mid];
+  bool placed = false, blocked = false;
+  double jct_total = 0.0;
+  int mdi = -1;
+
+  if (action != 0) {
+    mdi = P.model_a2md[(long)mid * D.A + action];
+    int n_union = -1;
+    if (mdi >= 0)
+      n_union = search_placement(P, D, mid, mdi, P.occ + (long)b * D.WW,
+                                 free_mem, op_servers, op_count, scratch,
+                                 out_union);
+    if (n_union >= 0) {
+      // memo probe (K4 open-addressing table in HBM)
+      long key = ((long)mid << 20) | (long)P.md_degree[mdi];
+      long slot = (long)(((unsigned long long)key * 0x9E3779B97F4A7C15ull
+                          >> 40) & (unsigned long long)(D.HS - 1));
+      const double* val = nullptr;
+      for (int probe = 0; probe < D.HS; ++probe) {
+        long kk = P.hash_keys[slot];
+        if (kk == key) { val = P.hash_vals + slot * 4; break; }
+        if (kk == -1) break;
+        slot = (slot + 1) & (D.HS - 1);
+      }
+      if (val == nullptr) { P.status[b] = ST_MISS; return; }
+      jct_total = val[0];
+      if (jct_total > P.sch_acc[sb + k]) {
+        blocked = true;               // lookahead JCT contract violation
+      } else {
+        placed = true;
+        int slot_i = P.n_running[b];
+        long so = ((long)b * D.K + slot_i) * D.WW;
+        for (int w = 0; w < D.WW; ++w) P.slot_occ[so + w] = 0ull;
+        for (int i = 0; i < n_union; ++i) {
+          int s = out_union[i];
+          unsigned long long bit = 1ull << (unsigned)(s & 63);
+          P.occ[(long)b * D.WW + (s >> 6)] |= bit;
+          P.slot_occ[so + (s >> 6)] |= bit;
+        }
+        long si = (long)b * D.K + slot_i;
+        P.slot_md[si] = mdi;
+        P.slot_sched[si] = k;
+        P.slot_start[si] = P.t[b];
+        P.slot_jct[si] = jct_total;
+        P.n_running[b] += 1;
+        P.log_status[nb + k] = LOG_RUNNING;
+        P.log_md[nb + k] = mdi;
+      }
+    } else {
+      blocked = true;                 // no feasible block placement
+    }
+    if (blocked) {
+      P.log_status[nb + k] = LOG_BLOCKED;
+      // lookahead-blocked (placement found, JCT contract violated) keeps its
+      // md for debugging; placement-fail logs -1.  Blocked STATS always use
+      // the original job either way (mirror parity).
+      P.log_md[nb + k] = (n_union >= 0) ? mdi : -1;
+      P.log_t_end[nb + k] = P.t[b];
+      P.log_order[nb + k] = P.order_counter[b]++;
+    }
+  } else {
+    blocked = true;                   // action 0: do not place
+    P.log_status[nb + k] = LOG_BLOCKED;
+    P.log_t_end[nb + k] = P.t[b];
+    P.log_order[nb + k] = P.order_counter[b]++;
+  }
+  P.queued[b] = -1;
+
+  // reward
+  double reward = 0.0;
+  if (D.r_jct_w != 0.0) {
+    double norm_seq = placed ? P.md_pjseq[mdi] : seq_jct;
+    reward += D.r_jct_w * jct_reward(D, jct_total, seq_jct, !placed,
+                                     norm_seq);
+  }
+  if (D.r_blk_w != 0.0)
+    reward += D.r_blk_w * (placed ? D.r_acc_success : D.r_acc_fail);
+  P.reward[b] = reward;
+  P.ep_return[b] += reward;
+  P.ep_len[b] += 1;
+
+  // ---- outer event loop (+ idle fast-forward) ----
+  bool done = false;
+  while (true) {
+    bool pool_empty = (!D.infinite_pool) && isinf(P.next_arrive[b]);
+    double tick = fmin(P.next_arrive[b] - P.t[b], D.max_sim - P.t[b]);
+    int nr = P.n_running[b];
+    for (int s = 0; s < nr; ++s) {
+      long si = (long)b * D.K + s;
+      double elapsed = P.t[b] - P.slot_start[si];
+      double remaining = P.slot_jct[si] - elapsed;
+      tick = fmin(tick, remaining);
+    }
+    int pc = 0;
+    for (int w = 0; w < D.WW; ++w)
+      pc += __popcll(P.occ[(long)b * D.WW + w]);
+    P.snapshot[b] = pc;
+    P.t[b] = P.t[b] + tick;
+    // completions (order-preserving compaction)
+    int keep = 0;
+    for (int s = 0; s < nr; ++s) {
+      long si = (long)b * D.K + s;
+      double elapsed = P.t[b] - P.slot_start[si];
+      double remaining = (P.slot_jct[si] - elapsed) - D.eps;
+      if (remaining <= 0.0) {
+        int kk = P.slot_sched[si];
+        P.log_status[nb + kk] = LOG_COMPLETED;
+        P.log_t_end[nb + kk] = P.t[b];
+        P.log_order[nb + kk] = P.order_counter[b]++;
+        long so = si * D.WW;
+        for (int w = 0; w < D.WW; ++w)
+          P.occ[(long)b * D.WW + w] &= ~P.slot_occ[so + w];
+      } else {
+        if (keep != s) {
+          long di = (long)b * D.K + keep;
+          P.slot_md[di] = P.slot_md[si];
+          P.slot_sched[di] = P.slot_sched[si];
+          P.slot_start[di] = P.slot_start[si];
+          P.slot_jct[di] = P.slot_jct[si];
+          for (int w = 0; w < D.WW; ++w)
+            P.slot_occ[di * D.WW + w] = P.slot_occ[si * D.WW + w];
+        }
+        keep++;
+      }
+    }
+    P.n_running[b] = keep;
+    // arrival
+    if (!pool_empty) {
+      if ((P.t[b] + D.eps) >= P.next_arrive[b]) {
+        int kk = P.arr_ptr[b];
+        P.log_t_arr[nb + kk] = P.t[b];
+        P.queued[b] = kk;
+        P.arr_ptr[b] += 1;
+        P.next_arrive[b] = P.sch_nominal[(long)b * (D.SCH + 1)
+                                         + P.arr_ptr[b]];
+      }
+    }
+    pool_empty = (!D.infinite_pool) && isinf(P.next_arrive[b]);
+    done = (P.t[b] >= D.max_sim) ||
+           (pool_empty && P.n_running[b] == 0 && P.queued[b] < 0);
+    if (P.queued[b] >= 0 || done) break;
+  }
+
+  if (done) {
+    // finalise: still-running jobs become blocked (reference :1111-1121)
+    int nr = P.n_running[b];
+    for (int s = 0; s < nr; ++s) {
+      long si = (long)b * D.K + s;
+      int kk = P.slot_sched[si];
+      P.log_status[nb + kk] = LOG_BLOCKED;
+      P.log_t_end[nb + kk] = P.t[b];
+      P.log_order[nb + kk] = P.order_counter[b]++;
+    }
+    P.done[b] = 1;
+    P.step_done[b] = 1;
+    P.status[b] = ST_OK;
+    return;
+  }
+  P.step_done[b] = 0;
+  write_obs(P, D, b);
+  P.status[b] = ST_OK;
+}
+
+// ---------------------------------------------------------------------------
+// host binding
+// ---------------------------------------------------------------------------
+
+void env_step_batch(std::vector<torch::Tensor> T,
+                    std::vector<double> fscal,
+                    std::vector<int64_t> iscal) {
+  TORCH_CHECK((int)T.size() == T_COUNT, "env_step_batch: tensor list size");
+  TORCH_CHECK((int)iscal.size() == I_COUNT && (int)fscal.size() == F_COUNT);
+  EnvDims D;
+  D.B = (int)iscal[I_B]; D.C = (int)iscal[I_C]; D.R = (int)iscal[I_R];
+  D.S = (int)iscal[I_S]; D.W = (int)iscal[I_W]; D.WW = (int)iscal[I_WW];
+  D.A = (int)iscal[I_A]; D.K = (int)iscal[I_K];
+  D.SEQ_CAP = (int)iscal[I_SEQ_CAP]; D.PAR_CAP = (int)iscal[I_PAR_CAP];
+  D.SCH = (int)iscal[I_SCH]; D.NJOBS = (int)iscal[I_NJOBS];
+  D.HS = (int)iscal[I_HS]; D.infinite_pool = (int)iscal[I_INFINITE];
+  D.r_inverse = (int)iscal[I_R_INVERSE]; D.r_log = (int)iscal[I_R_LOG];
+  D.r_normaliser = (int)iscal[I_R_NORMALISER];
+  D.r_fail_is_seq = (int)iscal[I_R_FAIL_IS_SEQ];
+  D.eps = fscal[F_EPS]; D.max_sim = fscal[F_MAX_SIM];
+  D.mem_cap = fscal[F_MEM_CAP]; D.r_sign = fscal[F_R_SIGN];
+  D.r_fail_factor = fscal[F_R_FAIL_FACTOR];
+  D.r_fail_const = fscal[F_R_FAIL_CONST];
+  D.r_acc_success = fscal[F_R_ACC_SUCCESS];
+  D.r_acc_fail = fscal[F_R_ACC_FAIL];
+  D.r_jct_w = fscal[F_R_JCT_W]; D.r_blk_w = fscal[F_R_BLK_W];
+  TORCH_CHECK(D.WW <= 16, "env_step: W > 1024 unsupported");
+
+  EnvPtrs P;
+  P.static_ok = T[T_STATIC_OK].data_ptr<uint8_t>();
+  P.shape_ptr = T[T_SHAPE_PTR].data_ptr<int>();
+  P.shapes = T[T_SHAPES].data_ptr<int>();
+  P.model_gf = T[T_MODEL_GF].data_ptr<double>();
+  P.model_seq = T[T_MODEL_SEQ].data_ptr<double>();
+  P.model_a2md = T[T_MODEL_A2MD].data_ptr<int>();
+  P.model_seq_len = T[T_MODEL_SEQ_LEN].data_ptr<int>();
+  P.op_mem = T[T_OP_MEM].data_ptr<double>();
+  P.par_ptr = T[T_PAR_PTR].data_ptr<int>();
+  P.par_idx = T[T_PAR_IDX].data_ptr<int>();
+  P.md_splits = T[T_MD_SPLITS].data_ptr<int>();
+  P.md_pjseq = T[T_MD_PJSEQ].data_ptr<double>();
+  P.md_degree = T[T_MD_DEGREE].data_ptr<int>();
+  P.md_model = T[T_MD_MODEL].data_ptr<int>();
+  P.hash_keys = T[T_HASH_KEYS].data_ptr<long>();
+  P.hash_vals = T[T_HASH_VALS].data_ptr<double>();
+  P.sch_model = T[T_SCH_MODEL].data_ptr<int>();
+  P.sch_frac = T[T_SCH_FRAC].data_ptr<double>();
+  P.sch_acc = T[T_SCH_ACC].data_ptr<double>();
+  P.sch_nominal = T[T_SCH_NOMINAL].data_ptr<double>();
+  P.sch_n = T[T_SCH_N].data_ptr<int>();
+  P.sch_params = T[T_SCH_PARAMS].data_ptr<double>();
+  P.t = T[T_T].data_ptr<double>();
+  P.next_arrive = T[T_NEXT_ARRIVE].data_ptr<double>();
+  P.arr_ptr = T[T_ARR_PTR].data_ptr<int>();
+  P.queued = T[T_QUEUED].data_ptr<int>();
+  P.n_running = T[T_N_RUNNING].data_ptr<int>();
+  P.slot_md = T[T_SLOT_MD].data_ptr<int>();
+  P.slot_sched = T[T_SLOT_SCHED].data_ptr<int>();
+  P.slot_start = T[T_SLOT_START].data_ptr<double>();
+  P.slot_jct = T[T_SLOT_JCT].data_ptr<double>();
+  P.slot_occ = (unsigned long long*)T[T_SLOT_OCC].data_ptr<int64_t>();
+  P.occ = (unsigned long long*)T[T_OCC].data_ptr<int64_t>();
+  P.snapshot = T[T_SNAPSHOT].data_ptr<int>();
+  P.ep_return = T[T_EP_RETURN].data_ptr<double>();
+  P.ep_len = T[T_EP_LEN].data_ptr<int>();
+  P.done = T[T_DONE].data_ptr<uint8_t>();
+  P.status = T[T_STATUS].data_ptr<int>();
+  P.log_status = T[T_LOG_STATUS].data_ptr<uint8_t>();
+  P.log_md = T[T_LOG_MD].data_ptr<int>();
+  P.log_t_arr = T[T_LOG_T_ARR].data_ptr<double>();
+  P.log_t_end = T[T_LOG_T_END].data_ptr<double>();
+  P.log_order = T[T_LOG_ORDER].data_ptr<int>();
+  P.order_counter = T[T_ORDER_COUNTER].data_ptr<int>();
+  P.obs_model = T[T_OBS_MODEL].data_ptr<int>();
+  P.obs_sched = T[T_OBS_SCHED].data_ptr<int>();
+  P.obs_gf = T[T_OBS_GF].data_ptr<float>();
+  P.obs_mask = T[T_OBS_MASK].data_ptr<float>();
+  P.reward = T[T_REWARD].data_ptr<double>();
+  P.step_done = T[T_STEP_DONE].data_ptr<uint8_t>();
+  P.actions = T[T_ACTIONS].data_ptr<int>();
+
+  const int MAXS = D.A - 1;
+  size_t shmem = (size_t)D.W * sizeof(double)
+      + ((size_t)D.SEQ_CAP * MAXS + D.SEQ_CAP + MAXS + D.W) * sizeof(short);
+  hipStream_t stream = at::cuda::getCurrentCUDAStream();
+  hipLaunchKernelGGL(env_step_kernel, dim3(D.B), dim3(64), shmem, stream,
+                     P, D);
+}
