@@ -1,14 +1,21 @@
 #!/usr/bin/env python3
 """Flagship benchmark: PPO training of the PAC-ML GNN partitioner on the
 32-worker RAMP cluster simulator (BASELINE.json metric: PPO env-steps/sec,
-whole node).
+whole node, + mean simulated JCT).
 
-One process per GPU (torchrun for N>1), vectorised envs per rank, fused
-gradient all-reduce with RCCL over xGMI.  A "step" is one PPO iteration:
-collect a fixed per-rank rollout (weak scaling) + the full SGD update.
+One process per GPU (torchrun for N>1), fused gradient all-reduce with RCCL
+over xGMI.  A "step" is one PPO iteration at the reference's tuned operating
+point (algo/ppo.yaml:26-54 + BASELINE configs[2]): 256 vectorised envs/rank,
+16 rollout steps -> train batch 4096/rank, 50 SGD epochs x minibatch 128.
 
-Synthetic workload: deterministic pipedream-format job graphs (the reference's
-"small_graphs" profile set is not distributed); random-init GNN weights.
+On GPU the envs are GPU-resident: the whole vectorised env step runs as one
+batched HIP kernel (cluster/gpu_engine.py) and the rollout
+policy uses a per-model GNN embedding cache (rl/engine_env.py).  --cpu-envs
+falls back to the round-1 subprocess-worker path.
+
+Synthetic workload: deterministic pipedream-format job graphs (the
+reference's "small_graphs" profile set is not distributed); random-init GNN
+weights.
 """
 import argparse
 import json
@@ -65,53 +72,71 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=6)
     ap.add_argument("--warmup", type=int, default=3)
-    ap.add_argument("--envs-per-rank", type=int, default=64)
+    # reference-tuned operating point (algo/ppo.yaml + BASELINE configs[2]):
+    # 256 envs x 16 rollout steps = 4096 train batch, 50 SGD epochs, mb 128
+    ap.add_argument("--envs-per-rank", type=int, default=256)
     ap.add_argument("--rollout-steps-per-env", type=int, default=16)
-    ap.add_argument("--env-workers", type=int, default=0,
-                    help="env worker processes per rank (0 = auto)")
-    ap.add_argument("--num-sgd-iter", type=int, default=8)
+    ap.add_argument("--num-sgd-iter", type=int, default=50)
     ap.add_argument("--sgd-minibatch-size", type=int, default=128)
+    ap.add_argument("--cpu-envs", action="store_true",
+                    help="round-1 CPU subprocess env workers instead of the "
+                         "GPU-resident engine")
+    ap.add_argument("--env-workers", type=int, default=0,
+                    help="(cpu-envs) worker processes per rank (0 = auto)")
     ap.add_argument("--no-precompute", action="store_true",
-                    help="skip the batched GPU lookahead memo precompute")
+                    help="(cpu-envs) skip the batched GPU lookahead memo "
+                         "precompute")
     args = ap.parse_args()
 
     from ddls_amd.models.gnn import GNNPolicy
     from ddls_amd.parallel import (get_rank, get_world_size,
                                    init_distributed_from_env, is_distributed)
     from ddls_amd.rl.ppo import PPOConfig, PPOTrainer
-    from ddls_amd.rl.subproc_env import SubprocVectorEnv
 
     rank = init_distributed_from_env()
     world_size = get_world_size()
     use_cuda = torch.cuda.is_available()
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
+    if use_cuda:
+        torch.cuda.set_device(device)
 
     torch.manual_seed(0)  # identical init weights on every rank
     policy = GNNPolicy(num_actions=17)
 
     env_fn = build_env_fn()
-    n_workers = args.env_workers
-    if n_workers <= 0:
-        n_workers = max(1, min(args.envs_per_rank,
-                               (os.cpu_count() or 2) // max(world_size, 1) - 1))
-    # fork env workers BEFORE any HIP context exists in this process
-    venv = SubprocVectorEnv(env_fn, num_envs=args.envs_per_rank,
-                            num_workers=n_workers,
-                            base_seed=1 + 100000 * rank)
-    if not args.no_precompute:
-        # precompute the whole (model x degree) lookahead memo table in one
-        # batched HIP kernel launch; ship the tables to the live workers
-        from ddls_amd.cluster.batched_lookahead import precompute_lookahead_memos
-        scratch = env_fn()
-        scratch.reset(seed=0)
-        memo_l, memo_i = precompute_lookahead_memos(
-            scratch, device=device if use_cuda else "cpu")
-        del scratch
-        venv.preload_memos(memo_l, memo_i)
-    if hasattr(venv, "warm"):
-        # pre-warm the worker-side pipeline caches (untimed setup)
-        venv.warm(96)
+    use_engine = use_cuda and not args.cpu_envs
+    jct_fn = None
+    if use_engine:
+        from ddls_amd.rl.engine_env import EngineVectorEnv
+        policy = policy.to(device)
+        venv = EngineVectorEnv(env_fn, num_envs=args.envs_per_rank,
+                               device=device,
+                               base_seed=1 + 100000 * rank, verbose=rank == 0)
+        jct_fn = venv.jct_running_stats
+    else:
+        from ddls_amd.rl.subproc_env import SubprocVectorEnv
+        n_workers = args.env_workers
+        if n_workers <= 0:
+            n_workers = max(1, min(args.envs_per_rank,
+                                   (os.cpu_count() or 2)
+                                   // max(world_size, 1) - 1))
+        # fork env workers BEFORE any HIP context exists in this process
+        venv = SubprocVectorEnv(env_fn, num_envs=args.envs_per_rank,
+                                num_workers=n_workers,
+                                base_seed=1 + 100000 * rank)
+        if not args.no_precompute:
+            from ddls_amd.cluster.batched_lookahead import \
+                precompute_lookahead_memos
+            scratch = env_fn()
+            scratch.reset(seed=0)
+            memo_l, memo_i = precompute_lookahead_memos(
+                scratch, device=device if use_cuda else "cpu")
+            del scratch
+            venv.preload_memos(memo_l, memo_i)
+        if hasattr(venv, "warm"):
+            venv.warm(96)
+
     per_step_env_steps = args.envs_per_rank * args.rollout_steps_per_env
     cfg = PPOConfig(train_batch_size=per_step_env_steps,
                     sgd_minibatch_size=args.sgd_minibatch_size,
@@ -128,6 +153,7 @@ def main():
         trainer.train(num_steps=args.rollout_steps_per_env)
 
     jct_samples = []
+    jct0 = jct_fn() if jct_fn else None
     barrier_sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
@@ -141,6 +167,7 @@ def main():
                   f"captures={st.get('hipgraph_captures', 0)}",
                   file=sys.stderr, flush=True)
     barrier_sync()
+    elapsed = time.perf_counter() - t0
     if rank == 0 and getattr(trainer, "_stepper", None) is not None:
         sp = trainer._stepper
         print(f"[bench] hipgraph: broken={sp.broken} captures={sp.capture_count} "
@@ -148,7 +175,17 @@ def main():
               f"fused_loss={getattr(sp, '_fused_loss', False)} "
               f"repairs={getattr(sp, 'repair_count', None)} "
               f"err={sp.last_error}", file=sys.stderr, flush=True)
-    elapsed = time.perf_counter() - t0
+
+    # mean simulated JCT over jobs completed INSIDE the timed window
+    # (BASELINE metric second half); fall back to full-episode means
+    mean_jct = None
+    if jct_fn is not None:
+        c1, s1 = jct_fn()
+        c0, s0 = jct0
+        if c1 > c0:
+            mean_jct = (s1 - s0) / (c1 - c0)
+    if mean_jct is None and jct_samples:
+        mean_jct = float(np.mean(jct_samples))
 
     # max over ranks
     if is_distributed():
@@ -183,14 +220,10 @@ def main():
                 "parallelism": f"dp{world_size}",
                 "ramp": "4x4x2_32workers",
                 "envs_per_rank": args.envs_per_rank,
-                "env_workers": n_workers,
+                "env_engine": "gpu_resident" if use_engine else "cpu_subproc",
                 "num_sgd_iter": args.num_sgd_iter,
-                # second half of the BASELINE metric ("mean simulated JCT"):
-                # mean over episodes COMPLETED inside the timed window (null
-                # on short runs — episodes are 1000 jobs long); the trained-
-                # policy JCT numbers live in docs/RESULTS.md
-                "mean_simulated_jct": (float(np.mean(jct_samples))
-                                       if jct_samples else None),
+                "sgd_minibatch_size": args.sgd_minibatch_size,
+                "mean_simulated_jct": mean_jct,
             },
         }))
     if venv_close is not None:
