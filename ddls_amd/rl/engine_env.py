@@ -105,7 +105,8 @@ class EngineVectorEnv:
                           edge_features=self._model_ef[mid],
                           edges_src=self._model_src[mid],
                           edges_dst=self._model_dst[mid],
-                          graph_features=gfull, action_mask=mask)
+                          graph_features=gfull, action_mask=mask,
+                          model_id=mid)
 
     def reset(self) -> List[CompactObs]:
         return self._obs_list()

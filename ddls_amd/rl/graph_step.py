@@ -79,11 +79,21 @@ class CapturedSGDStep:
     # ghn atomics
     RESERVE = 256
 
-    def __init__(self, policy, optimizer, config, device):
+    def __init__(self, policy, optimizer, config, device, models_batch=None):
         self.policy = policy
         self.optimizer = optimizer
         self.cfg = config
         self.device = device
+        # cached-models mode: every sample's node/edge features are static
+        # per workload model (vectorised-engine obs), so the GNN runs ONCE on
+        # the M model graphs per minibatch and per-sample gradients reach it
+        # summed through index_select's backward — exact by linearity, ~25x
+        # less GNN work, and every captured shape is static (no capacity
+        # management).
+        self.models_batch = models_batch
+        if models_batch is not None:
+            models_batch.csr_by_dst()   # materialise before capture
+            models_batch.node_ptr()
         self.B = config.sgd_minibatch_size
         self.A = policy.num_actions
         self.Fn = policy.config["in_features_node"]
@@ -133,6 +143,10 @@ class CapturedSGDStep:
         the minibatch loop never recaptures."""
         if self.broken:
             return False
+        if self.models_batch is not None:
+            if self.graph is not None:
+                return True
+            n_cap = e_cap = 0       # fixed-shape staging, single capture
         if (self.graph is not None and n_cap <= self.n_cap - self.RESERVE
                 and e_cap <= self.e_cap):
             return True
@@ -163,10 +177,14 @@ class CapturedSGDStep:
              vtarg: np.ndarray) -> bool:
         if self.broken or len(mb_obs) != self.B:
             return False
-        n = sum(len(o.node_features) for o in mb_obs)
-        e = sum(len(o.edges_src) for o in mb_obs)
-        if not self.ensure_capacity(n, e) or n > self.n_cap - self.RESERVE:
-            return False
+        if self.models_batch is not None:
+            if not self.ensure_capacity(0, 0):
+                return False
+        else:
+            n = sum(len(o.node_features) for o in mb_obs)
+            e = sum(len(o.edges_src) for o in mb_obs)
+            if not self.ensure_capacity(n, e) or n > self.n_cap - self.RESERVE:
+                return False
         self.commit(self.prepare(mb_obs, actions, old_logp, adv, vtarg))
         return True
 
@@ -200,6 +218,29 @@ class CapturedSGDStep:
         def pbuf(shape, dtype):
             return torch.zeros(shape, dtype=dtype, pin_memory=True)
 
+        if self.models_batch is not None:
+            # cached-models mode: tiny fixed-shape staging, no dummy graph
+            self.n_cap, self.e_cap = 0, 0
+            self.d = {
+                "model_ids": dbuf(B, torch.int64),
+                "gf": dbuf((B, self.Fg), torch.float32),
+                "mask": torch.ones((B, self.A), dtype=torch.float32,
+                                   device=dev),
+                "actions": dbuf(B, torch.int64),
+                "old_logp": dbuf(B, torch.float32),
+                "adv": dbuf(B, torch.float32),
+                "vtarg": dbuf(B, torch.float32),
+            }
+            self.pin, self.pin_np = [], []
+            for _ in range(2):
+                pset = {k: pbuf(tuple(v.shape), v.dtype)
+                        for k, v in self.d.items()}
+                self.pin.append(pset)
+                self.pin_np.append({k: v.numpy() for k, v in pset.items()})
+            self.copy_events = [torch.cuda.Event(), torch.cuda.Event()]
+            self._buf = 0
+            self.batch = None
+            return
         self.n_cap, self.e_cap = n_cap, e_cap
         self.d = {
             "z": dbuf((n_cap, self.Fn), torch.float32),
@@ -238,6 +279,18 @@ class CapturedSGDStep:
             _node_ptr=self.d["nptr"])
 
     def _fill(self, mb_obs, actions, old_logp, adv, vtarg):
+        if self.models_batch is not None:
+            j = self._buf = self._buf ^ 1
+            self.copy_events[j].synchronize()
+            p = self.pin_np[j]
+            p["model_ids"][:] = [o.model_id for o in mb_obs]
+            p["gf"][:] = np.stack([o.graph_features for o in mb_obs])
+            p["mask"][:] = np.stack([o.action_mask for o in mb_obs])
+            p["actions"][:] = actions
+            p["old_logp"][:] = old_logp
+            p["adv"][:] = adv
+            p["vtarg"][:] = vtarg
+            return j
         ns = np.array([len(o.node_features) for o in mb_obs], dtype=np.int64)
         offsets = np.concatenate([[0], np.cumsum(ns)[:-1]])
         n = int(ns.sum())
@@ -323,9 +376,25 @@ class CapturedSGDStep:
                 if (p_.grad is None
                         or p_.grad.data_ptr() != base + 4 * o):
                     p_.grad = self.flat_g[o:o + np_].view(p_.shape)
-        logits, values = self.policy.forward_flat(
-            self.batch, self.d["gf"], self.d["mask"])
-        logits, values = logits[:B], values[:B]
+        if self.models_batch is not None:
+            # GNN once over the M model graphs; per-sample grads reach it
+            # summed through index_select's backward (exact by linearity)
+            from ..models.gnn import graph_mean
+            node_emb = self.policy.gnn(self.models_batch)
+            pooled = graph_mean(node_emb, self.models_batch)     # [M, out]
+            emb = pooled.index_select(0, self.d["model_ids"])    # [B, out]
+            graph_emb = self.policy.graph_module(self.d["gf"])
+            final = torch.cat([emb, graph_emb], dim=-1)
+            logits = self.policy.policy_branch(final)
+            values = self.policy.value_branch(final).squeeze(-1)
+            if self.policy.config["apply_action_mask"]:
+                inf_mask = torch.clamp(torch.log(self.d["mask"]),
+                                       min=torch.finfo(torch.float32).min)
+                logits = logits + inf_mask
+        else:
+            logits, values = self.policy.forward_flat(
+                self.batch, self.d["gf"], self.d["mask"])
+            logits, values = logits[:B], values[:B]
         if self._fused_loss:
             loss, stats = _PPOLossFn.apply(
                 logits, values, self.d["actions"], self.d["old_logp"],

@@ -195,8 +195,15 @@ class PPOTrainer:
                 and torch.cuda.is_available()):
             if self._stepper is None:
                 from .graph_step import CapturedSGDStep
+                # cached-models mode when the env serves per-model static
+                # obs (vectorised engine): GNN once per model per minibatch
+                mb_static = getattr(self.env, "_models_batch", None)
+                if mb_static is not None and batch["obs"] \
+                        and getattr(batch["obs"][0], "model_id", -1) < 0:
+                    mb_static = None
                 self._stepper = CapturedSGDStep(self.policy, self.optimizer,
-                                                cfg, self.device)
+                                                cfg, self.device,
+                                                models_batch=mb_static)
             stepper = self._stepper
             stepper.reset_stats()
             stepper.set_kl_coeff(self.kl_coeff)
@@ -210,23 +217,27 @@ class PPOTrainer:
 
         captured = False
         if stepper is not None and not stepper.broken and n >= B and n % B == 0:
-            # the permutations are fixed up front, so capacity can be sized to
-            # the LARGEST ACTUAL minibatch (sum of 128 sample sizes
-            # concentrates near B x mean — sizing by B x max-sample would
-            # nearly double the padded kernel work)
             idx_lists = []
             for _ in range(cfg.num_sgd_iter):
                 perm = rng.permutation(n)
                 for start in range(0, n, B):
                     idx_lists.append(perm[start:start + B])
-            ns_arr = np.array([len(o.node_features) for o in batch["obs"]],
-                              dtype=np.int64)
-            es_arr = np.array([len(o.edges_src) for o in batch["obs"]],
-                              dtype=np.int64)
-            need_n = max(int(ns_arr[idx].sum()) for idx in idx_lists)
-            need_e = max(int(es_arr[idx].sum()) for idx in idx_lists)
-            captured = stepper.ensure_capacity(int(need_n * 1.05) + 8,
-                                               int(need_e * 1.05) + 8)
+            if stepper.models_batch is not None:
+                # cached-models mode: fixed shapes, single capture
+                captured = stepper.ensure_capacity(0, 0)
+            else:
+                # the permutations are fixed up front, so capacity can be
+                # sized to the LARGEST ACTUAL minibatch (sum of 128 sample
+                # sizes concentrates near B x mean — sizing by B x
+                # max-sample would nearly double the padded kernel work)
+                ns_arr = np.array([len(o.node_features)
+                                   for o in batch["obs"]], dtype=np.int64)
+                es_arr = np.array([len(o.edges_src) for o in batch["obs"]],
+                                  dtype=np.int64)
+                need_n = max(int(ns_arr[idx].sum()) for idx in idx_lists)
+                need_e = max(int(es_arr[idx].sum()) for idx in idx_lists)
+                captured = stepper.ensure_capacity(int(need_n * 1.05) + 8,
+                                                   int(need_e * 1.05) + 8)
             if not captured:  # replay the same permutation stream eagerly
                 rng = np.random.RandomState(self.iteration + 1234 * get_rank())
 

@@ -26,6 +26,11 @@ class CompactObs:
     edges_dst: np.ndarray       # [m]
     graph_features: np.ndarray  # [G]
     action_mask: np.ndarray     # [A]
+    # workload-model id when the obs comes from the vectorised engine (the
+    # node/edge arrays are then shared per-model views); -1 = unknown.
+    # Enables the cached-models SGD path (graph_step.py): GNN once per
+    # distinct model per minibatch, gradients summed by index_select.
+    model_id: int = -1
 
     @staticmethod
     def from_obs(obs: Dict[str, np.ndarray]) -> "CompactObs":

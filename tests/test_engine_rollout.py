@@ -71,3 +71,40 @@ def test_ppo_on_engine_env(multi_model_files):
     venv.drain_episode_stats()
     c, s = venv.jct_running_stats()
     assert c > 0 and np.isfinite(s)
+
+
+@pytest.mark.gpu
+def test_cached_models_sgd_matches_eager(multi_model_files):
+    """Cached-models captured SGD (GNN once per model, grads summed via
+    index_select) vs the eager per-sample path: same trajectories, updated
+    params must agree to fp-reordering tolerance."""
+    from ddls_amd.models.gnn import GNNPolicy
+    from ddls_amd.rl.engine_env import EngineVectorEnv
+    from ddls_amd.rl.ppo import PPOConfig, PPOTrainer
+
+    dev = torch.device("cuda:0")
+
+    def build(use_graphs):
+        torch.manual_seed(0)
+        policy = GNNPolicy(num_actions=17).to(dev)
+        venv = EngineVectorEnv(
+            lambda: make_env(multi_model_files, "remove_and_repeat", 2,
+                             3000, 15),
+            num_envs=16, device=dev, base_seed=21)
+        cfg = PPOConfig(train_batch_size=16 * 8, sgd_minibatch_size=32,
+                        num_sgd_iter=2, use_hip_graphs=use_graphs)
+        return PPOTrainer(venv, policy, cfg, device=dev)
+
+    t_cap = build(True)
+    t_eag = build(False)
+    torch.manual_seed(1234)
+    s_cap = t_cap.train(num_steps=8)
+    torch.manual_seed(1234)
+    s_eag = t_eag.train(num_steps=8)
+    assert s_cap.get("hipgraph_minibatches", 0) > 0, "captured path not used"
+    assert np.isclose(s_cap["mean_reward"], s_eag["mean_reward"]), \
+        "trajectories diverged (rollout should be identical)"
+    for (k1, p1), (k2, p2) in zip(t_cap.policy.state_dict().items(),
+                                  t_eag.policy.state_dict().items()):
+        np.testing.assert_allclose(p1.cpu().numpy(), p2.cpu().numpy(),
+                                   rtol=0, atol=5e-4, err_msg=k1)
