@@ -26,6 +26,7 @@ _SOURCES = [os.path.join(_HERE, "hip", "meanpool.hip"),
             os.path.join(_HERE, "hip", "policy_head.hip"),
             os.path.join(_HERE, "hip", "lookahead.hip"),
             os.path.join(_HERE, "hip", "env_step.hip"),
+            os.path.join(_HERE, "hip", "cached_step.hip"),
             os.path.join(_HERE, "hip", "bindings.hip")]
 
 

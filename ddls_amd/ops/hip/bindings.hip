@@ -63,6 +63,8 @@ std::vector<torch::Tensor> ppo_loss_bwd(
     torch::Tensor gl, double vf_clip, double vf_coef, double ent_coef);
 void env_step_batch(std::vector<torch::Tensor> T, std::vector<double> fscal,
                     std::vector<int64_t> iscal);
+void cached_step_fwd(std::vector<torch::Tensor> T, std::vector<double> fs);
+void cached_step_bwd(std::vector<torch::Tensor> T, std::vector<double> fs);
 std::vector<torch::Tensor> lookahead_batch(
     torch::Tensor descs, torch::Tensor op_remaining, torch::Tensor op_worker,
     torch::Tensor op_priority, torch::Tensor out_indptr, torch::Tensor out_edges,
@@ -103,6 +105,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     m.def("head_fwd", &head_fwd,
           "fused policy/value head forward (LN+graphMLP+concat+2 branches)");
     m.def("head_bwd", &head_bwd, "fused policy/value head backward");
+    m.def("cached_step_fwd", &cached_step_fwd,
+          "fully-fused cached-models PPO minibatch forward (GNN+head+loss)");
+    m.def("cached_step_bwd", &cached_step_bwd,
+          "analytic whole-net backward into the flat grad buffer");
     m.def("env_step_batch", &env_step_batch,
           "batched RAMP env step over vectorised envs (placement search + "
           "memo hash probe + event loop + obs encode)");

@@ -65,6 +65,125 @@ class _PPOLossFn(torch.autograd.Function):
         return (glogits, gvalues) + (None,) * 9
 
 
+class _FusedCachedEngine:
+    """Host side of ops/hip/cached_step.hip: the ENTIRE cached-models PPO
+    minibatch fwd+bwd as two kernels, weight grads written directly into the
+    flat grad buffer.  Eligible for the tuned PAC-ML policy shape
+    (module_depth 1, 2 rounds, one FC hidden, relu, masked)."""
+
+    @staticmethod
+    def eligible(policy, ext) -> bool:
+        cfg = policy.config
+        return (ext is not None and hasattr(ext, "cached_step_fwd")
+                and cfg["module_depth"] == 1 and cfg["num_rounds"] == 2
+                and len(cfg["fcnet_hiddens"]) == 1
+                and cfg["aggregator_activation"] == "relu"
+                and cfg["fcnet_activation"] == "relu"
+                and cfg["apply_action_mask"]
+                and policy.num_actions <= 64
+                and os.environ.get("DDLS_AMD_DISABLE_FUSED_STEP", "0") != "1")
+
+    # weight slots in the kernel's fixed semantic order (enum W_* in
+    # cached_step.hip)
+    _SLOTS = [
+        "gnn.layers.0.node_module.0.weight", "gnn.layers.0.node_module.0.bias",
+        "gnn.layers.0.node_module.1.weight", "gnn.layers.0.node_module.1.bias",
+        "gnn.layers.0.edge_module.0.weight", "gnn.layers.0.edge_module.0.bias",
+        "gnn.layers.0.edge_module.1.weight", "gnn.layers.0.edge_module.1.bias",
+        "gnn.layers.0.reduce_module.0.weight",
+        "gnn.layers.0.reduce_module.0.bias",
+        "gnn.layers.0.reduce_module.1.weight",
+        "gnn.layers.0.reduce_module.1.bias",
+        "gnn.layers.1.node_module.0.weight", "gnn.layers.1.node_module.0.bias",
+        "gnn.layers.1.node_module.1.weight", "gnn.layers.1.node_module.1.bias",
+        "gnn.layers.1.edge_module.0.weight", "gnn.layers.1.edge_module.0.bias",
+        "gnn.layers.1.edge_module.1.weight", "gnn.layers.1.edge_module.1.bias",
+        "gnn.layers.1.reduce_module.0.weight",
+        "gnn.layers.1.reduce_module.0.bias",
+        "gnn.layers.1.reduce_module.1.weight",
+        "gnn.layers.1.reduce_module.1.bias",
+        "graph_module.0.weight", "graph_module.0.bias",
+        "graph_module.1.weight", "graph_module.1.bias",
+        "policy_branch.0.weight", "policy_branch.0.bias",
+        "policy_branch.2.weight", "policy_branch.2.bias",
+        "value_branch.0.weight", "value_branch.0.bias",
+        "value_branch.2.weight", "value_branch.2.bias",
+    ]
+
+    def __init__(self, stepper, models_batch):
+        self.st = stepper
+        ext = stepper._ext
+        policy = stepper.policy
+        dev = stepper.device
+        cfg = policy.config
+        B, A = stepper.B, stepper.A
+        mb = models_batch
+        order, indptr = mb.csr_by_dst()
+        nptr = mb.node_ptr()
+        N, E = mb.z.shape[0], mb.src.shape[0]
+        M = mb.num_graphs
+        src_order = torch.argsort(mb.src, stable=True)
+        src_counts = torch.bincount(mb.src, minlength=N)
+        src_indptr = torch.zeros(N + 1, dtype=torch.int64, device=dev)
+        torch.cumsum(src_counts, 0, out=src_indptr[1:])
+
+        # flat offsets in the kernel's semantic slot order
+        name_to_off = {}
+        off = 0
+        for name, p in policy.named_parameters():
+            name_to_off[name] = off
+            off += p.numel()
+        offs = torch.tensor([name_to_off[n] for n in self._SLOTS],
+                            dtype=torch.int64, device=dev)
+
+        H = cfg["out_features_msg"] // 2
+        HID = cfg["out_features_hidden"]
+        OUT = cfg["out_features_node"]
+        GEMB = cfg["out_features_graph"]
+        FC = cfg["fcnet_hiddens"][0]
+        F0 = mb.z.shape[1]
+        FE = mb.e.shape[1]
+        GFin = cfg["in_features_graph"] + A
+        MSG = 2 * H
+
+        z = lambda *shape: torch.zeros(shape, device=dev)
+        self.scratch = [
+            # statics
+            mb.z.contiguous(), mb.e.contiguous(), mb.src, mb.dst,
+            order, indptr, src_order, src_indptr,
+            mb.graph_of_node, nptr,
+            # staged inputs (the stepper's d-buffers, filled per minibatch)
+            stepper.d["model_ids"], stepper.d["gf"], stepper.d["mask"],
+            stepper.d["actions"], stepper.d["old_logp"], stepper.d["adv"],
+            stepper.d["vtarg"], stepper.kl_coeff_t.reshape(1),
+            # weights
+            stepper.flat_p, stepper.flat_g, offs,
+            # fwd activations
+            z(N, H), z(E, H), z(E, HID), z(N, HID), z(N, HID),
+            z(N, H), z(E, H), z(E, OUT), z(N, OUT), z(N, OUT), z(M, OUT),
+            z(N, F0), z(E, FE), z(E, MSG), z(N, MSG), z(N), z(E), z(E), z(N),
+            z(N, HID), z(E, FE), z(E, MSG), z(N, MSG), z(N), z(E), z(E), z(N),
+            z(B, GFin), z(B, OUT + GEMB), z(B, FC), z(B, FC), z(B),
+            z(B, A), z(B, A), z(B), z(B),
+            stepper.stats_acc,
+            # bwd scratch
+            z(B, A), z(B), z(B, FC), z(B, FC), z(B, OUT + GEMB), z(B, GFin),
+            z(M, OUT), z(N, OUT), z(E, MSG), z(N, MSG), z(N, H), z(E, H),
+            z(N, HID),
+            z(E, MSG), z(N, MSG), z(N, H), z(E, H),
+            z(N, OUT), z(E, OUT), z(N, H), z(E, H),
+            z(E, HID), z(N, HID),
+        ]
+        c = stepper.cfg
+        self.fscal = [float(c.clip_param), float(c.vf_clip_param),
+                      float(c.vf_loss_coeff), float(c.entropy_coeff)]
+        self.ext = ext
+
+    def run(self):
+        self.ext.cached_step_fwd(self.scratch, self.fscal)
+        self.ext.cached_step_bwd(self.scratch, self.fscal)
+
+
 class CapturedSGDStep:
     """Replayable hipGraph of one PPO SGD minibatch step.
 
@@ -358,8 +477,14 @@ class CapturedSGDStep:
         return j
 
     # ------------------------------------------------------------------
+    _fused_cached = None
+
     def _body_fwd_bwd(self):
         cfg, B = self.cfg, self.B
+        if self._fused_cached is not None:
+            # fully-fused path: two kernels, grads straight into flat_g
+            self._fused_cached.run()
+            return
         if self.flat_p is None:
             # flat engine zeroes grads inside flat_adam instead
             grads = [pa.grad for pa in self.policy.parameters()
@@ -538,6 +663,10 @@ class CapturedSGDStep:
         saved_stats = self.stats_acc.clone()
         if self._flat_engine:
             self._flatten_params()
+            if (self.models_batch is not None and self._fused_cached is None
+                    and _FusedCachedEngine.eligible(self.policy, self._ext)):
+                self._fused_cached = _FusedCachedEngine(self,
+                                                        self.models_batch)
             # warmup (required before capture) runs REAL steps; snapshot the
             # flat training state and restore it in place afterwards so the
             # captured graph sees the live tensors but training resumes
