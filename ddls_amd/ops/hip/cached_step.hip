@@ -4,16 +4,16 @@
 // M <= 16 distinct workload-model graphs (~200 nodes total): the whole
 // fwd+bwd is ~10 MFLOP.  Run as torch autograd it is ~146 dispatches per
 // replay (42% zero-fill / grad-accumulate glue — profiles/
-// kernel_stats_r02_cachedsgd.csv); here it is TWO kernels:
+// kernel_stats_r02_cachedsgd.csv); here it is SIX kernels (3 fwd + 3 bwd)
+// sized to the work's parallelism, with every weight gradient computed as a
+// deterministic LDS-tiled A^T.B contraction (no atomics, replay-stable):
 //
-//   cached_step_fwd : GNN (2 MeanPool rounds) over the M static model
-//                     graphs -> per-model mean pooling -> per-sample head
-//                     (LN+graph MLP ++ pooled emb -> 2 FC branches + mask)
-//                     -> fused PPO loss (stats accumulated on device)
-//   cached_step_bwd : the ENTIRE analytic backward, weight grads written
-//                     DIRECTLY into the flat grad buffer (no AccumulateGrad,
-//                     no zero-fills) — deterministic (no atomics), so
-//                     hipGraph replays are bit-stable.
+//   fwd:  cs_fwd_gnn   (1 WG)   GNN rounds + pooling + per-sample `final`
+//         cs_fwd_head  (many)   FC branches over B x FC units
+//         cs_fwd_loss  (1 WG)   softmax/surrogate rows + device stats
+//   bwd:  cs_bwd_head  (many)   per-sample chain: glogits -> gh1 -> gfinal
+//         cs_bwd_gnn   (1 WG)   pool grads + both MeanPool rounds' data bwd
+//         cs_bwd_w     (many)   ALL weight/bias/LN grads into flat_g
 //
 // Exact-by-linearity gradient aggregation: per-sample dL/d emb reaches the
 // (shared) per-model GNN summed over the minibatch — the same contraction
@@ -22,22 +22,33 @@
 // Layer semantics mirror models/gnn.py (MeanPoolLayer / GNNPolicy): each
 // module is LayerNorm -> Linear -> ReLU (module_depth=1), message =
 // concat(hn[src], he), self-message = concat(hn[v], 0), out =
-// (sum msgs + self) / (deg+1), zero for nodes with no in-edges (DGL
-// update_all semantics); graph_module is LN -> Linear (no activation);
-// branches are Linear-ReLU-Linear; mask added as clamp(log(mask)).
-// Loss matches ppo_loss.hip (clipped surrogate + kl + clipped vf +
-// entropy).  LayerNorm eps = 1e-5 (torch default), biased variance.
-//
-// One 256-thread workgroup per kernel: every phase is a flat-index loop
-// with __syncthreads between phases; all row activations / rstd / xhat
-// needed by the backward are stored in global scratch.
+// (sum msgs + self) / (deg+1), zero for nodes with no in-edges; the
+// graph_module is LN -> Linear (no activation); branches Linear-ReLU-Linear;
+// mask added as clamp(log(mask)).  Loss matches ppo_loss.hip.  LayerNorm
+// eps = 1e-5 (torch default), biased variance.
 #include <hip/hip_runtime.h>
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 #include <vector>
 
-#define NTHREADS 256
 #define LN_EPS 1e-5f
+
+// Compile-time dims of the tuned PAC-ML policy (model/gnn.yaml values).
+// fill_ptrs TORCH_CHECKs the runtime tensors against these; the Python side
+// (graph_step._FusedCachedEngine.eligible) refuses other shapes.  Constant
+// bounds let hipcc fully unroll the row loops so per-thread scratch arrays
+// live in registers instead of spilled scratch memory.
+#define KF0 5
+#define KFE 2
+#define KH 16
+#define KMSG 32
+#define KHID 64
+#define KOUT 16
+#define KGF 34
+#define KGE 8
+#define KFIN 24
+#define KFC 256
+#define KA 17
 
 // ---- tensor-list indices (mirror rl/graph_step.py fused mode) ----
 enum {
@@ -63,6 +74,10 @@ enum {
   C_GME1, C_GMS1, C_GHN1, C_GHE1,
   C_GPN2, C_GPE2, C_GPN1, C_GPE1,
   C_GPRE1E, C_GPRE1S,
+  // LN-out grad rows (gu = W^T gpre), stored by the data-bwd kernels for
+  // the LN gamma/beta reductions in cs_bwd_w
+  C_GU_M2E, C_GU_M2S, C_GU_H2, C_GU_E2, C_GU_M1E, C_GU_M1S, C_GU_Z1,
+  C_GU_E1,
   C_NT
 };
 
@@ -105,272 +120,284 @@ struct CachedPtrs {
   float *gme1, *gms1, *ghn1, *ghe1;
   float *gpn2, *gpe2, *gpn1, *gpe1;
   float *gpre1_e, *gpre1_s;
+  float *gu_m2e, *gu_m2s, *gu_h2, *gu_e2, *gu_m1e, *gu_m1s, *gu_z1, *gu_e1;
 };
 
 #define WP(slot) (P.flat_p + P.offs[slot])
-#define WG(slot) (P.flat_g + P.offs[slot])
+#define WG_(slot) (P.flat_g + P.offs[slot])
 
 // LayerNorm + store xhat and rstd for the backward.  One thread per row.
-__device__ void ln_row(const float* x, int D, float* xhat, float* rstd_out) {
+__device__ __forceinline__ void ln_row(const float* x, int Dd, float* xhat, float* rstd_out) {
   float mu = 0.f;
-  for (int i = 0; i < D; ++i) mu += x[i];
-  mu /= D;
+  for (int i = 0; i < Dd; ++i) mu += x[i];
+  mu /= Dd;
   float var = 0.f;
-  for (int i = 0; i < D; ++i) { float d = x[i] - mu; var += d * d; }
-  var /= D;
+  for (int i = 0; i < Dd; ++i) { float d = x[i] - mu; var += d * d; }
+  var /= Dd;
   float rstd = rsqrtf(var + LN_EPS);
-  for (int i = 0; i < D; ++i) xhat[i] = (x[i] - mu) * rstd;
+  for (int i = 0; i < Dd; ++i) xhat[i] = (x[i] - mu) * rstd;
   *rstd_out = rstd;
 }
 
-// grad of LN wrt input, given upstream grad on the AFFINE output.
-// gin = rstd * (ghat - mean(ghat) - xhat * mean(ghat*xhat)),  ghat = go*gamma
-__device__ void ln_bwd_row(const float* go, const float* gamma,
-                           const float* xhat, float rstd, int D, float* gin) {
+// grad of LN wrt input given gu (grad on the affine output), also USES the
+// caller-provided gu buffer (stored for the gamma/beta reductions later)
+__device__ __forceinline__ void ln_bwd_row(const float* gu, const float* gamma,
+                           const float* xhat, float rstd, int Dd,
+                           float* gin) {
   float m1 = 0.f, m2 = 0.f;
-  for (int i = 0; i < D; ++i) {
-    float gh = go[i] * gamma[i];
+  for (int i = 0; i < Dd; ++i) {
+    float gh = gu[i] * gamma[i];
     m1 += gh;
     m2 += gh * xhat[i];
   }
-  m1 /= D;
-  m2 /= D;
-  for (int i = 0; i < D; ++i) {
-    float gh = go[i] * gamma[i];
+  m1 /= Dd;
+  m2 /= Dd;
+  for (int i = 0; i < Dd; ++i) {
+    float gh = gu[i] * gamma[i];
     gin[i] = rstd * (gh - m1 - xhat[i] * m2);
   }
 }
 
 // one MeanPool "row MLP": out = relu(Lin(LNaffine(x)));  LN xhat/rstd saved
-__device__ void row_mlp_fwd(const CachedPtrs& P, const float* x, int Din,
+__device__ __forceinline__ void row_mlp_fwd(const CachedPtrs& P, const float* x, int Din,
                             int Dout, int w_slot, float* xhat, float* rstd,
                             float* out) {
   ln_row(x, Din, xhat, rstd);
-  const float* gam = WP(w_slot);        // LN weight slot
+  const float* gam = WP(w_slot);
   const float* bet = WP(w_slot + 1);
-  const float* W = WP(w_slot + 2);      // Linear weight [Dout, Din]
+  const float* W = WP(w_slot + 2);
   const float* b = WP(w_slot + 3);
   for (int o = 0; o < Dout; ++o) {
     float acc = b[o];
+#pragma unroll 4
     for (int i = 0; i < Din; ++i)
       acc += W[o * Din + i] * (xhat[i] * gam[i] + bet[i]);
     out[o] = acc > 0.f ? acc : 0.f;
   }
 }
 
-// backward of one row MLP row: given gout (wrt relu output) and stored
-// (out, xhat, rstd): gpre = gout*(out>0); optional gin; gpre stored for the
-// weight-grad pass.
-__device__ void row_mlp_bwd_row(const CachedPtrs& P, const float* gout,
+// backward of one row-MLP row: gpre = gout*(out>0) stored; gu = W^T gpre
+// stored; optional input grad.
+__device__ __forceinline__ void row_mlp_bwd_row(const CachedPtrs& P, const float* gout,
                                 const float* out, const float* xhat,
                                 float rstd, int Din, int Dout, int w_slot,
-                                float* gpre_store, float* gin /*or null*/) {
+                                float* gpre_store, float* gu_store,
+                                float* gin /*or null*/) {
   float gpre[64];
   for (int o = 0; o < Dout; ++o)
     gpre[o] = out[o] > 0.f ? gout[o] : 0.f;
   for (int o = 0; o < Dout; ++o) gpre_store[o] = gpre[o];
+  const float* W = WP(w_slot + 2);
+  float gu[64];
+  for (int i = 0; i < Din; ++i) {
+    float acc = 0.f;
+#pragma unroll 4
+    for (int o = 0; o < Dout; ++o) acc += W[o * Din + i] * gpre[o];
+    gu[i] = acc;
+    gu_store[i] = acc;
+  }
   if (gin != nullptr) {
-    const float* W = WP(w_slot + 2);
     const float* gam = WP(w_slot);
-    float gu[64];
-    for (int i = 0; i < Din; ++i) {
-      float acc = 0.f;
-      for (int o = 0; o < Dout; ++o) acc += W[o * Din + i] * gpre[o];
-      gu[i] = acc;
-    }
     ln_bwd_row(gu, gam, xhat, rstd, Din, gin);
   }
 }
 
-// ---------------------------------------------------------------------------
-__global__ void __launch_bounds__(NTHREADS)
-cached_step_fwd_kernel(CachedPtrs P, CachedDims D) {
+// ===========================================================================
+// forward
+// ===========================================================================
+
+__global__ void __launch_bounds__(1024)
+cs_fwd_gnn_kernel(CachedPtrs P, CachedDims D) {
   const int tid = threadIdx.x;
-  const int NT = NTHREADS;
+  const int NT = blockDim.x;
 
-  // phase 1: round-1 node/edge MLPs (one thread per row)
+  // round-1 node/edge MLPs (thread per row)
   for (int v = tid; v < D.N; v += NT)
-    row_mlp_fwd(P, P.z0 + (long)v * D.F0, D.F0, D.H, W_LN_N1_W,
-                P.xh_z1 + (long)v * D.F0, P.rst_z1 + v,
-                P.hn1 + (long)v * D.H);
+    row_mlp_fwd(P, P.z0 + (long)v * KF0, KF0, KH, W_LN_N1_W,
+                P.xh_z1 + (long)v * KF0, P.rst_z1 + v,
+                P.hn1 + (long)v * KH);
   for (int k = tid; k < D.E; k += NT)
-    row_mlp_fwd(P, P.e + (long)k * D.FE, D.FE, D.H, W_LN_E1_W,
-                P.xh_e1 + (long)k * D.FE, P.rst_e1 + k,
-                P.he1 + (long)k * D.H);
+    row_mlp_fwd(P, P.e + (long)k * KFE, KFE, KH, W_LN_E1_W,
+                P.xh_e1 + (long)k * KFE, P.rst_e1 + k,
+                P.he1 + (long)k * KH);
   __syncthreads();
 
-  // phase 2: round-1 reduce MLP on edge + self messages
   for (int k = tid; k < D.E; k += NT) {
     const long s = P.src[k];
     float msg[64];
-    for (int i = 0; i < D.H; ++i) msg[i] = P.hn1[s * D.H + i];
-    for (int i = 0; i < D.H; ++i) msg[D.H + i] = P.he1[(long)k * D.H + i];
-    row_mlp_fwd(P, msg, D.MSG, D.HID, W_LN_R1_W,
-                P.xh_m1e + (long)k * D.MSG, P.rst_m1e + k,
-                P.re1 + (long)k * D.HID);
+    for (int i = 0; i < KH; ++i) msg[i] = P.hn1[s * KH + i];
+    for (int i = 0; i < KH; ++i) msg[KH + i] = P.he1[(long)k * KH + i];
+    row_mlp_fwd(P, msg, KMSG, KHID, W_LN_R1_W,
+                P.xh_m1e + (long)k * KMSG, P.rst_m1e + k,
+                P.re1 + (long)k * KHID);
   }
   for (int v = tid; v < D.N; v += NT) {
     float msg[64];
-    for (int i = 0; i < D.H; ++i) msg[i] = P.hn1[(long)v * D.H + i];
-    for (int i = 0; i < D.H; ++i) msg[D.H + i] = 0.f;
-    row_mlp_fwd(P, msg, D.MSG, D.HID, W_LN_R1_W,
-                P.xh_m1s + (long)v * D.MSG, P.rst_m1s + v,
-                P.rs1 + (long)v * D.HID);
+    for (int i = 0; i < KH; ++i) msg[i] = P.hn1[(long)v * KH + i];
+    for (int i = 0; i < KH; ++i) msg[KH + i] = 0.f;
+    row_mlp_fwd(P, msg, KMSG, KHID, W_LN_R1_W,
+                P.xh_m1s + (long)v * KMSG, P.rst_m1s + v,
+                P.rs1 + (long)v * KHID);
   }
   __syncthreads();
 
-  // phase 3: round-1 combine (CSR by dst): h1 = (sum + self)/(deg+1), or 0
-  for (long u = tid; u < (long)D.N * D.HID; u += NT) {
-    int v = (int)(u / D.HID), i = (int)(u % D.HID);
+  for (long u = tid; u < (long)D.N * KHID; u += NT) {
+    int v = (int)(u / KHID), i = (int)(u % KHID);
     long lo = P.indptr[v], hi = P.indptr[v + 1];
     float out = 0.f;
     if (hi > lo) {
-      float acc = P.rs1[(long)v * D.HID + i];
+      float acc = P.rs1[(long)v * KHID + i];
       for (long q = lo; q < hi; ++q)
-        acc += P.re1[P.order[q] * D.HID + i];
+        acc += P.re1[P.order[q] * KHID + i];
       out = acc / (float)(hi - lo + 1);
     }
-    P.h1[(long)v * D.HID + i] = out;
+    P.h1[(long)v * KHID + i] = out;
   }
   __syncthreads();
 
-  // phase 4: round-2 node/edge MLPs
   for (int v = tid; v < D.N; v += NT)
-    row_mlp_fwd(P, P.h1 + (long)v * D.HID, D.HID, D.H, W_LN_N2_W,
-                P.xh_h2 + (long)v * D.HID, P.rst_h2 + v,
-                P.hn2 + (long)v * D.H);
+    row_mlp_fwd(P, P.h1 + (long)v * KHID, KHID, KH, W_LN_N2_W,
+                P.xh_h2 + (long)v * KHID, P.rst_h2 + v,
+                P.hn2 + (long)v * KH);
   for (int k = tid; k < D.E; k += NT)
-    row_mlp_fwd(P, P.e + (long)k * D.FE, D.FE, D.H, W_LN_E2_W,
-                P.xh_e2 + (long)k * D.FE, P.rst_e2 + k,
-                P.he2 + (long)k * D.H);
+    row_mlp_fwd(P, P.e + (long)k * KFE, KFE, KH, W_LN_E2_W,
+                P.xh_e2 + (long)k * KFE, P.rst_e2 + k,
+                P.he2 + (long)k * KH);
   __syncthreads();
 
-  // phase 5: round-2 reduce MLP
   for (int k = tid; k < D.E; k += NT) {
     const long s = P.src[k];
     float msg[64];
-    for (int i = 0; i < D.H; ++i) msg[i] = P.hn2[s * D.H + i];
-    for (int i = 0; i < D.H; ++i) msg[D.H + i] = P.he2[(long)k * D.H + i];
-    row_mlp_fwd(P, msg, D.MSG, D.OUT, W_LN_R2_W,
-                P.xh_m2e + (long)k * D.MSG, P.rst_m2e + k,
-                P.re2 + (long)k * D.OUT);
+    for (int i = 0; i < KH; ++i) msg[i] = P.hn2[s * KH + i];
+    for (int i = 0; i < KH; ++i) msg[KH + i] = P.he2[(long)k * KH + i];
+    row_mlp_fwd(P, msg, KMSG, KOUT, W_LN_R2_W,
+                P.xh_m2e + (long)k * KMSG, P.rst_m2e + k,
+                P.re2 + (long)k * KOUT);
   }
   for (int v = tid; v < D.N; v += NT) {
     float msg[64];
-    for (int i = 0; i < D.H; ++i) msg[i] = P.hn2[(long)v * D.H + i];
-    for (int i = 0; i < D.H; ++i) msg[D.H + i] = 0.f;
-    row_mlp_fwd(P, msg, D.MSG, D.OUT, W_LN_R2_W,
-                P.xh_m2s + (long)v * D.MSG, P.rst_m2s + v,
-                P.rs2 + (long)v * D.OUT);
+    for (int i = 0; i < KH; ++i) msg[i] = P.hn2[(long)v * KH + i];
+    for (int i = 0; i < KH; ++i) msg[KH + i] = 0.f;
+    row_mlp_fwd(P, msg, KMSG, KOUT, W_LN_R2_W,
+                P.xh_m2s + (long)v * KMSG, P.rst_m2s + v,
+                P.rs2 + (long)v * KOUT);
   }
   __syncthreads();
 
-  // phase 6: round-2 combine -> h2
-  for (long u = tid; u < (long)D.N * D.OUT; u += NT) {
-    int v = (int)(u / D.OUT), i = (int)(u % D.OUT);
+  for (long u = tid; u < (long)D.N * KOUT; u += NT) {
+    int v = (int)(u / KOUT), i = (int)(u % KOUT);
     long lo = P.indptr[v], hi = P.indptr[v + 1];
     float out = 0.f;
     if (hi > lo) {
-      float acc = P.rs2[(long)v * D.OUT + i];
+      float acc = P.rs2[(long)v * KOUT + i];
       for (long q = lo; q < hi; ++q)
-        acc += P.re2[P.order[q] * D.OUT + i];
+        acc += P.re2[P.order[q] * KOUT + i];
       out = acc / (float)(hi - lo + 1);
     }
-    P.h2[(long)v * D.OUT + i] = out;
+    P.h2[(long)v * KOUT + i] = out;
   }
   __syncthreads();
 
-  // phase 7: per-model mean pooling
-  for (long u = tid; u < (long)D.M * D.OUT; u += NT) {
-    int m = (int)(u / D.OUT), i = (int)(u % D.OUT);
+  for (long u = tid; u < (long)D.M * KOUT; u += NT) {
+    int m = (int)(u / KOUT), i = (int)(u % KOUT);
     long lo = P.model_nptr[m], hi = P.model_nptr[m + 1];
     float acc = 0.f;
-    for (long v = lo; v < hi; ++v) acc += P.h2[v * D.OUT + i];
-    P.pooled[(long)m * D.OUT + i] = acc / (float)(hi - lo);
+    for (long v = lo; v < hi; ++v) acc += P.h2[v * KOUT + i];
+    P.pooled[(long)m * KOUT + i] = acc / (float)(hi - lo);
   }
   __syncthreads();
 
-  // phase 8: per-sample head: final = [pooled(model) ++ Lin(LN(gf))]
+  // per-sample `final` = [pooled(model) ++ Lin(LN(gf))] (thread per sample)
   for (int b = tid; b < D.B; b += NT) {
     const long mid = P.model_ids[b];
-    float* fin = P.fin + (long)b * D.FIN;
-    for (int i = 0; i < D.OUT; ++i) fin[i] = P.pooled[mid * D.OUT + i];
-    float* xh = P.xh34 + (long)b * D.GFin;
+    float* fin = P.fin + (long)b * KFIN;
+    for (int i = 0; i < KOUT; ++i) fin[i] = P.pooled[mid * KOUT + i];
+    float* xh = P.xh34 + (long)b * KGF;
     float rstd;
-    ln_row(P.gf + (long)b * D.GFin, D.GFin, xh, &rstd);  // rstd not needed
+    ln_row(P.gf + (long)b * KGF, KGF, xh, &rstd);
     const float* gam = WP(W_LN_G_W);
     const float* bet = WP(W_LN_G_B);
     const float* Wg = WP(W_G_W);
     const float* bg = WP(W_G_B);
-    for (int o = 0; o < D.GEMB; ++o) {
+    for (int o = 0; o < KGE; ++o) {
       float acc = bg[o];
-      for (int i = 0; i < D.GFin; ++i)
-        acc += Wg[o * D.GFin + i] * (xh[i] * gam[i] + bet[i]);
-      fin[D.OUT + o] = acc;               // graph_module has NO activation
+#pragma unroll 4
+      for (int i = 0; i < KGF; ++i)
+        acc += Wg[o * KGF + i] * (xh[i] * gam[i] + bet[i]);
+      fin[KOUT + o] = acc;               // graph_module has NO activation
     }
   }
-  __syncthreads();
+}
 
-  // phase 9: FC branches, unit-parallel: h1p/h1v
-  for (long u = tid; u < (long)D.B * D.FC; u += NT) {
-    int b = (int)(u / D.FC), j = (int)(u % D.FC);
-    const float* fin = P.fin + (long)b * D.FIN;
+// FC branches: one thread per (b, hidden j) unit; W1p/W1v rows via L1/L2
+__global__ void __launch_bounds__(256)
+cs_fwd_head_kernel(CachedPtrs P, CachedDims D) {
+  const long total = (long)D.B * KFC;
+  for (long u = blockIdx.x * blockDim.x + threadIdx.x; u < total;
+       u += (long)gridDim.x * blockDim.x) {
+    int b = (int)(u / KFC), j = (int)(u % KFC);
+    const float* fin = P.fin + (long)b * KFIN;
     const float* W1p = WP(W_P1_W);
     const float* W1v = WP(W_V1_W);
     float ap = WP(W_P1_B)[j], av = WP(W_V1_B)[j];
-    for (int i = 0; i < D.FIN; ++i) {
-      ap += W1p[j * D.FIN + i] * fin[i];
-      av += W1v[j * D.FIN + i] * fin[i];
+#pragma unroll 4
+    for (int i = 0; i < KFIN; ++i) {
+      ap += W1p[j * KFIN + i] * fin[i];
+      av += W1v[j * KFIN + i] * fin[i];
     }
     P.h1p[u] = ap > 0.f ? ap : 0.f;
     P.h1v[u] = av > 0.f ? av : 0.f;
   }
-  __syncthreads();
+}
 
-  // phase 10: logits (+mask) and value; then per-sample loss rows
-  for (long u = tid; u < (long)D.B * D.A; u += NT) {
-    int b = (int)(u / D.A), a = (int)(u % D.A);
+// logits+value rows + loss (1 WG so the stats reduction stays in-block)
+__global__ void __launch_bounds__(1024)
+cs_fwd_loss_kernel(CachedPtrs P, CachedDims D) {
+  const int tid = threadIdx.x;
+  const int NT = blockDim.x;
+  for (long u = tid; u < (long)D.B * KA; u += NT) {
+    int b = (int)(u / KA), a = (int)(u % KA);
     const float* W2p = WP(W_P2_W);
     float acc = WP(W_P2_B)[a];
-    const float* h = P.h1p + (long)b * D.FC;
-    for (int j = 0; j < D.FC; ++j) acc += W2p[a * D.FC + j] * h[j];
-    float mk = P.mask[(long)b * D.A + a];
+    const float* h = P.h1p + (long)b * KFC;
+#pragma unroll 8
+    for (int j = 0; j < KFC; ++j) acc += W2p[a * KFC + j] * h[j];
+    float mk = P.mask[(long)b * KA + a];
     float lm = logf(mk);
     const float fmin = -3.402823466e+38f;
-    if (!(lm > fmin)) lm = fmin;          // clamp(log(mask), min=f32 min)
-    P.p[u] = acc + lm;                    // reuse p as raw masked logits
+    if (!(lm > fmin)) lm = fmin;
+    P.p[u] = acc + lm;                    // raw masked logits (overwritten)
   }
   for (int b = tid; b < D.B; b += NT) {
     const float* W2v = WP(W_V2_W);
     float acc = WP(W_V2_B)[0];
-    const float* h = P.h1v + (long)b * D.FC;
-    for (int j = 0; j < D.FC; ++j) acc += W2v[j] * h[j];
+#pragma unroll 8
+    for (int j = 0; j < KFC; ++j)
+      acc += W2v[j] * P.h1v[(long)b * KFC + j];
     P.values[b] = acc;
   }
   __syncthreads();
 
-  // phase 11: loss per sample (softmax + surrogate), block-accumulate stats
   __shared__ float acc_s[4];
   if (tid < 4) acc_s[tid] = 0.f;
   __syncthreads();
   for (int b = tid; b < D.B; b += NT) {
-    float* row = P.p + (long)b * D.A;
+    float* row = P.p + (long)b * KA;
     float mx = -3.0e38f;
-    for (int a = 0; a < D.A; ++a) mx = fmaxf(mx, row[a]);
+    for (int a = 0; a < KA; ++a) mx = fmaxf(mx, row[a]);
     float Z = 0.f;
-    for (int a = 0; a < D.A; ++a) Z += __expf(row[a] - mx);
+    for (int a = 0; a < KA; ++a) Z += __expf(row[a] - mx);
     float lse = mx + __logf(Z);
     float ent = 0.f;
-    for (int a = 0; a < D.A; ++a) {
+    for (int a = 0; a < KA; ++a) {
       float lp = row[a] - lse;
-      float p = __expf(lp);
-      P.lp[(long)b * D.A + a] = lp;
-      ent -= p * lp;
+      P.lp[(long)b * KA + a] = lp;
+      ent -= __expf(lp) * lp;
     }
-    for (int a = 0; a < D.A; ++a)
-      P.p[(long)b * D.A + a] = __expf(P.lp[(long)b * D.A + a]);
+    for (int a = 0; a < KA; ++a)
+      row[a] = __expf(P.lp[(long)b * KA + a]);
     const long a = P.actions[b];
-    const float logp_a = P.lp[(long)b * D.A + a];
+    const float logp_a = P.lp[(long)b * KA + a];
     const float ratio = __expf(logp_a - P.old_logp[b]);
     const float A_b = P.adv[b];
     const float r_cl = fminf(fmaxf(ratio, 1.f - D.clip), 1.f + D.clip);
@@ -396,509 +423,477 @@ cached_step_fwd_kernel(CachedPtrs P, CachedDims D) {
     const float vf = acc_s[1] * inv;
     const float kl = acc_s[2] * inv;
     const float ent = acc_s[3] * inv;
-    const float loss = pl + P.kl[0] * kl + D.vf_coef * vf
-                       - D.ent_coef * ent;
     P.stats[0] += pl;
     P.stats[1] += vf;
     P.stats[2] += kl;
     P.stats[3] += ent;
-    P.stats[4] += loss;
+    P.stats[4] += pl + P.kl[0] * kl + D.vf_coef * vf - D.ent_coef * ent;
   }
 }
 
-// ---------------------------------------------------------------------------
-__global__ void __launch_bounds__(NTHREADS)
-cached_step_bwd_kernel(CachedPtrs P, CachedDims D) {
+// ===========================================================================
+// backward
+// ===========================================================================
+
+// per-SAMPLE chain: each block owns a contiguous sample range, so every
+// cross-unit dependency (glogits row -> gh1 row -> gfinal row) stays inside
+// the block.  256 threads cooperate FC-wise within each sample.
+__global__ void __launch_bounds__(256)
+cs_bwd_head_kernel(CachedPtrs P, CachedDims D) {
   const int tid = threadIdx.x;
-  const int NT = NTHREADS;
-
-  // phase 0: loss backward rows -> glogits, gvalue  (upstream grad = 1)
-  for (long u = tid; u < (long)D.B * D.A; u += NT) {
-    int b = (int)(u / D.A), a = (int)(u % D.A);
-    const long act = P.actions[b];
-    const float c = P.coef[b] / D.B;
-    const float pj = P.p[u];
-    const float lpj = P.lp[u];
-    const float delta = (a == (int)act) ? 1.f : 0.f;
-    P.glogits[u] = c * (delta - pj)
-                   + (D.ent_coef / D.B) * pj * (lpj + P.hent[b]);
-  }
-  for (int b = tid; b < D.B; b += NT) {
-    const float verr = P.values[b] - P.vtarg[b];
-    const float mk = (verr * verr <= D.vf_clip) ? 1.f : 0.f;
-    P.gvalue[b] = D.vf_coef * 2.f * verr * mk / D.B;
-  }
-  __syncthreads();
-
-  // phase 1: branch hidden grads
-  for (long u = tid; u < (long)D.B * D.FC; u += NT) {
-    int b = (int)(u / D.FC), j = (int)(u % D.FC);
-    const float* W2p = WP(W_P2_W);
-    float acc = 0.f;
-    for (int a = 0; a < D.A; ++a)
-      acc += W2p[a * D.FC + j] * P.glogits[(long)b * D.A + a];
-    P.gh1p[u] = P.h1p[u] > 0.f ? acc : 0.f;
-    const float* W2v = WP(W_V2_W);
-    float av = W2v[j] * P.gvalue[b];
-    P.gh1v[u] = P.h1v[u] > 0.f ? av : 0.f;
-  }
-  __syncthreads();
-
-  // phase 2: gfinal
-  for (long u = tid; u < (long)D.B * D.FIN; u += NT) {
-    int b = (int)(u / D.FIN), i = (int)(u % D.FIN);
-    const float* W1p = WP(W_P1_W);
-    const float* W1v = WP(W_V1_W);
-    float acc = 0.f;
-    for (int j = 0; j < D.FC; ++j) {
-      acc += W1p[j * D.FIN + i] * P.gh1p[(long)b * D.FC + j];
-      acc += W1v[j * D.FIN + i] * P.gh1v[(long)b * D.FC + j];
+  const int NT = blockDim.x;
+  const int per = (D.B + gridDim.x - 1) / gridDim.x;
+  const int b0 = blockIdx.x * per;
+  const int b1 = min(D.B, b0 + per);
+  for (int b = b0; b < b1; ++b) {
+    // loss backward row (upstream grad = 1)
+    for (int a = tid; a < KA; a += NT) {
+      const long u = (long)b * KA + a;
+      const long act = P.actions[b];
+      const float c = P.coef[b] / D.B;
+      const float pj = P.p[u];
+      const float lpj = P.lp[u];
+      const float delta = (a == (int)act) ? 1.f : 0.f;
+      P.glogits[u] = c * (delta - pj)
+                     + (D.ent_coef / D.B) * pj * (lpj + P.hent[b]);
     }
-    P.gfinal[u] = acc;
+    if (tid == 0) {
+      const float verr = P.values[b] - P.vtarg[b];
+      const float mk = (verr * verr <= D.vf_clip) ? 1.f : 0.f;
+      P.gvalue[b] = D.vf_coef * 2.f * verr * mk / D.B;
+    }
+    __syncthreads();
+    // hidden grads
+    for (int j = tid; j < KFC; j += NT) {
+      const long u = (long)b * KFC + j;
+      const float* W2p = WP(W_P2_W);
+      float acc = 0.f;
+      for (int a = 0; a < KA; ++a)
+        acc += W2p[a * KFC + j] * P.glogits[(long)b * KA + a];
+      P.gh1p[u] = P.h1p[u] > 0.f ? acc : 0.f;
+      const float av = WP(W_V2_W)[j] * P.gvalue[b];
+      P.gh1v[u] = P.h1v[u] > 0.f ? av : 0.f;
+    }
+    __syncthreads();
+    // gfinal + graph-module LN-out grads
+    for (int i = tid; i < KFIN; i += NT) {
+      const float* W1p = WP(W_P1_W);
+      const float* W1v = WP(W_V1_W);
+      float acc = 0.f;
+#pragma unroll 8
+      for (int j = 0; j < KFC; ++j) {
+        acc += W1p[j * KFIN + i] * P.gh1p[(long)b * KFC + j];
+        acc += W1v[j * KFIN + i] * P.gh1v[(long)b * KFC + j];
+      }
+      P.gfinal[(long)b * KFIN + i] = acc;
+    }
+    __syncthreads();
+    for (int i = tid; i < KGF; i += NT) {
+      const float* Wg = WP(W_G_W);
+      float acc = 0.f;
+      for (int o = 0; o < KGE; ++o)
+        acc += Wg[o * KGF + i] * P.gfinal[(long)b * KFIN + KOUT + o];
+      P.gu34[(long)b * KGF + i] = acc;
+    }
+    __syncthreads();
   }
-  __syncthreads();
+}
 
-  // phase 3: graph-module LN-output grads (gu34) + pooled grads
-  for (long u = tid; u < (long)D.B * D.GFin; u += NT) {
-    int b = (int)(u / D.GFin), i = (int)(u % D.GFin);
-    const float* Wg = WP(W_G_W);
-    float acc = 0.f;
-    for (int o = 0; o < D.GEMB; ++o)
-      acc += Wg[o * D.GFin + i] * P.gfinal[(long)b * D.FIN + D.OUT + o];
-    P.gu34[u] = acc;
-  }
-  for (long u = tid; u < (long)D.M * D.OUT; u += NT) {
-    int m = (int)(u / D.OUT), i = (int)(u % D.OUT);
+// pool grads + GNN data backward (graph-sized; 1 WG)
+__global__ void __launch_bounds__(1024)
+cs_bwd_gnn_kernel(CachedPtrs P, CachedDims D) {
+  const int tid = threadIdx.x;
+  const int NT = blockDim.x;
+
+  for (long u = tid; u < (long)D.M * KOUT; u += NT) {
+    int m = (int)(u / KOUT), i = (int)(u % KOUT);
     float acc = 0.f;
     for (int b = 0; b < D.B; ++b)
       if ((int)P.model_ids[b] == m)
-        acc += P.gfinal[(long)b * D.FIN + i];
+        acc += P.gfinal[(long)b * KFIN + i];
     P.gpool[u] = acc;
   }
   __syncthreads();
-
-  // phase 4: gh2 (pool-mean broadcast)
-  for (long u = tid; u < (long)D.N * D.OUT; u += NT) {
-    int v = (int)(u / D.OUT), i = (int)(u % D.OUT);
+  for (long u = tid; u < (long)D.N * KOUT; u += NT) {
+    int v = (int)(u / KOUT), i = (int)(u % KOUT);
     const long m = P.node_model[v];
     const float nn = (float)(P.model_nptr[m + 1] - P.model_nptr[m]);
-    P.gh2[u] = P.gpool[m * D.OUT + i] / nn;
+    P.gh2[u] = P.gpool[m * KOUT + i] / nn;
   }
   __syncthreads();
 
-  // phase 5: round-2 message grads
+  // round-2 message grads
   for (int k = tid; k < D.E; k += NT) {
     const int v = (int)P.dst[k];
     const long deg = P.indptr[v + 1] - P.indptr[v];
     const float f = 1.f / (float)(deg + 1);
     float gr[64];
-    for (int i = 0; i < D.OUT; ++i)
-      gr[i] = P.gh2[(long)v * D.OUT + i] * f;
-    row_mlp_bwd_row(P, gr, P.re2 + (long)k * D.OUT,
-                    P.xh_m2e + (long)k * D.MSG, P.rst_m2e[k], D.MSG, D.OUT,
-                    W_LN_R2_W, P.gpe2 + (long)k * D.OUT,
-                    P.gme2 + (long)k * D.MSG);
+    for (int i = 0; i < KOUT; ++i)
+      gr[i] = P.gh2[(long)v * KOUT + i] * f;
+    row_mlp_bwd_row(P, gr, P.re2 + (long)k * KOUT,
+                    P.xh_m2e + (long)k * KMSG, P.rst_m2e[k], KMSG, KOUT,
+                    W_LN_R2_W, P.gpe2 + (long)k * KOUT,
+                    P.gu_m2e + (long)k * KMSG,
+                    P.gme2 + (long)k * KMSG);
   }
   for (int v = tid; v < D.N; v += NT) {
     const long deg = P.indptr[v + 1] - P.indptr[v];
     float gr[64];
     if (deg > 0) {
       const float f = 1.f / (float)(deg + 1);
-      for (int i = 0; i < D.OUT; ++i)
-        gr[i] = P.gh2[(long)v * D.OUT + i] * f;
+      for (int i = 0; i < KOUT; ++i)
+        gr[i] = P.gh2[(long)v * KOUT + i] * f;
     } else {
-      for (int i = 0; i < D.OUT; ++i) gr[i] = 0.f;   // zero-filled output
+      for (int i = 0; i < KOUT; ++i) gr[i] = 0.f;
     }
-    row_mlp_bwd_row(P, gr, P.rs2 + (long)v * D.OUT,
-                    P.xh_m2s + (long)v * D.MSG, P.rst_m2s[v], D.MSG, D.OUT,
-                    W_LN_R2_W, P.gpn2 + (long)v * D.OUT,
-                    P.gms2 + (long)v * D.MSG);
+    row_mlp_bwd_row(P, gr, P.rs2 + (long)v * KOUT,
+                    P.xh_m2s + (long)v * KMSG, P.rst_m2s[v], KMSG, KOUT,
+                    W_LN_R2_W, P.gpn2 + (long)v * KOUT,
+                    P.gu_m2s + (long)v * KMSG,
+                    P.gms2 + (long)v * KMSG);
   }
   __syncthreads();
 
-  // phase 6: scatter message grads to hn2 / he2 (CSR by src for hn)
-  for (long u = tid; u < (long)D.N * D.H; u += NT) {
-    int v = (int)(u / D.H), i = (int)(u % D.H);
+  // scatter to hn2 / he2
+  for (long u = tid; u < (long)D.N * KH; u += NT) {
+    int v = (int)(u / KH), i = (int)(u % KH);
     long lo = P.src_indptr[v], hi = P.src_indptr[v + 1];
-    float acc = P.gms2[(long)v * D.MSG + i];      // self-message front half
+    float acc = P.gms2[(long)v * KMSG + i];
     for (long q = lo; q < hi; ++q)
-      acc += P.gme2[P.src_order[q] * D.MSG + i];
+      acc += P.gme2[P.src_order[q] * KMSG + i];
     P.ghn2[u] = acc;
   }
-  for (long u = tid; u < (long)D.E * D.H; u += NT) {
-    int k = (int)(u / D.H), i = (int)(u % D.H);
-    P.ghe2[u] = P.gme2[(long)k * D.MSG + D.H + i];
+  for (long u = tid; u < (long)D.E * KH; u += NT) {
+    int k = (int)(u / KH), i = (int)(u % KH);
+    P.ghe2[u] = P.gme2[(long)k * KMSG + KH + i];
   }
   __syncthreads();
 
-  // phase 7: round-2 node/edge row-MLP backward (node input = h1 grads)
+  // round-2 node/edge row-MLP data backward
   for (int v = tid; v < D.N; v += NT)
-    row_mlp_bwd_row(P, P.ghn2 + (long)v * D.H, P.hn2 + (long)v * D.H,
-                    P.xh_h2 + (long)v * D.HID, P.rst_h2[v], D.HID, D.H,
-                    W_LN_N2_W, P.gpn1 + (long)v * D.H /*reuse as gpre_n2*/,
-                    P.gh1b + (long)v * D.HID);
+    row_mlp_bwd_row(P, P.ghn2 + (long)v * KH, P.hn2 + (long)v * KH,
+                    P.xh_h2 + (long)v * KHID, P.rst_h2[v], KHID, KH,
+                    W_LN_N2_W, P.gpn1 + (long)v * KH,
+                    P.gu_h2 + (long)v * KHID,
+                    P.gh1b + (long)v * KHID);
   for (int k = tid; k < D.E; k += NT)
-    row_mlp_bwd_row(P, P.ghe2 + (long)k * D.H, P.he2 + (long)k * D.H,
-                    P.xh_e2 + (long)k * D.FE, P.rst_e2[k], D.FE, D.H,
-                    W_LN_E2_W, P.gpe1 + (long)k * D.H /*reuse as gpre_e2m*/,
-                    nullptr);
+    row_mlp_bwd_row(P, P.ghe2 + (long)k * KH, P.he2 + (long)k * KH,
+                    P.xh_e2 + (long)k * KFE, P.rst_e2[k], KFE, KH,
+                    W_LN_E2_W, P.gpe1 + (long)k * KH,
+                    P.gu_e2 + (long)k * KFE, nullptr);
   __syncthreads();
 
-  // ---- weight grads for round 2 + head are computed in phase W below;
-  // first finish round-1 data flow (needs gh1) ----
-
-  // phase 8: round-1 message grads (upstream gh1b); round-1 needs BOTH
-  // gpre rows (HID wide, for the weight-grad pass) and gmsg rows (MSG wide)
-  float* gpre1_e = P.gpre1_e;   // [E, HID]
-  float* gpre1_s = P.gpre1_s;   // [N, HID]
+  // round-1 message grads
   for (int k = tid; k < D.E; k += NT) {
     const int v = (int)P.dst[k];
     const long deg = P.indptr[v + 1] - P.indptr[v];
     const float f = 1.f / (float)(deg + 1);
     float gr[64];
-    for (int i = 0; i < D.HID; ++i)
-      gr[i] = P.gh1b[(long)v * D.HID + i] * f;
-    row_mlp_bwd_row(P, gr, P.re1 + (long)k * D.HID,
-                    P.xh_m1e + (long)k * D.MSG, P.rst_m1e[k], D.MSG, D.HID,
-                    W_LN_R1_W, gpre1_e + (long)k * D.HID,
-                    P.gme1 + (long)k * D.MSG);
+    for (int i = 0; i < KHID; ++i)
+      gr[i] = P.gh1b[(long)v * KHID + i] * f;
+    row_mlp_bwd_row(P, gr, P.re1 + (long)k * KHID,
+                    P.xh_m1e + (long)k * KMSG, P.rst_m1e[k], KMSG, KHID,
+                    W_LN_R1_W, P.gpre1_e + (long)k * KHID,
+                    P.gu_m1e + (long)k * KMSG,
+                    P.gme1 + (long)k * KMSG);
   }
   for (int v = tid; v < D.N; v += NT) {
     const long deg = P.indptr[v + 1] - P.indptr[v];
     float gr[64];
     if (deg > 0) {
       const float f = 1.f / (float)(deg + 1);
-      for (int i = 0; i < D.HID; ++i)
-        gr[i] = P.gh1b[(long)v * D.HID + i] * f;
+      for (int i = 0; i < KHID; ++i)
+        gr[i] = P.gh1b[(long)v * KHID + i] * f;
     } else {
-      for (int i = 0; i < D.HID; ++i) gr[i] = 0.f;
+      for (int i = 0; i < KHID; ++i) gr[i] = 0.f;
     }
-    row_mlp_bwd_row(P, gr, P.rs1 + (long)v * D.HID,
-                    P.xh_m1s + (long)v * D.MSG, P.rst_m1s[v], D.MSG, D.HID,
-                    W_LN_R1_W, gpre1_s + (long)v * D.HID,
-                    P.gms1 + (long)v * D.MSG);
+    row_mlp_bwd_row(P, gr, P.rs1 + (long)v * KHID,
+                    P.xh_m1s + (long)v * KMSG, P.rst_m1s[v], KMSG, KHID,
+                    W_LN_R1_W, P.gpre1_s + (long)v * KHID,
+                    P.gu_m1s + (long)v * KMSG,
+                    P.gms1 + (long)v * KMSG);
   }
   __syncthreads();
 
-  // phase 9: scatter to hn1/he1 grads, then node/edge row-MLP gpre rows
-  for (long u = tid; u < (long)D.N * D.H; u += NT) {
-    int v = (int)(u / D.H), i = (int)(u % D.H);
+  // scatter to hn1 / he1; relu-mask in place; store gu rows for LN grads
+  for (long u = tid; u < (long)D.N * KH; u += NT) {
+    int v = (int)(u / KH), i = (int)(u % KH);
     long lo = P.src_indptr[v], hi = P.src_indptr[v + 1];
-    float acc = P.gms1[(long)v * D.MSG + i];
+    float acc = P.gms1[(long)v * KMSG + i];
     for (long q = lo; q < hi; ++q)
-      acc += P.gme1[P.src_order[q] * D.MSG + i];
-    P.ghn1[u] = acc;
+      acc += P.gme1[P.src_order[q] * KMSG + i];
+    P.ghn1[u] = P.hn1[u] > 0.f ? acc : 0.f;   // gpre rows for node module 1
   }
-  for (long u = tid; u < (long)D.E * D.H; u += NT) {
-    int k = (int)(u / D.H), i = (int)(u % D.H);
-    P.ghe1[u] = P.gme1[(long)k * D.MSG + D.H + i];
-  }
-  __syncthreads();
-  // gpre rows for node1/edge1 (inputs static: no input grads) — store into
-  // ghn1/ghe1 in place (relu mask applied)
-  for (long u = tid; u < (long)D.N * D.H; u += NT)
-    P.ghn1[u] = P.hn1[u] > 0.f ? P.ghn1[u] : 0.f;
-  for (long u = tid; u < (long)D.E * D.H; u += NT)
-    P.ghe1[u] = P.he1[u] > 0.f ? P.ghe1[u] : 0.f;
-  __syncthreads();
-
-  // ---- phase W: ALL weight gradients, deterministic unit-per-thread ----
-  // helper lambdas are spelled out per weight for clarity.
-
-  // FC branch 2: W2p[A,FC], b2p[A]; W2v[1,FC], b2v[1]
-  for (long u = tid; u < (long)D.A * D.FC; u += NT) {
-    int a = (int)(u / D.FC), j = (int)(u % D.FC);
-    float acc = 0.f;
-    for (int b = 0; b < D.B; ++b)
-      acc += P.glogits[(long)b * D.A + a] * P.h1p[(long)b * D.FC + j];
-    WG(W_P2_W)[u] = acc;
-  }
-  for (int a = tid; a < D.A; a += NT) {
-    float acc = 0.f;
-    for (int b = 0; b < D.B; ++b) acc += P.glogits[(long)b * D.A + a];
-    WG(W_P2_B)[a] = acc;
-  }
-  for (int j = tid; j < D.FC; j += NT) {
-    float acc = 0.f, accb = 0.f;
-    for (int b = 0; b < D.B; ++b) {
-      acc += P.gvalue[b] * P.h1v[(long)b * D.FC + j];
-      accb += P.gvalue[b];
-    }
-    WG(W_V2_W)[j] = acc;
-    if (j == 0) WG(W_V2_B)[0] = accb;
-  }
-  // FC branch 1: W1p[FC,FIN], b1p[FC]; W1v, b1v
-  for (long u = tid; u < (long)D.FC * D.FIN; u += NT) {
-    int j = (int)(u / D.FIN), i = (int)(u % D.FIN);
-    float ap = 0.f, av = 0.f;
-    for (int b = 0; b < D.B; ++b) {
-      const float fin = P.fin[(long)b * D.FIN + i];
-      ap += P.gh1p[(long)b * D.FC + j] * fin;
-      av += P.gh1v[(long)b * D.FC + j] * fin;
-    }
-    WG(W_P1_W)[u] = ap;
-    WG(W_V1_W)[u] = av;
-  }
-  for (int j = tid; j < D.FC; j += NT) {
-    float ap = 0.f, av = 0.f;
-    for (int b = 0; b < D.B; ++b) {
-      ap += P.gh1p[(long)b * D.FC + j];
-      av += P.gh1v[(long)b * D.FC + j];
-    }
-    WG(W_P1_B)[j] = ap;
-    WG(W_V1_B)[j] = av;
-  }
-  // graph module: Wg[GEMB,GFin] += ggemb x u34 (u = xhat*gam+bet), bg, LN
-  {
-    const float* gam = WP(W_LN_G_W);
-    const float* bet = WP(W_LN_G_B);
-    for (long u = tid; u < (long)D.GEMB * D.GFin; u += NT) {
-      int o = (int)(u / D.GFin), i = (int)(u % D.GFin);
-      float acc = 0.f;
-      for (int b = 0; b < D.B; ++b) {
-        const float xh = P.xh34[(long)b * D.GFin + i];
-        acc += P.gfinal[(long)b * D.FIN + D.OUT + o]
-               * (xh * gam[i] + bet[i]);
-      }
-      WG(W_G_W)[u] = acc;
-    }
-    for (int o = tid; o < D.GEMB; o += NT) {
-      float acc = 0.f;
-      for (int b = 0; b < D.B; ++b)
-        acc += P.gfinal[(long)b * D.FIN + D.OUT + o];
-      WG(W_G_B)[o] = acc;
-    }
-    for (int i = tid; i < D.GFin; i += NT) {
-      float gw = 0.f, gb = 0.f;
-      for (int b = 0; b < D.B; ++b) {
-        const float gu = P.gu34[(long)b * D.GFin + i];
-        gw += gu * P.xh34[(long)b * D.GFin + i];
-        gb += gu;
-      }
-      WG(W_LN_G_W)[i] = gw;
-      WG(W_LN_G_B)[i] = gb;
-    }
+  for (long u = tid; u < (long)D.E * KH; u += NT) {
+    int k = (int)(u / KH), i = (int)(u % KH);
+    float g = P.gme1[(long)k * KMSG + KH + i];
+    P.ghe1[u] = P.he1[u] > 0.f ? g : 0.f;     // gpre rows for edge module 1
   }
   __syncthreads();
-
-  // reduce MLP 2: Wr2[OUT,MSG] over E edge rows + N self rows
-  {
-    const float* gam = WP(W_LN_R2_W);
-    const float* bet = WP(W_LN_R2_B);
-    for (long u = tid; u < (long)D.OUT * D.MSG; u += NT) {
-      int o = (int)(u / D.MSG), i = (int)(u % D.MSG);
+  // gu rows for node/edge module 1 (inputs static, no input grads)
+  for (int v = tid; v < D.N; v += NT) {
+    const float* W = WP(W_N1_W);
+    for (int i = 0; i < KF0; ++i) {
       float acc = 0.f;
-      for (int k = 0; k < D.E; ++k)
-        acc += P.gpe2[(long)k * D.OUT + o]
-               * (P.xh_m2e[(long)k * D.MSG + i] * gam[i] + bet[i]);
-      for (int v = 0; v < D.N; ++v)
-        acc += P.gpn2[(long)v * D.OUT + o]
-               * (P.xh_m2s[(long)v * D.MSG + i] * gam[i] + bet[i]);
-      WG(W_R2_W)[u] = acc;
-    }
-    for (int o = tid; o < D.OUT; o += NT) {
-      float acc = 0.f;
-      for (int k = 0; k < D.E; ++k) acc += P.gpe2[(long)k * D.OUT + o];
-      for (int v = 0; v < D.N; ++v) acc += P.gpn2[(long)v * D.OUT + o];
-      WG(W_R2_B)[o] = acc;
-    }
-    const float* Wr = WP(W_R2_W);
-    for (int i = tid; i < D.MSG; i += NT) {
-      float gw = 0.f, gb = 0.f;
-      for (int k = 0; k < D.E; ++k) {
-        float gu = 0.f;
-        for (int o = 0; o < D.OUT; ++o)
-          gu += Wr[o * D.MSG + i] * P.gpe2[(long)k * D.OUT + o];
-        gw += gu * P.xh_m2e[(long)k * D.MSG + i];
-        gb += gu;
-      }
-      for (int v = 0; v < D.N; ++v) {
-        float gu = 0.f;
-        for (int o = 0; o < D.OUT; ++o)
-          gu += Wr[o * D.MSG + i] * P.gpn2[(long)v * D.OUT + o];
-        gw += gu * P.xh_m2s[(long)v * D.MSG + i];
-        gb += gu;
-      }
-      WG(W_LN_R2_W)[i] = gw;
-      WG(W_LN_R2_B)[i] = gb;
+      for (int o = 0; o < KH; ++o)
+        acc += W[o * KF0 + i] * P.ghn1[(long)v * KH + o];
+      P.gu_z1[(long)v * KF0 + i] = acc;
     }
   }
-  // node module 2: Wn2[H,HID] over N rows (gpre in gpn1 buffer)
-  {
-    const float* gam = WP(W_LN_N2_W);
-    const float* bet = WP(W_LN_N2_B);
-    for (long u = tid; u < (long)D.H * D.HID; u += NT) {
-      int o = (int)(u / D.HID), i = (int)(u % D.HID);
+  for (int k = tid; k < D.E; k += NT) {
+    const float* W = WP(W_E1_W);
+    for (int i = 0; i < KFE; ++i) {
       float acc = 0.f;
-      for (int v = 0; v < D.N; ++v)
-        acc += P.gpn1[(long)v * D.H + o]
-               * (P.xh_h2[(long)v * D.HID + i] * gam[i] + bet[i]);
-      WG(W_N2_W)[u] = acc;
-    }
-    for (int o = tid; o < D.H; o += NT) {
-      float acc = 0.f;
-      for (int v = 0; v < D.N; ++v) acc += P.gpn1[(long)v * D.H + o];
-      WG(W_N2_B)[o] = acc;
-    }
-    const float* Wn = WP(W_N2_W);
-    for (int i = tid; i < D.HID; i += NT) {
-      float gw = 0.f, gb = 0.f;
-      for (int v = 0; v < D.N; ++v) {
-        float gu = 0.f;
-        for (int o = 0; o < D.H; ++o)
-          gu += Wn[o * D.HID + i] * P.gpn1[(long)v * D.H + o];
-        gw += gu * P.xh_h2[(long)v * D.HID + i];
-        gb += gu;
-      }
-      WG(W_LN_N2_W)[i] = gw;
-      WG(W_LN_N2_B)[i] = gb;
-    }
-  }
-  // edge module 2: We2[H,FE] over E rows (gpre in gpe1 buffer)
-  {
-    const float* gam = WP(W_LN_E2_W);
-    const float* bet = WP(W_LN_E2_B);
-    for (long u = tid; u < (long)D.H * D.FE; u += NT) {
-      int o = (int)(u / D.FE), i = (int)(u % D.FE);
-      float acc = 0.f;
-      for (int k = 0; k < D.E; ++k)
-        acc += P.gpe1[(long)k * D.H + o]
-               * (P.xh_e2[(long)k * D.FE + i] * gam[i] + bet[i]);
-      WG(W_E2_W)[u] = acc;
-    }
-    for (int o = tid; o < D.H; o += NT) {
-      float acc = 0.f;
-      for (int k = 0; k < D.E; ++k) acc += P.gpe1[(long)k * D.H + o];
-      WG(W_E2_B)[o] = acc;
-    }
-    const float* We = WP(W_E2_W);
-    for (int i = tid; i < D.FE; i += NT) {
-      float gw = 0.f, gb = 0.f;
-      for (int k = 0; k < D.E; ++k) {
-        float gu = 0.f;
-        for (int o = 0; o < D.H; ++o)
-          gu += We[o * D.FE + i] * P.gpe1[(long)k * D.H + o];
-        gw += gu * P.xh_e2[(long)k * D.FE + i];
-        gb += gu;
-      }
-      WG(W_LN_E2_W)[i] = gw;
-      WG(W_LN_E2_B)[i] = gb;
-    }
-  }
-  // reduce MLP 1: Wr1[HID,MSG] over E + N rows (gpre in gpre1_e/gpre1_s)
-  {
-    const float* gam = WP(W_LN_R1_W);
-    const float* bet = WP(W_LN_R1_B);
-    for (long u = tid; u < (long)D.HID * D.MSG; u += NT) {
-      int o = (int)(u / D.MSG), i = (int)(u % D.MSG);
-      float acc = 0.f;
-      for (int k = 0; k < D.E; ++k)
-        acc += gpre1_e[(long)k * D.HID + o]
-               * (P.xh_m1e[(long)k * D.MSG + i] * gam[i] + bet[i]);
-      for (int v = 0; v < D.N; ++v)
-        acc += gpre1_s[(long)v * D.HID + o]
-               * (P.xh_m1s[(long)v * D.MSG + i] * gam[i] + bet[i]);
-      WG(W_R1_W)[u] = acc;
-    }
-    for (int o = tid; o < D.HID; o += NT) {
-      float acc = 0.f;
-      for (int k = 0; k < D.E; ++k) acc += gpre1_e[(long)k * D.HID + o];
-      for (int v = 0; v < D.N; ++v) acc += gpre1_s[(long)v * D.HID + o];
-      WG(W_R1_B)[o] = acc;
-    }
-    const float* Wr = WP(W_R1_W);
-    for (int i = tid; i < D.MSG; i += NT) {
-      float gw = 0.f, gb = 0.f;
-      for (int k = 0; k < D.E; ++k) {
-        float gu = 0.f;
-        for (int o = 0; o < D.HID; ++o)
-          gu += Wr[o * D.MSG + i] * gpre1_e[(long)k * D.HID + o];
-        gw += gu * P.xh_m1e[(long)k * D.MSG + i];
-        gb += gu;
-      }
-      for (int v = 0; v < D.N; ++v) {
-        float gu = 0.f;
-        for (int o = 0; o < D.HID; ++o)
-          gu += Wr[o * D.MSG + i] * gpre1_s[(long)v * D.HID + o];
-        gw += gu * P.xh_m1s[(long)v * D.MSG + i];
-        gb += gu;
-      }
-      WG(W_LN_R1_W)[i] = gw;
-      WG(W_LN_R1_B)[i] = gb;
-    }
-  }
-  // node module 1: Wn1[H,F0] over N rows (gpre in ghn1, relu-masked)
-  {
-    const float* gam = WP(W_LN_N1_W);
-    const float* bet = WP(W_LN_N1_B);
-    for (long u = tid; u < (long)D.H * D.F0; u += NT) {
-      int o = (int)(u / D.F0), i = (int)(u % D.F0);
-      float acc = 0.f;
-      for (int v = 0; v < D.N; ++v)
-        acc += P.ghn1[(long)v * D.H + o]
-               * (P.xh_z1[(long)v * D.F0 + i] * gam[i] + bet[i]);
-      WG(W_N1_W)[u] = acc;
-    }
-    for (int o = tid; o < D.H; o += NT) {
-      float acc = 0.f;
-      for (int v = 0; v < D.N; ++v) acc += P.ghn1[(long)v * D.H + o];
-      WG(W_N1_B)[o] = acc;
-    }
-    const float* Wn = WP(W_N1_W);
-    for (int i = tid; i < D.F0; i += NT) {
-      float gw = 0.f, gb = 0.f;
-      for (int v = 0; v < D.N; ++v) {
-        float gu = 0.f;
-        for (int o = 0; o < D.H; ++o)
-          gu += Wn[o * D.F0 + i] * P.ghn1[(long)v * D.H + o];
-        gw += gu * P.xh_z1[(long)v * D.F0 + i];
-        gb += gu;
-      }
-      WG(W_LN_N1_W)[i] = gw;
-      WG(W_LN_N1_B)[i] = gb;
-    }
-  }
-  // edge module 1: We1[H,FE] over E rows (gpre in ghe1)
-  {
-    const float* gam = WP(W_LN_E1_W);
-    const float* bet = WP(W_LN_E1_B);
-    for (long u = tid; u < (long)D.H * D.FE; u += NT) {
-      int o = (int)(u / D.FE), i = (int)(u % D.FE);
-      float acc = 0.f;
-      for (int k = 0; k < D.E; ++k)
-        acc += P.ghe1[(long)k * D.H + o]
-               * (P.xh_e1[(long)k * D.FE + i] * gam[i] + bet[i]);
-      WG(W_E1_W)[u] = acc;
-    }
-    for (int o = tid; o < D.H; o += NT) {
-      float acc = 0.f;
-      for (int k = 0; k < D.E; ++k) acc += P.ghe1[(long)k * D.H + o];
-      WG(W_E1_B)[o] = acc;
-    }
-    const float* We = WP(W_E1_W);
-    for (int i = tid; i < D.FE; i += NT) {
-      float gw = 0.f, gb = 0.f;
-      for (int k = 0; k < D.E; ++k) {
-        float gu = 0.f;
-        for (int o = 0; o < D.H; ++o)
-          gu += We[o * D.FE + i] * P.ghe1[(long)k * D.H + o];
-        gw += gu * P.xh_e1[(long)k * D.FE + i];
-        gb += gu;
-      }
-      WG(W_LN_E1_W)[i] = gw;
-      WG(W_LN_E1_B)[i] = gb;
+      for (int o = 0; o < KH; ++o)
+        acc += W[o * KFE + i] * P.ghe1[(long)k * KH + o];
+      P.gu_e1[(long)k * KFE + i] = acc;
     }
   }
 }
 
 // ---------------------------------------------------------------------------
-// host bindings
+// weight gradients: one block per weight-matrix job, LDS-tiled over rows.
+// Each job: gW[Dout,Din] = sum_r gpre[r,:] (x) u[r,:]  with u = xh*gam+bet,
+// rows drawn from one or two sources (edge rows ++ self rows); bias grads
+// accumulated from the same tiles; LN gamma/beta grads from stored gu rows.
+// Deterministic: fixed tile order, no atomics.
 // ---------------------------------------------------------------------------
+
+#define TILE_K 32
+
+struct WJob {
+  const float *gpre0, *xh0;   // source 0 rows
+  const float *gpre1, *xh1;   // source 1 rows (or null)
+  const float *gu0, *gu1;     // LN-out grad rows (for gamma/beta)
+  int rows0, rows1, Din, Dout;
+  int w_slot;                 // base slot (LN w, LN b, W, b)
+  int has_affine_u;           // u = xh*gam+bet (always true here)
+};
+
+__global__ void __launch_bounds__(256)
+cs_bwd_w_kernel(CachedPtrs P, CachedDims D) {
+  const int tid = threadIdx.x;
+  const int NT = blockDim.x;
+  // build this block's job
+  WJob J;
+  J.gpre1 = nullptr; J.xh1 = nullptr; J.gu1 = nullptr; J.rows1 = 0;
+  J.has_affine_u = 1;
+  switch (blockIdx.x) {
+    case 0:  // node module 1
+      J.gpre0 = P.ghn1; J.xh0 = P.xh_z1; J.gu0 = P.gu_z1;
+      J.rows0 = D.N; J.Din = KF0; J.Dout = KH; J.w_slot = W_LN_N1_W;
+      break;
+    case 1:  // edge module 1
+      J.gpre0 = P.ghe1; J.xh0 = P.xh_e1; J.gu0 = P.gu_e1;
+      J.rows0 = D.E; J.Din = KFE; J.Dout = KH; J.w_slot = W_LN_E1_W;
+      break;
+    case 2:  // reduce module 1 (edge ++ self rows)
+      J.gpre0 = P.gpre1_e; J.xh0 = P.xh_m1e; J.gu0 = P.gu_m1e;
+      J.rows0 = D.E;
+      J.gpre1 = P.gpre1_s; J.xh1 = P.xh_m1s; J.gu1 = P.gu_m1s;
+      J.rows1 = D.N;
+      J.Din = KMSG; J.Dout = KHID; J.w_slot = W_LN_R1_W;
+      break;
+    case 3:  // node module 2
+      J.gpre0 = P.gpn1; J.xh0 = P.xh_h2; J.gu0 = P.gu_h2;
+      J.rows0 = D.N; J.Din = KHID; J.Dout = KH; J.w_slot = W_LN_N2_W;
+      break;
+    case 4:  // edge module 2
+      J.gpre0 = P.gpe1; J.xh0 = P.xh_e2; J.gu0 = P.gu_e2;
+      J.rows0 = D.E; J.Din = KFE; J.Dout = KH; J.w_slot = W_LN_E2_W;
+      break;
+    case 5:  // reduce module 2
+      J.gpre0 = P.gpe2; J.xh0 = P.xh_m2e; J.gu0 = P.gu_m2e;
+      J.rows0 = D.E;
+      J.gpre1 = P.gpn2; J.xh1 = P.xh_m2s; J.gu1 = P.gu_m2s;
+      J.rows1 = D.N;
+      J.Din = KMSG; J.Dout = KOUT; J.w_slot = W_LN_R2_W;
+      break;
+    case 6:  // graph module (u rows over samples)
+      J.gpre0 = nullptr; break;   // handled specially below
+    default:
+      J.gpre0 = nullptr; break;   // FC branches handled specially below
+  }
+
+  __shared__ float tG[TILE_K][64 + 1];
+  __shared__ float tU[TILE_K][64 + 1];
+  __shared__ float tGU[TILE_K][64 + 1];
+  __shared__ float tXH[TILE_K][64 + 1];
+
+  if (blockIdx.x <= 5) {
+    const float* gam = WP(J.w_slot);
+    const float* bet = WP(J.w_slot + 1);
+    const int units = J.Dout * J.Din;
+    float acc[32];
+    const int my_units = (units + NT - 1) / NT;
+    for (int q = 0; q < my_units; ++q) acc[q] = 0.f;
+    float bacc = 0.f;                 // bias grad for o = tid (if tid < Dout)
+    float lacc_w = 0.f, lacc_b = 0.f; // LN grads for i = tid (if tid < Din)
+    const int rows = J.rows0 + J.rows1;
+    for (int r0 = 0; r0 < rows; r0 += TILE_K) {
+      const int rt = min(TILE_K, rows - r0);
+      // cooperative tile load
+      for (int x = tid; x < rt * J.Dout; x += NT) {
+        int t = x / J.Dout, o = x % J.Dout;
+        int r = r0 + t;
+        tG[t][o] = (r < J.rows0)
+            ? J.gpre0[(long)r * J.Dout + o]
+            : J.gpre1[(long)(r - J.rows0) * J.Dout + o];
+      }
+      for (int x = tid; x < rt * J.Din; x += NT) {
+        int t = x / J.Din, i = x % J.Din;
+        int r = r0 + t;
+        float xh = (r < J.rows0)
+            ? J.xh0[(long)r * J.Din + i]
+            : J.xh1[(long)(r - J.rows0) * J.Din + i];
+        tU[t][i] = xh * gam[i] + bet[i];
+        tXH[t][i] = xh;
+        float gu = (r < J.rows0)
+            ? J.gu0[(long)r * J.Din + i]
+            : J.gu1[(long)(r - J.rows0) * J.Din + i];
+        tGU[t][i] = gu;
+      }
+      __syncthreads();
+      for (int q = 0; q < my_units; ++q) {
+        int u = tid + q * NT;
+        if (u < units) {
+          int o = u / J.Din, i = u % J.Din;
+          float a = acc[q];
+#pragma unroll 8
+          for (int t = 0; t < rt; ++t) a += tG[t][o] * tU[t][i];
+          acc[q] = a;
+        }
+      }
+      if (tid < J.Dout) {
+        for (int t = 0; t < rt; ++t) bacc += tG[t][tid];
+      }
+      if (tid < J.Din) {
+        for (int t = 0; t < rt; ++t) {
+          lacc_w += tGU[t][tid] * tXH[t][tid];
+          lacc_b += tGU[t][tid];
+        }
+      }
+      __syncthreads();
+    }
+    for (int q = 0; q < my_units; ++q) {
+      int u = tid + q * NT;
+      if (u < units) WG_(J.w_slot + 2)[u] = acc[q];
+    }
+    if (tid < J.Dout) WG_(J.w_slot + 3)[tid] = bacc;
+    if (tid < J.Din) {
+      WG_(J.w_slot)[tid] = lacc_w;
+      WG_(J.w_slot + 1)[tid] = lacc_b;
+    }
+    return;
+  }
+
+  // ---- special jobs over the B sample rows ----
+  if (blockIdx.x == 6) {
+    // graph module: Wg[GEMB,GFin], bg; LN gamma/beta from gu34 + xh34
+    const float* gam = WP(W_LN_G_W);
+    const float* bet = WP(W_LN_G_B);
+    const int units = KGE * KGF;
+    for (int u = tid; u < units; u += NT) {
+      int o = u / KGF, i = u % KGF;
+      float a = 0.f;
+      for (int b = 0; b < D.B; ++b)
+        a += P.gfinal[(long)b * KFIN + KOUT + o]
+             * (P.xh34[(long)b * KGF + i] * gam[i] + bet[i]);
+      WG_(W_G_W)[u] = a;
+    }
+    for (int o = tid; o < KGE; o += NT) {
+      float a = 0.f;
+      for (int b = 0; b < D.B; ++b)
+        a += P.gfinal[(long)b * KFIN + KOUT + o];
+      WG_(W_G_B)[o] = a;
+    }
+    for (int i = tid; i < KGF; i += NT) {
+      float gw = 0.f, gb = 0.f;
+      for (int b = 0; b < D.B; ++b) {
+        const float gu = P.gu34[(long)b * KGF + i];
+        gw += gu * P.xh34[(long)b * KGF + i];
+        gb += gu;
+      }
+      WG_(W_LN_G_W)[i] = gw;
+      WG_(W_LN_G_B)[i] = gb;
+    }
+    return;
+  }
+  if (blockIdx.x == 7) {
+    // W2p[A,FC], b2p; W2v[FC], b2v — K = B, rows gh/h1
+    for (long u = tid; u < (long)KA * KFC; u += NT) {
+      int a = (int)(u / KFC), j = (int)(u % KFC);
+      float acc = 0.f;
+      for (int b = 0; b < D.B; ++b)
+        acc += P.glogits[(long)b * KA + a] * P.h1p[(long)b * KFC + j];
+      WG_(W_P2_W)[u] = acc;
+    }
+    for (int a = tid; a < KA; a += NT) {
+      float acc = 0.f;
+      for (int b = 0; b < D.B; ++b) acc += P.glogits[(long)b * KA + a];
+      WG_(W_P2_B)[a] = acc;
+    }
+    for (int j = tid; j < KFC; j += NT) {
+      float acc = 0.f;
+      for (int b = 0; b < D.B; ++b)
+        acc += P.gvalue[b] * P.h1v[(long)b * KFC + j];
+      WG_(W_V2_W)[j] = acc;
+    }
+    if (tid == 0) {
+      float acc = 0.f;
+      for (int b = 0; b < D.B; ++b) acc += P.gvalue[b];
+      WG_(W_V2_B)[0] = acc;
+    }
+    return;
+  }
+  // blocks 8..: W1p/W1v [FC,FIN] split FC-rows across blocks; K = B via LDS
+  {
+    const int nb = gridDim.x - 8;
+    const int jb = blockIdx.x - 8;
+    const int per = (KFC + nb - 1) / nb;
+    const int j0 = jb * per, j1 = min(KFC, j0 + per);
+    __shared__ float tF[TILE_K][32];
+    for (int r0 = 0; r0 < D.B; r0 += TILE_K) {
+      const int rt = min(TILE_K, D.B - r0);
+      for (int x = tid; x < rt * KFIN; x += NT) {
+        int t = x / KFIN, i = x % KFIN;
+        tF[t][i] = P.fin[(long)(r0 + t) * KFIN + i];
+      }
+      __syncthreads();
+      for (long u = tid; u < (long)(j1 - j0) * KFIN; u += NT) {
+        int j = j0 + (int)(u / KFIN), i = (int)(u % KFIN);
+        float ap = (r0 == 0) ? 0.f : WG_(W_P1_W)[(long)j * KFIN + i];
+        float av = (r0 == 0) ? 0.f : WG_(W_V1_W)[(long)j * KFIN + i];
+        for (int t = 0; t < rt; ++t) {
+          ap += P.gh1p[(long)(r0 + t) * KFC + j] * tF[t][i];
+          av += P.gh1v[(long)(r0 + t) * KFC + j] * tF[t][i];
+        }
+        WG_(W_P1_W)[(long)j * KFIN + i] = ap;
+        WG_(W_V1_W)[(long)j * KFIN + i] = av;
+      }
+      __syncthreads();
+    }
+    for (int j = j0 + tid; j < j1; j += NT) {
+      float ap = 0.f, av = 0.f;
+      for (int b = 0; b < D.B; ++b) {
+        ap += P.gh1p[(long)b * KFC + j];
+        av += P.gh1v[(long)b * KFC + j];
+      }
+      WG_(W_P1_B)[j] = ap;
+      WG_(W_V1_B)[j] = av;
+    }
+  }
+}
+
+// ===========================================================================
+// host bindings
+// ===========================================================================
 
 static void fill_ptrs(CachedPtrs& P, CachedDims& D,
                       std::vector<torch::Tensor>& T,
                       std::vector<double>& fs) {
-  TORCH_CHECK((int)T.size() == C_NT, "cached_step: tensor list size");
+  TORCH_CHECK((int)T.size() == C_NT, "cached_step: tensor list size ",
+              T.size(), " != ", (int)C_NT);
   D.N = (int)T[C_Z0].size(0);
   D.F0 = (int)T[C_Z0].size(1);
   D.E = (int)T[C_E].size(0);
@@ -918,9 +913,10 @@ static void fill_ptrs(CachedPtrs& P, CachedDims& D,
   D.vf_clip = (float)fs[1];
   D.vf_coef = (float)fs[2];
   D.ent_coef = (float)fs[3];
-  TORCH_CHECK(D.MSG <= 64 && D.HID <= 64 && D.OUT <= 64 && D.A <= 64
-              && D.F0 <= 64 && D.FE <= 64 && D.GFin <= 64 && D.FIN <= 64,
-              "cached_step: dims exceed 64");
+  TORCH_CHECK(D.F0 == KF0 && D.FE == KFE && D.H == KH && D.MSG == KMSG
+              && D.HID == KHID && D.OUT == KOUT && D.GFin == KGF
+              && D.GEMB == KGE && D.FIN == KFIN && D.FC == KFC && D.A == KA,
+              "cached_step: compiled for the tuned PAC-ML dims only");
   P.z0 = T[C_Z0].data_ptr<float>();
   P.e = T[C_E].data_ptr<float>();
   P.src = T[C_SRC].data_ptr<long>();
@@ -1002,6 +998,14 @@ static void fill_ptrs(CachedPtrs& P, CachedDims& D,
   P.gpe1 = T[C_GPE1].data_ptr<float>();
   P.gpre1_e = T[C_GPRE1E].data_ptr<float>();
   P.gpre1_s = T[C_GPRE1S].data_ptr<float>();
+  P.gu_m2e = T[C_GU_M2E].data_ptr<float>();
+  P.gu_m2s = T[C_GU_M2S].data_ptr<float>();
+  P.gu_h2 = T[C_GU_H2].data_ptr<float>();
+  P.gu_e2 = T[C_GU_E2].data_ptr<float>();
+  P.gu_m1e = T[C_GU_M1E].data_ptr<float>();
+  P.gu_m1s = T[C_GU_M1S].data_ptr<float>();
+  P.gu_z1 = T[C_GU_Z1].data_ptr<float>();
+  P.gu_e1 = T[C_GU_E1].data_ptr<float>();
 }
 
 void cached_step_fwd(std::vector<torch::Tensor> T, std::vector<double> fs) {
@@ -1009,8 +1013,13 @@ void cached_step_fwd(std::vector<torch::Tensor> T, std::vector<double> fs) {
   CachedDims D;
   fill_ptrs(P, D, T, fs);
   hipStream_t stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(cached_step_fwd_kernel, dim3(1), dim3(NTHREADS), 0,
+  hipLaunchKernelGGL(cs_fwd_gnn_kernel, dim3(1), dim3(1024), 0, stream, P, D);
+  int head_blocks = ((long)D.B * D.FC + 255) / 256;
+  if (head_blocks > 128) head_blocks = 128;
+  hipLaunchKernelGGL(cs_fwd_head_kernel, dim3(head_blocks), dim3(256), 0,
                      stream, P, D);
+  hipLaunchKernelGGL(cs_fwd_loss_kernel, dim3(1), dim3(1024), 0, stream,
+                     P, D);
 }
 
 void cached_step_bwd(std::vector<torch::Tensor> T, std::vector<double> fs) {
@@ -1018,6 +1027,11 @@ void cached_step_bwd(std::vector<torch::Tensor> T, std::vector<double> fs) {
   CachedDims D;
   fill_ptrs(P, D, T, fs);
   hipStream_t stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(cached_step_bwd_kernel, dim3(1), dim3(NTHREADS), 0,
-                     stream, P, D);
+  int hb = D.B < 32 ? D.B : 32;
+  hipLaunchKernelGGL(cs_bwd_head_kernel, dim3(hb), dim3(256), 0, stream,
+                     P, D);
+  hipLaunchKernelGGL(cs_bwd_gnn_kernel, dim3(1), dim3(1024), 0, stream,
+                     P, D);
+  hipLaunchKernelGGL(cs_bwd_w_kernel, dim3(8 + 16), dim3(256), 0, stream,
+                     P, D);
 }

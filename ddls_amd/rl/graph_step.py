@@ -74,13 +74,22 @@ class _FusedCachedEngine:
     @staticmethod
     def eligible(policy, ext) -> bool:
         cfg = policy.config
+        # the kernels are compiled for the tuned PAC-ML dims (cached_step.hip
+        # K* constants); other shapes take the autograd cached path
         return (ext is not None and hasattr(ext, "cached_step_fwd")
                 and cfg["module_depth"] == 1 and cfg["num_rounds"] == 2
-                and len(cfg["fcnet_hiddens"]) == 1
+                and cfg["fcnet_hiddens"] == [256]
                 and cfg["aggregator_activation"] == "relu"
                 and cfg["fcnet_activation"] == "relu"
                 and cfg["apply_action_mask"]
-                and policy.num_actions <= 64
+                and cfg["in_features_node"] == 5
+                and cfg["in_features_edge"] == 2
+                and cfg["in_features_graph"] == 17
+                and cfg["out_features_msg"] == 32
+                and cfg["out_features_hidden"] == 64
+                and cfg["out_features_node"] == 16
+                and cfg["out_features_graph"] == 8
+                and policy.num_actions == 17
                 and os.environ.get("DDLS_AMD_DISABLE_FUSED_STEP", "0") != "1")
 
     # weight slots in the kernel's fixed semantic order (enum W_* in
@@ -173,6 +182,9 @@ class _FusedCachedEngine:
             z(E, MSG), z(N, MSG), z(N, H), z(E, H),
             z(N, OUT), z(E, OUT), z(N, H), z(E, H),
             z(E, HID), z(N, HID),
+            # LN-out grad rows (gu) stored by the data-bwd kernels
+            z(E, MSG), z(N, MSG), z(N, HID), z(E, FE),
+            z(E, MSG), z(N, MSG), z(N, F0), z(E, FE),
         ]
         c = stepper.cfg
         self.fscal = [float(c.clip_param), float(c.vf_clip_param),
