@@ -126,79 +126,98 @@ struct CachedPtrs {
 #define WP(slot) (P.flat_p + P.offs[slot])
 #define WG_(slot) (P.flat_g + P.offs[slot])
 
-// LayerNorm + store xhat and rstd for the backward.  One thread per row.
-__device__ __forceinline__ void ln_row(const float* x, int Dd, float* xhat, float* rstd_out) {
+// LayerNorm into a REGISTER xhat array (fully unrolled: constant Dd), also
+// spilled to the global xhat row for the backward.  One thread per row.
+template <int Dd>
+__device__ __forceinline__ float ln_row_reg(const float* __restrict__ x,
+                                            float* __restrict__ xhat_reg,
+                                            float* __restrict__ xhat_out) {
   float mu = 0.f;
+#pragma unroll
   for (int i = 0; i < Dd; ++i) mu += x[i];
   mu /= Dd;
   float var = 0.f;
+#pragma unroll
   for (int i = 0; i < Dd; ++i) { float d = x[i] - mu; var += d * d; }
   var /= Dd;
   float rstd = rsqrtf(var + LN_EPS);
-  for (int i = 0; i < Dd; ++i) xhat[i] = (x[i] - mu) * rstd;
-  *rstd_out = rstd;
+#pragma unroll
+  for (int i = 0; i < Dd; ++i) {
+    xhat_reg[i] = (x[i] - mu) * rstd;
+    xhat_out[i] = xhat_reg[i];
+  }
+  return rstd;
 }
 
-// grad of LN wrt input given gu (grad on the affine output), also USES the
-// caller-provided gu buffer (stored for the gamma/beta reductions later)
-__device__ __forceinline__ void ln_bwd_row(const float* gu, const float* gamma,
-                           const float* xhat, float rstd, int Dd,
-                           float* gin) {
-  float m1 = 0.f, m2 = 0.f;
-  for (int i = 0; i < Dd; ++i) {
-    float gh = gu[i] * gamma[i];
-    m1 += gh;
-    m2 += gh * xhat[i];
-  }
-  m1 /= Dd;
-  m2 /= Dd;
-  for (int i = 0; i < Dd; ++i) {
-    float gh = gu[i] * gamma[i];
-    gin[i] = rstd * (gh - m1 - xhat[i] * m2);
-  }
-}
-
-// one MeanPool "row MLP": out = relu(Lin(LNaffine(x)));  LN xhat/rstd saved
-__device__ __forceinline__ void row_mlp_fwd(const CachedPtrs& P, const float* x, int Din,
-                            int Dout, int w_slot, float* xhat, float* rstd,
-                            float* out) {
-  ln_row(x, Din, xhat, rstd);
-  const float* gam = WP(w_slot);
-  const float* bet = WP(w_slot + 1);
-  const float* W = WP(w_slot + 2);
-  const float* b = WP(w_slot + 3);
+// one MeanPool "row MLP": out = relu(Lin(LNaffine(x))); x in registers
+template <int Din, int Dout>
+__device__ __forceinline__ void row_mlp_fwd_t(const CachedPtrs& P,
+                                              const float* __restrict__ x,
+                                              int w_slot,
+                                              float* __restrict__ xhat_out,
+                                              float* __restrict__ rstd_out,
+                                              float* __restrict__ out) {
+  float xh[Din];
+  *rstd_out = ln_row_reg<Din>(x, xh, xhat_out);
+  const float* __restrict__ gam = WP(w_slot);
+  const float* __restrict__ bet = WP(w_slot + 1);
+  const float* __restrict__ W = WP(w_slot + 2);
+  const float* __restrict__ b = WP(w_slot + 3);
+  float u[Din];
+#pragma unroll
+  for (int i = 0; i < Din; ++i) u[i] = xh[i] * gam[i] + bet[i];
+#pragma unroll 2
   for (int o = 0; o < Dout; ++o) {
     float acc = b[o];
-#pragma unroll 4
-    for (int i = 0; i < Din; ++i)
-      acc += W[o * Din + i] * (xhat[i] * gam[i] + bet[i]);
+#pragma unroll
+    for (int i = 0; i < Din; ++i) acc += W[o * Din + i] * u[i];
     out[o] = acc > 0.f ? acc : 0.f;
   }
 }
 
 // backward of one row-MLP row: gpre = gout*(out>0) stored; gu = W^T gpre
-// stored; optional input grad.
-__device__ __forceinline__ void row_mlp_bwd_row(const CachedPtrs& P, const float* gout,
-                                const float* out, const float* xhat,
-                                float rstd, int Din, int Dout, int w_slot,
-                                float* gpre_store, float* gu_store,
-                                float* gin /*or null*/) {
-  float gpre[64];
-  for (int o = 0; o < Dout; ++o)
+// stored; optional input grad via the LN backward.
+template <int Din, int Dout>
+__device__ __forceinline__ void row_mlp_bwd_row_t(
+    const CachedPtrs& P, const float* __restrict__ gout,
+    const float* __restrict__ out, const float* __restrict__ xhat,
+    float rstd, int w_slot, float* __restrict__ gpre_store,
+    float* __restrict__ gu_store, float* __restrict__ gin /*or null*/) {
+  float gpre[Dout];
+#pragma unroll
+  for (int o = 0; o < Dout; ++o) {
     gpre[o] = out[o] > 0.f ? gout[o] : 0.f;
-  for (int o = 0; o < Dout; ++o) gpre_store[o] = gpre[o];
-  const float* W = WP(w_slot + 2);
-  float gu[64];
+    gpre_store[o] = gpre[o];
+  }
+  const float* __restrict__ W = WP(w_slot + 2);
+  float gu[Din];
+#pragma unroll 2
   for (int i = 0; i < Din; ++i) {
     float acc = 0.f;
-#pragma unroll 4
+#pragma unroll
     for (int o = 0; o < Dout; ++o) acc += W[o * Din + i] * gpre[o];
     gu[i] = acc;
     gu_store[i] = acc;
   }
   if (gin != nullptr) {
-    const float* gam = WP(w_slot);
-    ln_bwd_row(gu, gam, xhat, rstd, Din, gin);
+    const float* __restrict__ gam = WP(w_slot);
+    float xh[Din];
+#pragma unroll
+    for (int i = 0; i < Din; ++i) xh[i] = xhat[i];
+    float m1 = 0.f, m2 = 0.f;
+#pragma unroll
+    for (int i = 0; i < Din; ++i) {
+      float gh = gu[i] * gam[i];
+      m1 += gh;
+      m2 += gh * xh[i];
+    }
+    m1 /= Din;
+    m2 /= Din;
+#pragma unroll
+    for (int i = 0; i < Din; ++i) {
+      float gh = gu[i] * gam[i];
+      gin[i] = rstd * (gh - m1 - xh[i] * m2);
+    }
   }
 }
 
@@ -213,31 +232,35 @@ cs_fwd_gnn_kernel(CachedPtrs P, CachedDims D) {
 
   // round-1 node/edge MLPs (thread per row)
   for (int v = tid; v < D.N; v += NT)
-    row_mlp_fwd(P, P.z0 + (long)v * KF0, KF0, KH, W_LN_N1_W,
-                P.xh_z1 + (long)v * KF0, P.rst_z1 + v,
-                P.hn1 + (long)v * KH);
+    row_mlp_fwd_t<KF0, KH>(P, P.z0 + (long)v * KF0, W_LN_N1_W,
+                           P.xh_z1 + (long)v * KF0, P.rst_z1 + v,
+                           P.hn1 + (long)v * KH);
   for (int k = tid; k < D.E; k += NT)
-    row_mlp_fwd(P, P.e + (long)k * KFE, KFE, KH, W_LN_E1_W,
-                P.xh_e1 + (long)k * KFE, P.rst_e1 + k,
-                P.he1 + (long)k * KH);
+    row_mlp_fwd_t<KFE, KH>(P, P.e + (long)k * KFE, W_LN_E1_W,
+                           P.xh_e1 + (long)k * KFE, P.rst_e1 + k,
+                           P.he1 + (long)k * KH);
   __syncthreads();
 
   for (int k = tid; k < D.E; k += NT) {
     const long s = P.src[k];
-    float msg[64];
+    float msg[KMSG];
+#pragma unroll
     for (int i = 0; i < KH; ++i) msg[i] = P.hn1[s * KH + i];
+#pragma unroll
     for (int i = 0; i < KH; ++i) msg[KH + i] = P.he1[(long)k * KH + i];
-    row_mlp_fwd(P, msg, KMSG, KHID, W_LN_R1_W,
-                P.xh_m1e + (long)k * KMSG, P.rst_m1e + k,
-                P.re1 + (long)k * KHID);
+    row_mlp_fwd_t<KMSG, KHID>(P, msg, W_LN_R1_W,
+                              P.xh_m1e + (long)k * KMSG, P.rst_m1e + k,
+                              P.re1 + (long)k * KHID);
   }
   for (int v = tid; v < D.N; v += NT) {
-    float msg[64];
+    float msg[KMSG];
+#pragma unroll
     for (int i = 0; i < KH; ++i) msg[i] = P.hn1[(long)v * KH + i];
+#pragma unroll
     for (int i = 0; i < KH; ++i) msg[KH + i] = 0.f;
-    row_mlp_fwd(P, msg, KMSG, KHID, W_LN_R1_W,
-                P.xh_m1s + (long)v * KMSG, P.rst_m1s + v,
-                P.rs1 + (long)v * KHID);
+    row_mlp_fwd_t<KMSG, KHID>(P, msg, W_LN_R1_W,
+                              P.xh_m1s + (long)v * KMSG, P.rst_m1s + v,
+                              P.rs1 + (long)v * KHID);
   }
   __syncthreads();
 
@@ -256,31 +279,35 @@ cs_fwd_gnn_kernel(CachedPtrs P, CachedDims D) {
   __syncthreads();
 
   for (int v = tid; v < D.N; v += NT)
-    row_mlp_fwd(P, P.h1 + (long)v * KHID, KHID, KH, W_LN_N2_W,
-                P.xh_h2 + (long)v * KHID, P.rst_h2 + v,
-                P.hn2 + (long)v * KH);
+    row_mlp_fwd_t<KHID, KH>(P, P.h1 + (long)v * KHID, W_LN_N2_W,
+                            P.xh_h2 + (long)v * KHID, P.rst_h2 + v,
+                            P.hn2 + (long)v * KH);
   for (int k = tid; k < D.E; k += NT)
-    row_mlp_fwd(P, P.e + (long)k * KFE, KFE, KH, W_LN_E2_W,
-                P.xh_e2 + (long)k * KFE, P.rst_e2 + k,
-                P.he2 + (long)k * KH);
+    row_mlp_fwd_t<KFE, KH>(P, P.e + (long)k * KFE, W_LN_E2_W,
+                           P.xh_e2 + (long)k * KFE, P.rst_e2 + k,
+                           P.he2 + (long)k * KH);
   __syncthreads();
 
   for (int k = tid; k < D.E; k += NT) {
     const long s = P.src[k];
-    float msg[64];
+    float msg[KMSG];
+#pragma unroll
     for (int i = 0; i < KH; ++i) msg[i] = P.hn2[s * KH + i];
+#pragma unroll
     for (int i = 0; i < KH; ++i) msg[KH + i] = P.he2[(long)k * KH + i];
-    row_mlp_fwd(P, msg, KMSG, KOUT, W_LN_R2_W,
-                P.xh_m2e + (long)k * KMSG, P.rst_m2e + k,
-                P.re2 + (long)k * KOUT);
+    row_mlp_fwd_t<KMSG, KOUT>(P, msg, W_LN_R2_W,
+                              P.xh_m2e + (long)k * KMSG, P.rst_m2e + k,
+                              P.re2 + (long)k * KOUT);
   }
   for (int v = tid; v < D.N; v += NT) {
-    float msg[64];
+    float msg[KMSG];
+#pragma unroll
     for (int i = 0; i < KH; ++i) msg[i] = P.hn2[(long)v * KH + i];
+#pragma unroll
     for (int i = 0; i < KH; ++i) msg[KH + i] = 0.f;
-    row_mlp_fwd(P, msg, KMSG, KOUT, W_LN_R2_W,
-                P.xh_m2s + (long)v * KMSG, P.rst_m2s + v,
-                P.rs2 + (long)v * KOUT);
+    row_mlp_fwd_t<KMSG, KOUT>(P, msg, W_LN_R2_W,
+                              P.xh_m2s + (long)v * KMSG, P.rst_m2s + v,
+                              P.rs2 + (long)v * KOUT);
   }
   __syncthreads();
 
@@ -311,19 +338,22 @@ cs_fwd_gnn_kernel(CachedPtrs P, CachedDims D) {
   for (int b = tid; b < D.B; b += NT) {
     const long mid = P.model_ids[b];
     float* fin = P.fin + (long)b * KFIN;
+#pragma unroll
     for (int i = 0; i < KOUT; ++i) fin[i] = P.pooled[mid * KOUT + i];
-    float* xh = P.xh34 + (long)b * KGF;
-    float rstd;
-    ln_row(P.gf + (long)b * KGF, KGF, xh, &rstd);
-    const float* gam = WP(W_LN_G_W);
-    const float* bet = WP(W_LN_G_B);
-    const float* Wg = WP(W_G_W);
-    const float* bg = WP(W_G_B);
+    float xh[KGF];
+    ln_row_reg<KGF>(P.gf + (long)b * KGF, xh, P.xh34 + (long)b * KGF);
+    const float* __restrict__ gam = WP(W_LN_G_W);
+    const float* __restrict__ bet = WP(W_LN_G_B);
+    const float* __restrict__ Wg = WP(W_G_W);
+    const float* __restrict__ bg = WP(W_G_B);
+    float u[KGF];
+#pragma unroll
+    for (int i = 0; i < KGF; ++i) u[i] = xh[i] * gam[i] + bet[i];
+#pragma unroll
     for (int o = 0; o < KGE; ++o) {
       float acc = bg[o];
-#pragma unroll 4
-      for (int i = 0; i < KGF; ++i)
-        acc += Wg[o * KGF + i] * (xh[i] * gam[i] + bet[i]);
+#pragma unroll
+      for (int i = 0; i < KGF; ++i) acc += Wg[o * KGF + i] * u[i];
       fin[KOUT + o] = acc;               // graph_module has NO activation
     }
   }
@@ -355,26 +385,35 @@ __global__ void __launch_bounds__(1024)
 cs_fwd_loss_kernel(CachedPtrs P, CachedDims D) {
   const int tid = threadIdx.x;
   const int NT = blockDim.x;
-  for (long u = tid; u < (long)D.B * KA; u += NT) {
-    int b = (int)(u / KA), a = (int)(u % KA);
-    const float* W2p = WP(W_P2_W);
-    float acc = WP(W_P2_B)[a];
-    const float* h = P.h1p + (long)b * KFC;
-#pragma unroll 8
-    for (int j = 0; j < KFC; ++j) acc += W2p[a * KFC + j] * h[j];
-    float mk = P.mask[(long)b * KA + a];
-    float lm = logf(mk);
-    const float fmin = -3.402823466e+38f;
-    if (!(lm > fmin)) lm = fmin;
-    P.p[u] = acc + lm;                    // raw masked logits (overwritten)
-  }
+  // stage W2p [A,FC] + W2v [FC] in LDS: each sample's h rows then stream
+  // ONCE from L2 while all A+1 output accumulators live in registers
+  __shared__ float sW2p[KA * KFC];
+  __shared__ float sW2v[KFC];
+  for (int x = tid; x < KA * KFC; x += NT) sW2p[x] = WP(W_P2_W)[x];
+  for (int x = tid; x < KFC; x += NT) sW2v[x] = WP(W_V2_W)[x];
+  __syncthreads();
   for (int b = tid; b < D.B; b += NT) {
-    const float* W2v = WP(W_V2_W);
-    float acc = WP(W_V2_B)[0];
-#pragma unroll 8
-    for (int j = 0; j < KFC; ++j)
-      acc += W2v[j] * P.h1v[(long)b * KFC + j];
-    P.values[b] = acc;
+    float acc[KA];
+#pragma unroll
+    for (int a = 0; a < KA; ++a) acc[a] = WP(W_P2_B)[a];
+    float av = WP(W_V2_B)[0];
+    const float* __restrict__ hp = P.h1p + (long)b * KFC;
+    const float* __restrict__ hv = P.h1v + (long)b * KFC;
+#pragma unroll 4
+    for (int j = 0; j < KFC; ++j) {
+      const float h = hp[j];
+#pragma unroll
+      for (int a = 0; a < KA; ++a) acc[a] += sW2p[a * KFC + j] * h;
+      av += sW2v[j] * hv[j];
+    }
+    const float fmin = -3.402823466e+38f;
+#pragma unroll
+    for (int a = 0; a < KA; ++a) {
+      float lm = logf(P.mask[(long)b * KA + a]);
+      if (!(lm > fmin)) lm = fmin;
+      P.p[(long)b * KA + a] = acc[a] + lm;   // raw masked logits
+    }
+    P.values[b] = av;
   }
   __syncthreads();
 
@@ -527,30 +566,28 @@ cs_bwd_gnn_kernel(CachedPtrs P, CachedDims D) {
     const int v = (int)P.dst[k];
     const long deg = P.indptr[v + 1] - P.indptr[v];
     const float f = 1.f / (float)(deg + 1);
-    float gr[64];
+    float gr[KOUT];
+#pragma unroll
     for (int i = 0; i < KOUT; ++i)
       gr[i] = P.gh2[(long)v * KOUT + i] * f;
-    row_mlp_bwd_row(P, gr, P.re2 + (long)k * KOUT,
-                    P.xh_m2e + (long)k * KMSG, P.rst_m2e[k], KMSG, KOUT,
-                    W_LN_R2_W, P.gpe2 + (long)k * KOUT,
-                    P.gu_m2e + (long)k * KMSG,
-                    P.gme2 + (long)k * KMSG);
+    row_mlp_bwd_row_t<KMSG, KOUT>(P, gr, P.re2 + (long)k * KOUT,
+                                  P.xh_m2e + (long)k * KMSG, P.rst_m2e[k],
+                                  W_LN_R2_W, P.gpe2 + (long)k * KOUT,
+                                  P.gu_m2e + (long)k * KMSG,
+                                  P.gme2 + (long)k * KMSG);
   }
   for (int v = tid; v < D.N; v += NT) {
     const long deg = P.indptr[v + 1] - P.indptr[v];
-    float gr[64];
-    if (deg > 0) {
-      const float f = 1.f / (float)(deg + 1);
-      for (int i = 0; i < KOUT; ++i)
-        gr[i] = P.gh2[(long)v * KOUT + i] * f;
-    } else {
-      for (int i = 0; i < KOUT; ++i) gr[i] = 0.f;
-    }
-    row_mlp_bwd_row(P, gr, P.rs2 + (long)v * KOUT,
-                    P.xh_m2s + (long)v * KMSG, P.rst_m2s[v], KMSG, KOUT,
-                    W_LN_R2_W, P.gpn2 + (long)v * KOUT,
-                    P.gu_m2s + (long)v * KMSG,
-                    P.gms2 + (long)v * KMSG);
+    float gr[KOUT];
+    const float f = (deg > 0) ? 1.f / (float)(deg + 1) : 0.f;
+#pragma unroll
+    for (int i = 0; i < KOUT; ++i)
+      gr[i] = P.gh2[(long)v * KOUT + i] * f;
+    row_mlp_bwd_row_t<KMSG, KOUT>(P, gr, P.rs2 + (long)v * KOUT,
+                                  P.xh_m2s + (long)v * KMSG, P.rst_m2s[v],
+                                  W_LN_R2_W, P.gpn2 + (long)v * KOUT,
+                                  P.gu_m2s + (long)v * KMSG,
+                                  P.gms2 + (long)v * KMSG);
   }
   __syncthreads();
 
@@ -571,16 +608,18 @@ cs_bwd_gnn_kernel(CachedPtrs P, CachedDims D) {
 
   // round-2 node/edge row-MLP data backward
   for (int v = tid; v < D.N; v += NT)
-    row_mlp_bwd_row(P, P.ghn2 + (long)v * KH, P.hn2 + (long)v * KH,
-                    P.xh_h2 + (long)v * KHID, P.rst_h2[v], KHID, KH,
-                    W_LN_N2_W, P.gpn1 + (long)v * KH,
-                    P.gu_h2 + (long)v * KHID,
-                    P.gh1b + (long)v * KHID);
+    row_mlp_bwd_row_t<KHID, KH>(P, P.ghn2 + (long)v * KH,
+                                P.hn2 + (long)v * KH,
+                                P.xh_h2 + (long)v * KHID, P.rst_h2[v],
+                                W_LN_N2_W, P.gpn1 + (long)v * KH,
+                                P.gu_h2 + (long)v * KHID,
+                                P.gh1b + (long)v * KHID);
   for (int k = tid; k < D.E; k += NT)
-    row_mlp_bwd_row(P, P.ghe2 + (long)k * KH, P.he2 + (long)k * KH,
-                    P.xh_e2 + (long)k * KFE, P.rst_e2[k], KFE, KH,
-                    W_LN_E2_W, P.gpe1 + (long)k * KH,
-                    P.gu_e2 + (long)k * KFE, nullptr);
+    row_mlp_bwd_row_t<KFE, KH>(P, P.ghe2 + (long)k * KH,
+                               P.he2 + (long)k * KH,
+                               P.xh_e2 + (long)k * KFE, P.rst_e2[k],
+                               W_LN_E2_W, P.gpe1 + (long)k * KH,
+                               P.gu_e2 + (long)k * KFE, nullptr);
   __syncthreads();
 
   // round-1 message grads
@@ -588,30 +627,28 @@ cs_bwd_gnn_kernel(CachedPtrs P, CachedDims D) {
     const int v = (int)P.dst[k];
     const long deg = P.indptr[v + 1] - P.indptr[v];
     const float f = 1.f / (float)(deg + 1);
-    float gr[64];
+    float gr[KHID];
+#pragma unroll
     for (int i = 0; i < KHID; ++i)
       gr[i] = P.gh1b[(long)v * KHID + i] * f;
-    row_mlp_bwd_row(P, gr, P.re1 + (long)k * KHID,
-                    P.xh_m1e + (long)k * KMSG, P.rst_m1e[k], KMSG, KHID,
-                    W_LN_R1_W, P.gpre1_e + (long)k * KHID,
-                    P.gu_m1e + (long)k * KMSG,
-                    P.gme1 + (long)k * KMSG);
+    row_mlp_bwd_row_t<KMSG, KHID>(P, gr, P.re1 + (long)k * KHID,
+                                  P.xh_m1e + (long)k * KMSG, P.rst_m1e[k],
+                                  W_LN_R1_W, P.gpre1_e + (long)k * KHID,
+                                  P.gu_m1e + (long)k * KMSG,
+                                  P.gme1 + (long)k * KMSG);
   }
   for (int v = tid; v < D.N; v += NT) {
     const long deg = P.indptr[v + 1] - P.indptr[v];
-    float gr[64];
-    if (deg > 0) {
-      const float f = 1.f / (float)(deg + 1);
-      for (int i = 0; i < KHID; ++i)
-        gr[i] = P.gh1b[(long)v * KHID + i] * f;
-    } else {
-      for (int i = 0; i < KHID; ++i) gr[i] = 0.f;
-    }
-    row_mlp_bwd_row(P, gr, P.rs1 + (long)v * KHID,
-                    P.xh_m1s + (long)v * KMSG, P.rst_m1s[v], KMSG, KHID,
-                    W_LN_R1_W, P.gpre1_s + (long)v * KHID,
-                    P.gu_m1s + (long)v * KMSG,
-                    P.gms1 + (long)v * KMSG);
+    float gr[KHID];
+    const float f = (deg > 0) ? 1.f / (float)(deg + 1) : 0.f;
+#pragma unroll
+    for (int i = 0; i < KHID; ++i)
+      gr[i] = P.gh1b[(long)v * KHID + i] * f;
+    row_mlp_bwd_row_t<KMSG, KHID>(P, gr, P.rs1 + (long)v * KHID,
+                                  P.xh_m1s + (long)v * KMSG, P.rst_m1s[v],
+                                  W_LN_R1_W, P.gpre1_s + (long)v * KHID,
+                                  P.gu_m1s + (long)v * KMSG,
+                                  P.gms1 + (long)v * KMSG);
   }
   __syncthreads();
 
@@ -632,20 +669,28 @@ cs_bwd_gnn_kernel(CachedPtrs P, CachedDims D) {
   __syncthreads();
   // gu rows for node/edge module 1 (inputs static, no input grads)
   for (int v = tid; v < D.N; v += NT) {
-    const float* W = WP(W_N1_W);
+    const float* __restrict__ W = WP(W_N1_W);
+    float g[KH];
+#pragma unroll
+    for (int o = 0; o < KH; ++o) g[o] = P.ghn1[(long)v * KH + o];
+#pragma unroll
     for (int i = 0; i < KF0; ++i) {
       float acc = 0.f;
-      for (int o = 0; o < KH; ++o)
-        acc += W[o * KF0 + i] * P.ghn1[(long)v * KH + o];
+#pragma unroll
+      for (int o = 0; o < KH; ++o) acc += W[o * KF0 + i] * g[o];
       P.gu_z1[(long)v * KF0 + i] = acc;
     }
   }
   for (int k = tid; k < D.E; k += NT) {
-    const float* W = WP(W_E1_W);
+    const float* __restrict__ W = WP(W_E1_W);
+    float g[KH];
+#pragma unroll
+    for (int o = 0; o < KH; ++o) g[o] = P.ghe1[(long)k * KH + o];
+#pragma unroll
     for (int i = 0; i < KFE; ++i) {
       float acc = 0.f;
-      for (int o = 0; o < KH; ++o)
-        acc += W[o * KFE + i] * P.ghe1[(long)k * KH + o];
+#pragma unroll
+      for (int o = 0; o < KH; ++o) acc += W[o * KFE + i] * g[o];
       P.gu_e1[(long)k * KFE + i] = acc;
     }
   }
@@ -661,6 +706,73 @@ cs_bwd_gnn_kernel(CachedPtrs P, CachedDims D) {
 
 #define TILE_K 32
 
+template <int Din, int Dout>
+__device__ __forceinline__ void wgrad_tiled(
+    const CachedPtrs& P, int tid, int NT, int rows0, int rows1,
+    const float* __restrict__ gpre0, const float* __restrict__ gpre1,
+    const float* __restrict__ xh0, const float* __restrict__ xh1,
+    const float* __restrict__ gu0, const float* __restrict__ gu1,
+    int w_slot, float (*tG)[65], float (*tU)[65], float (*tGU)[65],
+    float (*tXH)[65]) {
+  const float* __restrict__ gam = WP(w_slot);
+  const float* __restrict__ bet = WP(w_slot + 1);
+  constexpr int UNITS = Din * Dout;
+  constexpr int MYU = (UNITS + 255) / 256;
+  float acc[MYU];
+#pragma unroll
+  for (int q = 0; q < MYU; ++q) acc[q] = 0.f;
+  float bacc = 0.f, lacc_w = 0.f, lacc_b = 0.f;
+  const int rows = rows0 + rows1;
+  for (int r0 = 0; r0 < rows; r0 += TILE_K) {
+    const int rt = min(TILE_K, rows - r0);
+    for (int x = tid; x < rt * Dout; x += NT) {
+      int t = x / Dout, o = x % Dout;
+      int r = r0 + t;
+      tG[t][o] = (r < rows0) ? gpre0[(long)r * Dout + o]
+                             : gpre1[(long)(r - rows0) * Dout + o];
+    }
+    for (int x = tid; x < rt * Din; x += NT) {
+      int t = x / Din, i = x % Din;
+      int r = r0 + t;
+      float xh = (r < rows0) ? xh0[(long)r * Din + i]
+                             : xh1[(long)(r - rows0) * Din + i];
+      tU[t][i] = xh * gam[i] + bet[i];
+      tXH[t][i] = xh;
+      tGU[t][i] = (r < rows0) ? gu0[(long)r * Din + i]
+                              : gu1[(long)(r - rows0) * Din + i];
+    }
+    __syncthreads();
+#pragma unroll
+    for (int q = 0; q < MYU; ++q) {
+      int u = tid + q * 256;
+      if (u < UNITS) {
+        int o = u / Din, i = u % Din;
+        float a = acc[q];
+        for (int t = 0; t < rt; ++t) a += tG[t][o] * tU[t][i];
+        acc[q] = a;
+      }
+    }
+    if (tid < Dout)
+      for (int t = 0; t < rt; ++t) bacc += tG[t][tid];
+    if (tid < Din)
+      for (int t = 0; t < rt; ++t) {
+        lacc_w += tGU[t][tid] * tXH[t][tid];
+        lacc_b += tGU[t][tid];
+      }
+    __syncthreads();
+  }
+#pragma unroll
+  for (int q = 0; q < MYU; ++q) {
+    int u = tid + q * 256;
+    if (u < UNITS) WG_(w_slot + 2)[u] = acc[q];
+  }
+  if (tid < Dout) WG_(w_slot + 3)[tid] = bacc;
+  if (tid < Din) {
+    WG_(w_slot)[tid] = lacc_w;
+    WG_(w_slot + 1)[tid] = lacc_b;
+  }
+}
+
 struct WJob {
   const float *gpre0, *xh0;   // source 0 rows
   const float *gpre1, *xh1;   // source 1 rows (or null)
@@ -674,117 +786,44 @@ __global__ void __launch_bounds__(256)
 cs_bwd_w_kernel(CachedPtrs P, CachedDims D) {
   const int tid = threadIdx.x;
   const int NT = blockDim.x;
-  // build this block's job
-  WJob J;
-  J.gpre1 = nullptr; J.xh1 = nullptr; J.gu1 = nullptr; J.rows1 = 0;
-  J.has_affine_u = 1;
+  __shared__ float tG[TILE_K][65];
+  __shared__ float tU[TILE_K][65];
+  __shared__ float tGU[TILE_K][65];
+  __shared__ float tXH[TILE_K][65];
+
   switch (blockIdx.x) {
     case 0:  // node module 1
-      J.gpre0 = P.ghn1; J.xh0 = P.xh_z1; J.gu0 = P.gu_z1;
-      J.rows0 = D.N; J.Din = KF0; J.Dout = KH; J.w_slot = W_LN_N1_W;
-      break;
+      wgrad_tiled<KF0, KH>(P, tid, NT, D.N, 0, P.ghn1, nullptr, P.xh_z1,
+                           nullptr, P.gu_z1, nullptr, W_LN_N1_W, tG, tU,
+                           tGU, tXH);
+      return;
     case 1:  // edge module 1
-      J.gpre0 = P.ghe1; J.xh0 = P.xh_e1; J.gu0 = P.gu_e1;
-      J.rows0 = D.E; J.Din = KFE; J.Dout = KH; J.w_slot = W_LN_E1_W;
-      break;
+      wgrad_tiled<KFE, KH>(P, tid, NT, D.E, 0, P.ghe1, nullptr, P.xh_e1,
+                           nullptr, P.gu_e1, nullptr, W_LN_E1_W, tG, tU,
+                           tGU, tXH);
+      return;
     case 2:  // reduce module 1 (edge ++ self rows)
-      J.gpre0 = P.gpre1_e; J.xh0 = P.xh_m1e; J.gu0 = P.gu_m1e;
-      J.rows0 = D.E;
-      J.gpre1 = P.gpre1_s; J.xh1 = P.xh_m1s; J.gu1 = P.gu_m1s;
-      J.rows1 = D.N;
-      J.Din = KMSG; J.Dout = KHID; J.w_slot = W_LN_R1_W;
-      break;
+      wgrad_tiled<KMSG, KHID>(P, tid, NT, D.E, D.N, P.gpre1_e, P.gpre1_s,
+                              P.xh_m1e, P.xh_m1s, P.gu_m1e, P.gu_m1s,
+                              W_LN_R1_W, tG, tU, tGU, tXH);
+      return;
     case 3:  // node module 2
-      J.gpre0 = P.gpn1; J.xh0 = P.xh_h2; J.gu0 = P.gu_h2;
-      J.rows0 = D.N; J.Din = KHID; J.Dout = KH; J.w_slot = W_LN_N2_W;
-      break;
+      wgrad_tiled<KHID, KH>(P, tid, NT, D.N, 0, P.gpn1, nullptr, P.xh_h2,
+                            nullptr, P.gu_h2, nullptr, W_LN_N2_W, tG, tU,
+                            tGU, tXH);
+      return;
     case 4:  // edge module 2
-      J.gpre0 = P.gpe1; J.xh0 = P.xh_e2; J.gu0 = P.gu_e2;
-      J.rows0 = D.E; J.Din = KFE; J.Dout = KH; J.w_slot = W_LN_E2_W;
-      break;
+      wgrad_tiled<KFE, KH>(P, tid, NT, D.E, 0, P.gpe1, nullptr, P.xh_e2,
+                           nullptr, P.gu_e2, nullptr, W_LN_E2_W, tG, tU,
+                           tGU, tXH);
+      return;
     case 5:  // reduce module 2
-      J.gpre0 = P.gpe2; J.xh0 = P.xh_m2e; J.gu0 = P.gu_m2e;
-      J.rows0 = D.E;
-      J.gpre1 = P.gpn2; J.xh1 = P.xh_m2s; J.gu1 = P.gu_m2s;
-      J.rows1 = D.N;
-      J.Din = KMSG; J.Dout = KOUT; J.w_slot = W_LN_R2_W;
-      break;
-    case 6:  // graph module (u rows over samples)
-      J.gpre0 = nullptr; break;   // handled specially below
+      wgrad_tiled<KMSG, KOUT>(P, tid, NT, D.E, D.N, P.gpe2, P.gpn2,
+                              P.xh_m2e, P.xh_m2s, P.gu_m2e, P.gu_m2s,
+                              W_LN_R2_W, tG, tU, tGU, tXH);
+      return;
     default:
-      J.gpre0 = nullptr; break;   // FC branches handled specially below
-  }
-
-  __shared__ float tG[TILE_K][64 + 1];
-  __shared__ float tU[TILE_K][64 + 1];
-  __shared__ float tGU[TILE_K][64 + 1];
-  __shared__ float tXH[TILE_K][64 + 1];
-
-  if (blockIdx.x <= 5) {
-    const float* gam = WP(J.w_slot);
-    const float* bet = WP(J.w_slot + 1);
-    const int units = J.Dout * J.Din;
-    float acc[32];
-    const int my_units = (units + NT - 1) / NT;
-    for (int q = 0; q < my_units; ++q) acc[q] = 0.f;
-    float bacc = 0.f;                 // bias grad for o = tid (if tid < Dout)
-    float lacc_w = 0.f, lacc_b = 0.f; // LN grads for i = tid (if tid < Din)
-    const int rows = J.rows0 + J.rows1;
-    for (int r0 = 0; r0 < rows; r0 += TILE_K) {
-      const int rt = min(TILE_K, rows - r0);
-      // cooperative tile load
-      for (int x = tid; x < rt * J.Dout; x += NT) {
-        int t = x / J.Dout, o = x % J.Dout;
-        int r = r0 + t;
-        tG[t][o] = (r < J.rows0)
-            ? J.gpre0[(long)r * J.Dout + o]
-            : J.gpre1[(long)(r - J.rows0) * J.Dout + o];
-      }
-      for (int x = tid; x < rt * J.Din; x += NT) {
-        int t = x / J.Din, i = x % J.Din;
-        int r = r0 + t;
-        float xh = (r < J.rows0)
-            ? J.xh0[(long)r * J.Din + i]
-            : J.xh1[(long)(r - J.rows0) * J.Din + i];
-        tU[t][i] = xh * gam[i] + bet[i];
-        tXH[t][i] = xh;
-        float gu = (r < J.rows0)
-            ? J.gu0[(long)r * J.Din + i]
-            : J.gu1[(long)(r - J.rows0) * J.Din + i];
-        tGU[t][i] = gu;
-      }
-      __syncthreads();
-      for (int q = 0; q < my_units; ++q) {
-        int u = tid + q * NT;
-        if (u < units) {
-          int o = u / J.Din, i = u % J.Din;
-          float a = acc[q];
-#pragma unroll 8
-          for (int t = 0; t < rt; ++t) a += tG[t][o] * tU[t][i];
-          acc[q] = a;
-        }
-      }
-      if (tid < J.Dout) {
-        for (int t = 0; t < rt; ++t) bacc += tG[t][tid];
-      }
-      if (tid < J.Din) {
-        for (int t = 0; t < rt; ++t) {
-          lacc_w += tGU[t][tid] * tXH[t][tid];
-          lacc_b += tGU[t][tid];
-        }
-      }
-      __syncthreads();
-    }
-    for (int q = 0; q < my_units; ++q) {
-      int u = tid + q * NT;
-      if (u < units) WG_(J.w_slot + 2)[u] = acc[q];
-    }
-    if (tid < J.Dout) WG_(J.w_slot + 3)[tid] = bacc;
-    if (tid < J.Din) {
-      WG_(J.w_slot)[tid] = lacc_w;
-      WG_(J.w_slot + 1)[tid] = lacc_b;
-    }
-    return;
+      break;
   }
 
   // ---- special jobs over the B sample rows ----
