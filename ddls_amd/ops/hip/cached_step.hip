@@ -225,7 +225,7 @@ __device__ __forceinline__ void row_mlp_bwd_row_t(
 // forward
 // ===========================================================================
 
-__global__ void __launch_bounds__(1024)
+__global__ void __launch_bounds__(512)
 cs_fwd_gnn_kernel(CachedPtrs P, CachedDims D) {
   const int tid = threadIdx.x;
   const int NT = blockDim.x;
@@ -381,7 +381,7 @@ cs_fwd_head_kernel(CachedPtrs P, CachedDims D) {
 }
 
 // logits+value rows + loss (1 WG so the stats reduction stays in-block)
-__global__ void __launch_bounds__(1024)
+__global__ void __launch_bounds__(512)
 cs_fwd_loss_kernel(CachedPtrs P, CachedDims D) {
   const int tid = threadIdx.x;
   const int NT = blockDim.x;
@@ -539,7 +539,7 @@ cs_bwd_head_kernel(CachedPtrs P, CachedDims D) {
 }
 
 // pool grads + GNN data backward (graph-sized; 1 WG)
-__global__ void __launch_bounds__(1024)
+__global__ void __launch_bounds__(512)
 cs_bwd_gnn_kernel(CachedPtrs P, CachedDims D) {
   const int tid = threadIdx.x;
   const int NT = blockDim.x;
@@ -773,6 +773,53 @@ __device__ __forceinline__ void wgrad_tiled(
   }
 }
 
+template <int Din, int Dout>
+__device__ __forceinline__ void wgrad_plain(
+    const CachedPtrs& P, int tid, int NT, int rows,
+    const float* __restrict__ g, int g_stride,
+    const float* __restrict__ u, int u_stride,
+    int w_off_w, int w_off_b, float* lds) {
+  float* tG = lds;                       // [TILE_K][Dout]
+  float* tU = lds + TILE_K * Dout;       // [TILE_K][Din]
+  constexpr int UNITS = Din * Dout;
+  constexpr int MYU = (UNITS + 255) / 256;
+  float acc[MYU];
+#pragma unroll
+  for (int q = 0; q < MYU; ++q) acc[q] = 0.f;
+  float bacc = 0.f;
+  for (int r0 = 0; r0 < rows; r0 += TILE_K) {
+    const int rt = min(TILE_K, rows - r0);
+    for (int x = tid; x < rt * Dout; x += NT) {
+      int t = x / Dout, o = x % Dout;
+      tG[t * Dout + o] = g[(long)(r0 + t) * g_stride + o];
+    }
+    for (int x = tid; x < rt * Din; x += NT) {
+      int t = x / Din, i = x % Din;
+      tU[t * Din + i] = u[(long)(r0 + t) * u_stride + i];
+    }
+    __syncthreads();
+#pragma unroll
+    for (int q = 0; q < MYU; ++q) {
+      int uu = tid + q * 256;
+      if (uu < UNITS) {
+        int o = uu / Din, i = uu % Din;
+        float a = acc[q];
+        for (int t = 0; t < rt; ++t) a += tG[t * Dout + o] * tU[t * Din + i];
+        acc[q] = a;
+      }
+    }
+    if (tid < Dout)
+      for (int t = 0; t < rt; ++t) bacc += tG[t * Dout + tid];
+    __syncthreads();
+  }
+#pragma unroll
+  for (int q = 0; q < MYU; ++q) {
+    int uu = tid + q * 256;
+    if (uu < UNITS) (P.flat_g + w_off_w)[uu] = acc[q];
+  }
+  if (tid < Dout) (P.flat_g + w_off_b)[tid] = bacc;
+}
+
 struct WJob {
   const float *gpre0, *xh0;   // source 0 rows
   const float *gpre1, *xh1;   // source 1 rows (or null)
@@ -859,19 +906,11 @@ cs_bwd_w_kernel(CachedPtrs P, CachedDims D) {
     return;
   }
   if (blockIdx.x == 7) {
-    // W2p[A,FC], b2p; W2v[FC], b2v — K = B, rows gh/h1
-    for (long u = tid; u < (long)KA * KFC; u += NT) {
-      int a = (int)(u / KFC), j = (int)(u % KFC);
-      float acc = 0.f;
-      for (int b = 0; b < D.B; ++b)
-        acc += P.glogits[(long)b * KA + a] * P.h1p[(long)b * KFC + j];
-      WG_(W_P2_W)[u] = acc;
-    }
-    for (int a = tid; a < KA; a += NT) {
-      float acc = 0.f;
-      for (int b = 0; b < D.B; ++b) acc += P.glogits[(long)b * KA + a];
-      WG_(W_P2_B)[a] = acc;
-    }
+    // W2p[A,FC] = sum_b glogits (x) h1p (LDS-tiled); b2p folded in
+    __shared__ float lbuf[TILE_K * (KA + KFC)];
+    wgrad_plain<KFC, KA>(P, tid, NT, D.B, P.glogits, KA, P.h1p, KFC,
+                         (int)P.offs[W_P2_W], (int)P.offs[W_P2_B], lbuf);
+    __syncthreads();
     for (int j = tid; j < KFC; j += NT) {
       float acc = 0.f;
       for (int b = 0; b < D.B; ++b)
@@ -1052,12 +1091,12 @@ void cached_step_fwd(std::vector<torch::Tensor> T, std::vector<double> fs) {
   CachedDims D;
   fill_ptrs(P, D, T, fs);
   hipStream_t stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(cs_fwd_gnn_kernel, dim3(1), dim3(1024), 0, stream, P, D);
+  hipLaunchKernelGGL(cs_fwd_gnn_kernel, dim3(1), dim3(512), 0, stream, P, D);
   int head_blocks = ((long)D.B * D.FC + 255) / 256;
   if (head_blocks > 128) head_blocks = 128;
   hipLaunchKernelGGL(cs_fwd_head_kernel, dim3(head_blocks), dim3(256), 0,
                      stream, P, D);
-  hipLaunchKernelGGL(cs_fwd_loss_kernel, dim3(1), dim3(1024), 0, stream,
+  hipLaunchKernelGGL(cs_fwd_loss_kernel, dim3(1), dim3(512), 0, stream,
                      P, D);
 }
 
@@ -1069,7 +1108,7 @@ void cached_step_bwd(std::vector<torch::Tensor> T, std::vector<double> fs) {
   int hb = D.B < 32 ? D.B : 32;
   hipLaunchKernelGGL(cs_bwd_head_kernel, dim3(hb), dim3(256), 0, stream,
                      P, D);
-  hipLaunchKernelGGL(cs_bwd_gnn_kernel, dim3(1), dim3(1024), 0, stream,
+  hipLaunchKernelGGL(cs_bwd_gnn_kernel, dim3(1), dim3(512), 0, stream,
                      P, D);
   hipLaunchKernelGGL(cs_bwd_w_kernel, dim3(8 + 16), dim3(256), 0, stream,
                      P, D);
