@@ -225,106 +225,103 @@ __device__ __forceinline__ void row_mlp_bwd_row_t(
 // forward
 // ===========================================================================
 
+#define GSTRIDE for (long u = (long)blockIdx.x * blockDim.x + threadIdx.x; \
+                            u < total; u += (long)gridDim.x * blockDim.x)
+
+// round MLPs: rows 0..N-1 are node rows, N..N+E-1 are edge rows
+template <int ROUND>
+__global__ void __launch_bounds__(256)
+cs_fwd_mlp_kernel(CachedPtrs P, CachedDims D) {
+  const long total = D.N + D.E;
+  GSTRIDE {
+    if (u < D.N) {
+      const long v = u;
+      if (ROUND == 1)
+        row_mlp_fwd_t<KF0, KH>(P, P.z0 + v * KF0, W_LN_N1_W,
+                               P.xh_z1 + v * KF0, P.rst_z1 + v,
+                               P.hn1 + v * KH);
+      else
+        row_mlp_fwd_t<KHID, KH>(P, P.h1 + v * KHID, W_LN_N2_W,
+                                P.xh_h2 + v * KHID, P.rst_h2 + v,
+                                P.hn2 + v * KH);
+    } else {
+      const long k = u - D.N;
+      if (ROUND == 1)
+        row_mlp_fwd_t<KFE, KH>(P, P.e + k * KFE, W_LN_E1_W,
+                               P.xh_e1 + k * KFE, P.rst_e1 + k,
+                               P.he1 + k * KH);
+      else
+        row_mlp_fwd_t<KFE, KH>(P, P.e + k * KFE, W_LN_E2_W,
+                               P.xh_e2 + k * KFE, P.rst_e2 + k,
+                               P.he2 + k * KH);
+    }
+  }
+}
+
+template <int ROUND>
+__global__ void __launch_bounds__(256)
+cs_fwd_reduce_kernel(CachedPtrs P, CachedDims D) {
+  const float* hn = (ROUND == 1) ? P.hn1 : P.hn2;
+  const float* he = (ROUND == 1) ? P.he1 : P.he2;
+  const long total = D.E + D.N;
+  GSTRIDE {
+    float msg[KMSG];
+    if (u < D.E) {
+      const long k = u;
+      const long sv = P.src[k];
+#pragma unroll
+      for (int i = 0; i < KH; ++i) msg[i] = hn[sv * KH + i];
+#pragma unroll
+      for (int i = 0; i < KH; ++i) msg[KH + i] = he[k * KH + i];
+      if (ROUND == 1)
+        row_mlp_fwd_t<KMSG, KHID>(P, msg, W_LN_R1_W, P.xh_m1e + k * KMSG,
+                                  P.rst_m1e + k, P.re1 + k * KHID);
+      else
+        row_mlp_fwd_t<KMSG, KOUT>(P, msg, W_LN_R2_W, P.xh_m2e + k * KMSG,
+                                  P.rst_m2e + k, P.re2 + k * KOUT);
+    } else {
+      const long v = u - D.E;
+#pragma unroll
+      for (int i = 0; i < KH; ++i) msg[i] = hn[v * KH + i];
+#pragma unroll
+      for (int i = 0; i < KH; ++i) msg[KH + i] = 0.f;
+      if (ROUND == 1)
+        row_mlp_fwd_t<KMSG, KHID>(P, msg, W_LN_R1_W, P.xh_m1s + v * KMSG,
+                                  P.rst_m1s + v, P.rs1 + v * KHID);
+      else
+        row_mlp_fwd_t<KMSG, KOUT>(P, msg, W_LN_R2_W, P.xh_m2s + v * KMSG,
+                                  P.rst_m2s + v, P.rs2 + v * KOUT);
+    }
+  }
+}
+
+template <int ROUND>
+__global__ void __launch_bounds__(256)
+cs_fwd_combine_kernel(CachedPtrs P, CachedDims D) {
+  const int DO = (ROUND == 1) ? KHID : KOUT;
+  const float* re = (ROUND == 1) ? P.re1 : P.re2;
+  const float* rs = (ROUND == 1) ? P.rs1 : P.rs2;
+  float* h = (ROUND == 1) ? P.h1 : P.h2;
+  const long total = (long)D.N * DO;
+  GSTRIDE {
+    int v = (int)(u / DO), i = (int)(u % DO);
+    long lo = P.indptr[v], hi = P.indptr[v + 1];
+    float out = 0.f;
+    if (hi > lo) {
+      float acc = rs[(long)v * DO + i];
+      for (long q = lo; q < hi; ++q)
+        acc += re[P.order[q] * DO + i];
+      out = acc / (float)(hi - lo + 1);
+    }
+    h[(long)v * DO + i] = out;
+  }
+}
+
+// pool (M x OUT) then per-sample `final` (one WG: final depends on pooled)
 __global__ void __launch_bounds__(512)
-cs_fwd_gnn_kernel(CachedPtrs P, CachedDims D) {
+cs_fwd_pool_kernel(CachedPtrs P, CachedDims D) {
   const int tid = threadIdx.x;
   const int NT = blockDim.x;
-
-  // round-1 node/edge MLPs (thread per row)
-  for (int v = tid; v < D.N; v += NT)
-    row_mlp_fwd_t<KF0, KH>(P, P.z0 + (long)v * KF0, W_LN_N1_W,
-                           P.xh_z1 + (long)v * KF0, P.rst_z1 + v,
-                           P.hn1 + (long)v * KH);
-  for (int k = tid; k < D.E; k += NT)
-    row_mlp_fwd_t<KFE, KH>(P, P.e + (long)k * KFE, W_LN_E1_W,
-                           P.xh_e1 + (long)k * KFE, P.rst_e1 + k,
-                           P.he1 + (long)k * KH);
-  __syncthreads();
-
-  for (int k = tid; k < D.E; k += NT) {
-    const long s = P.src[k];
-    float msg[KMSG];
-#pragma unroll
-    for (int i = 0; i < KH; ++i) msg[i] = P.hn1[s * KH + i];
-#pragma unroll
-    for (int i = 0; i < KH; ++i) msg[KH + i] = P.he1[(long)k * KH + i];
-    row_mlp_fwd_t<KMSG, KHID>(P, msg, W_LN_R1_W,
-                              P.xh_m1e + (long)k * KMSG, P.rst_m1e + k,
-                              P.re1 + (long)k * KHID);
-  }
-  for (int v = tid; v < D.N; v += NT) {
-    float msg[KMSG];
-#pragma unroll
-    for (int i = 0; i < KH; ++i) msg[i] = P.hn1[(long)v * KH + i];
-#pragma unroll
-    for (int i = 0; i < KH; ++i) msg[KH + i] = 0.f;
-    row_mlp_fwd_t<KMSG, KHID>(P, msg, W_LN_R1_W,
-                              P.xh_m1s + (long)v * KMSG, P.rst_m1s + v,
-                              P.rs1 + (long)v * KHID);
-  }
-  __syncthreads();
-
-  for (long u = tid; u < (long)D.N * KHID; u += NT) {
-    int v = (int)(u / KHID), i = (int)(u % KHID);
-    long lo = P.indptr[v], hi = P.indptr[v + 1];
-    float out = 0.f;
-    if (hi > lo) {
-      float acc = P.rs1[(long)v * KHID + i];
-      for (long q = lo; q < hi; ++q)
-        acc += P.re1[P.order[q] * KHID + i];
-      out = acc / (float)(hi - lo + 1);
-    }
-    P.h1[(long)v * KHID + i] = out;
-  }
-  __syncthreads();
-
-  for (int v = tid; v < D.N; v += NT)
-    row_mlp_fwd_t<KHID, KH>(P, P.h1 + (long)v * KHID, W_LN_N2_W,
-                            P.xh_h2 + (long)v * KHID, P.rst_h2 + v,
-                            P.hn2 + (long)v * KH);
-  for (int k = tid; k < D.E; k += NT)
-    row_mlp_fwd_t<KFE, KH>(P, P.e + (long)k * KFE, W_LN_E2_W,
-                           P.xh_e2 + (long)k * KFE, P.rst_e2 + k,
-                           P.he2 + (long)k * KH);
-  __syncthreads();
-
-  for (int k = tid; k < D.E; k += NT) {
-    const long s = P.src[k];
-    float msg[KMSG];
-#pragma unroll
-    for (int i = 0; i < KH; ++i) msg[i] = P.hn2[s * KH + i];
-#pragma unroll
-    for (int i = 0; i < KH; ++i) msg[KH + i] = P.he2[(long)k * KH + i];
-    row_mlp_fwd_t<KMSG, KOUT>(P, msg, W_LN_R2_W,
-                              P.xh_m2e + (long)k * KMSG, P.rst_m2e + k,
-                              P.re2 + (long)k * KOUT);
-  }
-  for (int v = tid; v < D.N; v += NT) {
-    float msg[KMSG];
-#pragma unroll
-    for (int i = 0; i < KH; ++i) msg[i] = P.hn2[(long)v * KH + i];
-#pragma unroll
-    for (int i = 0; i < KH; ++i) msg[KH + i] = 0.f;
-    row_mlp_fwd_t<KMSG, KOUT>(P, msg, W_LN_R2_W,
-                              P.xh_m2s + (long)v * KMSG, P.rst_m2s + v,
-                              P.rs2 + (long)v * KOUT);
-  }
-  __syncthreads();
-
-  for (long u = tid; u < (long)D.N * KOUT; u += NT) {
-    int v = (int)(u / KOUT), i = (int)(u % KOUT);
-    long lo = P.indptr[v], hi = P.indptr[v + 1];
-    float out = 0.f;
-    if (hi > lo) {
-      float acc = P.rs2[(long)v * KOUT + i];
-      for (long q = lo; q < hi; ++q)
-        acc += P.re2[P.order[q] * KOUT + i];
-      out = acc / (float)(hi - lo + 1);
-    }
-    P.h2[(long)v * KOUT + i] = out;
-  }
-  __syncthreads();
-
   for (long u = tid; u < (long)D.M * KOUT; u += NT) {
     int m = (int)(u / KOUT), i = (int)(u % KOUT);
     long lo = P.model_nptr[m], hi = P.model_nptr[m + 1];
@@ -333,8 +330,6 @@ cs_fwd_gnn_kernel(CachedPtrs P, CachedDims D) {
     P.pooled[(long)m * KOUT + i] = acc / (float)(hi - lo);
   }
   __syncthreads();
-
-  // per-sample `final` = [pooled(model) ++ Lin(LN(gf))] (thread per sample)
   for (int b = tid; b < D.B; b += NT) {
     const long mid = P.model_ids[b];
     float* fin = P.fin + (long)b * KFIN;
@@ -346,14 +341,14 @@ cs_fwd_gnn_kernel(CachedPtrs P, CachedDims D) {
     const float* __restrict__ bet = WP(W_LN_G_B);
     const float* __restrict__ Wg = WP(W_G_W);
     const float* __restrict__ bg = WP(W_G_B);
-    float u[KGF];
+    float uu[KGF];
 #pragma unroll
-    for (int i = 0; i < KGF; ++i) u[i] = xh[i] * gam[i] + bet[i];
+    for (int i = 0; i < KGF; ++i) uu[i] = xh[i] * gam[i] + bet[i];
 #pragma unroll
     for (int o = 0; o < KGE; ++o) {
       float acc = bg[o];
 #pragma unroll
-      for (int i = 0; i < KGF; ++i) acc += Wg[o * KGF + i] * u[i];
+      for (int i = 0; i < KGF; ++i) acc += Wg[o * KGF + i] * uu[i];
       fin[KOUT + o] = acc;               // graph_module has NO activation
     }
   }
@@ -538,12 +533,11 @@ cs_bwd_head_kernel(CachedPtrs P, CachedDims D) {
   }
 }
 
-// pool grads + GNN data backward (graph-sized; 1 WG)
+// pool grads -> gh2 (1 WG: gh2 depends on gpool; both tiny)
 __global__ void __launch_bounds__(512)
-cs_bwd_gnn_kernel(CachedPtrs P, CachedDims D) {
+cs_bwd_pool_kernel(CachedPtrs P, CachedDims D) {
   const int tid = threadIdx.x;
   const int NT = blockDim.x;
-
   for (long u = tid; u < (long)D.M * KOUT; u += NT) {
     int m = (int)(u / KOUT), i = (int)(u % KOUT);
     float acc = 0.f;
@@ -559,139 +553,157 @@ cs_bwd_gnn_kernel(CachedPtrs P, CachedDims D) {
     const float nn = (float)(P.model_nptr[m + 1] - P.model_nptr[m]);
     P.gh2[u] = P.gpool[m * KOUT + i] / nn;
   }
-  __syncthreads();
+}
 
-  // round-2 message grads
-  for (int k = tid; k < D.E; k += NT) {
-    const int v = (int)P.dst[k];
-    const long deg = P.indptr[v + 1] - P.indptr[v];
-    const float f = 1.f / (float)(deg + 1);
-    float gr[KOUT];
+// reduce-module backward rows (edge rows then self rows)
+template <int ROUND>
+__global__ void __launch_bounds__(256)
+cs_bwd_reduce_kernel(CachedPtrs P, CachedDims D) {
+  const long total = D.E + D.N;
+  GSTRIDE {
+    if (ROUND == 2) {
+      float gr[KOUT];
+      if (u < D.E) {
+        const long k = u;
+        const int v = (int)P.dst[k];
+        const long deg = P.indptr[v + 1] - P.indptr[v];
+        const float f = 1.f / (float)(deg + 1);
 #pragma unroll
-    for (int i = 0; i < KOUT; ++i)
-      gr[i] = P.gh2[(long)v * KOUT + i] * f;
-    row_mlp_bwd_row_t<KMSG, KOUT>(P, gr, P.re2 + (long)k * KOUT,
-                                  P.xh_m2e + (long)k * KMSG, P.rst_m2e[k],
-                                  W_LN_R2_W, P.gpe2 + (long)k * KOUT,
-                                  P.gu_m2e + (long)k * KMSG,
-                                  P.gme2 + (long)k * KMSG);
-  }
-  for (int v = tid; v < D.N; v += NT) {
-    const long deg = P.indptr[v + 1] - P.indptr[v];
-    float gr[KOUT];
-    const float f = (deg > 0) ? 1.f / (float)(deg + 1) : 0.f;
+        for (int i = 0; i < KOUT; ++i)
+          gr[i] = P.gh2[(long)v * KOUT + i] * f;
+        row_mlp_bwd_row_t<KMSG, KOUT>(P, gr, P.re2 + k * KOUT,
+                                      P.xh_m2e + k * KMSG, P.rst_m2e[k],
+                                      W_LN_R2_W, P.gpe2 + k * KOUT,
+                                      P.gu_m2e + k * KMSG,
+                                      P.gme2 + k * KMSG);
+      } else {
+        const long v = u - D.E;
+        const long deg = P.indptr[v + 1] - P.indptr[v];
+        const float f = (deg > 0) ? 1.f / (float)(deg + 1) : 0.f;
 #pragma unroll
-    for (int i = 0; i < KOUT; ++i)
-      gr[i] = P.gh2[(long)v * KOUT + i] * f;
-    row_mlp_bwd_row_t<KMSG, KOUT>(P, gr, P.rs2 + (long)v * KOUT,
-                                  P.xh_m2s + (long)v * KMSG, P.rst_m2s[v],
-                                  W_LN_R2_W, P.gpn2 + (long)v * KOUT,
-                                  P.gu_m2s + (long)v * KMSG,
-                                  P.gms2 + (long)v * KMSG);
-  }
-  __syncthreads();
-
-  // scatter to hn2 / he2
-  for (long u = tid; u < (long)D.N * KH; u += NT) {
-    int v = (int)(u / KH), i = (int)(u % KH);
-    long lo = P.src_indptr[v], hi = P.src_indptr[v + 1];
-    float acc = P.gms2[(long)v * KMSG + i];
-    for (long q = lo; q < hi; ++q)
-      acc += P.gme2[P.src_order[q] * KMSG + i];
-    P.ghn2[u] = acc;
-  }
-  for (long u = tid; u < (long)D.E * KH; u += NT) {
-    int k = (int)(u / KH), i = (int)(u % KH);
-    P.ghe2[u] = P.gme2[(long)k * KMSG + KH + i];
-  }
-  __syncthreads();
-
-  // round-2 node/edge row-MLP data backward
-  for (int v = tid; v < D.N; v += NT)
-    row_mlp_bwd_row_t<KHID, KH>(P, P.ghn2 + (long)v * KH,
-                                P.hn2 + (long)v * KH,
-                                P.xh_h2 + (long)v * KHID, P.rst_h2[v],
-                                W_LN_N2_W, P.gpn1 + (long)v * KH,
-                                P.gu_h2 + (long)v * KHID,
-                                P.gh1b + (long)v * KHID);
-  for (int k = tid; k < D.E; k += NT)
-    row_mlp_bwd_row_t<KFE, KH>(P, P.ghe2 + (long)k * KH,
-                               P.he2 + (long)k * KH,
-                               P.xh_e2 + (long)k * KFE, P.rst_e2[k],
-                               W_LN_E2_W, P.gpe1 + (long)k * KH,
-                               P.gu_e2 + (long)k * KFE, nullptr);
-  __syncthreads();
-
-  // round-1 message grads
-  for (int k = tid; k < D.E; k += NT) {
-    const int v = (int)P.dst[k];
-    const long deg = P.indptr[v + 1] - P.indptr[v];
-    const float f = 1.f / (float)(deg + 1);
-    float gr[KHID];
+        for (int i = 0; i < KOUT; ++i)
+          gr[i] = P.gh2[v * KOUT + i] * f;
+        row_mlp_bwd_row_t<KMSG, KOUT>(P, gr, P.rs2 + v * KOUT,
+                                      P.xh_m2s + v * KMSG, P.rst_m2s[v],
+                                      W_LN_R2_W, P.gpn2 + v * KOUT,
+                                      P.gu_m2s + v * KMSG,
+                                      P.gms2 + v * KMSG);
+      }
+    } else {
+      float gr[KHID];
+      if (u < D.E) {
+        const long k = u;
+        const int v = (int)P.dst[k];
+        const long deg = P.indptr[v + 1] - P.indptr[v];
+        const float f = 1.f / (float)(deg + 1);
 #pragma unroll
-    for (int i = 0; i < KHID; ++i)
-      gr[i] = P.gh1b[(long)v * KHID + i] * f;
-    row_mlp_bwd_row_t<KMSG, KHID>(P, gr, P.re1 + (long)k * KHID,
-                                  P.xh_m1e + (long)k * KMSG, P.rst_m1e[k],
-                                  W_LN_R1_W, P.gpre1_e + (long)k * KHID,
-                                  P.gu_m1e + (long)k * KMSG,
-                                  P.gme1 + (long)k * KMSG);
-  }
-  for (int v = tid; v < D.N; v += NT) {
-    const long deg = P.indptr[v + 1] - P.indptr[v];
-    float gr[KHID];
-    const float f = (deg > 0) ? 1.f / (float)(deg + 1) : 0.f;
+        for (int i = 0; i < KHID; ++i)
+          gr[i] = P.gh1b[(long)v * KHID + i] * f;
+        row_mlp_bwd_row_t<KMSG, KHID>(P, gr, P.re1 + k * KHID,
+                                      P.xh_m1e + k * KMSG, P.rst_m1e[k],
+                                      W_LN_R1_W, P.gpre1_e + k * KHID,
+                                      P.gu_m1e + k * KMSG,
+                                      P.gme1 + k * KMSG);
+      } else {
+        const long v = u - D.E;
+        const long deg = P.indptr[v + 1] - P.indptr[v];
+        const float f = (deg > 0) ? 1.f / (float)(deg + 1) : 0.f;
 #pragma unroll
-    for (int i = 0; i < KHID; ++i)
-      gr[i] = P.gh1b[(long)v * KHID + i] * f;
-    row_mlp_bwd_row_t<KMSG, KHID>(P, gr, P.rs1 + (long)v * KHID,
-                                  P.xh_m1s + (long)v * KMSG, P.rst_m1s[v],
-                                  W_LN_R1_W, P.gpre1_s + (long)v * KHID,
-                                  P.gu_m1s + (long)v * KMSG,
-                                  P.gms1 + (long)v * KMSG);
-  }
-  __syncthreads();
-
-  // scatter to hn1 / he1; relu-mask in place; store gu rows for LN grads
-  for (long u = tid; u < (long)D.N * KH; u += NT) {
-    int v = (int)(u / KH), i = (int)(u % KH);
-    long lo = P.src_indptr[v], hi = P.src_indptr[v + 1];
-    float acc = P.gms1[(long)v * KMSG + i];
-    for (long q = lo; q < hi; ++q)
-      acc += P.gme1[P.src_order[q] * KMSG + i];
-    P.ghn1[u] = P.hn1[u] > 0.f ? acc : 0.f;   // gpre rows for node module 1
-  }
-  for (long u = tid; u < (long)D.E * KH; u += NT) {
-    int k = (int)(u / KH), i = (int)(u % KH);
-    float g = P.gme1[(long)k * KMSG + KH + i];
-    P.ghe1[u] = P.he1[u] > 0.f ? g : 0.f;     // gpre rows for edge module 1
-  }
-  __syncthreads();
-  // gu rows for node/edge module 1 (inputs static, no input grads)
-  for (int v = tid; v < D.N; v += NT) {
-    const float* __restrict__ W = WP(W_N1_W);
-    float g[KH];
-#pragma unroll
-    for (int o = 0; o < KH; ++o) g[o] = P.ghn1[(long)v * KH + o];
-#pragma unroll
-    for (int i = 0; i < KF0; ++i) {
-      float acc = 0.f;
-#pragma unroll
-      for (int o = 0; o < KH; ++o) acc += W[o * KF0 + i] * g[o];
-      P.gu_z1[(long)v * KF0 + i] = acc;
+        for (int i = 0; i < KHID; ++i)
+          gr[i] = P.gh1b[v * KHID + i] * f;
+        row_mlp_bwd_row_t<KMSG, KHID>(P, gr, P.rs1 + v * KHID,
+                                      P.xh_m1s + v * KMSG, P.rst_m1s[v],
+                                      W_LN_R1_W, P.gpre1_s + v * KHID,
+                                      P.gu_m1s + v * KMSG,
+                                      P.gms1 + v * KMSG);
+      }
     }
   }
-  for (int k = tid; k < D.E; k += NT) {
-    const float* __restrict__ W = WP(W_E1_W);
-    float g[KH];
+}
+
+// scatter message grads back to hn/he (ROUND 2 plain; ROUND 1 also applies
+// the relu mask in place, producing the module-1 gpre rows)
+template <int ROUND>
+__global__ void __launch_bounds__(256)
+cs_bwd_scatter_kernel(CachedPtrs P, CachedDims D) {
+  const float* gme = (ROUND == 1) ? P.gme1 : P.gme2;
+  const float* gms = (ROUND == 1) ? P.gms1 : P.gms2;
+  const long total = (long)(D.N + D.E) * KH;
+  GSTRIDE {
+    if (u < (long)D.N * KH) {
+      int v = (int)(u / KH), i = (int)(u % KH);
+      long lo = P.src_indptr[v], hi = P.src_indptr[v + 1];
+      float acc = gms[(long)v * KMSG + i];
+      for (long q = lo; q < hi; ++q)
+        acc += gme[P.src_order[q] * KMSG + i];
+      if (ROUND == 1)
+        P.ghn1[u] = P.hn1[u] > 0.f ? acc : 0.f;
+      else
+        P.ghn2[u] = acc;
+    } else {
+      const long x = u - (long)D.N * KH;
+      int k = (int)(x / KH), i = (int)(x % KH);
+      float g = gme[(long)k * KMSG + KH + i];
+      if (ROUND == 1)
+        P.ghe1[x] = P.he1[x] > 0.f ? g : 0.f;
+      else
+        P.ghe2[x] = g;
+    }
+  }
+}
+
+// round-2 node/edge module data backward (node rows produce gh1)
+__global__ void __launch_bounds__(256)
+cs_bwd_mlp2_kernel(CachedPtrs P, CachedDims D) {
+  const long total = D.N + D.E;
+  GSTRIDE {
+    if (u < D.N) {
+      const long v = u;
+      row_mlp_bwd_row_t<KHID, KH>(P, P.ghn2 + v * KH, P.hn2 + v * KH,
+                                  P.xh_h2 + v * KHID, P.rst_h2[v],
+                                  W_LN_N2_W, P.gpn1 + v * KH,
+                                  P.gu_h2 + v * KHID, P.gh1b + v * KHID);
+    } else {
+      const long k = u - D.N;
+      row_mlp_bwd_row_t<KFE, KH>(P, P.ghe2 + k * KH, P.he2 + k * KH,
+                                 P.xh_e2 + k * KFE, P.rst_e2[k],
+                                 W_LN_E2_W, P.gpe1 + k * KH,
+                                 P.gu_e2 + k * KFE, nullptr);
+    }
+  }
+}
+
+// module-1 LN-out grad rows (gu = W^T gpre; inputs static, no input grads)
+__global__ void __launch_bounds__(256)
+cs_bwd_gu1_kernel(CachedPtrs P, CachedDims D) {
+  const long total = D.N + D.E;
+  GSTRIDE {
+    if (u < D.N) {
+      const long v = u;
+      const float* __restrict__ W = WP(W_N1_W);
+      float g[KH];
 #pragma unroll
-    for (int o = 0; o < KH; ++o) g[o] = P.ghe1[(long)k * KH + o];
+      for (int o = 0; o < KH; ++o) g[o] = P.ghn1[v * KH + o];
 #pragma unroll
-    for (int i = 0; i < KFE; ++i) {
-      float acc = 0.f;
+      for (int i = 0; i < KF0; ++i) {
+        float acc = 0.f;
 #pragma unroll
-      for (int o = 0; o < KH; ++o) acc += W[o * KFE + i] * g[o];
-      P.gu_e1[(long)k * KFE + i] = acc;
+        for (int o = 0; o < KH; ++o) acc += W[o * KF0 + i] * g[o];
+        P.gu_z1[v * KF0 + i] = acc;
+      }
+    } else {
+      const long k = u - D.N;
+      const float* __restrict__ W = WP(W_E1_W);
+      float g[KH];
+#pragma unroll
+      for (int o = 0; o < KH; ++o) g[o] = P.ghe1[k * KH + o];
+#pragma unroll
+      for (int i = 0; i < KFE; ++i) {
+        float acc = 0.f;
+#pragma unroll
+        for (int o = 0; o < KH; ++o) acc += W[o * KFE + i] * g[o];
+        P.gu_e1[k * KFE + i] = acc;
+      }
     }
   }
 }
@@ -1091,7 +1103,23 @@ void cached_step_fwd(std::vector<torch::Tensor> T, std::vector<double> fs) {
   CachedDims D;
   fill_ptrs(P, D, T, fs);
   hipStream_t stream = at::cuda::getCurrentCUDAStream();
-  hipLaunchKernelGGL(cs_fwd_gnn_kernel, dim3(1), dim3(512), 0, stream, P, D);
+  const int rows_b = (D.N + D.E + 255) / 256;
+  const int comb1_b = ((long)D.N * D.HID + 255) / 256;
+  const int comb2_b = ((long)D.N * D.OUT + 255) / 256;
+  hipLaunchKernelGGL(cs_fwd_mlp_kernel<1>, dim3(rows_b), dim3(256), 0,
+                     stream, P, D);
+  hipLaunchKernelGGL(cs_fwd_reduce_kernel<1>, dim3(rows_b), dim3(256), 0,
+                     stream, P, D);
+  hipLaunchKernelGGL(cs_fwd_combine_kernel<1>, dim3(comb1_b), dim3(256), 0,
+                     stream, P, D);
+  hipLaunchKernelGGL(cs_fwd_mlp_kernel<2>, dim3(rows_b), dim3(256), 0,
+                     stream, P, D);
+  hipLaunchKernelGGL(cs_fwd_reduce_kernel<2>, dim3(rows_b), dim3(256), 0,
+                     stream, P, D);
+  hipLaunchKernelGGL(cs_fwd_combine_kernel<2>, dim3(comb2_b), dim3(256), 0,
+                     stream, P, D);
+  hipLaunchKernelGGL(cs_fwd_pool_kernel, dim3(1), dim3(512), 0, stream,
+                     P, D);
   int head_blocks = ((long)D.B * D.FC + 255) / 256;
   if (head_blocks > 128) head_blocks = 128;
   hipLaunchKernelGGL(cs_fwd_head_kernel, dim3(head_blocks), dim3(256), 0,
@@ -1106,9 +1134,23 @@ void cached_step_bwd(std::vector<torch::Tensor> T, std::vector<double> fs) {
   fill_ptrs(P, D, T, fs);
   hipStream_t stream = at::cuda::getCurrentCUDAStream();
   int hb = D.B < 32 ? D.B : 32;
+  const int rows_b = (D.N + D.E + 255) / 256;
+  const int scat_b = ((long)(D.N + D.E) * D.H + 255) / 256;
   hipLaunchKernelGGL(cs_bwd_head_kernel, dim3(hb), dim3(256), 0, stream,
                      P, D);
-  hipLaunchKernelGGL(cs_bwd_gnn_kernel, dim3(1), dim3(512), 0, stream,
+  hipLaunchKernelGGL(cs_bwd_pool_kernel, dim3(1), dim3(512), 0, stream,
+                     P, D);
+  hipLaunchKernelGGL(cs_bwd_reduce_kernel<2>, dim3(rows_b), dim3(256), 0,
+                     stream, P, D);
+  hipLaunchKernelGGL(cs_bwd_scatter_kernel<2>, dim3(scat_b), dim3(256), 0,
+                     stream, P, D);
+  hipLaunchKernelGGL(cs_bwd_mlp2_kernel, dim3(rows_b), dim3(256), 0, stream,
+                     P, D);
+  hipLaunchKernelGGL(cs_bwd_reduce_kernel<1>, dim3(rows_b), dim3(256), 0,
+                     stream, P, D);
+  hipLaunchKernelGGL(cs_bwd_scatter_kernel<1>, dim3(scat_b), dim3(256), 0,
+                     stream, P, D);
+  hipLaunchKernelGGL(cs_bwd_gu1_kernel, dim3(rows_b), dim3(256), 0, stream,
                      P, D);
   hipLaunchKernelGGL(cs_bwd_w_kernel, dim3(8 + 16), dim3(256), 0, stream,
                      P, D);
