@@ -149,20 +149,15 @@ __device__ __forceinline__ float ln_row_reg(const float* __restrict__ x,
   return rstd;
 }
 
-// one MeanPool "row MLP": out = relu(Lin(LNaffine(x))); x in registers
+// one MeanPool "row MLP": out = relu(Lin(LNaffine(x))); weights from LDS
 template <int Din, int Dout>
-__device__ __forceinline__ void row_mlp_fwd_t(const CachedPtrs& P,
-                                              const float* __restrict__ x,
-                                              int w_slot,
-                                              float* __restrict__ xhat_out,
-                                              float* __restrict__ rstd_out,
-                                              float* __restrict__ out) {
+__device__ __forceinline__ void row_mlp_fwd_t(
+    const float* __restrict__ x, const float* __restrict__ gam,
+    const float* __restrict__ bet, const float* __restrict__ W,
+    const float* __restrict__ b, float* __restrict__ xhat_out,
+    float* __restrict__ rstd_out, float* __restrict__ out) {
   float xh[Din];
   *rstd_out = ln_row_reg<Din>(x, xh, xhat_out);
-  const float* __restrict__ gam = WP(w_slot);
-  const float* __restrict__ bet = WP(w_slot + 1);
-  const float* __restrict__ W = WP(w_slot + 2);
-  const float* __restrict__ b = WP(w_slot + 3);
   float u[Din];
 #pragma unroll
   for (int i = 0; i < Din; ++i) u[i] = xh[i] * gam[i] + bet[i];
@@ -175,13 +170,28 @@ __device__ __forceinline__ void row_mlp_fwd_t(const CachedPtrs& P,
   }
 }
 
+// cooperative LDS stage of one module's (gam, bet, W, b)
+template <int Din, int Dout>
+__device__ __forceinline__ void stage_module(const CachedPtrs& P, int w_slot,
+                                             int tid, int NT, float* sGam,
+                                             float* sBet, float* sW,
+                                             float* sB) {
+  for (int i = tid; i < Din; i += NT) {
+    sGam[i] = WP(w_slot)[i];
+    sBet[i] = WP(w_slot + 1)[i];
+  }
+  for (int x = tid; x < Din * Dout; x += NT) sW[x] = WP(w_slot + 2)[x];
+  for (int o = tid; o < Dout; o += NT) sB[o] = WP(w_slot + 3)[o];
+}
+
 // backward of one row-MLP row: gpre = gout*(out>0) stored; gu = W^T gpre
 // stored; optional input grad via the LN backward.
 template <int Din, int Dout>
 __device__ __forceinline__ void row_mlp_bwd_row_t(
-    const CachedPtrs& P, const float* __restrict__ gout,
+    const float* __restrict__ gout,
     const float* __restrict__ out, const float* __restrict__ xhat,
-    float rstd, int w_slot, float* __restrict__ gpre_store,
+    float rstd, const float* __restrict__ gam,
+    const float* __restrict__ W, float* __restrict__ gpre_store,
     float* __restrict__ gu_store, float* __restrict__ gin /*or null*/) {
   float gpre[Dout];
 #pragma unroll
@@ -189,7 +199,6 @@ __device__ __forceinline__ void row_mlp_bwd_row_t(
     gpre[o] = out[o] > 0.f ? gout[o] : 0.f;
     gpre_store[o] = gpre[o];
   }
-  const float* __restrict__ W = WP(w_slot + 2);
   float gu[Din];
 #pragma unroll 2
   for (int i = 0; i < Din; ++i) {
@@ -200,7 +209,6 @@ __device__ __forceinline__ void row_mlp_bwd_row_t(
     gu_store[i] = acc;
   }
   if (gin != nullptr) {
-    const float* __restrict__ gam = WP(w_slot);
     float xh[Din];
 #pragma unroll
     for (int i = 0; i < Din; ++i) xh[i] = xhat[i];
@@ -232,26 +240,35 @@ __device__ __forceinline__ void row_mlp_bwd_row_t(
 template <int ROUND>
 __global__ void __launch_bounds__(256)
 cs_fwd_mlp_kernel(CachedPtrs P, CachedDims D) {
+  constexpr int DN = (ROUND == 1) ? KF0 : KHID;
+  __shared__ float sGamN[DN], sBetN[DN], sWN[KH * DN], sBN[KH];
+  __shared__ float sGamE[KFE], sBetE[KFE], sWE[KH * KFE], sBE[KH];
+  const int tid = threadIdx.x, NT = blockDim.x;
+  stage_module<DN, KH>(P, (ROUND == 1) ? W_LN_N1_W : W_LN_N2_W, tid, NT,
+                       sGamN, sBetN, sWN, sBN);
+  stage_module<KFE, KH>(P, (ROUND == 1) ? W_LN_E1_W : W_LN_E2_W, tid, NT,
+                        sGamE, sBetE, sWE, sBE);
+  __syncthreads();
   const long total = D.N + D.E;
   GSTRIDE {
     if (u < D.N) {
       const long v = u;
       if (ROUND == 1)
-        row_mlp_fwd_t<KF0, KH>(P, P.z0 + v * KF0, W_LN_N1_W,
+        row_mlp_fwd_t<KF0, KH>(P.z0 + v * KF0, sGamN, sBetN, sWN, sBN,
                                P.xh_z1 + v * KF0, P.rst_z1 + v,
                                P.hn1 + v * KH);
       else
-        row_mlp_fwd_t<KHID, KH>(P, P.h1 + v * KHID, W_LN_N2_W,
+        row_mlp_fwd_t<KHID, KH>(P.h1 + v * KHID, sGamN, sBetN, sWN, sBN,
                                 P.xh_h2 + v * KHID, P.rst_h2 + v,
                                 P.hn2 + v * KH);
     } else {
       const long k = u - D.N;
       if (ROUND == 1)
-        row_mlp_fwd_t<KFE, KH>(P, P.e + k * KFE, W_LN_E1_W,
+        row_mlp_fwd_t<KFE, KH>(P.e + k * KFE, sGamE, sBetE, sWE, sBE,
                                P.xh_e1 + k * KFE, P.rst_e1 + k,
                                P.he1 + k * KH);
       else
-        row_mlp_fwd_t<KFE, KH>(P, P.e + k * KFE, W_LN_E2_W,
+        row_mlp_fwd_t<KFE, KH>(P.e + k * KFE, sGamE, sBetE, sWE, sBE,
                                P.xh_e2 + k * KFE, P.rst_e2 + k,
                                P.he2 + k * KH);
     }
@@ -261,6 +278,12 @@ cs_fwd_mlp_kernel(CachedPtrs P, CachedDims D) {
 template <int ROUND>
 __global__ void __launch_bounds__(256)
 cs_fwd_reduce_kernel(CachedPtrs P, CachedDims D) {
+  constexpr int DO = (ROUND == 1) ? KHID : KOUT;
+  __shared__ float sGam[KMSG], sBet[KMSG], sW[DO * KMSG], sB[DO];
+  const int tid = threadIdx.x, NT = blockDim.x;
+  stage_module<KMSG, DO>(P, (ROUND == 1) ? W_LN_R1_W : W_LN_R2_W, tid, NT,
+                         sGam, sBet, sW, sB);
+  __syncthreads();
   const float* hn = (ROUND == 1) ? P.hn1 : P.hn2;
   const float* he = (ROUND == 1) ? P.he1 : P.he2;
   const long total = D.E + D.N;
@@ -274,10 +297,12 @@ cs_fwd_reduce_kernel(CachedPtrs P, CachedDims D) {
 #pragma unroll
       for (int i = 0; i < KH; ++i) msg[KH + i] = he[k * KH + i];
       if (ROUND == 1)
-        row_mlp_fwd_t<KMSG, KHID>(P, msg, W_LN_R1_W, P.xh_m1e + k * KMSG,
+        row_mlp_fwd_t<KMSG, KHID>(msg, sGam, sBet, sW, sB,
+                                  P.xh_m1e + k * KMSG,
                                   P.rst_m1e + k, P.re1 + k * KHID);
       else
-        row_mlp_fwd_t<KMSG, KOUT>(P, msg, W_LN_R2_W, P.xh_m2e + k * KMSG,
+        row_mlp_fwd_t<KMSG, KOUT>(msg, sGam, sBet, sW, sB,
+                                  P.xh_m2e + k * KMSG,
                                   P.rst_m2e + k, P.re2 + k * KOUT);
     } else {
       const long v = u - D.E;
@@ -286,10 +311,12 @@ cs_fwd_reduce_kernel(CachedPtrs P, CachedDims D) {
 #pragma unroll
       for (int i = 0; i < KH; ++i) msg[KH + i] = 0.f;
       if (ROUND == 1)
-        row_mlp_fwd_t<KMSG, KHID>(P, msg, W_LN_R1_W, P.xh_m1s + v * KMSG,
+        row_mlp_fwd_t<KMSG, KHID>(msg, sGam, sBet, sW, sB,
+                                  P.xh_m1s + v * KMSG,
                                   P.rst_m1s + v, P.rs1 + v * KHID);
       else
-        row_mlp_fwd_t<KMSG, KOUT>(P, msg, W_LN_R2_W, P.xh_m2s + v * KMSG,
+        row_mlp_fwd_t<KMSG, KOUT>(msg, sGam, sBet, sW, sB,
+                                  P.xh_m2s + v * KMSG,
                                   P.rst_m2s + v, P.rs2 + v * KOUT);
     }
   }
@@ -474,8 +501,18 @@ cs_fwd_loss_kernel(CachedPtrs P, CachedDims D) {
 // the block.  256 threads cooperate FC-wise within each sample.
 __global__ void __launch_bounds__(256)
 cs_bwd_head_kernel(CachedPtrs P, CachedDims D) {
+  __shared__ float sW2p[KA * KFC], sW2v[KFC];
+  __shared__ float sW1p[KFC * KFIN], sW1v[KFC * KFIN], sWg[KGE * KGF];
   const int tid = threadIdx.x;
   const int NT = blockDim.x;
+  for (int x = tid; x < KA * KFC; x += NT) sW2p[x] = WP(W_P2_W)[x];
+  for (int x = tid; x < KFC; x += NT) sW2v[x] = WP(W_V2_W)[x];
+  for (int x = tid; x < KFC * KFIN; x += NT) {
+    sW1p[x] = WP(W_P1_W)[x];
+    sW1v[x] = WP(W_V1_W)[x];
+  }
+  for (int x = tid; x < KGE * KGF; x += NT) sWg[x] = WP(W_G_W)[x];
+  __syncthreads();
   const int per = (D.B + gridDim.x - 1) / gridDim.x;
   const int b0 = blockIdx.x * per;
   const int b1 = min(D.B, b0 + per);
@@ -500,33 +537,31 @@ cs_bwd_head_kernel(CachedPtrs P, CachedDims D) {
     // hidden grads
     for (int j = tid; j < KFC; j += NT) {
       const long u = (long)b * KFC + j;
-      const float* W2p = WP(W_P2_W);
       float acc = 0.f;
+#pragma unroll
       for (int a = 0; a < KA; ++a)
-        acc += W2p[a * KFC + j] * P.glogits[(long)b * KA + a];
+        acc += sW2p[a * KFC + j] * P.glogits[(long)b * KA + a];
       P.gh1p[u] = P.h1p[u] > 0.f ? acc : 0.f;
-      const float av = WP(W_V2_W)[j] * P.gvalue[b];
+      const float av = sW2v[j] * P.gvalue[b];
       P.gh1v[u] = P.h1v[u] > 0.f ? av : 0.f;
     }
     __syncthreads();
     // gfinal + graph-module LN-out grads
     for (int i = tid; i < KFIN; i += NT) {
-      const float* W1p = WP(W_P1_W);
-      const float* W1v = WP(W_V1_W);
       float acc = 0.f;
 #pragma unroll 8
       for (int j = 0; j < KFC; ++j) {
-        acc += W1p[j * KFIN + i] * P.gh1p[(long)b * KFC + j];
-        acc += W1v[j * KFIN + i] * P.gh1v[(long)b * KFC + j];
+        acc += sW1p[j * KFIN + i] * P.gh1p[(long)b * KFC + j];
+        acc += sW1v[j * KFIN + i] * P.gh1v[(long)b * KFC + j];
       }
       P.gfinal[(long)b * KFIN + i] = acc;
     }
     __syncthreads();
     for (int i = tid; i < KGF; i += NT) {
-      const float* Wg = WP(W_G_W);
       float acc = 0.f;
+#pragma unroll
       for (int o = 0; o < KGE; ++o)
-        acc += Wg[o * KGF + i] * P.gfinal[(long)b * KFIN + KOUT + o];
+        acc += sWg[o * KGF + i] * P.gfinal[(long)b * KFIN + KOUT + o];
       P.gu34[(long)b * KGF + i] = acc;
     }
     __syncthreads();
@@ -559,6 +594,13 @@ cs_bwd_pool_kernel(CachedPtrs P, CachedDims D) {
 template <int ROUND>
 __global__ void __launch_bounds__(256)
 cs_bwd_reduce_kernel(CachedPtrs P, CachedDims D) {
+  constexpr int DO = (ROUND == 1) ? KHID : KOUT;
+  constexpr int SLOT = (ROUND == 1) ? W_LN_R1_W : W_LN_R2_W;
+  __shared__ float sGam[KMSG], sW[DO * KMSG];
+  const int tid = threadIdx.x, NT = blockDim.x;
+  for (int i = tid; i < KMSG; i += NT) sGam[i] = WP(SLOT)[i];
+  for (int x = tid; x < DO * KMSG; x += NT) sW[x] = WP(SLOT + 2)[x];
+  __syncthreads();
   const long total = D.E + D.N;
   GSTRIDE {
     if (ROUND == 2) {
@@ -571,9 +613,9 @@ cs_bwd_reduce_kernel(CachedPtrs P, CachedDims D) {
 #pragma unroll
         for (int i = 0; i < KOUT; ++i)
           gr[i] = P.gh2[(long)v * KOUT + i] * f;
-        row_mlp_bwd_row_t<KMSG, KOUT>(P, gr, P.re2 + k * KOUT,
+        row_mlp_bwd_row_t<KMSG, KOUT>(gr, P.re2 + k * KOUT,
                                       P.xh_m2e + k * KMSG, P.rst_m2e[k],
-                                      W_LN_R2_W, P.gpe2 + k * KOUT,
+                                      sGam, sW, P.gpe2 + k * KOUT,
                                       P.gu_m2e + k * KMSG,
                                       P.gme2 + k * KMSG);
       } else {
@@ -583,9 +625,9 @@ cs_bwd_reduce_kernel(CachedPtrs P, CachedDims D) {
 #pragma unroll
         for (int i = 0; i < KOUT; ++i)
           gr[i] = P.gh2[v * KOUT + i] * f;
-        row_mlp_bwd_row_t<KMSG, KOUT>(P, gr, P.rs2 + v * KOUT,
+        row_mlp_bwd_row_t<KMSG, KOUT>(gr, P.rs2 + v * KOUT,
                                       P.xh_m2s + v * KMSG, P.rst_m2s[v],
-                                      W_LN_R2_W, P.gpn2 + v * KOUT,
+                                      sGam, sW, P.gpn2 + v * KOUT,
                                       P.gu_m2s + v * KMSG,
                                       P.gms2 + v * KMSG);
       }
@@ -599,9 +641,9 @@ cs_bwd_reduce_kernel(CachedPtrs P, CachedDims D) {
 #pragma unroll
         for (int i = 0; i < KHID; ++i)
           gr[i] = P.gh1b[(long)v * KHID + i] * f;
-        row_mlp_bwd_row_t<KMSG, KHID>(P, gr, P.re1 + k * KHID,
+        row_mlp_bwd_row_t<KMSG, KHID>(gr, P.re1 + k * KHID,
                                       P.xh_m1e + k * KMSG, P.rst_m1e[k],
-                                      W_LN_R1_W, P.gpre1_e + k * KHID,
+                                      sGam, sW, P.gpre1_e + k * KHID,
                                       P.gu_m1e + k * KMSG,
                                       P.gme1 + k * KMSG);
       } else {
@@ -611,9 +653,9 @@ cs_bwd_reduce_kernel(CachedPtrs P, CachedDims D) {
 #pragma unroll
         for (int i = 0; i < KHID; ++i)
           gr[i] = P.gh1b[v * KHID + i] * f;
-        row_mlp_bwd_row_t<KMSG, KHID>(P, gr, P.rs1 + v * KHID,
+        row_mlp_bwd_row_t<KMSG, KHID>(gr, P.rs1 + v * KHID,
                                       P.xh_m1s + v * KMSG, P.rst_m1s[v],
-                                      W_LN_R1_W, P.gpre1_s + v * KHID,
+                                      sGam, sW, P.gpre1_s + v * KHID,
                                       P.gu_m1s + v * KMSG,
                                       P.gms1 + v * KMSG);
       }
@@ -655,19 +697,27 @@ cs_bwd_scatter_kernel(CachedPtrs P, CachedDims D) {
 // round-2 node/edge module data backward (node rows produce gh1)
 __global__ void __launch_bounds__(256)
 cs_bwd_mlp2_kernel(CachedPtrs P, CachedDims D) {
+  __shared__ float sGamN[KHID], sWN[KH * KHID];
+  __shared__ float sGamE[KFE], sWE[KH * KFE];
+  const int tid = threadIdx.x, NT = blockDim.x;
+  for (int i = tid; i < KHID; i += NT) sGamN[i] = WP(W_LN_N2_W)[i];
+  for (int x = tid; x < KH * KHID; x += NT) sWN[x] = WP(W_N2_W)[x];
+  for (int i = tid; i < KFE; i += NT) sGamE[i] = WP(W_LN_E2_W)[i];
+  for (int x = tid; x < KH * KFE; x += NT) sWE[x] = WP(W_E2_W)[x];
+  __syncthreads();
   const long total = D.N + D.E;
   GSTRIDE {
     if (u < D.N) {
       const long v = u;
-      row_mlp_bwd_row_t<KHID, KH>(P, P.ghn2 + v * KH, P.hn2 + v * KH,
+      row_mlp_bwd_row_t<KHID, KH>(P.ghn2 + v * KH, P.hn2 + v * KH,
                                   P.xh_h2 + v * KHID, P.rst_h2[v],
-                                  W_LN_N2_W, P.gpn1 + v * KH,
+                                  sGamN, sWN, P.gpn1 + v * KH,
                                   P.gu_h2 + v * KHID, P.gh1b + v * KHID);
     } else {
       const long k = u - D.N;
-      row_mlp_bwd_row_t<KFE, KH>(P, P.ghe2 + k * KH, P.he2 + k * KH,
+      row_mlp_bwd_row_t<KFE, KH>(P.ghe2 + k * KH, P.he2 + k * KH,
                                  P.xh_e2 + k * KFE, P.rst_e2[k],
-                                 W_LN_E2_W, P.gpe1 + k * KH,
+                                 sGamE, sWE, P.gpe1 + k * KH,
                                  P.gu_e2 + k * KFE, nullptr);
     }
   }
@@ -676,11 +726,16 @@ cs_bwd_mlp2_kernel(CachedPtrs P, CachedDims D) {
 // module-1 LN-out grad rows (gu = W^T gpre; inputs static, no input grads)
 __global__ void __launch_bounds__(256)
 cs_bwd_gu1_kernel(CachedPtrs P, CachedDims D) {
+  __shared__ float sWN[KH * KF0], sWE[KH * KFE];
+  const int tid = threadIdx.x, NT = blockDim.x;
+  for (int x = tid; x < KH * KF0; x += NT) sWN[x] = WP(W_N1_W)[x];
+  for (int x = tid; x < KH * KFE; x += NT) sWE[x] = WP(W_E1_W)[x];
+  __syncthreads();
   const long total = D.N + D.E;
   GSTRIDE {
     if (u < D.N) {
       const long v = u;
-      const float* __restrict__ W = WP(W_N1_W);
+      const float* __restrict__ W = sWN;
       float g[KH];
 #pragma unroll
       for (int o = 0; o < KH; ++o) g[o] = P.ghn1[v * KH + o];
@@ -693,7 +748,7 @@ cs_bwd_gu1_kernel(CachedPtrs P, CachedDims D) {
       }
     } else {
       const long k = u - D.N;
-      const float* __restrict__ W = WP(W_E1_W);
+      const float* __restrict__ W = sWE;
       float g[KH];
 #pragma unroll
       for (int o = 0; o < KH; ++o) g[o] = P.ghe1[k * KH + o];
