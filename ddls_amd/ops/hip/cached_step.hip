@@ -403,10 +403,12 @@ cs_fwd_head_kernel(CachedPtrs P, CachedDims D) {
 }
 
 // logits+value rows + loss (1 WG so the stats reduction stays in-block)
-__global__ void __launch_bounds__(512)
+__global__ void __launch_bounds__(256)
 cs_fwd_loss_kernel(CachedPtrs P, CachedDims D) {
   const int tid = threadIdx.x;
   const int NT = blockDim.x;
+  const int b_lo = blockIdx.x * 32;
+  const int b_hi = min(D.B, b_lo + 32);
   // stage W2p [A,FC] + W2v [FC] in LDS: each sample's h rows then stream
   // ONCE from L2 while all A+1 output accumulators live in registers
   __shared__ float sW2p[KA * KFC];
@@ -414,7 +416,7 @@ cs_fwd_loss_kernel(CachedPtrs P, CachedDims D) {
   for (int x = tid; x < KA * KFC; x += NT) sW2p[x] = WP(W_P2_W)[x];
   for (int x = tid; x < KFC; x += NT) sW2v[x] = WP(W_V2_W)[x];
   __syncthreads();
-  for (int b = tid; b < D.B; b += NT) {
+  for (int b = b_lo + tid; b < b_hi; b += NT) {
     float acc[KA];
 #pragma unroll
     for (int a = 0; a < KA; ++a) acc[a] = WP(W_P2_B)[a];
@@ -442,7 +444,7 @@ cs_fwd_loss_kernel(CachedPtrs P, CachedDims D) {
   __shared__ float acc_s[4];
   if (tid < 4) acc_s[tid] = 0.f;
   __syncthreads();
-  for (int b = tid; b < D.B; b += NT) {
+  for (int b = b_lo + tid; b < b_hi; b += NT) {
     float* row = P.p + (long)b * KA;
     float mx = -3.0e38f;
     for (int a = 0; a < KA; ++a) mx = fmaxf(mx, row[a]);
@@ -479,16 +481,19 @@ cs_fwd_loss_kernel(CachedPtrs P, CachedDims D) {
   }
   __syncthreads();
   if (tid == 0) {
+    // cross-block accumulation: float atomics (stats are logging-only; the
+    // gradient path is deterministic elsewhere)
     const float inv = 1.f / D.B;
     const float pl = acc_s[0] * inv;
     const float vf = acc_s[1] * inv;
     const float kl = acc_s[2] * inv;
     const float ent = acc_s[3] * inv;
-    P.stats[0] += pl;
-    P.stats[1] += vf;
-    P.stats[2] += kl;
-    P.stats[3] += ent;
-    P.stats[4] += pl + P.kl[0] * kl + D.vf_coef * vf - D.ent_coef * ent;
+    atomicAdd(&P.stats[0], pl);
+    atomicAdd(&P.stats[1], vf);
+    atomicAdd(&P.stats[2], kl);
+    atomicAdd(&P.stats[3], ent);
+    atomicAdd(&P.stats[4],
+              pl + P.kl[0] * kl + D.vf_coef * vf - D.ent_coef * ent);
   }
 }
 
@@ -573,12 +578,15 @@ __global__ void __launch_bounds__(512)
 cs_bwd_pool_kernel(CachedPtrs P, CachedDims D) {
   const int tid = threadIdx.x;
   const int NT = blockDim.x;
+  __shared__ int smid[1024];
+  for (int b = tid; b < D.B; b += NT) smid[b] = (int)P.model_ids[b];
+  __syncthreads();
   for (long u = tid; u < (long)D.M * KOUT; u += NT) {
     int m = (int)(u / KOUT), i = (int)(u % KOUT);
     float acc = 0.f;
+#pragma unroll 8
     for (int b = 0; b < D.B; ++b)
-      if ((int)P.model_ids[b] == m)
-        acc += P.gfinal[(long)b * KFIN + i];
+      acc += (smid[b] == m) ? P.gfinal[(long)b * KFIN + i] : 0.f;
     P.gpool[u] = acc;
   }
   __syncthreads();
@@ -815,7 +823,12 @@ __device__ __forceinline__ void wgrad_tiled(
       if (u < UNITS) {
         int o = u / Din, i = u % Din;
         float a = acc[q];
-        for (int t = 0; t < rt; ++t) a += tG[t][o] * tU[t][i];
+        if (rt == TILE_K) {
+#pragma unroll
+          for (int t = 0; t < TILE_K; ++t) a += tG[t][o] * tU[t][i];
+        } else {
+          for (int t = 0; t < rt; ++t) a += tG[t][o] * tU[t][i];
+        }
         acc[q] = a;
       }
     }
@@ -871,7 +884,14 @@ __device__ __forceinline__ void wgrad_plain(
       if (uu < UNITS) {
         int o = uu / Din, i = uu % Din;
         float a = acc[q];
-        for (int t = 0; t < rt; ++t) a += tG[t * Dout + o] * tU[t * Din + i];
+        if (rt == TILE_K) {
+#pragma unroll
+          for (int t = 0; t < TILE_K; ++t)
+            a += tG[t * Dout + o] * tU[t * Din + i];
+        } else {
+          for (int t = 0; t < rt; ++t)
+            a += tG[t * Dout + o] * tU[t * Din + i];
+        }
         acc[q] = a;
       }
     }
@@ -991,32 +1011,64 @@ cs_bwd_w_kernel(CachedPtrs P, CachedDims D) {
     }
     return;
   }
-  // blocks 8..: W1p/W1v [FC,FIN] split FC-rows across blocks; K = B via LDS
+  // blocks 8..: W1p/W1v [FC,FIN] split FC-rows across blocks; LDS tiles of
+  // fin AND the gh1p/gh1v column slices
   {
     const int nb = gridDim.x - 8;
     const int jb = blockIdx.x - 8;
     const int per = (KFC + nb - 1) / nb;
     const int j0 = jb * per, j1 = min(KFC, j0 + per);
-    __shared__ float tF[TILE_K][32];
+    const int JW = j1 - j0;
+    __shared__ float tF[TILE_K][KFIN];
+    __shared__ float tGp[TILE_K][16], tGv[TILE_K][16];
+    constexpr int MYU = (16 * KFIN + 255) / 256;   // JW <= 16
+    float accp[MYU], accv[MYU];
+#pragma unroll
+    for (int q = 0; q < MYU; ++q) { accp[q] = 0.f; accv[q] = 0.f; }
     for (int r0 = 0; r0 < D.B; r0 += TILE_K) {
       const int rt = min(TILE_K, D.B - r0);
       for (int x = tid; x < rt * KFIN; x += NT) {
         int t = x / KFIN, i = x % KFIN;
         tF[t][i] = P.fin[(long)(r0 + t) * KFIN + i];
       }
-      __syncthreads();
-      for (long u = tid; u < (long)(j1 - j0) * KFIN; u += NT) {
-        int j = j0 + (int)(u / KFIN), i = (int)(u % KFIN);
-        float ap = (r0 == 0) ? 0.f : WG_(W_P1_W)[(long)j * KFIN + i];
-        float av = (r0 == 0) ? 0.f : WG_(W_V1_W)[(long)j * KFIN + i];
-        for (int t = 0; t < rt; ++t) {
-          ap += P.gh1p[(long)(r0 + t) * KFC + j] * tF[t][i];
-          av += P.gh1v[(long)(r0 + t) * KFC + j] * tF[t][i];
-        }
-        WG_(W_P1_W)[(long)j * KFIN + i] = ap;
-        WG_(W_V1_W)[(long)j * KFIN + i] = av;
+      for (int x = tid; x < rt * JW; x += NT) {
+        int t = x / JW, j = x % JW;
+        tGp[t][j] = P.gh1p[(long)(r0 + t) * KFC + j0 + j];
+        tGv[t][j] = P.gh1v[(long)(r0 + t) * KFC + j0 + j];
       }
       __syncthreads();
+#pragma unroll
+      for (int q = 0; q < MYU; ++q) {
+        int u = tid + q * NT;
+        if (u < JW * KFIN) {
+          int j = u / KFIN, i = u % KFIN;
+          float ap = accp[q], av = accv[q];
+          if (rt == TILE_K) {
+#pragma unroll
+            for (int t = 0; t < TILE_K; ++t) {
+              ap += tGp[t][j] * tF[t][i];
+              av += tGv[t][j] * tF[t][i];
+            }
+          } else {
+            for (int t = 0; t < rt; ++t) {
+              ap += tGp[t][j] * tF[t][i];
+              av += tGv[t][j] * tF[t][i];
+            }
+          }
+          accp[q] = ap;
+          accv[q] = av;
+        }
+      }
+      __syncthreads();
+    }
+#pragma unroll
+    for (int q = 0; q < MYU; ++q) {
+      int u = tid + q * NT;
+      if (u < JW * KFIN) {
+        int j = u / KFIN, i = u % KFIN;
+        WG_(W_P1_W)[(long)(j0 + j) * KFIN + i] = accp[q];
+        WG_(W_V1_W)[(long)(j0 + j) * KFIN + i] = accv[q];
+      }
     }
     for (int j = j0 + tid; j < j1; j += NT) {
       float ap = 0.f, av = 0.f;
@@ -1179,8 +1231,8 @@ void cached_step_fwd(std::vector<torch::Tensor> T, std::vector<double> fs) {
   if (head_blocks > 128) head_blocks = 128;
   hipLaunchKernelGGL(cs_fwd_head_kernel, dim3(head_blocks), dim3(256), 0,
                      stream, P, D);
-  hipLaunchKernelGGL(cs_fwd_loss_kernel, dim3(1), dim3(512), 0, stream,
-                     P, D);
+  hipLaunchKernelGGL(cs_fwd_loss_kernel, dim3((D.B + 31) / 32), dim3(256),
+                     0, stream, P, D);
 }
 
 void cached_step_bwd(std::vector<torch::Tensor> T, std::vector<double> fs) {
