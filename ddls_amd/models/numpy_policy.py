@@ -171,10 +171,19 @@ class NumpyGNNPolicy:
     def act(self, obs: CompactObs, rng: np.random.RandomState
             ) -> Tuple[int, float, float]:
         """Sample a masked action; returns (action, logp, value)."""
+        a, logp, value, _ = self.act_full(obs, rng)
+        return a, logp, value
+
+    def act_full(self, obs: CompactObs, rng: np.random.RandomState
+                 ) -> Tuple[int, float, float, np.ndarray]:
+        """act() plus the full behavior log-prob vector (for the analytic
+        KL(old || new) used by the adaptive KL coefficient)."""
         logits, value = self.forward(obs)
         x = logits - logits.max()
         p = np.exp(x)
-        p /= p.sum()
+        Z = p.sum()
+        p /= Z
+        lp_all = (x - np.log(Z)).astype(np.float32)
         a = int(rng.choice(len(p), p=p))
         logp = float(np.log(max(p[a], 1e-45)))
-        return a, logp, value
+        return a, logp, value, lp_all

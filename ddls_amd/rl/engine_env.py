@@ -140,6 +140,7 @@ class EngineVectorEnv:
         buf_gfull = torch.empty((T, B, 17 + A), device=dev)
         buf_model = torch.empty((T, B), dtype=torch.int64, device=dev)
         buf_actions = torch.empty((T, B), dtype=torch.int64, device=dev)
+        buf_lp_all = torch.empty((T, B, A), device=dev)
         buf_logp = torch.empty((T, B), device=dev)
         buf_values = torch.empty((T, B), device=dev)
         buf_rewards = torch.empty((T, B), device=dev)
@@ -157,6 +158,7 @@ class EngineVectorEnv:
             buf_gfull[t] = gfull
             buf_model[t] = obs_model
             buf_actions[t] = actions
+            buf_lp_all[t] = dist.logits   # normalized log-probs (Categorical)
             buf_logp[t] = dist.log_prob(actions)
             buf_values[t] = value
             eng.step(actions)          # syncs status internally
@@ -191,6 +193,7 @@ class EngineVectorEnv:
         return {
             "obs": obs_flat,
             "actions": buf_actions.cpu().numpy(),
+            "lp_all": buf_lp_all.cpu().numpy(),
             "logp": buf_logp.cpu().numpy(),
             "values": buf_values.cpu().numpy(),
             "rewards": buf_rewards.cpu().numpy(),

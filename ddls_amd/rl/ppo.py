@@ -110,7 +110,7 @@ class PPOTrainer:
                 lastgaelam = (delta + cfg.gamma * cfg.lambda_ * nonterminal
                               * lastgaelam)
                 adv[t] = lastgaelam
-            return {
+            out = {
                 "obs": data["obs"],
                 "actions": actions.reshape(-1),
                 "logp": logps.reshape(-1),
@@ -119,12 +119,20 @@ class PPOTrainer:
                 "values": values.reshape(-1),
                 "rewards": rewards.reshape(-1),
             }
+            if "lp_all" in data:
+                lp = np.asarray(data["lp_all"])
+                out["lp_all"] = lp.reshape(-1, lp.shape[-1])
+            return out
 
         obs_buf: List[CompactObs] = []
-        act_buf, logp_buf = [], []
+        act_buf, logp_buf, lp_all_buf = [], [], []
         rew_buf, done_buf, val_buf = [], [], []
         for _ in range(steps):
-            actions, logp, values, _ = self._policy_step(self.obs)
+            actions, logp, values, logits = self._policy_step(self.obs)
+            x = logits - logits.max(axis=-1, keepdims=True)
+            lp_all_buf.append(
+                (x - np.log(np.exp(x).sum(axis=-1, keepdims=True)))
+                .astype(np.float32))
             obs_buf.extend(self.obs)
             act_buf.append(actions)
             logp_buf.append(logp)
@@ -155,6 +163,7 @@ class PPOTrainer:
             adv[t] = lastgaelam
         value_targets = adv + values
 
+        lp_all = np.stack(lp_all_buf)
         return {
             "obs": obs_buf,
             "actions": actions.reshape(-1),
@@ -163,6 +172,7 @@ class PPOTrainer:
             "value_targets": value_targets.reshape(-1),
             "values": values.reshape(-1),
             "rewards": rewards.reshape(-1),
+            "lp_all": lp_all.reshape(-1, lp_all.shape[-1]),
         }
 
     # ------------------------------------------------------------------
@@ -336,16 +346,64 @@ class PPOTrainer:
         for k in stats:
             stats[k] /= max(num_updates, 1)
 
-        # adaptive KL coefficient (RLlib rule)
-        if stats["kl"] > 2.0 * cfg.kl_target:
+        # adaptive KL coefficient (RLlib rule) driven by the ANALYTIC
+        # KL(old || new) over the full old/new distributions (ADVICE r01:
+        # the taken-action sample estimate is noisy and can go negative)
+        kl_for_adapt = stats["kl"]
+        if batch.get("lp_all") is not None:
+            try:
+                kl_an = self._analytic_kl(batch)
+                stats["kl_analytic"] = kl_an
+                kl_for_adapt = kl_an
+            except Exception:
+                pass
+        if kl_for_adapt > 2.0 * cfg.kl_target:
             self.kl_coeff *= 1.5
-        elif stats["kl"] < 0.5 * cfg.kl_target:
+        elif kl_for_adapt < 0.5 * cfg.kl_target:
             self.kl_coeff *= 0.5
         stats["kl_coeff"] = self.kl_coeff
         if stepper is not None:
             stats["hipgraph_minibatches"] = num_captured
             stats["hipgraph_captures"] = stepper.capture_count
         return stats
+
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def _analytic_kl(self, batch: Dict) -> float:
+        """KL(old || new) = E_old[sum_a p_old (lp_old - lp_new)] over the
+        train batch, with the post-update policy."""
+        obs = batch["obs"]
+        old_lp = torch.as_tensor(batch["lp_all"], device=self.device)
+        mb_static = getattr(self.env, "_models_batch", None)
+        if mb_static is not None and getattr(obs[0], "model_id", -1) >= 0:
+            # cached-models forward: GNN once over the model graphs
+            from ..models.gnn import graph_mean
+            node_emb = self.policy.gnn(mb_static)
+            model_emb = graph_mean(node_emb, mb_static)
+            mids = torch.as_tensor([o.model_id for o in obs],
+                                   device=self.device)
+            gfull = torch.as_tensor(
+                np.stack([o.graph_features for o in obs]),
+                device=self.device)
+            mask = torch.as_tensor(np.stack([o.action_mask for o in obs]),
+                                   device=self.device)
+            graph_emb = self.policy.graph_module(gfull)
+            final = torch.cat([model_emb[mids], graph_emb], dim=-1)
+            logits = self.policy.policy_branch(final)
+            if self.policy.config["apply_action_mask"]:
+                logits = logits + torch.clamp(
+                    torch.log(mask), min=torch.finfo(torch.float32).min)
+            new_lp = F.log_softmax(logits, dim=-1)
+        else:
+            chunks = []
+            for start in range(0, len(obs), 1024):
+                inputs = collate(obs[start:start + 1024], self.device)
+                logits, _ = self._forward_flat(inputs)
+                chunks.append(F.log_softmax(logits, dim=-1))
+            new_lp = torch.cat(chunks)
+        p_old = torch.exp(old_lp)
+        kl = (p_old * (old_lp - new_lp)).sum(-1).mean()
+        return float(kl.item())
 
     # ------------------------------------------------------------------
     def train(self, num_steps: Optional[int] = None) -> Dict[str, float]:

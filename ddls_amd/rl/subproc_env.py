@@ -82,15 +82,19 @@ def _worker_loop(conn, env_fn_count: int, env_fn, base_seed: int,
                     obs_cache = [to_compact(env.reset(seed=base_seed + 1000 * i))
                                  for i, env in enumerate(envs)]
                 traj = {"obs": [], "actions": [], "logp": [], "values": [],
-                        "rewards": [], "dones": []}
+                        "rewards": [], "dones": [], "lp_all": []}
                 stats_out = []
                 n = len(envs)
+                num_actions = len(envs[0].action_set)
                 for _t in range(steps):
                     actions = np.zeros(n, dtype=np.int64)
                     logps = np.zeros(n, dtype=np.float32)
                     values = np.zeros(n, dtype=np.float32)
+                    lp_all = np.zeros((n, num_actions), dtype=np.float32)
                     for i, o in enumerate(obs_cache):
-                        actions[i], logps[i], values[i] = policy.act(o, rng)
+                        actions[i], logps[i], values[i], lp_all[i] = \
+                            policy.act_full(o, rng)
+                    traj["lp_all"].append(lp_all)
                     traj["obs"].append(list(obs_cache))
                     traj["actions"].append(actions)
                     traj["logp"].append(logps)
@@ -292,7 +296,7 @@ class SubprocVectorEnv:
             for traj in per_worker:
                 obs_flat.extend(traj["obs"][t])
         out = {"obs": obs_flat}
-        for key in ("actions", "logp", "values", "rewards", "dones"):
+        for key in ("actions", "logp", "values", "rewards", "dones", "lp_all"):
             out[key] = np.stack(
                 [np.concatenate([traj[key][t] for traj in per_worker])
                  for t in range(T)])
