@@ -152,6 +152,20 @@ def build_trainer_from_config(cfg: dict, device=None):
     torch.manual_seed(seed)
     policy = GNNPolicy(num_actions=num_actions,
                        config=model_cfg.get("custom_model_config"))
+    if algo.get("name", "ppo") == "impala":
+        from ..rl.impala import ImpalaConfig, ImpalaTrainer
+        imp_cfg = ImpalaConfig(
+            lr=algo.get("lr", 5e-4),
+            gamma=algo.get("gamma", 0.997),
+            vf_loss_coeff=algo.get("vf_loss_coeff", 0.5),
+            entropy_coeff=algo.get("entropy_coeff", 0.01),
+            grad_clip=algo.get("grad_clip", 40.0),
+            rho_clip=algo.get("vtrace_clip_rho_threshold", 1.0),
+            pg_rho_clip=algo.get("vtrace_clip_pg_rho_threshold", 1.0),
+            c_clip=algo.get("vtrace_clip_c_threshold", 1.0),
+            train_batch_size=algo.get("train_batch_size", 4000),
+            vtrace_drop_last_ts=algo.get("vtrace_drop_last_ts", True))
+        return ImpalaTrainer(venv, policy, imp_cfg, device=device)
     ppo_cfg = PPOConfig(
         lr=algo.get("lr", 2.785e-4),
         gamma=algo.get("gamma", 0.997),
