@@ -101,6 +101,7 @@ class PolicyActor:
                         config: Optional[dict] = None, **kwargs):
         from ..models.gnn import GNNPolicy
         state = Checkpointer.read(checkpoint_path)
+        state = state.get("trainer", state)   # launcher full-state layout
         policy = GNNPolicy(num_actions=num_actions, config=config)
         policy.load_state_dict(state["policy"])
         return cls(policy, **kwargs)
@@ -132,6 +133,50 @@ class Launcher:
         self.results_log = defaultdict(list)
         self.best_eval_return = -float("inf")
         self.best_checkpoint: Optional[str] = None
+        self.epoch = 0
+
+    # ------------------------------------------------------------------
+    # full-state resume (reference launcher.py:97-184
+    # restarts from the RLlib checkpoint — here the launcher's own
+    # bookkeeping rides along in the same checkpoint file)
+    def _full_state(self) -> Dict:
+        return {
+            "trainer": self.epoch_loop.state_dict(),
+            "launcher": {
+                "epoch": self.epoch,
+                "results_log": {k: list(v)
+                                for k, v in self.results_log.items()},
+                "best_eval_return": self.best_eval_return,
+                "best_checkpoint": self.best_checkpoint,
+            },
+        }
+
+    def _write_checkpoint(self, index: int) -> Optional[str]:
+        if self.checkpointer is None:
+            return None
+        return self.checkpointer.write(self._full_state(), index=index)
+
+    def resume(self, checkpoint_path: Optional[str] = None) -> bool:
+        """Restore trainer + launcher bookkeeping from a checkpoint (the
+        latest one by default).  Returns True if a checkpoint was loaded."""
+        if checkpoint_path is None:
+            if self.checkpointer is None:
+                return False
+            checkpoint_path = self.checkpointer.latest()
+            if checkpoint_path is None:
+                return False
+        state = Checkpointer.read(checkpoint_path)
+        trainer_state = state.get("trainer", state)   # old flat layout too
+        self.epoch_loop.trainer.load_state_dict(trainer_state)
+        ls = state.get("launcher")
+        if ls is not None:
+            self.epoch = int(ls.get("epoch", 0))
+            self.results_log = defaultdict(list)
+            for k, v in ls.get("results_log", {}).items():
+                self.results_log[k] = list(v)
+            self.best_eval_return = ls.get("best_eval_return", -float("inf"))
+            self.best_checkpoint = ls.get("best_checkpoint")
+        return True
 
     def _should_stop(self, epoch: int, total_steps: int) -> bool:
         if self.num_epochs is not None and epoch >= self.num_epochs:
@@ -141,13 +186,14 @@ class Launcher:
         return False
 
     def run(self) -> Dict:
-        epoch = 0
-        if self.checkpointer is not None:
-            self.checkpointer.write(self.epoch_loop.state_dict(), index=0)
+        epoch = self.epoch
+        if self.checkpointer is not None and epoch == 0:
+            self._write_checkpoint(index=0)
         while not self._should_stop(epoch, getattr(self.epoch_loop.trainer,
                                                    "total_env_steps", 0)):
             stats = self.epoch_loop.run()
             epoch += 1
+            self.epoch = epoch
             for k, v in stats.items():
                 if isinstance(v, (int, float, np.floating, np.integer)):
                     self.results_log[k].append(float(v))
@@ -161,15 +207,11 @@ class Launcher:
                     self.results_log["eval_episode_return"].append(ret)
                     if ret is not None and ret > self.best_eval_return:
                         self.best_eval_return = ret
-                        if self.checkpointer is not None:
-                            self.best_checkpoint = self.checkpointer.write(
-                                self.epoch_loop.state_dict(), index=epoch)
-                    elif self.checkpointer is not None:
-                        self.checkpointer.write(self.epoch_loop.state_dict(),
-                                                index=epoch)
-                elif self.checkpointer is not None:
-                    self.checkpointer.write(self.epoch_loop.state_dict(),
-                                            index=epoch)
+                        self.best_checkpoint = self._write_checkpoint(epoch)
+                    else:
+                        self._write_checkpoint(epoch)
+                else:
+                    self._write_checkpoint(epoch)
                 if self.logger is not None:
                     self.logger.write({"train_log": dict(self.results_log)})
             if self.verbose:

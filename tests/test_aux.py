@@ -312,3 +312,46 @@ def test_torus_topology_channels():
     assert t.num_nodes == 8
     # 2D torus: each node has 4 neighbours (2 per dim), directed channels
     assert len(t.channel_id_to_channel) > 0
+
+
+def test_launcher_full_state_resume(tiny_model_files, tmp_path):
+    """Kill/resume: a fresh Launcher resumed from the latest checkpoint
+    continues with the same epoch counter, results log, and best-checkpoint
+    bookkeeping."""
+    import torch
+
+    from ddls_amd.models.gnn import GNNPolicy
+    from ddls_amd.rl.ppo import PPOConfig, PPOTrainer
+    from ddls_amd.rl.rollout import VectorEnv
+    from ddls_amd.runtime.loops import EpochLoop, Launcher
+    from tests.conftest import make_env as mk
+
+    def build(num_epochs):
+        torch.manual_seed(0)
+        venv = VectorEnv([lambda i=i: mk(tiny_model_files, replication=2)
+                          for i in range(2)], base_seed=5)
+        policy = GNNPolicy(num_actions=17)
+        tr = PPOTrainer(venv, policy,
+                        PPOConfig(train_batch_size=8, sgd_minibatch_size=4,
+                                  num_sgd_iter=2, use_hip_graphs=False),
+                        device=torch.device("cpu"))
+        return Launcher(EpochLoop(tr), num_epochs=num_epochs,
+                        evaluation_interval=2,
+                        eval_fn=lambda: {"episode_return": -1.0},
+                        path_to_save=str(tmp_path / "run"), verbose=False)
+
+    l1 = build(num_epochs=4)
+    l1.run()
+    assert l1.epoch == 4
+    ret_log = dict(l1.results_log)
+
+    l2 = build(num_epochs=6)
+    assert l2.resume()                      # latest checkpoint
+    assert l2.epoch == 4
+    assert l2.best_eval_return == l1.best_eval_return
+    assert l2.best_checkpoint == l1.best_checkpoint
+    assert dict(l2.results_log) == ret_log
+    assert l2.epoch_loop.trainer.iteration == l1.epoch_loop.trainer.iteration
+    l2.run()                                # continues 2 more epochs
+    assert l2.epoch == 6
+    assert len(l2.results_log["epoch_counter"]) == 6
