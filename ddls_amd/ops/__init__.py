@@ -36,11 +36,18 @@ def build_extensions(verbose: bool = False):
     os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
     os.makedirs(_BUILD_DIR, exist_ok=True)
     from torch.utils.cpp_extension import load
-    _EXT = load(name="ddls_amd_kernels",
+
+    from ..utils.debug import debug_build_flags, debug_enabled, \
+        enable_debug_mode
+    if debug_enabled():
+        # serialize kernels/copies + XNACK faults BEFORE the HIP context
+        enable_debug_mode()
+    dbg = debug_build_flags()
+    _EXT = load(name="ddls_amd_kernels" + ("_debug" if dbg else ""),
                 sources=_SOURCES,
                 build_directory=_BUILD_DIR,
-                extra_cflags=["-O3"],
-                extra_cuda_cflags=["-O3"],
+                extra_cflags=["-O3"] + dbg,
+                extra_cuda_cflags=["-O3"] + dbg,
                 verbose=verbose)
     _EXT_ERR = None
     return _EXT

@@ -30,6 +30,15 @@
 #define ST_MISS 3
 #define ST_ERR 4
 
+// debug builds (-DDDLS_AMD_DEVICE_ASSERT via DDLS_AMD_DEBUG_BUILD=1) turn
+// engine invariants into device-side aborts with file:line
+#ifdef DDLS_AMD_DEVICE_ASSERT
+#define ENGINE_ASSERT(x) assert(x)
+#include <cassert>
+#else
+#define ENGINE_ASSERT(x)
+#endif
+
 // ---- tensor-list indices (mirror gpu_engine.py TENSOR_LAYOUT) ----
 enum {
   T_STATIC_OK = 0, T_SHAPE_PTR, T_SHAPES,
@@ -350,6 +359,7 @@ env_step_kernel(EnvPtrs P, EnvDims D) {
       } else {
         placed = true;
         int slot_i = P.n_running[b];
+        ENGINE_ASSERT(slot_i < D.K && "running-slot overflow");
         long so = ((long)b * D.K + slot_i) * D.WW;
         for (int w = 0; w < D.WW; ++w) P.slot_occ[so + w] = 0ull;
         for (int i = 0; i < n_union; ++i) {
@@ -449,6 +459,7 @@ env_step_kernel(EnvPtrs P, EnvDims D) {
     if (!pool_empty) {
       if ((P.t[b] + D.eps) >= P.next_arrive[b]) {
         int kk = P.arr_ptr[b];
+        ENGINE_ASSERT(kk < D.NJOBS && "job-log overflow");
         P.log_t_arr[nb + kk] = P.t[b];
         P.queued[b] = kk;
         P.arr_ptr[b] += 1;

@@ -113,3 +113,39 @@ class Logger:
             return out
         with gzip.open(path + ".pkl", "rb") as f:
             return pickle.load(f)
+
+
+class ExternalLogger:
+    """Switchable external metric logger seam (the reference pushes every
+    epoch's unpacked results to wandb, ``rllib_epoch_loop.py:144-230``; this
+    rebuild logs to local sqlite/pkl tables by default and mirrors to any
+    wandb-API-compatible backend when one is attached).
+
+    Backend contract: an object with ``log(metrics: dict, step: int)`` and
+    optionally ``finish()``.  ``ExternalLogger.wandb(...)`` builds one from
+    the wandb package if it is importable (no-op otherwise — this image has
+    no network, so the seam is exercised with fakes in tests).
+    """
+
+    def __init__(self, backend=None):
+        self.backend = backend
+
+    @classmethod
+    def wandb(cls, project: str = "ddls_amd", **init_kwargs):
+        try:
+            import wandb  # noqa: F401
+        except Exception:
+            return cls(backend=None)
+        run = wandb.init(project=project, **init_kwargs)
+        return cls(backend=run)
+
+    def log(self, metrics: Dict, step: Optional[int] = None):
+        if self.backend is None:
+            return
+        scalars = {k: v for k, v in metrics.items()
+                   if isinstance(v, (int, float))}
+        self.backend.log(scalars, step=step)
+
+    def finish(self):
+        if self.backend is not None and hasattr(self.backend, "finish"):
+            self.backend.finish()

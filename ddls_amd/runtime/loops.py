@@ -120,6 +120,7 @@ class Launcher:
                  eval_fn: Optional[Callable[[], Dict]] = None,
                  path_to_save: Optional[str] = None,
                  use_sqlite_database: bool = False,
+                 external_logger=None,
                  verbose: bool = True):
         self.epoch_loop = epoch_loop
         self.num_epochs = num_epochs
@@ -129,6 +130,7 @@ class Launcher:
         self.logger = (Logger(path_to_save, use_sqlite_database=use_sqlite_database)
                        if path_to_save else None)
         self.checkpointer = Checkpointer(path_to_save) if path_to_save else None
+        self.external_logger = external_logger   # runtime.logger.ExternalLogger
         self.verbose = verbose
         self.results_log = defaultdict(list)
         self.best_eval_return = -float("inf")
@@ -198,6 +200,8 @@ class Launcher:
                 if isinstance(v, (int, float, np.floating, np.integer)):
                     self.results_log[k].append(float(v))
             self.results_log["epoch_counter"].append(epoch)
+            if self.external_logger is not None:
+                self.external_logger.log(stats, step=epoch)
 
             if epoch % self.evaluation_interval == 0 or self._should_stop(
                     epoch, getattr(self.epoch_loop.trainer, "total_env_steps", 0)):

@@ -355,3 +355,36 @@ def test_launcher_full_state_resume(tiny_model_files, tmp_path):
     l2.run()                                # continues 2 more epochs
     assert l2.epoch == 6
     assert len(l2.results_log["epoch_counter"]) == 6
+
+
+def test_external_logger_seam(tiny_model_files, tmp_path):
+    """Launcher mirrors per-epoch scalar stats to any wandb-API-compatible
+    backend attached via ExternalLogger."""
+    import torch
+
+    from ddls_amd.models.gnn import GNNPolicy
+    from ddls_amd.rl.ppo import PPOConfig, PPOTrainer
+    from ddls_amd.rl.rollout import VectorEnv
+    from ddls_amd.runtime.logger import ExternalLogger
+    from ddls_amd.runtime.loops import EpochLoop, Launcher
+    from tests.conftest import make_env as mk
+
+    class FakeRun:
+        def __init__(self):
+            self.rows = []
+
+        def log(self, metrics, step=None):
+            self.rows.append((step, metrics))
+
+    run = FakeRun()
+    torch.manual_seed(0)
+    venv = VectorEnv([lambda: mk(tiny_model_files, replication=2)],
+                     base_seed=5)
+    tr = PPOTrainer(venv, GNNPolicy(num_actions=17),
+                    PPOConfig(train_batch_size=4, sgd_minibatch_size=2,
+                              num_sgd_iter=1, use_hip_graphs=False),
+                    device=torch.device("cpu"))
+    Launcher(EpochLoop(tr), num_epochs=3, verbose=False,
+             external_logger=ExternalLogger(backend=run)).run()
+    assert len(run.rows) == 3
+    assert run.rows[0][0] == 1 and "total_loss" in run.rows[0][1]
