@@ -319,7 +319,7 @@ def compile_engine_spec(env, lookahead_device=None,
         infinite_pool=infinite,
         static_shape_ok=_static_mask_ok(A, topo.shape),
         reward=_reward_spec_from_env(env),
-        max_running=min(W, 64),
+        max_running=min(W, 256),
         device_type=cluster.device_type,
     )
     spec.shape_ptr, spec.shapes = _shape_lists(A - 1, topo.shape)
@@ -826,6 +826,9 @@ def cpu_step_env(spec: EngineSpec, st: EngineState, b: int,
                 placed = True
                 # commit: occupy servers, append running slot
                 slot = int(st.n_running[b])
+                if slot >= st.K:
+                    st.status[b] = ST_ERR
+                    return ST_ERR
                 st.slot_occ[b, slot, :] = 0
                 for s in union:
                     w, bit = int(s) >> 6, np.uint64(1) << np.uint64(int(s) & 63)

@@ -360,6 +360,7 @@ env_step_kernel(EnvPtrs P, EnvDims D) {
         placed = true;
         int slot_i = P.n_running[b];
         ENGINE_ASSERT(slot_i < D.K && "running-slot overflow");
+        if (slot_i >= D.K) { P.status[b] = ST_ERR; return; }
         long so = ((long)b * D.K + slot_i) * D.WW;
         for (int w = 0; w < D.WW; ++w) P.slot_occ[so + w] = 0ull;
         for (int i = 0; i < n_union; ++i) {

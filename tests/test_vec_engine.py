@@ -318,3 +318,59 @@ def test_gpu_kernel_memo_miss_path(multi_model_files):
     results stay bitwise-identical to the mirror (warm dict)."""
     _lockstep_engines(multi_model_files, B=2, steps=60, seeds=[3, 5],
                       preload=False)
+
+
+def test_mirror_matches_real_env_1024_workers(multi_model_files):
+    """configs[3]: 1024-worker RAMP — mirror vs real env over a short
+    congested prefix (the block search runs over a (16,16,4) mesh)."""
+    from ddls_amd.envs import RampJobPartitioningEnvironment
+
+    def mk():
+        return RampJobPartitioningEnvironment(
+            topology_config={"type": "ramp", "kwargs": {
+                "num_communication_groups": 16,
+                "num_racks_per_communication_group": 16,
+                "num_servers_per_rack": 4,
+                "num_channels": 1,
+                "total_node_bandwidth": 1.6e12,
+                "intra_gpu_propagation_latency": 50e-9,
+                "worker_io_latency": 100e-9}},
+            node_config={"type_1": {"num_nodes": 1024, "workers_config": [
+                {"num_workers": 1, "worker": "ddls_amd.devices.A100"}]}},
+            jobs_config={"path_to_files": multi_model_files,
+                         "replication_factor": 2,
+                         "job_sampling_mode": "remove_and_repeat",
+                         "job_interarrival_time_dist": {
+                             "_target_": "ddls_amd.distributions.Exponential",
+                             "mean": 20},
+                         "max_acceptable_job_completion_time_frac_dist": {
+                             "_target_": "ddls_amd.distributions.Uniform",
+                             "min_val": 0.1, "max_val": 1, "decimals": 2},
+                         "num_training_steps": 10},
+            max_partitions_per_op=16,
+            min_op_run_time_quantum=0.01,
+            pad_obs_kwargs=None,
+            max_simulation_run_time=2000)
+
+    env_c = mk()
+    env_c.reset(seed=12345)
+    spec = compile_engine_spec(env_c)
+    assert spec.W == 1024
+    memo_preload = _memo_preload_for_env(spec)
+    gen = env_c.cluster.jobs_generator
+    sched = drain_episode_schedule(gen, spec, seed=5)
+    st = EngineState(spec, B=1, n_jobs_cap=sched.n + 2)
+    st.reset_env(spec, 0, sched)
+    env = mk()
+    env.lookahead_memo_preload = memo_preload
+    obs = env.reset(seed=5)
+    for t in range(15):
+        np.testing.assert_array_equal(
+            obs["action_mask"].astype(np.float32), st.obs_mask[0])
+        a = scripted_action(obs["action_mask"], t)
+        obs, reward, done, _ = env.step(int(a))
+        assert cpu_step_env(spec, st, 0, sched, int(a)) == ST_OK
+        assert st.reward[0] == reward
+        assert bool(st.done[0]) == bool(done)
+        if done:
+            break
