@@ -97,6 +97,10 @@ def main():
     world_size = get_world_size()
     use_cuda = torch.cuda.is_available()
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    if use_cuda:
+        # single-GPU rehearsal of the dp path (gloo backend) maps every rank
+        # onto the available device(s)
+        local_rank = local_rank % max(torch.cuda.device_count(), 1)
     device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
     if use_cuda:
         torch.cuda.set_device(device)

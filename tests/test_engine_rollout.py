@@ -181,3 +181,34 @@ def test_fused_cached_step_grads_match_autograd(multi_model_files):
     scale = np.abs(gr_).max()
     np.testing.assert_allclose(gf_, gr_, rtol=0, atol=max(scale, 1.0) * 2e-5,
                                err_msg="flat gradients")
+
+
+@pytest.mark.gpu
+def test_bench_dp2_gloo_single_gpu_rehearsal(tmp_path):
+    """World-size-2 rehearsal of the data-parallel bench on ONE GPU (gloo
+    backend, host-staged flat all-reduce; the real 8-GPU run uses RCCL with
+    the same code path).  Validates the engine + split-capture stepper under
+    torch.distributed end to end."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env.update({"DDLS_AMD_DIST_BACKEND": "gloo",
+                "MASTER_ADDR": "127.0.0.1"})
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+           "--master-port", "29731", "bench.py", "--gpus", "2",
+           "--steps", "2", "--warmup", "1", "--envs-per-rank", "16",
+           "--rollout-steps-per-env", "4", "--num-sgd-iter", "2",
+           "--sgd-minibatch-size", "32"]
+    out = subprocess.run(cmd, cwd=root, env=env, capture_output=True,
+                         text=True, timeout=600)
+    assert out.returncode == 0, out.stderr[-3000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    rec = json.loads(line)
+    assert rec["n_gpus"] == 2
+    assert rec["config"]["env_engine"] == "gpu_resident"
+    assert rec["value"] > 0
