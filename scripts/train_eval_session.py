@@ -33,6 +33,10 @@ WORKLOAD = os.environ.get("WORKLOAD", "medium_graphs")
 ITERS = int(os.environ.get("ITERS", "300"))
 OUT = os.environ.get("OUT", f"gpurun_out/session_{WORKLOAD}")
 EVAL_STEPS = int(os.environ.get("EVAL_STEPS", "0")) or 100000  # full episode
+# per-workload exploration override (the reference's entropy_coeff was tuned
+# on its own workload; medium_graphs needs longer exploration)
+ENTROPY = float(os.environ.get("ENTROPY", "0.003"))
+LR = float(os.environ.get("LR", "2.785e-4"))
 DATA = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
                     "data", WORKLOAD)
 
@@ -77,7 +81,8 @@ def main():
                            base_seed=1, verbose=True)
     trainer = PPOTrainer(venv, policy,
                          PPOConfig(train_batch_size=4096,
-                                   sgd_minibatch_size=128, num_sgd_iter=50),
+                                   sgd_minibatch_size=128, num_sgd_iter=50,
+                                   entropy_coeff=ENTROPY, lr=LR),
                          device=device)
     log = []
     t0 = time.time()
