@@ -191,7 +191,8 @@ class GpuEngine:
     def reset_env(self, b: int, sched: EpisodeSchedule):
         """Upload the episode schedule + mirror of EngineState.reset_env."""
         if not self._state_alloc_done:
-            self._alloc_state(sch_cap=sched.n + 8, njobs_cap=sched.n + 8)
+            cap = max(self.NJOBS, sched.n + 8)
+            self._alloc_state(sch_cap=cap, njobs_cap=cap)
         if sched.n + 1 > self.SCH:
             raise RuntimeError(
                 f"schedule length {sched.n} exceeds engine capacity "

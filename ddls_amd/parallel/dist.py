@@ -35,7 +35,9 @@ def init_distributed_from_env(device: Optional[torch.device] = None) -> int:
     dist.init_process_group(backend=backend)
     if torch.cuda.is_available():
         local_rank = int(os.environ.get("LOCAL_RANK", "0"))
-        torch.cuda.set_device(local_rank)
+        # modulo lets a multi-rank job rehearse on fewer GPUs (gloo backend,
+        # e.g. the world-size-2 single-GPU smoke test)
+        torch.cuda.set_device(local_rank % max(torch.cuda.device_count(), 1))
     return dist.get_rank()
 
 
