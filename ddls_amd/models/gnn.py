@@ -241,14 +241,26 @@ class MeanPoolLayer(nn.Module):
         ln_n, lin_n = self.node_module[0], self.node_module[1]
         ln_e, lin_e = self.edge_module[0], self.edge_module[1]
         ln_r, lin_r = self.reduce_module[0], self.reduce_module[1]
-        hn = ext.row_mlp(z.contiguous(), ln_n.weight, ln_n.bias,
-                         lin_n.weight.contiguous(), lin_n.bias)
-        he = ext.row_mlp(batch.e.contiguous(), ln_e.weight, ln_e.bias,
-                         lin_e.weight.contiguous(), lin_e.bias)
+        zc, ec = z.contiguous(), batch.e.contiguous()
+        Wn = lin_n.weight.contiguous()
+        We = lin_e.weight.contiguous()
+        Wr = lin_r.weight.contiguous()
         order, indptr = batch.csr_by_dst()
+        half, out_dim = Wn.shape[0], Wr.shape[0]
+        if (half == 16 and Wr.shape[1] == 32 and out_dim % 16 == 0
+                and out_dim <= 64 and zc.shape[1] <= 64 and ec.shape[1] <= 64
+                and os.environ.get("DDLS_AMD_DISABLE_MFMA", "0") != "1"):
+            # matrix-core inference path (v_mfma_f32_16x16x4_f32 tiles) —
+            # the rollout embedding cache runs through here every iteration
+            hn = ext.row_mlp_mfma(zc, ln_n.weight, ln_n.bias, Wn, lin_n.bias)
+            he = ext.row_mlp_mfma(ec, ln_e.weight, ln_e.bias, We, lin_e.bias)
+            r_edge, r_self = ext.message_mlp_mfma(
+                hn, he, batch.src, ln_r.weight, ln_r.bias, Wr, lin_r.bias)
+            return ext.segment_combine(r_edge, r_self, order, indptr)
+        hn = ext.row_mlp(zc, ln_n.weight, ln_n.bias, Wn, lin_n.bias)
+        he = ext.row_mlp(ec, ln_e.weight, ln_e.bias, We, lin_e.bias)
         return ext.message_reduce(hn, he, batch.src, order, indptr,
-                                  ln_r.weight, ln_r.bias,
-                                  lin_r.weight.contiguous(), lin_r.bias)
+                                  ln_r.weight, ln_r.bias, Wr, lin_r.bias)
 
     def forward(self, z: torch.Tensor, e: torch.Tensor, src: torch.Tensor,
                 dst: torch.Tensor) -> torch.Tensor:
