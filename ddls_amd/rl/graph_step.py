@@ -328,8 +328,12 @@ class CapturedSGDStep:
 
     def commit(self, j: int):
         """Enqueue H2D copies from pinned set j + replay the graph."""
-        for k, dst_t in self.d.items():
-            dst_t.copy_(self.pin[j][k], non_blocking=True)
+        if self.models_batch is not None:
+            self._flat_f32.copy_(self.pin[j]["_flat_f32"], non_blocking=True)
+            self._flat_i64.copy_(self.pin[j]["_flat_i64"], non_blocking=True)
+        else:
+            for k, dst_t in self.d.items():
+                dst_t.copy_(self.pin[j][k], non_blocking=True)
         self.copy_events[j].record()
         self.graph.replay()
         if self.graph_opt is not None:
@@ -350,24 +354,51 @@ class CapturedSGDStep:
             return torch.zeros(shape, dtype=dtype, pin_memory=True)
 
         if self.models_batch is not None:
-            # cached-models mode: tiny fixed-shape staging, no dummy graph
+            # cached-models mode: tiny fixed-shape staging, no dummy graph.
+            # All staged tensors are VIEWS of two packed flats (one f32, one
+            # i64) so each commit is TWO H2D copies instead of seven.
             self.n_cap, self.e_cap = 0, 0
-            self.d = {
-                "model_ids": dbuf(B, torch.int64),
-                "gf": dbuf((B, self.Fg), torch.float32),
-                "mask": torch.ones((B, self.A), dtype=torch.float32,
-                                   device=dev),
-                "actions": dbuf(B, torch.int64),
-                "old_logp": dbuf(B, torch.float32),
-                "adv": dbuf(B, torch.float32),
-                "vtarg": dbuf(B, torch.float32),
-            }
+            f32_specs = [("gf", (B, self.Fg)), ("mask", (B, self.A)),
+                         ("old_logp", (B,)), ("adv", (B,)), ("vtarg", (B,))]
+            i64_specs = [("model_ids", (B,)), ("actions", (B,))]
+            nf = sum(int(np.prod(sh)) for _k, sh in f32_specs)
+            ni = sum(int(np.prod(sh)) for _k, sh in i64_specs)
+            self._flat_f32 = dbuf(nf, torch.float32)
+            self._flat_i64 = dbuf(ni, torch.int64)
+            self.d = {}
+            off = 0
+            for k, sh in f32_specs:
+                n = int(np.prod(sh))
+                self.d[k] = self._flat_f32[off:off + n].view(sh)
+                off += n
+            off = 0
+            for k, sh in i64_specs:
+                n = int(np.prod(sh))
+                self.d[k] = self._flat_i64[off:off + n].view(sh)
+                off += n
             self.pin, self.pin_np = [], []
             for _ in range(2):
-                pset = {k: pbuf(tuple(v.shape), v.dtype)
-                        for k, v in self.d.items()}
+                pf = pbuf((nf,), torch.float32)
+                pi = pbuf((ni,), torch.int64)
+                pset = {"_flat_f32": pf, "_flat_i64": pi}
+                npset = {}
+                off = 0
+                for k, sh in f32_specs:
+                    n = int(np.prod(sh))
+                    npset[k] = pf[off:off + n].view(sh).numpy()
+                    off += n
+                off = 0
+                for k, sh in i64_specs:
+                    n = int(np.prod(sh))
+                    npset[k] = pi[off:off + n].view(sh).numpy()
+                    off += n
                 self.pin.append(pset)
-                self.pin_np.append({k: v.numpy() for k, v in pset.items()})
+                self.pin_np.append(npset)
+            # benign pre-first-commit content for capture warmup (all-ones
+            # mask keeps log(mask) finite)
+            self.d["mask"].fill_(1.0)
+            for npset in self.pin_np:
+                npset["mask"][:] = 1.0
             self.copy_events = [torch.cuda.Event(), torch.cuda.Event()]
             self._buf = 0
             self.batch = None
