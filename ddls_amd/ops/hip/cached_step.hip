@@ -907,6 +907,113 @@ __device__ __forceinline__ void wgrad_plain(
   if (tid < Dout) (P.flat_g + w_off_b)[tid] = bacc;
 }
 
+
+// ---------------------------------------------------------------------------
+// MFMA weight-grad contraction: gW[Dout,Din] = sum_r gpre[r,:] (x) u[r,:]
+// on v_mfma_f32_16x16x4_f32 tiles (exact f32: the result is a k-ordered fmaf
+// chain).  A[m][k=r] = gpre[r][m], B[k=r][i] = u[r][i]; per-lane fragments
+// A[l&15][l>>4], B[l>>4][l&15]; D col = l&15, row = (l>>4)*4 + reg.
+// Rows stream through zero-padded LDS tiles; each WAVE owns a set of 16x16
+// C tiles and accumulates across row tiles.  LN gamma/beta + bias grads ride
+// along as VALU reductions on the same tiles.
+// ---------------------------------------------------------------------------
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+template <int Din, int Dout, int DinP, int DoutP>
+__device__ __forceinline__ void wgrad_tiled_mfma(
+    const CachedPtrs& P, int tid, int NT, int rows0, int rows1,
+    const float* __restrict__ gpre0, const float* __restrict__ gpre1,
+    const float* __restrict__ xh0, const float* __restrict__ xh1,
+    const float* __restrict__ gu0, const float* __restrict__ gu1,
+    int w_slot, float* ldsG, float* ldsU, float* ldsGU, float* ldsXH) {
+  // lds row stride = padded dim (+1 would break MFMA b128 patterns; plain)
+  const float* __restrict__ gam = WP(w_slot);
+  const float* __restrict__ bet = WP(w_slot + 1);
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int n_waves = NT >> 6;
+  constexpr int MT = DoutP / 16;
+  constexpr int NTL = DinP / 16;
+  constexpr int CT = MT * NTL;                 // 16x16 C tiles
+  constexpr int MY_CT = (CT + 3) / 4;          // per wave (4 waves)
+  f32x4 acc[MY_CT];
+#pragma unroll
+  for (int q = 0; q < MY_CT; ++q) acc[q] = (f32x4){0.f, 0.f, 0.f, 0.f};
+  float bacc = 0.f, lacc_w = 0.f, lacc_b = 0.f;
+  const int rows = rows0 + rows1;
+  for (int r0 = 0; r0 < rows; r0 += TILE_K) {
+    const int rt = min(TILE_K, rows - r0);
+    // zero-fill then stage (padding rows/cols contribute 0 to the MFMA)
+    for (int x = tid; x < TILE_K * DoutP; x += NT) ldsG[x] = 0.f;
+    for (int x = tid; x < TILE_K * DinP; x += NT) ldsU[x] = 0.f;
+    __syncthreads();
+    for (int x = tid; x < rt * Dout; x += NT) {
+      int t = x / Dout, o = x % Dout;
+      int r = r0 + t;
+      ldsG[t * DoutP + o] = (r < rows0)
+          ? gpre0[(long)r * Dout + o]
+          : gpre1[(long)(r - rows0) * Dout + o];
+    }
+    for (int x = tid; x < rt * Din; x += NT) {
+      int t = x / Din, i = x % Din;
+      int r = r0 + t;
+      float xh = (r < rows0) ? xh0[(long)r * Din + i]
+                             : xh1[(long)(r - rows0) * Din + i];
+      ldsU[t * DinP + i] = xh * gam[i] + bet[i];
+      ldsXH[t * DinP + i] = xh;
+      ldsGU[t * DinP + i] = (r < rows0)
+          ? gu0[(long)r * Din + i]
+          : gu1[(long)(r - rows0) * Din + i];
+    }
+    __syncthreads();
+#pragma unroll
+    for (int q = 0; q < MY_CT; ++q) {
+      const int ct = wave + q * n_waves;
+      if (ct < CT) {
+        const int m0 = (ct / NTL) * 16;
+        const int n0 = (ct % NTL) * 16;
+        f32x4 a = acc[q];
+#pragma unroll
+        for (int kk = 0; kk < TILE_K; kk += 4) {
+          const float af = ldsG[(kk + (lane >> 4)) * DoutP + m0 + (lane & 15)];
+          const float bf = ldsU[(kk + (lane >> 4)) * DinP + n0 + (lane & 15)];
+          a = __builtin_amdgcn_mfma_f32_16x16x4f32(af, bf, a, 0, 0, 0);
+        }
+        acc[q] = a;
+      }
+    }
+    // bias + LN grads (VALU, from the same tiles)
+    if (tid < Dout)
+      for (int t = 0; t < rt; ++t) bacc += ldsG[t * DoutP + tid];
+    if (tid < Din)
+      for (int t = 0; t < rt; ++t) {
+        lacc_w += ldsGU[t * DinP + tid] * ldsXH[t * DinP + tid];
+        lacc_b += ldsGU[t * DinP + tid];
+      }
+    __syncthreads();
+  }
+  // write C tiles (skip padded rows/cols)
+#pragma unroll
+  for (int q = 0; q < MY_CT; ++q) {
+    const int ct = wave + q * n_waves;
+    if (ct >= CT) continue;
+    const int m0 = (ct / NTL) * 16;
+    const int n0 = (ct % NTL) * 16;
+    const int col = n0 + (lane & 15);
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int row = m0 + (lane >> 4) * 4 + reg;
+      if (row < Dout && col < Din)
+        WG_(w_slot + 2)[(long)row * Din + col] = acc[q][reg];
+    }
+  }
+  if (tid < Dout) WG_(w_slot + 3)[tid] = bacc;
+  if (tid < Din) {
+    WG_(w_slot)[tid] = lacc_w;
+    WG_(w_slot + 1)[tid] = lacc_b;
+  }
+}
+
 struct WJob {
   const float *gpre0, *xh0;   // source 0 rows
   const float *gpre1, *xh1;   // source 1 rows (or null)
@@ -924,6 +1031,8 @@ cs_bwd_w_kernel(CachedPtrs P, CachedDims D) {
   __shared__ float tU[TILE_K][65];
   __shared__ float tGU[TILE_K][65];
   __shared__ float tXH[TILE_K][65];
+  __shared__ float ldsG[TILE_K * 64], ldsU[TILE_K * 64];
+  __shared__ float ldsGU[TILE_K * 64], ldsXH[TILE_K * 64];
 
   switch (blockIdx.x) {
     case 0:  // node module 1
@@ -936,25 +1045,27 @@ cs_bwd_w_kernel(CachedPtrs P, CachedDims D) {
                            nullptr, P.gu_e1, nullptr, W_LN_E1_W, tG, tU,
                            tGU, tXH);
       return;
-    case 2:  // reduce module 1 (edge ++ self rows)
-      wgrad_tiled<KMSG, KHID>(P, tid, NT, D.E, D.N, P.gpre1_e, P.gpre1_s,
-                              P.xh_m1e, P.xh_m1s, P.gu_m1e, P.gu_m1s,
-                              W_LN_R1_W, tG, tU, tGU, tXH);
+    case 2:  // reduce module 1 (edge ++ self rows) — matrix cores
+      wgrad_tiled_mfma<KMSG, KHID, KMSG, KHID>(
+          P, tid, NT, D.E, D.N, P.gpre1_e, P.gpre1_s,
+          P.xh_m1e, P.xh_m1s, P.gu_m1e, P.gu_m1s,
+          W_LN_R1_W, ldsG, ldsU, ldsGU, ldsXH);
       return;
-    case 3:  // node module 2
-      wgrad_tiled<KHID, KH>(P, tid, NT, D.N, 0, P.gpn1, nullptr, P.xh_h2,
-                            nullptr, P.gu_h2, nullptr, W_LN_N2_W, tG, tU,
-                            tGU, tXH);
+    case 3:  // node module 2 — matrix cores
+      wgrad_tiled_mfma<KHID, KH, KHID, 16>(
+          P, tid, NT, D.N, 0, P.gpn1, nullptr, P.xh_h2,
+          nullptr, P.gu_h2, nullptr, W_LN_N2_W, ldsG, ldsU, ldsGU, ldsXH);
       return;
     case 4:  // edge module 2
       wgrad_tiled<KFE, KH>(P, tid, NT, D.E, 0, P.gpe1, nullptr, P.xh_e2,
                            nullptr, P.gu_e2, nullptr, W_LN_E2_W, tG, tU,
                            tGU, tXH);
       return;
-    case 5:  // reduce module 2
-      wgrad_tiled<KMSG, KOUT>(P, tid, NT, D.E, D.N, P.gpe2, P.gpn2,
-                              P.xh_m2e, P.xh_m2s, P.gu_m2e, P.gu_m2s,
-                              W_LN_R2_W, tG, tU, tGU, tXH);
+    case 5:  // reduce module 2 — matrix cores
+      wgrad_tiled_mfma<KMSG, KOUT, KMSG, 16>(
+          P, tid, NT, D.E, D.N, P.gpe2, P.gpn2,
+          P.xh_m2e, P.xh_m2s, P.gu_m2e, P.gu_m2s,
+          W_LN_R2_W, ldsG, ldsU, ldsGU, ldsXH);
       return;
     default:
       break;
