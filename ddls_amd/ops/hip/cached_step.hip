@@ -98,6 +98,12 @@ enum {
 struct CachedDims {
   int N, E, M, B, A;
   int F0, FE, H, MSG, HID, OUT, GFin, GEMB, FIN, FC;
+  int wgrad_mfma;   // DDLS_AMD_WGRAD_MFMA=1: matrix-core weight grads.
+                    // Default VALU: measured A/B at the tuned tile sizes
+                    // (gW <= 64x64, ds_read-latency-bound) has the tiled
+                    // VALU contraction ~3% faster end-to-end (6.68k vs
+                    // 6.50k env-steps/s) — MFMA's 16x16x4 chains leave the
+                    // wave latency-exposed at 1-2 accumulators.
   float clip, vf_clip, vf_coef, ent_coef;
 };
 
@@ -1045,27 +1051,42 @@ cs_bwd_w_kernel(CachedPtrs P, CachedDims D) {
                            nullptr, P.gu_e1, nullptr, W_LN_E1_W, tG, tU,
                            tGU, tXH);
       return;
-    case 2:  // reduce module 1 (edge ++ self rows) — matrix cores
-      wgrad_tiled_mfma<KMSG, KHID, KMSG, KHID>(
-          P, tid, NT, D.E, D.N, P.gpre1_e, P.gpre1_s,
-          P.xh_m1e, P.xh_m1s, P.gu_m1e, P.gu_m1s,
-          W_LN_R1_W, ldsG, ldsU, ldsGU, ldsXH);
+    case 2:  // reduce module 1 (edge ++ self rows)
+      if (D.wgrad_mfma)
+        wgrad_tiled_mfma<KMSG, KHID, KMSG, KHID>(
+            P, tid, NT, D.E, D.N, P.gpre1_e, P.gpre1_s,
+            P.xh_m1e, P.xh_m1s, P.gu_m1e, P.gu_m1s,
+            W_LN_R1_W, ldsG, ldsU, ldsGU, ldsXH);
+      else
+        wgrad_tiled<KMSG, KHID>(P, tid, NT, D.E, D.N, P.gpre1_e, P.gpre1_s,
+                                P.xh_m1e, P.xh_m1s, P.gu_m1e, P.gu_m1s,
+                                W_LN_R1_W, tG, tU, tGU, tXH);
       return;
-    case 3:  // node module 2 — matrix cores
-      wgrad_tiled_mfma<KHID, KH, KHID, 16>(
-          P, tid, NT, D.N, 0, P.gpn1, nullptr, P.xh_h2,
-          nullptr, P.gu_h2, nullptr, W_LN_N2_W, ldsG, ldsU, ldsGU, ldsXH);
+    case 3:  // node module 2
+      if (D.wgrad_mfma)
+        wgrad_tiled_mfma<KHID, KH, KHID, 16>(
+            P, tid, NT, D.N, 0, P.gpn1, nullptr, P.xh_h2,
+            nullptr, P.gu_h2, nullptr, W_LN_N2_W, ldsG, ldsU, ldsGU, ldsXH);
+      else
+        wgrad_tiled<KHID, KH>(P, tid, NT, D.N, 0, P.gpn1, nullptr, P.xh_h2,
+                              nullptr, P.gu_h2, nullptr, W_LN_N2_W, tG, tU,
+                              tGU, tXH);
       return;
     case 4:  // edge module 2
       wgrad_tiled<KFE, KH>(P, tid, NT, D.E, 0, P.gpe1, nullptr, P.xh_e2,
                            nullptr, P.gu_e2, nullptr, W_LN_E2_W, tG, tU,
                            tGU, tXH);
       return;
-    case 5:  // reduce module 2 — matrix cores
-      wgrad_tiled_mfma<KMSG, KOUT, KMSG, 16>(
-          P, tid, NT, D.E, D.N, P.gpe2, P.gpn2,
-          P.xh_m2e, P.xh_m2s, P.gu_m2e, P.gu_m2s,
-          W_LN_R2_W, ldsG, ldsU, ldsGU, ldsXH);
+    case 5:  // reduce module 2
+      if (D.wgrad_mfma)
+        wgrad_tiled_mfma<KMSG, KOUT, KMSG, 16>(
+            P, tid, NT, D.E, D.N, P.gpe2, P.gpn2,
+            P.xh_m2e, P.xh_m2s, P.gu_m2e, P.gu_m2s,
+            W_LN_R2_W, ldsG, ldsU, ldsGU, ldsXH);
+      else
+        wgrad_tiled<KMSG, KOUT>(P, tid, NT, D.E, D.N, P.gpe2, P.gpn2,
+                                P.xh_m2e, P.xh_m2s, P.gu_m2e, P.gu_m2s,
+                                W_LN_R2_W, tG, tU, tGU, tXH);
       return;
     default:
       break;
@@ -1217,6 +1238,8 @@ static void fill_ptrs(CachedPtrs& P, CachedDims& D,
   D.GEMB = (int)T[C_FINAL].size(1) - D.OUT;
   D.FIN = (int)T[C_FINAL].size(1);
   D.FC = (int)T[C_H1P].size(1);
+  const char* wm = getenv("DDLS_AMD_WGRAD_MFMA");
+  D.wgrad_mfma = (wm != nullptr && wm[0] == '1') ? 1 : 0;
   D.clip = (float)fs[0];
   D.vf_clip = (float)fs[1];
   D.vf_coef = (float)fs[2];
