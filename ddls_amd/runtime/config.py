@@ -152,6 +152,13 @@ def build_trainer_from_config(cfg: dict, device=None):
     torch.manual_seed(seed)
     policy = GNNPolicy(num_actions=num_actions,
                        config=model_cfg.get("custom_model_config"))
+    if algo.get("name", "ppo") == "pg":
+        from ..rl.pg import PGConfig, PGTrainer
+        pg_cfg = PGConfig(lr=algo.get("lr", 4e-4),
+                          gamma=algo.get("gamma", 0.99),
+                          train_batch_size=algo.get("train_batch_size", 200),
+                          grad_clip=algo.get("grad_clip"))
+        return PGTrainer(venv, policy, pg_cfg, device=device)
     if algo.get("name", "ppo") == "impala":
         from ..rl.impala import ImpalaConfig, ImpalaTrainer
         imp_cfg = ImpalaConfig(
