@@ -152,6 +152,30 @@ def build_trainer_from_config(cfg: dict, device=None):
     torch.manual_seed(seed)
     policy = GNNPolicy(num_actions=num_actions,
                        config=model_cfg.get("custom_model_config"))
+    if algo.get("name", "ppo") == "es":
+        from ..rl.es import ESConfig, ESTrainer
+        es_cfg = ESConfig(
+            noise_stdev=algo.get("noise_stdev", 0.02),
+            stepsize=algo.get("stepsize", 0.01),
+            l2_coeff=algo.get("l2_coeff", 0.005),
+            perturbation_pairs=algo.get("perturbation_pairs", 16),
+            fragment_steps=algo.get("fragment_steps", 16))
+        return ESTrainer(venv, policy, es_cfg, device=device)
+    if algo.get("name", "ppo") == "dqn":
+        from ..rl.dqn import DQNConfig, DQNTrainer
+        dq = DQNConfig(
+            gamma=algo.get("gamma", 0.999), lr=algo.get("lr", 4.121e-7),
+            n_step=algo.get("n_step", 3),
+            double_q=algo.get("double_q", True),
+            dueling=algo.get("dueling", True),
+            target_network_update_freq=algo.get(
+                "target_network_update_freq", 100000),
+            v_min=algo.get("v_min", -1000.0),
+            v_max=algo.get("v_max", 1000.0),
+            train_batch_size=algo.get("train_batch_size", 4096),
+            sgd_minibatch_size=algo.get("sgd_minibatch_size", 256),
+            num_sgd_iter=algo.get("num_sgd_iter", 4))
+        return DQNTrainer(venv, policy, dq, device=device)
     if algo.get("name", "ppo") == "pg":
         from ..rl.pg import PGConfig, PGTrainer
         pg_cfg = PGConfig(lr=algo.get("lr", 4e-4),
