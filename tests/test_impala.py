@@ -249,3 +249,30 @@ def test_es_improves_on_bandit(multi_model_files):
     st = tr.train()
     assert np.isfinite(st["fitness_mean"])
     assert not torch.equal(before, tr._get_flat())
+
+
+@pytest.mark.gpu
+def test_dqn_and_es_on_engine_env(multi_model_files):
+    """DQN and ES end-to-end on the GPU-resident envs (cached-models
+    forward path)."""
+    from ddls_amd.models.gnn import GNNPolicy
+    from ddls_amd.rl.dqn import DQNConfig, DQNTrainer
+    from ddls_amd.rl.engine_env import EngineVectorEnv
+    from ddls_amd.rl.es import ESConfig, ESTrainer
+
+    dev = torch.device("cuda:0")
+    torch.manual_seed(0)
+    venv = EngineVectorEnv(
+        lambda: make_env(multi_model_files, "remove_and_repeat", 2, 3000, 15),
+        num_envs=16, device=dev, base_seed=23)
+    dqn = DQNTrainer(venv, GNNPolicy(num_actions=17).to(dev),
+                     DQNConfig(learning_starts=50, sgd_minibatch_size=32,
+                               num_sgd_iter=2, lr=1e-4), device=dev)
+    for _ in range(2):
+        st = dqn.train(num_steps=8)
+        assert np.isfinite(st["total_loss"])
+    es = ESTrainer(venv, GNNPolicy(num_actions=17).to(dev),
+                   ESConfig(perturbation_pairs=2, fragment_steps=4),
+                   device=dev)
+    st = es.train()
+    assert np.isfinite(st["fitness_mean"])
