@@ -164,35 +164,14 @@ __device__ __forceinline__ void row_mlp_fwd_t(
     float* __restrict__ rstd_out, float* __restrict__ out) {
   float xh[Din];
   *rstd_out = ln_row_reg<Din>(x, xh, xhat_out);
+  float u[Din];
+#pragma unroll
+  for (int i = 0; i < Din; ++i) u[i] = xh[i] * gam[i] + bet[i];
 #pragma unroll 2
   for (int o = 0; o < Dout; ++o) {
     float acc = b[o];
 #pragma unroll
-    for (int i = 0; i < Din; ++i)
-      acc += W[o * Din + i] * (xh[i] * gam[i] + bet[i]);
-    out[o] = acc > 0.f ? acc : 0.f;
-  }
-}
-
-// chunked variant: thread (row, chunk) computes outputs [o0, o0+OC) —
-// 4x the threads at ~1/4 the registers of the full-row form (the LN is
-// recomputed per chunk; identical xhat/rstd double-writes are benign)
-template <int Din, int Dout, int OC>
-__device__ __forceinline__ void row_mlp_fwd_chunk(
-    const float* __restrict__ x, const float* __restrict__ gam,
-    const float* __restrict__ bet, const float* __restrict__ W,
-    const float* __restrict__ b, int o0, float* __restrict__ xhat_out,
-    float* __restrict__ rstd_out, float* __restrict__ out) {
-  float xh[Din];
-  float rstd = ln_row_reg<Din>(x, xh, xhat_out);
-  if (o0 == 0) *rstd_out = rstd;
-#pragma unroll
-  for (int oo = 0; oo < OC; ++oo) {
-    const int o = o0 + oo;
-    float acc = b[o];
-#pragma unroll
-    for (int i = 0; i < Din; ++i)
-      acc += W[o * Din + i] * (xh[i] * gam[i] + bet[i]);
+    for (int i = 0; i < Din; ++i) acc += W[o * Din + i] * u[i];
     out[o] = acc > 0.f ? acc : 0.f;
   }
 }
@@ -226,26 +205,32 @@ __device__ __forceinline__ void row_mlp_bwd_row_t(
     gpre[o] = out[o] > 0.f ? gout[o] : 0.f;
     gpre_store[o] = gpre[o];
   }
-  float m1 = 0.f, m2 = 0.f;
+  float gu[Din];
 #pragma unroll 2
   for (int i = 0; i < Din; ++i) {
     float acc = 0.f;
 #pragma unroll
     for (int o = 0; o < Dout; ++o) acc += W[o * Din + i] * gpre[o];
+    gu[i] = acc;
     gu_store[i] = acc;
-    const float gh = acc * gam[i];
-    m1 += gh;
-    m2 += gh * xhat[i];
   }
   if (gin != nullptr) {
+    float xh[Din];
+#pragma unroll
+    for (int i = 0; i < Din; ++i) xh[i] = xhat[i];
+    float m1 = 0.f, m2 = 0.f;
+#pragma unroll
+    for (int i = 0; i < Din; ++i) {
+      float gh = gu[i] * gam[i];
+      m1 += gh;
+      m2 += gh * xh[i];
+    }
     m1 /= Din;
     m2 /= Din;
-#pragma unroll 4
+#pragma unroll
     for (int i = 0; i < Din; ++i) {
-      // gu re-read from the just-written store (L1/L2-hot) rather than a
-      // Din-wide register array — keeps the fat rows under the spill line
-      const float gh = gu_store[i] * gam[i];
-      gin[i] = rstd * (gh - m1 - xhat[i] * m2);
+      float gh = gu[i] * gam[i];
+      gin[i] = rstd * (gh - m1 - xh[i] * m2);
     }
   }
 }
@@ -307,42 +292,38 @@ cs_fwd_reduce_kernel(CachedPtrs P, CachedDims D) {
   __syncthreads();
   const float* hn = (ROUND == 1) ? P.hn1 : P.hn2;
   const float* he = (ROUND == 1) ? P.he1 : P.he2;
-  constexpr int OC = 16;                   // outputs per thread
-  constexpr int CH = DO / OC;              // chunks per row
-  const long total = (long)(D.E + D.N) * CH;
+  const long total = D.E + D.N;
   GSTRIDE {
-    const long row = u / CH;
-    const int o0 = (int)(u % CH) * OC;
     float msg[KMSG];
-    if (row < D.E) {
-      const long k = row;
+    if (u < D.E) {
+      const long k = u;
       const long sv = P.src[k];
 #pragma unroll
       for (int i = 0; i < KH; ++i) msg[i] = hn[sv * KH + i];
 #pragma unroll
       for (int i = 0; i < KH; ++i) msg[KH + i] = he[k * KH + i];
       if (ROUND == 1)
-        row_mlp_fwd_chunk<KMSG, KHID, OC>(msg, sGam, sBet, sW, sB, o0,
-                                          P.xh_m1e + k * KMSG,
-                                          P.rst_m1e + k, P.re1 + k * KHID);
+        row_mlp_fwd_t<KMSG, KHID>(msg, sGam, sBet, sW, sB,
+                                  P.xh_m1e + k * KMSG,
+                                  P.rst_m1e + k, P.re1 + k * KHID);
       else
-        row_mlp_fwd_chunk<KMSG, KOUT, OC>(msg, sGam, sBet, sW, sB, o0,
-                                          P.xh_m2e + k * KMSG,
-                                          P.rst_m2e + k, P.re2 + k * KOUT);
+        row_mlp_fwd_t<KMSG, KOUT>(msg, sGam, sBet, sW, sB,
+                                  P.xh_m2e + k * KMSG,
+                                  P.rst_m2e + k, P.re2 + k * KOUT);
     } else {
-      const long v = row - D.E;
+      const long v = u - D.E;
 #pragma unroll
       for (int i = 0; i < KH; ++i) msg[i] = hn[v * KH + i];
 #pragma unroll
       for (int i = 0; i < KH; ++i) msg[KH + i] = 0.f;
       if (ROUND == 1)
-        row_mlp_fwd_chunk<KMSG, KHID, OC>(msg, sGam, sBet, sW, sB, o0,
-                                          P.xh_m1s + v * KMSG,
-                                          P.rst_m1s + v, P.rs1 + v * KHID);
+        row_mlp_fwd_t<KMSG, KHID>(msg, sGam, sBet, sW, sB,
+                                  P.xh_m1s + v * KMSG,
+                                  P.rst_m1s + v, P.rs1 + v * KHID);
       else
-        row_mlp_fwd_chunk<KMSG, KOUT, OC>(msg, sGam, sBet, sW, sB, o0,
-                                          P.xh_m2s + v * KMSG,
-                                          P.rst_m2s + v, P.rs2 + v * KOUT);
+        row_mlp_fwd_t<KMSG, KOUT>(msg, sGam, sBet, sW, sB,
+                                  P.xh_m2s + v * KMSG,
+                                  P.rst_m2s + v, P.rs2 + v * KOUT);
     }
   }
 }
@@ -1368,20 +1349,14 @@ void cached_step_fwd(std::vector<torch::Tensor> T, std::vector<double> fs) {
   const int comb2_b = ((long)D.N * D.OUT + 255) / 256;
   hipLaunchKernelGGL(cs_fwd_mlp_kernel<1>, dim3(rows_b), dim3(256), 0,
                      stream, P, D);
-  {
-    const int r1_b = ((long)(D.N + D.E) * (D.HID / 16) + 255) / 256;
-    hipLaunchKernelGGL(cs_fwd_reduce_kernel<1>, dim3(r1_b), dim3(256), 0,
-                       stream, P, D);
-  }
+  hipLaunchKernelGGL(cs_fwd_reduce_kernel<1>, dim3(rows_b), dim3(256), 0,
+                     stream, P, D);
   hipLaunchKernelGGL(cs_fwd_combine_kernel<1>, dim3(comb1_b), dim3(256), 0,
                      stream, P, D);
   hipLaunchKernelGGL(cs_fwd_mlp_kernel<2>, dim3(rows_b), dim3(256), 0,
                      stream, P, D);
-  {
-    const int r2_b = ((long)(D.N + D.E) * (D.OUT / 16) + 255) / 256;
-    hipLaunchKernelGGL(cs_fwd_reduce_kernel<2>, dim3(r2_b), dim3(256), 0,
-                       stream, P, D);
-  }
+  hipLaunchKernelGGL(cs_fwd_reduce_kernel<2>, dim3(rows_b), dim3(256), 0,
+                     stream, P, D);
   hipLaunchKernelGGL(cs_fwd_combine_kernel<2>, dim3(comb2_b), dim3(256), 0,
                      stream, P, D);
   hipLaunchKernelGGL(cs_fwd_pool_kernel, dim3(1), dim3(512), 0, stream,
