@@ -390,6 +390,48 @@ def test_external_logger_seam(tiny_model_files, tmp_path):
     assert run.rows[0][0] == 1 and "total_loss" in run.rows[0][1]
 
 
+def _learned_matches_optimum(workload, ckpt_name):
+    import os
+
+    import torch
+
+    from ddls_amd.envs.actors import ACTORS
+    from ddls_amd.runtime.loops import EvalLoop, PolicyActor
+    from ddls_amd.utils import seed_everything
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    ckpt = os.path.join(root, "artifacts", "round2", ckpt_name)
+    os.environ["WORKLOAD"] = workload
+    import sys
+    sys.path.insert(0, os.path.join(root, "scripts"))
+    import importlib
+    import train_eval_session
+    importlib.reload(train_eval_session)
+    make_env = train_eval_session.make_env
+
+    def run(actor, steps=250):
+        seed_everything(1799)
+        env = make_env()
+        return EvalLoop(actor, env, max_steps=steps).run(seed=1799)
+
+    learned = run(PolicyActor.from_checkpoint(ckpt,
+                                              device=torch.device("cpu")))
+    sipml = run(ACTORS["sip_ml"](max_partitions_per_op=8))
+    acceptable = run(ACTORS["acceptable_jct"]())
+    assert learned["episode_return"] == sipml["episode_return"], (
+        "learned policy should equal the optimal heuristic exactly")
+    assert learned["episode_return"] > acceptable["episode_return"]
+    assert learned["blocking_rate"] == sipml["blocking_rate"]
+
+
+def test_small_workload_learned_policy_matches_optimum():
+    """Round-2 stack on data/small_graphs: degree 8 is argmin-JCT for all
+    five models (degrees > 8 unplaceable on the 32-worker RAMP under the
+    reference's block-shape rules), so SiP-ML@8 is optimal; the committed
+    checkpoint's greedy policy ties it exactly."""
+    _learned_matches_optimum("small_graphs", "checkpoint-small-500")
+
+
 def test_medium_workload_learned_policy_matches_optimum():
     """Round-2 quality eval on data/medium_graphs: the
     committed checkpoint's greedy policy attains the OPTIMAL per-job rule.
