@@ -45,6 +45,36 @@ def main():
           f"{tr.total_env_steps/dt:,.0f} steps/s")
     with open(f"{OUT}/train_log.json", "w") as f:
         json.dump(log, f)
+    torch.save({"policy": policy.state_dict()}, f"{OUT}/policy.pt")
+
+    # greedy full-episode eval on the REAL env vs the two strongest
+    # heuristics, same protocol as scripts/train_eval_session.py
+    from ddls_amd.envs.actors import ACTORS
+    from ddls_amd.runtime.loops import EvalLoop, PolicyActor
+    from ddls_amd.utils import seed_everything
+
+    def run_eval(actor, steps=1000):
+        seed_everything(1799)
+        env = build_env_fn()()
+        r = EvalLoop(actor, env, max_steps=steps).run(seed=1799)
+        es = env.cluster.episode_stats
+        arrived = max(es["num_jobs_arrived"], 1)
+        return {
+            "episode_return": r["episode_return"],
+            "mean_jct": r["mean_job_completion_time"],
+            "jct_speedup": r["mean_job_completion_time_speedup"],
+            "blocking_rate": es["num_jobs_blocked"] / arrived,
+            "num_actor_steps": r["num_actor_steps"],
+        }
+
+    evals = {"learned_impala": run_eval(
+        PolicyActor(policy, device=torch.device("cpu")))}
+    for name, kw in (("sip_ml", {"max_partitions_per_op": 8}),
+                     ("acceptable_jct", {})):
+        evals[name] = run_eval(ACTORS[name](**kw))
+    print(json.dumps(evals, indent=1))
+    with open(f"{OUT}/eval.json", "w") as f:
+        json.dump(evals, f, indent=1)
 
 
 if __name__ == "__main__":
