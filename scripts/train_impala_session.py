@@ -1,7 +1,7 @@
 #!/usr/bin/env python3
-"""IMPALA (V-trace) training session on the GPU-resident env engine —
-demonstrates the second algorithm end-to-end on the flagship stack.
-Env: ITERS=N OUT=dir"""
+"""IMPALA (V-trace) / PG training session on the GPU-resident env engine —
+demonstrates the non-PPO algorithms end-to-end on the flagship stack.
+Env: ALGO=impala|pg ITERS=N OUT=dir"""
 import json
 import os
 import sys
@@ -15,9 +15,11 @@ from bench import build_env_fn
 from ddls_amd.models.gnn import GNNPolicy
 from ddls_amd.rl.engine_env import EngineVectorEnv
 from ddls_amd.rl.impala import ImpalaConfig, ImpalaTrainer
+from ddls_amd.rl.pg import PGConfig, PGTrainer
 
+ALGO = os.environ.get("ALGO", "impala")
 ITERS = int(os.environ.get("ITERS", "40"))
-OUT = os.environ.get("OUT", "gpurun_out/impala_session")
+OUT = os.environ.get("OUT", f"gpurun_out/{ALGO}_session")
 
 
 def main():
@@ -27,8 +29,12 @@ def main():
     policy = GNNPolicy(num_actions=17).to(dev)
     venv = EngineVectorEnv(build_env_fn(), num_envs=256, device=dev,
                            base_seed=1)
-    tr = ImpalaTrainer(venv, policy, ImpalaConfig(train_batch_size=4096),
+    if ALGO == "pg":
+        tr = PGTrainer(venv, policy, PGConfig(train_batch_size=4096),
                        device=dev)
+    else:
+        tr = ImpalaTrainer(venv, policy, ImpalaConfig(train_batch_size=4096),
+                           device=dev)
     log = []
     t0 = time.time()
     for i in range(ITERS):
@@ -38,10 +44,11 @@ def main():
                                            "entropy")})
         if (i + 1) % 10 == 0:
             print(f"iter {i+1}: reward {st['mean_reward']:.1f} "
-                  f"loss {st['total_loss']:.1f} rho {st['mean_rho']:.3f} "
-                  f"entropy {st['entropy']:.2f}", flush=True)
+                  f"loss {st['total_loss']:.1f} "
+                  f"rho {st.get('mean_rho', 0):.3f} "
+                  f"entropy {st.get('entropy', 0):.2f}", flush=True)
     dt = time.time() - t0
-    print(f"IMPALA: {tr.total_env_steps} env steps in {dt:.0f}s = "
+    print(f"{ALGO}: {tr.total_env_steps} env steps in {dt:.0f}s = "
           f"{tr.total_env_steps/dt:,.0f} steps/s")
     with open(f"{OUT}/train_log.json", "w") as f:
         json.dump(log, f)
@@ -67,7 +74,7 @@ def main():
             "num_actor_steps": r["num_actor_steps"],
         }
 
-    evals = {"learned_impala": run_eval(
+    evals = {f"learned_{ALGO}": run_eval(
         PolicyActor(policy, device=torch.device("cpu")))}
     for name, kw in (("sip_ml", {"max_partitions_per_op": 8}),
                      ("acceptable_jct", {})):
