@@ -78,8 +78,19 @@ enum {
   // the LN gamma/beta reductions in cs_bwd_w
   C_GU_M2E, C_GU_M2S, C_GU_H2, C_GU_E2, C_GU_M1E, C_GU_M1S, C_GU_Z1,
   C_GU_E1,
+  // row-split wgrad partial sums: [6 jobs][WSPLIT][WG_JSTRIDE]
+  C_WG_SCRATCH,
   C_NT
 };
+
+// The six row-heavy wgrad jobs (module weight grads over N/E rows) are
+// row-split WSPLIT ways — a single workgroup per job leaves one CU
+// streaming ~4 MB from HBM while the other 255 idle (measured 85 us, 22%
+// of the fused step).  Partials land in wg_scratch in a FIXED layout and
+// a 6-block reduce kernel sums them in fixed split order: deterministic,
+// no atomics.  WG_JSTRIDE = max(Din*Dout) + max(Dout) + 2*max(Din).
+#define WSPLIT 16
+#define WG_JSTRIDE (KMSG * KHID + KHID + KHID + KHID)
 
 // weight slots within OFFS (order fixed; mirror _fused_offsets in Python)
 enum {
@@ -127,6 +138,7 @@ struct CachedPtrs {
   float *gpn2, *gpe2, *gpn1, *gpe1;
   float *gpre1_e, *gpre1_s;
   float *gu_m2e, *gu_m2s, *gu_h2, *gu_e2, *gu_m1e, *gu_m1s, *gu_z1, *gu_e1;
+  float *wg_scratch;
 };
 
 #define WP(slot) (P.flat_p + P.offs[slot])
@@ -409,44 +421,57 @@ cs_fwd_head_kernel(CachedPtrs P, CachedDims D) {
 }
 
 // logits+value rows + loss (1 WG so the stats reduction stays in-block)
+// final logit/value projection, one block per SAMPLE: the previous
+// 32-samples-per-block layout ran the whole [B,17]x[17,256] contraction on
+// 4 workgroups (one serial 256-wide loop per sample-thread, 39 us); here
+// 128 blocks tile (action x 16-chunk) over 272 threads with an LDS
+// partial-sum reduce — fixed order, deterministic
+__global__ void __launch_bounds__(512)
+cs_fwd_logits_kernel(CachedPtrs P, CachedDims D) {
+  const int b = blockIdx.x;
+  const int tid = threadIdx.x;
+  __shared__ float shp[KFC], shv[KFC];
+  __shared__ float part[KA][16];
+  __shared__ float vred[256];
+  const float* __restrict__ hp = P.h1p + (long)b * KFC;
+  const float* __restrict__ hv = P.h1v + (long)b * KFC;
+  for (int x = tid; x < KFC; x += 512) {
+    shp[x] = hp[x];
+    shv[x] = hv[x];
+  }
+  __syncthreads();
+  if (tid < KA * 16) {
+    const int a = tid / 16, c = tid % 16;
+    const float* __restrict__ w = WP(W_P2_W) + (long)a * KFC + c * 16;
+    float s = 0.f;
+#pragma unroll
+    for (int i = 0; i < 16; ++i) s += w[i] * shp[c * 16 + i];
+    part[a][c] = s;
+  }
+  if (tid < 256) vred[tid] = WP(W_V2_W)[tid] * shv[tid];
+  __syncthreads();
+  for (int sft = 128; sft > 0; sft >>= 1) {
+    if (tid < sft) vred[tid] += vred[tid + sft];
+    __syncthreads();
+  }
+  if (tid < KA) {
+    float acc = WP(W_P2_B)[tid];
+#pragma unroll
+    for (int c = 0; c < 16; ++c) acc += part[tid][c];
+    const float fmin = -3.402823466e+38f;
+    float lm = logf(P.mask[(long)b * KA + tid]);
+    if (!(lm > fmin)) lm = fmin;
+    P.p[(long)b * KA + tid] = acc + lm;   // raw masked logits
+  }
+  if (tid == 0) P.values[b] = vred[0] + WP(W_V2_B)[0];
+}
+
 __global__ void __launch_bounds__(256)
 cs_fwd_loss_kernel(CachedPtrs P, CachedDims D) {
   const int tid = threadIdx.x;
   const int NT = blockDim.x;
   const int b_lo = blockIdx.x * 32;
   const int b_hi = min(D.B, b_lo + 32);
-  // stage W2p [A,FC] + W2v [FC] in LDS: each sample's h rows then stream
-  // ONCE from L2 while all A+1 output accumulators live in registers
-  __shared__ float sW2p[KA * KFC];
-  __shared__ float sW2v[KFC];
-  for (int x = tid; x < KA * KFC; x += NT) sW2p[x] = WP(W_P2_W)[x];
-  for (int x = tid; x < KFC; x += NT) sW2v[x] = WP(W_V2_W)[x];
-  __syncthreads();
-  for (int b = b_lo + tid; b < b_hi; b += NT) {
-    float acc[KA];
-#pragma unroll
-    for (int a = 0; a < KA; ++a) acc[a] = WP(W_P2_B)[a];
-    float av = WP(W_V2_B)[0];
-    const float* __restrict__ hp = P.h1p + (long)b * KFC;
-    const float* __restrict__ hv = P.h1v + (long)b * KFC;
-#pragma unroll 4
-    for (int j = 0; j < KFC; ++j) {
-      const float h = hp[j];
-#pragma unroll
-      for (int a = 0; a < KA; ++a) acc[a] += sW2p[a * KFC + j] * h;
-      av += sW2v[j] * hv[j];
-    }
-    const float fmin = -3.402823466e+38f;
-#pragma unroll
-    for (int a = 0; a < KA; ++a) {
-      float lm = logf(P.mask[(long)b * KA + a]);
-      if (!(lm > fmin)) lm = fmin;
-      P.p[(long)b * KA + a] = acc[a] + lm;   // raw masked logits
-    }
-    P.values[b] = av;
-  }
-  __syncthreads();
-
   __shared__ float acc_s[4];
   if (tid < 4) acc_s[tid] = 0.f;
   __syncthreads();
@@ -794,7 +819,8 @@ __device__ __forceinline__ void wgrad_tiled(
     const float* __restrict__ xh0, const float* __restrict__ xh1,
     const float* __restrict__ gu0, const float* __restrict__ gu1,
     int w_slot, float (*tG)[65], float (*tU)[65], float (*tGU)[65],
-    float (*tXH)[65]) {
+    float (*tXH)[65], int split = 0, int nsplit = 1,
+    float* __restrict__ pout = nullptr) {
   const float* __restrict__ gam = WP(w_slot);
   const float* __restrict__ bet = WP(w_slot + 1);
   constexpr int UNITS = Din * Dout;
@@ -804,7 +830,7 @@ __device__ __forceinline__ void wgrad_tiled(
   for (int q = 0; q < MYU; ++q) acc[q] = 0.f;
   float bacc = 0.f, lacc_w = 0.f, lacc_b = 0.f;
   const int rows = rows0 + rows1;
-  for (int r0 = 0; r0 < rows; r0 += TILE_K) {
+  for (int r0 = split * TILE_K; r0 < rows; r0 += nsplit * TILE_K) {
     const int rt = min(TILE_K, rows - r0);
     for (int x = tid; x < rt * Dout; x += NT) {
       int t = x / Dout, o = x % Dout;
@@ -846,6 +872,20 @@ __device__ __forceinline__ void wgrad_tiled(
         lacc_b += tGU[t][tid];
       }
     __syncthreads();
+  }
+  if (pout != nullptr) {
+    // partial layout: [w UNITS][b Dout][ln_w Din][ln_b Din]
+#pragma unroll
+    for (int q = 0; q < MYU; ++q) {
+      int u = tid + q * 256;
+      if (u < UNITS) pout[u] = acc[q];
+    }
+    if (tid < Dout) pout[UNITS + tid] = bacc;
+    if (tid < Din) {
+      pout[UNITS + Dout + tid] = lacc_w;
+      pout[UNITS + Dout + Din + tid] = lacc_b;
+    }
+    return;
   }
 #pragma unroll
   for (int q = 0; q < MYU; ++q) {
@@ -1040,60 +1080,72 @@ cs_bwd_w_kernel(CachedPtrs P, CachedDims D) {
   __shared__ float ldsG[TILE_K * 64], ldsU[TILE_K * 64];
   __shared__ float ldsGU[TILE_K * 64], ldsXH[TILE_K * 64];
 
-  switch (blockIdx.x) {
-    case 0:  // node module 1
-      wgrad_tiled<KF0, KH>(P, tid, NT, D.N, 0, P.ghn1, nullptr, P.xh_z1,
-                           nullptr, P.gu_z1, nullptr, W_LN_N1_W, tG, tU,
-                           tGU, tXH);
-      return;
-    case 1:  // edge module 1
-      wgrad_tiled<KFE, KH>(P, tid, NT, D.E, 0, P.ghe1, nullptr, P.xh_e1,
-                           nullptr, P.gu_e1, nullptr, W_LN_E1_W, tG, tU,
-                           tGU, tXH);
-      return;
-    case 2:  // reduce module 1 (edge ++ self rows)
-      if (D.wgrad_mfma)
-        wgrad_tiled_mfma<KMSG, KHID, KMSG, KHID>(
-            P, tid, NT, D.E, D.N, P.gpre1_e, P.gpre1_s,
-            P.xh_m1e, P.xh_m1s, P.gu_m1e, P.gu_m1s,
-            W_LN_R1_W, ldsG, ldsU, ldsGU, ldsXH);
-      else
-        wgrad_tiled<KMSG, KHID>(P, tid, NT, D.E, D.N, P.gpre1_e, P.gpre1_s,
-                                P.xh_m1e, P.xh_m1s, P.gu_m1e, P.gu_m1s,
-                                W_LN_R1_W, tG, tU, tGU, tXH);
-      return;
-    case 3:  // node module 2
-      if (D.wgrad_mfma)
-        wgrad_tiled_mfma<KHID, KH, KHID, 16>(
-            P, tid, NT, D.N, 0, P.gpn1, nullptr, P.xh_h2,
-            nullptr, P.gu_h2, nullptr, W_LN_N2_W, ldsG, ldsU, ldsGU, ldsXH);
-      else
-        wgrad_tiled<KHID, KH>(P, tid, NT, D.N, 0, P.gpn1, nullptr, P.xh_h2,
-                              nullptr, P.gu_h2, nullptr, W_LN_N2_W, tG, tU,
-                              tGU, tXH);
-      return;
-    case 4:  // edge module 2
-      wgrad_tiled<KFE, KH>(P, tid, NT, D.E, 0, P.gpe1, nullptr, P.xh_e2,
-                           nullptr, P.gu_e2, nullptr, W_LN_E2_W, tG, tU,
-                           tGU, tXH);
-      return;
-    case 5:  // reduce module 2
-      if (D.wgrad_mfma)
-        wgrad_tiled_mfma<KMSG, KOUT, KMSG, 16>(
-            P, tid, NT, D.E, D.N, P.gpe2, P.gpn2,
-            P.xh_m2e, P.xh_m2s, P.gu_m2e, P.gu_m2s,
-            W_LN_R2_W, ldsG, ldsU, ldsGU, ldsXH);
-      else
-        wgrad_tiled<KMSG, KOUT>(P, tid, NT, D.E, D.N, P.gpe2, P.gpn2,
-                                P.xh_m2e, P.xh_m2s, P.gu_m2e, P.gu_m2s,
-                                W_LN_R2_W, tG, tU, tGU, tXH);
-      return;
-    default:
-      break;
+  // jobs 0..5 are row-split nsplit ways (one WG per (job, split)); the
+  // MFMA variants (env opt-in) keep single-WG jobs
+  const int nsplit = D.wgrad_mfma ? 1 : WSPLIT;
+  const int jobs_end = 6 * nsplit;
+  if ((int)blockIdx.x < jobs_end) {
+    const int job = blockIdx.x / nsplit;
+    const int sp = blockIdx.x % nsplit;
+    float* pout = (nsplit > 1)
+        ? P.wg_scratch + ((long)job * WSPLIT + sp) * WG_JSTRIDE : nullptr;
+    switch (job) {
+      case 0:  // node module 1
+        wgrad_tiled<KF0, KH>(P, tid, NT, D.N, 0, P.ghn1, nullptr, P.xh_z1,
+                             nullptr, P.gu_z1, nullptr, W_LN_N1_W, tG, tU,
+                             tGU, tXH, sp, nsplit, pout);
+        return;
+      case 1:  // edge module 1
+        wgrad_tiled<KFE, KH>(P, tid, NT, D.E, 0, P.ghe1, nullptr, P.xh_e1,
+                             nullptr, P.gu_e1, nullptr, W_LN_E1_W, tG, tU,
+                             tGU, tXH, sp, nsplit, pout);
+        return;
+      case 2:  // reduce module 1 (edge ++ self rows)
+        if (D.wgrad_mfma)
+          wgrad_tiled_mfma<KMSG, KHID, KMSG, KHID>(
+              P, tid, NT, D.E, D.N, P.gpre1_e, P.gpre1_s,
+              P.xh_m1e, P.xh_m1s, P.gu_m1e, P.gu_m1s,
+              W_LN_R1_W, ldsG, ldsU, ldsGU, ldsXH);
+        else
+          wgrad_tiled<KMSG, KHID>(P, tid, NT, D.E, D.N, P.gpre1_e,
+                                  P.gpre1_s, P.xh_m1e, P.xh_m1s, P.gu_m1e,
+                                  P.gu_m1s, W_LN_R1_W, tG, tU, tGU, tXH,
+                                  sp, nsplit, pout);
+        return;
+      case 3:  // node module 2
+        if (D.wgrad_mfma)
+          wgrad_tiled_mfma<KHID, KH, KHID, 16>(
+              P, tid, NT, D.N, 0, P.gpn1, nullptr, P.xh_h2,
+              nullptr, P.gu_h2, nullptr, W_LN_N2_W, ldsG, ldsU, ldsGU,
+              ldsXH);
+        else
+          wgrad_tiled<KHID, KH>(P, tid, NT, D.N, 0, P.gpn1, nullptr,
+                                P.xh_h2, nullptr, P.gu_h2, nullptr,
+                                W_LN_N2_W, tG, tU, tGU, tXH, sp, nsplit,
+                                pout);
+        return;
+      case 4:  // edge module 2
+        wgrad_tiled<KFE, KH>(P, tid, NT, D.E, 0, P.gpe1, nullptr, P.xh_e2,
+                             nullptr, P.gu_e2, nullptr, W_LN_E2_W, tG, tU,
+                             tGU, tXH, sp, nsplit, pout);
+        return;
+      default:  // 5: reduce module 2
+        if (D.wgrad_mfma)
+          wgrad_tiled_mfma<KMSG, KOUT, KMSG, 16>(
+              P, tid, NT, D.E, D.N, P.gpe2, P.gpn2,
+              P.xh_m2e, P.xh_m2s, P.gu_m2e, P.gu_m2s,
+              W_LN_R2_W, ldsG, ldsU, ldsGU, ldsXH);
+        else
+          wgrad_tiled<KMSG, KOUT>(P, tid, NT, D.E, D.N, P.gpe2, P.gpn2,
+                                  P.xh_m2e, P.xh_m2s, P.gu_m2e, P.gu_m2s,
+                                  W_LN_R2_W, tG, tU, tGU, tXH, sp, nsplit,
+                                  pout);
+        return;
+    }
   }
 
   // ---- special jobs over the B sample rows ----
-  if (blockIdx.x == 6) {
+  if ((int)blockIdx.x == jobs_end) {
     // graph module: Wg[GEMB,GFin], bg; LN gamma/beta from gu34 + xh34
     const float* gam = WP(W_LN_G_W);
     const float* bet = WP(W_LN_G_B);
@@ -1124,7 +1176,7 @@ cs_bwd_w_kernel(CachedPtrs P, CachedDims D) {
     }
     return;
   }
-  if (blockIdx.x == 7) {
+  if ((int)blockIdx.x == jobs_end + 1) {
     // W2p[A,FC] = sum_b glogits (x) h1p (LDS-tiled); b2p folded in
     __shared__ float lbuf[TILE_K * (KA + KFC)];
     wgrad_plain<KFC, KA>(P, tid, NT, D.B, P.glogits, KA, P.h1p, KFC,
@@ -1143,11 +1195,11 @@ cs_bwd_w_kernel(CachedPtrs P, CachedDims D) {
     }
     return;
   }
-  // blocks 8..: W1p/W1v [FC,FIN] split FC-rows across blocks; LDS tiles of
-  // fin AND the gh1p/gh1v column slices
+  // remaining blocks: W1p/W1v [FC,FIN] split FC-rows across blocks; LDS
+  // tiles of fin AND the gh1p/gh1v column slices
   {
-    const int nb = gridDim.x - 8;
-    const int jb = blockIdx.x - 8;
+    const int nb = gridDim.x - jobs_end - 2;
+    const int jb = blockIdx.x - jobs_end - 2;
     const int per = (KFC + nb - 1) / nb;
     const int j0 = jb * per, j1 = min(KFC, j0 + per);
     const int JW = j1 - j0;
@@ -1211,6 +1263,45 @@ cs_bwd_w_kernel(CachedPtrs P, CachedDims D) {
       WG_(W_P1_B)[j] = ap;
       WG_(W_V1_B)[j] = av;
     }
+  }
+}
+
+// sum the row-split partials (fixed split order: deterministic) into the
+// flat gradient; one block per job
+__global__ void __launch_bounds__(256)
+cs_bwd_w_reduce_kernel(CachedPtrs P, CachedDims D) {
+  int Din, Dout, slot;
+  switch (blockIdx.x) {
+    case 0: Din = KF0;  Dout = KH;   slot = W_LN_N1_W; break;
+    case 1: Din = KFE;  Dout = KH;   slot = W_LN_E1_W; break;
+    case 2: Din = KMSG; Dout = KHID; slot = W_LN_R1_W; break;
+    case 3: Din = KHID; Dout = KH;   slot = W_LN_N2_W; break;
+    case 4: Din = KFE;  Dout = KH;   slot = W_LN_E2_W; break;
+    default: Din = KMSG; Dout = KOUT; slot = W_LN_R2_W; break;
+  }
+  const float* __restrict__ base =
+      P.wg_scratch + (long)blockIdx.x * WSPLIT * WG_JSTRIDE;
+  const int tid = threadIdx.x, NT = blockDim.x;
+  const int units = Din * Dout;
+  for (int u = tid; u < units; u += NT) {
+    float a = 0.f;
+    for (int s = 0; s < WSPLIT; ++s) a += base[(long)s * WG_JSTRIDE + u];
+    WG_(slot + 2)[u] = a;
+  }
+  for (int o = tid; o < Dout; o += NT) {
+    float a = 0.f;
+    for (int s = 0; s < WSPLIT; ++s)
+      a += base[(long)s * WG_JSTRIDE + units + o];
+    WG_(slot + 3)[o] = a;
+  }
+  for (int i = tid; i < Din; i += NT) {
+    float aw = 0.f, ab = 0.f;
+    for (int s = 0; s < WSPLIT; ++s) {
+      aw += base[(long)s * WG_JSTRIDE + units + Dout + i];
+      ab += base[(long)s * WG_JSTRIDE + units + Dout + Din + i];
+    }
+    WG_(slot)[i] = aw;
+    WG_(slot + 1)[i] = ab;
   }
 }
 
@@ -1337,6 +1428,9 @@ static void fill_ptrs(CachedPtrs& P, CachedDims& D,
   P.gu_m1s = T[C_GU_M1S].data_ptr<float>();
   P.gu_z1 = T[C_GU_Z1].data_ptr<float>();
   P.gu_e1 = T[C_GU_E1].data_ptr<float>();
+  P.wg_scratch = T[C_WG_SCRATCH].data_ptr<float>();
+  TORCH_CHECK(T[C_WG_SCRATCH].numel() >= (long)6 * WSPLIT * WG_JSTRIDE,
+              "cached_step: wg_scratch too small");
 }
 
 void cached_step_fwd(std::vector<torch::Tensor> T, std::vector<double> fs) {
@@ -1364,6 +1458,8 @@ void cached_step_fwd(std::vector<torch::Tensor> T, std::vector<double> fs) {
   int head_blocks = ((long)D.B * D.FC + 255) / 256;
   if (head_blocks > 128) head_blocks = 128;
   hipLaunchKernelGGL(cs_fwd_head_kernel, dim3(head_blocks), dim3(256), 0,
+                     stream, P, D);
+  hipLaunchKernelGGL(cs_fwd_logits_kernel, dim3(D.B), dim3(512), 0,
                      stream, P, D);
   hipLaunchKernelGGL(cs_fwd_loss_kernel, dim3((D.B + 31) / 32), dim3(256),
                      0, stream, P, D);
@@ -1393,6 +1489,10 @@ void cached_step_bwd(std::vector<torch::Tensor> T, std::vector<double> fs) {
                      stream, P, D);
   hipLaunchKernelGGL(cs_bwd_gu1_kernel, dim3(rows_b), dim3(256), 0, stream,
                      P, D);
-  hipLaunchKernelGGL(cs_bwd_w_kernel, dim3(8 + 16), dim3(256), 0, stream,
-                     P, D);
+  const int nsplit = D.wgrad_mfma ? 1 : WSPLIT;
+  hipLaunchKernelGGL(cs_bwd_w_kernel, dim3(6 * nsplit + 2 + 16), dim3(256),
+                     0, stream, P, D);
+  if (nsplit > 1)
+    hipLaunchKernelGGL(cs_bwd_w_reduce_kernel, dim3(6), dim3(256), 0,
+                       stream, P, D);
 }

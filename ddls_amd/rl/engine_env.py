@@ -14,6 +14,7 @@ per-model views — zero copies).
 """
 from __future__ import annotations
 
+import sys
 import time
 from typing import Dict, List, Optional
 
@@ -63,8 +64,9 @@ class EngineVectorEnv:
         self._jct_base_count = 0
         self._jct_base_sum = 0.0
         if verbose:
+            # stderr: bench.py's stdout must stay a single JSON line
             print(f"[engine_env] init {num_envs} envs in "
-                  f"{time.perf_counter() - t0:.2f}s")
+                  f"{time.perf_counter() - t0:.2f}s", file=sys.stderr)
 
     # ------------------------------------------------------------------
     def _drain(self, b: int):
