@@ -78,31 +78,19 @@ enum {
   // the LN gamma/beta reductions in cs_bwd_w
   C_GU_M2E, C_GU_M2S, C_GU_H2, C_GU_E2, C_GU_M1E, C_GU_M1S, C_GU_Z1,
   C_GU_E1,
-  // row-split wgrad partial sums: [WG_SLOTS_TOTAL][WG_JSTRIDE]
+  // row-split wgrad partial sums: [6 jobs][WSPLIT][WG_JSTRIDE]
   C_WG_SCRATCH,
   C_NT
 };
 
 // The six row-heavy wgrad jobs (module weight grads over N/E rows) are
-// row-split — a single workgroup per job leaves one CU streaming ~4 MB
-// from HBM while the other 255 idle (measured 85 us, 22% of the fused
-// step).  The two (edge ++ self)-row jobs (reduce modules, N+E rows) are
-// the bandwidth hogs and get 128 splits; the rest 32.  Partials land in
-// wg_scratch in a FIXED layout and a reduce kernel sums them in fixed
-// split order: deterministic, no atomics.
-// WG_JSTRIDE = max(Din*Dout) + max(Dout) + 2*max(Din).
-#define WG_NSPLIT_BIG 128
-#define WG_NSPLIT_SMALL 32
-#define WG_SLOTS_TOTAL (4 * WG_NSPLIT_SMALL + 2 * WG_NSPLIT_BIG)
+// row-split WSPLIT ways — a single workgroup per job leaves one CU
+// streaming ~4 MB from HBM while the other 255 idle (measured 85 us, 22%
+// of the fused step).  Partials land in wg_scratch in a FIXED layout and
+// a 6-block reduce kernel sums them in fixed split order: deterministic,
+// no atomics.  WG_JSTRIDE = max(Din*Dout) + max(Dout) + 2*max(Din).
+#define WSPLIT 16
 #define WG_JSTRIDE (KMSG * KHID + KHID + KHID + KHID)
-__device__ __forceinline__ int wg_nsplit(int job) {
-  return (job == 2 || job == 5) ? WG_NSPLIT_BIG : WG_NSPLIT_SMALL;
-}
-__device__ __forceinline__ int wg_slot_off(int job) {
-  int o = 0;
-  for (int j = 0; j < job; ++j) o += wg_nsplit(j);
-  return o;
-}
 
 // weight slots within OFFS (order fixed; mirror _fused_offsets in Python)
 enum {
@@ -1092,25 +1080,15 @@ cs_bwd_w_kernel(CachedPtrs P, CachedDims D) {
   __shared__ float ldsG[TILE_K * 64], ldsU[TILE_K * 64];
   __shared__ float ldsGU[TILE_K * 64], ldsXH[TILE_K * 64];
 
-  // jobs 0..5 are row-split (one WG per (job, split)); the MFMA variants
-  // (env opt-in) keep single-WG jobs
-  const int jobs_end = D.wgrad_mfma ? 6 : WG_SLOTS_TOTAL;
+  // jobs 0..5 are row-split nsplit ways (one WG per (job, split)); the
+  // MFMA variants (env opt-in) keep single-WG jobs
+  const int nsplit = D.wgrad_mfma ? 1 : WSPLIT;
+  const int jobs_end = 6 * nsplit;
   if ((int)blockIdx.x < jobs_end) {
-    int job, sp, nsplit;
-    if (D.wgrad_mfma) {
-      job = blockIdx.x;
-      sp = 0;
-      nsplit = 1;
-    } else {
-      int jb = blockIdx.x;
-      job = 0;
-      while (jb >= wg_nsplit(job)) { jb -= wg_nsplit(job); ++job; }
-      sp = jb;
-      nsplit = wg_nsplit(job);
-    }
+    const int job = blockIdx.x / nsplit;
+    const int sp = blockIdx.x % nsplit;
     float* pout = (nsplit > 1)
-        ? P.wg_scratch + ((long)wg_slot_off(job) + sp) * WG_JSTRIDE
-        : nullptr;
+        ? P.wg_scratch + ((long)job * WSPLIT + sp) * WG_JSTRIDE : nullptr;
     switch (job) {
       case 0:  // node module 1
         wgrad_tiled<KF0, KH>(P, tid, NT, D.N, 0, P.ghn1, nullptr, P.xh_z1,
@@ -1289,15 +1267,11 @@ cs_bwd_w_kernel(CachedPtrs P, CachedDims D) {
 }
 
 // sum the row-split partials (fixed split order: deterministic) into the
-// flat gradient; 8 unit-chunk blocks per job so the 128-split jobs don't
-// serialize the sum on one CU
-#define WG_RED_CHUNKS 8
+// flat gradient; one block per job
 __global__ void __launch_bounds__(256)
 cs_bwd_w_reduce_kernel(CachedPtrs P, CachedDims D) {
-  const int job = blockIdx.x / WG_RED_CHUNKS;
-  const int chunk = blockIdx.x % WG_RED_CHUNKS;
   int Din, Dout, slot;
-  switch (job) {
+  switch (blockIdx.x) {
     case 0: Din = KF0;  Dout = KH;   slot = W_LN_N1_W; break;
     case 1: Din = KFE;  Dout = KH;   slot = W_LN_E1_W; break;
     case 2: Din = KMSG; Dout = KHID; slot = W_LN_R1_W; break;
@@ -1305,28 +1279,24 @@ cs_bwd_w_reduce_kernel(CachedPtrs P, CachedDims D) {
     case 4: Din = KFE;  Dout = KH;   slot = W_LN_E2_W; break;
     default: Din = KMSG; Dout = KOUT; slot = W_LN_R2_W; break;
   }
-  const int nsplit = wg_nsplit(job);
   const float* __restrict__ base =
-      P.wg_scratch + (long)wg_slot_off(job) * WG_JSTRIDE;
+      P.wg_scratch + (long)blockIdx.x * WSPLIT * WG_JSTRIDE;
   const int tid = threadIdx.x, NT = blockDim.x;
   const int units = Din * Dout;
-  const int per = (units + WG_RED_CHUNKS - 1) / WG_RED_CHUNKS;
-  const int u_lo = chunk * per, u_hi = min(units, u_lo + per);
-  for (int u = u_lo + tid; u < u_hi; u += NT) {
+  for (int u = tid; u < units; u += NT) {
     float a = 0.f;
-    for (int s = 0; s < nsplit; ++s) a += base[(long)s * WG_JSTRIDE + u];
+    for (int s = 0; s < WSPLIT; ++s) a += base[(long)s * WG_JSTRIDE + u];
     WG_(slot + 2)[u] = a;
   }
-  if (chunk != 0) return;
   for (int o = tid; o < Dout; o += NT) {
     float a = 0.f;
-    for (int s = 0; s < nsplit; ++s)
+    for (int s = 0; s < WSPLIT; ++s)
       a += base[(long)s * WG_JSTRIDE + units + o];
     WG_(slot + 3)[o] = a;
   }
   for (int i = tid; i < Din; i += NT) {
     float aw = 0.f, ab = 0.f;
-    for (int s = 0; s < nsplit; ++s) {
+    for (int s = 0; s < WSPLIT; ++s) {
       aw += base[(long)s * WG_JSTRIDE + units + Dout + i];
       ab += base[(long)s * WG_JSTRIDE + units + Dout + Din + i];
     }
@@ -1459,8 +1429,7 @@ static void fill_ptrs(CachedPtrs& P, CachedDims& D,
   P.gu_z1 = T[C_GU_Z1].data_ptr<float>();
   P.gu_e1 = T[C_GU_E1].data_ptr<float>();
   P.wg_scratch = T[C_WG_SCRATCH].data_ptr<float>();
-  TORCH_CHECK(T[C_WG_SCRATCH].numel()
-                  >= (long)WG_SLOTS_TOTAL * WG_JSTRIDE,
+  TORCH_CHECK(T[C_WG_SCRATCH].numel() >= (long)6 * WSPLIT * WG_JSTRIDE,
               "cached_step: wg_scratch too small");
 }
 
@@ -1520,10 +1489,10 @@ void cached_step_bwd(std::vector<torch::Tensor> T, std::vector<double> fs) {
                      stream, P, D);
   hipLaunchKernelGGL(cs_bwd_gu1_kernel, dim3(rows_b), dim3(256), 0, stream,
                      P, D);
-  const int wg_jobs = D.wgrad_mfma ? 6 : WG_SLOTS_TOTAL;
-  hipLaunchKernelGGL(cs_bwd_w_kernel, dim3(wg_jobs + 2 + 16), dim3(256),
+  const int nsplit = D.wgrad_mfma ? 1 : WSPLIT;
+  hipLaunchKernelGGL(cs_bwd_w_kernel, dim3(6 * nsplit + 2 + 16), dim3(256),
                      0, stream, P, D);
-  if (!D.wgrad_mfma)
-    hipLaunchKernelGGL(cs_bwd_w_reduce_kernel, dim3(6 * WG_RED_CHUNKS),
-                       dim3(256), 0, stream, P, D);
+  if (nsplit > 1)
+    hipLaunchKernelGGL(cs_bwd_w_reduce_kernel, dim3(6), dim3(256), 0,
+                       stream, P, D);
 }

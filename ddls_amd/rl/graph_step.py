@@ -185,9 +185,9 @@ class _FusedCachedEngine:
             # LN-out grad rows (gu) stored by the data-bwd kernels
             z(E, MSG), z(N, MSG), z(N, HID), z(E, FE),
             z(E, MSG), z(N, MSG), z(N, F0), z(E, FE),
-            # row-split wgrad partials: WG_SLOTS_TOTAL(384) x WG_JSTRIDE
-            # (2240) — mirrors cached_step.hip's split tables
-            z(384 * 2240),
+            # row-split wgrad partials: 6 jobs x WSPLIT(16) x WG_JSTRIDE
+            # (2240) — mirrors cached_step.hip's WSPLIT/WG_JSTRIDE
+            z(6 * 16 * 2240),
         ]
         c = stepper.cfg
         self.fscal = [float(c.clip_param), float(c.vf_clip_param),
