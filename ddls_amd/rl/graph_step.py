@@ -119,6 +119,10 @@ class _FusedCachedEngine:
         "value_branch.2.weight", "value_branch.2.bias",
     ]
 
+    # must equal the C_* enum length in ops/hip/cached_step.hip (C_NT);
+    # pinned by tests/test_aux.py::test_fused_scratch_contract
+    N_SCRATCH = 90
+
     def __init__(self, stepper, models_batch):
         self.st = stepper
         ext = stepper._ext
@@ -189,6 +193,7 @@ class _FusedCachedEngine:
             # (2240) — mirrors cached_step.hip's WSPLIT/WG_JSTRIDE
             z(6 * 16 * 2240),
         ]
+        assert len(self.scratch) == self.N_SCRATCH, len(self.scratch)
         c = stepper.cfg
         self.fscal = [float(c.clip_param), float(c.vf_clip_param),
                       float(c.vf_loss_coeff), float(c.entropy_coeff)]

@@ -473,3 +473,34 @@ def test_medium_workload_learned_policy_matches_optimum():
         "learned policy should equal the optimal heuristic exactly")
     assert learned["episode_return"] > acceptable["episode_return"] * 1.0
     assert learned["blocking_rate"] == sipml["blocking_rate"]
+
+
+def test_fused_scratch_contract():
+    """The fused cached-step tensor list (graph_step._FusedCachedEngine)
+    must match cached_step.hip's C_* enum length and split constants —
+    guards against one side drifting (a mismatched list is a silent
+    wrong-pointer bug on the GPU, caught here on CPU)."""
+    import os
+    import re
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    src = open(os.path.join(root, "ddls_amd", "ops", "hip",
+                            "cached_step.hip")).read()
+    body = re.search(r"enum \{(.*?)\bC_NT\b", src, re.S).group(1)
+    body = re.sub(r"//[^\n]*", "", body)
+    names = re.findall(r"\bC_[A-Z0-9_]+\b", body)
+    from ddls_amd.rl.graph_step import _FusedCachedEngine
+    assert len(names) == _FusedCachedEngine.N_SCRATCH
+    assert len(set(names)) == len(names)
+    # split constants: the Python wg_scratch allocation (6 * WSPLIT *
+    # WG_JSTRIDE) must cover what the kernel indexes
+    wsplit = int(re.search(r"#define WSPLIT (\d+)", src).group(1))
+    kmsg = int(re.search(r"#define KMSG (\d+)", src).group(1))
+    khid = int(re.search(r"#define KHID (\d+)", src).group(1))
+    jstride = kmsg * khid + 3 * khid
+    assert wsplit == 16 and jstride == 2240, (wsplit, jstride)
+    # the W_* weight-slot enum must match the Python _SLOTS order length
+    wbody = re.search(r"enum \{\s*(W_LN_N1_W.*?)\bW_COUNT\b", src,
+                      re.S).group(1)
+    wnames = re.findall(r"\bW_[A-Z0-9_]+\b", re.sub(r"//[^\n]*", "", wbody))
+    assert len(wnames) == len(_FusedCachedEngine._SLOTS) == 36
