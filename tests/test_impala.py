@@ -276,3 +276,37 @@ def test_dqn_and_es_on_engine_env(multi_model_files):
                    device=dev)
     st = es.train()
     assert np.isfinite(st["fitness_mean"])
+
+
+@pytest.mark.parametrize("group,cls_path", [
+    ("pg", "ddls_amd.rl.pg.PGTrainer"),
+    ("apex_dqn", "ddls_amd.rl.dqn.DQNTrainer"),
+    ("es", "ddls_amd.rl.es.ESTrainer"),
+])
+def test_remaining_config_groups_build(multi_model_files, group, cls_path):
+    """Each remaining reference algo config group
+    (scripts/ramp_job_partitioning_configs/algo/{pg,apex_dqn,es}.yaml)
+    routes build_trainer_from_config to its trainer and runs an
+    iteration on CPU."""
+    import importlib
+    import os
+
+    import yaml
+
+    from ddls_amd.runtime.config import build_trainer_from_config, load_config
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cfg = load_config(os.path.join(root, "configs", "train_config.yaml"))
+    with open(os.path.join(root, "configs", "algo", f"{group}.yaml")) as f:
+        cfg["algo"] = yaml.safe_load(f)
+    cfg["env_config"]["jobs_config"]["path_to_files"] = multi_model_files
+    cfg["env_config"]["jobs_config"]["replication_factor"] = 2
+    cfg["epoch_loop"]["num_envs"] = 2
+    cfg["epoch_loop"]["num_env_workers"] = 1
+    cfg["epoch_loop"]["precompute_lookaheads"] = False
+    mod, name = cls_path.rsplit(".", 1)
+    cls = getattr(importlib.import_module(mod), name)
+    tr = build_trainer_from_config(cfg, device=torch.device("cpu"))
+    assert isinstance(tr, cls)
+    st = tr.train(num_steps=4)
+    assert np.isfinite(st.get("total_loss", 0.0))
+    tr.env.close()
