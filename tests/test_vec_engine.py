@@ -83,7 +83,8 @@ def _memo_preload_for_env(spec):
 
 def run_parity(jobs_dir, seed, steps, mode="remove", replication=3,
                max_sim=1e6, interarrival=40, reward_kwargs=None,
-               reward="lookahead_job_completion_time"):
+               reward="lookahead_job_completion_time",
+               action_fn=scripted_action):
     # spec compiled from a scratch env (empty-cluster pipeline runs)
     env_c = make_env(jobs_dir, mode, replication, max_sim, interarrival,
                      reward_kwargs=reward_kwargs, reward=reward)
@@ -119,7 +120,7 @@ def run_parity(jobs_dir, seed, steps, mode="remove", replication=3,
         np.testing.assert_array_equal(obs["edge_features"],
                                       spec.models[mid].edge_features)
 
-        a = scripted_action(obs["action_mask"], t)
+        a = action_fn(obs["action_mask"], t)
         obs, reward, done, _ = env.step(int(a))
         status = cpu_step_env(spec, st, 0, sched, int(a))
         assert status == ST_OK
@@ -374,3 +375,25 @@ def test_mirror_matches_real_env_1024_workers(multi_model_files):
         assert bool(st.done[0]) == bool(done)
         if done:
             break
+
+
+def test_mirror_parity_random_action_streams(multi_model_files):
+    """Property-style sweep: the mirror must match the real env bitwise
+    under RANDOM valid-action streams (not just the scripted cycle), across
+    episode seeds and arrival intensities — broadens coverage of placement/
+    queue interleavings the scripted pattern cannot reach."""
+    cases = [(101, 5001, 15), (202, 5002, 40), (303, 5003, 90),
+             (404, 5004, 25), (505, 5005, 60)]
+    for ep_seed, act_seed, interarrival in cases:
+        rng = np.random.RandomState(act_seed)
+
+        def rand_action(mask, t, rng=rng):
+            valid = [a for a in range(len(mask)) if mask[a]]
+            if rng.rand() < 0.15:
+                return 0
+            return int(valid[rng.randint(len(valid))])
+
+        env, spec, sched, st_, n = run_parity(
+            multi_model_files, seed=ep_seed, steps=30,
+            interarrival=interarrival, action_fn=rand_action)
+        assert n > 0
