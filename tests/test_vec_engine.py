@@ -397,3 +397,27 @@ def test_mirror_parity_random_action_streams(multi_model_files):
             multi_model_files, seed=ep_seed, steps=30,
             interarrival=interarrival, action_fn=rand_action)
         assert n > 0
+
+
+def test_mirror_job_acceptance_reward(multi_model_files):
+    """JobAcceptance (+success/-fail on accept/block, reference
+    ``job_acceptance.py:9-33``): mirror reward bitwise-equal to the real
+    env over a full episode, including blocked steps."""
+    env, spec, sched, st_, n = run_parity(
+        multi_model_files, seed=11, steps=40, interarrival=15,
+        reward="job_acceptance",
+        reward_kwargs={"fail_reward": -2.0, "success_reward": 0.5})
+    assert n > 0
+
+
+def test_mirror_multi_objective_reward(multi_model_files):
+    """MultiObjectiveJCTBlocking (weighted -JCT + acceptance, reference
+    ``multi_objective_jct_blocking.py:9-90``): mirror reward bitwise-equal
+    to the real env."""
+    env, spec, sched, st_, n = run_parity(
+        multi_model_files, seed=13, steps=40, interarrival=20,
+        reward="multi_objective_jct_blocking",
+        reward_kwargs={"jct_weight": 0.7, "blocking_weight": 0.3,
+                       "transform_with_log": True,
+                       "fail_reward_factor": 2.0})
+    assert n > 0
