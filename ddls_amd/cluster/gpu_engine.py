@@ -252,6 +252,10 @@ class GpuEngine:
     def _scalars(self):
         spec = self.spec
         r = spec.reward
+        if getattr(r, "tp_kind", 0):
+            raise ValueError(
+                "GpuEngine: throughput rewards are CPU-mirror-only for now "
+                "(env_step.hip port pending) — use the CPU env path")
         iscal = [self.B, spec.C, spec.R, spec.S, spec.W, self.WW, spec.A,
                  self.K, self.T["_SEQ_CAP"], self.T["_PAR_CAP"], self.SCH,
                  self.NJOBS, self.HS, int(spec.infinite_pool),

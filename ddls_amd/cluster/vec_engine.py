@@ -118,6 +118,13 @@ class RewardSpec:
     acc_fail: float = -1.0
     jct_weight: float = 1.0
     blocking_weight: float = 0.0  # 0 -> pure JCT reward
+    # throughput family (reference mean_compute_throughput.py:9-57 etc.):
+    # 0 none, 1 compute (partitioned op memory), 2 cluster (op memory +
+    # dep size), 3 demand_total (ORIGINAL job's totals).  Reward = the
+    # first cluster step's info_processed / step_time (the RL env extracts
+    # the reward right after cluster.step(action), before fast-forward).
+    # CPU-mirror support; the GPU kernel port is pending (GpuEngine gates).
+    tp_kind: int = 0
 
 
 @dataclass
@@ -235,6 +242,8 @@ def _graph_features_static(job, cluster) -> np.ndarray:
 
 def _reward_spec_from_env(env) -> RewardSpec:
     from ..envs.rewards import (JobAcceptance, LookaheadJobCompletionTime,
+                                MeanClusterThroughput, MeanComputeThroughput,
+                                MeanDemandTotalThroughput,
                                 MultiObjectiveJCTBlocking)
     rf = env.reward_function
     norm_map = {None: 0, "job_sequential_completion_time": 1,
@@ -270,11 +279,18 @@ def _reward_spec_from_env(env) -> RewardSpec:
         out.acc_fail = float(rf.acceptance.fail_reward)
         out.jct_weight = float(rf.jct_weight)
         out.blocking_weight = float(rf.blocking_weight)
+    elif isinstance(rf, MeanComputeThroughput):
+        out.kind, out.tp_kind = "mean_compute_throughput", 1
+        out.jct_weight = out.blocking_weight = 0.0
+    elif isinstance(rf, MeanClusterThroughput):
+        out.kind, out.tp_kind = "mean_cluster_throughput", 2
+        out.jct_weight = out.blocking_weight = 0.0
+    elif isinstance(rf, MeanDemandTotalThroughput):
+        out.kind, out.tp_kind = "mean_demand_total_throughput", 3
+        out.jct_weight = out.blocking_weight = 0.0
     else:
         raise ValueError(
-            f"engine: unsupported reward function {type(rf).__name__} "
-            "(throughput rewards need the full per-tick sim — use the CPU "
-            "env path)")
+            f"engine: unsupported reward function {type(rf).__name__}")
     return out
 
 
@@ -858,8 +874,10 @@ def cpu_step_env(spec: EngineSpec, st: EngineState, b: int,
         st.order_counter[b] += 1
     st.queued[b] = -1
 
-    # reward (RL env computes it straight after the first cluster.step;
-    # nothing in the event loop below affects it)
+    # reward (the RL env extracts it straight after the FIRST cluster.step;
+    # JCT/acceptance parts depend only on the placement decision, the
+    # throughput family on that step's info_processed/step_time — which the
+    # event loop below accrues tick by tick)
     r = spec.reward
     reward = 0.0
     if r.jct_weight != 0.0:
@@ -868,9 +886,10 @@ def cpu_step_env(spec: EngineSpec, st: EngineState, b: int,
                                              not placed, norm_seq)
     if r.blocking_weight != 0.0:
         reward += r.blocking_weight * (r.acc_success if placed else r.acc_fail)
-    st.reward[b] = reward
-    st.ep_return[b] += reward
     st.ep_len[b] += 1
+    tp_acc = 0.0
+    tp_time = None          # set when the first cluster step closes
+    t_step_start = float(st.t[b])
 
     # ---- outer event loop (+ idle fast-forward) ----
     eps = spec.eps
@@ -882,6 +901,20 @@ def cpu_step_env(spec: EngineSpec, st: EngineState, b: int,
             remaining = st.slot_jct[b, s] - elapsed
             tick = min(tick, remaining)
         st.snapshot[b] = int(sum(bin(int(w)).count("1") for w in st.occ[b]))
+        if r.tp_kind and tp_time is None:
+            # mirror of environment.py:658-674: per running job (mount
+            # order), frac = tick/jct, info += quantity * frac — same f64
+            # accumulation order, so the ratio is bitwise-equal
+            for s_ in range(int(st.n_running[b])):
+                frac = float(tick) / float(st.slot_jct[b, s_])
+                md_ = spec.mds[int(st.slot_md[b, s_])]
+                if r.tp_kind == 1:
+                    tp_acc += md_.pj_mem * frac
+                elif r.tp_kind == 2:
+                    tp_acc += (md_.pj_mem + md_.pj_dep) * frac
+                else:
+                    ms_ = spec.models[md_.model_id]
+                    tp_acc += (ms_.mem_total + ms_.dep_total) * frac
         st.t[b] = st.t[b] + tick
         # completions (in running order; removal preserves order)
         nr = int(st.n_running[b])
@@ -918,8 +951,22 @@ def cpu_step_env(spec: EngineSpec, st: EngineState, b: int,
         pool_empty = (not spec.infinite_pool) and math.isinf(st.next_arrive[b])
         done = (st.t[b] >= spec.max_sim) or (
             pool_empty and st.n_running[b] == 0 and st.queued[b] < 0)
+        # a cluster step ends at the first COMPLETION or ARRIVAL (or done):
+        # the RL env's reward covers only that first cluster step; later
+        # iterations here belong to its idle fast-forward
+        if (r.tp_kind and tp_time is None
+                and (len(keep) != nr or st.queued[b] >= 0 or done)):
+            tp_time = float(st.t[b]) - t_step_start
         if st.queued[b] >= 0 or done:
             break
+
+    if r.tp_kind:
+        step_time = (tp_time if tp_time is not None
+                     else float(st.t[b]) - t_step_start)
+        reward = (tp_acc / step_time
+                  if (tp_acc != 0 and step_time != 0) else 0.0)
+    st.reward[b] = reward
+    st.ep_return[b] += reward
 
     if done:
         # finalise: still-running jobs are blocked (reference :1111-1121)

@@ -421,3 +421,18 @@ def test_mirror_multi_objective_reward(multi_model_files):
                        "transform_with_log": True,
                        "fail_reward_factor": 2.0})
     assert n > 0
+
+
+@pytest.mark.parametrize("reward", ["mean_compute_throughput",
+                                    "mean_cluster_throughput",
+                                    "mean_demand_total_throughput"])
+def test_mirror_throughput_rewards(multi_model_files, reward):
+    """Throughput reward family (reference mean_compute_throughput.py:9-57,
+    mean_cluster_throughput.py, mean_demand_total_throughput.py): the
+    mirror accrues per-tick info_processed with the env's exact f64
+    summation order, so the first-cluster-step info/step_time reward is
+    bitwise-equal over whole episodes."""
+    env, spec, sched, st_, n = run_parity(
+        multi_model_files, seed=17, steps=60, interarrival=25,
+        replication=4, reward=reward)
+    assert n > 0
