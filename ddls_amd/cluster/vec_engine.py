@@ -1153,3 +1153,61 @@ def build_episode_stats(spec: EngineSpec, sched: EpisodeSchedule,
         rates.append((ms.mem_total + ms.dep_total) / denom)
     es["mean_load_rate"] = float(np.mean(rates)) if rates else 0
     return es
+
+
+# ---------------------------------------------------------------------------
+# CPU engine backend: the mirror as a production vectorised engine
+# ---------------------------------------------------------------------------
+
+class CpuEngine:
+    """Drop-in CPU counterpart of ``gpu_engine.GpuEngine`` built on the
+    (bitwise parity-tested) mirror ``cpu_step_env``: the same ``T`` tensor
+    surface (zero-copy torch views of the numpy SoA state), ``step``,
+    ``reset_env`` and ``episode_stats`` — so ``EngineVectorEnv`` runs at
+    engine speed on CPU-only machines instead of requiring subprocess env
+    workers.  No RampJobPartitioningEnvironment objects are stepped after
+    spec compile; the per-env step is the compact-state mirror."""
+
+    def __init__(self, spec: EngineSpec, B: int, device=None,
+                 n_jobs_cap: int = 64, sch_cap: Optional[int] = None):
+        import torch
+        self.spec = spec
+        self.B = B
+        self.st = EngineState(spec, B=B, n_jobs_cap=n_jobs_cap)
+        self.schedules: List[Optional[EpisodeSchedule]] = [None] * B
+        st = self.st
+        self.T = {
+            "obs_gf": torch.from_numpy(st.obs_gf),
+            "obs_mask": torch.from_numpy(st.obs_mask),
+            "obs_model": torch.from_numpy(st.obs_model),
+            "reward": torch.from_numpy(st.reward),
+            "done": torch.from_numpy(st.done),
+            "status": torch.from_numpy(st.status),
+            "log_status": torch.from_numpy(st.log_status),
+            "log_t_arr": torch.from_numpy(st.log_t_arr),
+            "log_t_end": torch.from_numpy(st.log_t_end),
+        }
+
+    def reset_env(self, b: int, sched: EpisodeSchedule):
+        if sched.n + 1 > self.st.n_jobs_cap:
+            raise ValueError(
+                f"CpuEngine: schedule length {sched.n} exceeds capacity "
+                f"{self.st.n_jobs_cap} — construct with a larger n_jobs_cap")
+        self.schedules[b] = sched
+        self.st.reset_env(self.spec, b, sched)
+
+    def step(self, actions, active: Optional[List[int]] = None):
+        acts = actions.detach().cpu().numpy()
+        idx = range(self.B) if active is None else active
+        for b in idx:
+            if self.st.done[b]:
+                self.st.status[b] = 0
+                continue
+            status = cpu_step_env(self.spec, self.st, b,
+                                  self.schedules[b], int(acts[b]))
+            if status != ST_OK:
+                raise RuntimeError(f"CpuEngine env {b}: status {status}")
+        return self.T["status"]
+
+    def episode_stats(self, b: int) -> Dict:
+        return build_episode_stats(self.spec, self.schedules[b], self.st, b)

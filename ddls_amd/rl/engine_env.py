@@ -49,8 +49,14 @@ class EngineVectorEnv:
         self.episode_counters = [0] * num_envs
         self.scheds = [self._drain(i) for i in range(num_envs)]
         cap = max(s.n for s in self.scheds) + 8
-        self.eng = GpuEngine(spec, B=num_envs, device=self.device,
-                             n_jobs_cap=cap)
+        if self.device.type == "cuda":
+            self.eng = GpuEngine(spec, B=num_envs, device=self.device,
+                                 n_jobs_cap=cap)
+        else:
+            # CPU backend: the parity-tested mirror as a vectorised engine
+            # (engine-speed rollouts without subprocess env workers)
+            from ..cluster.vec_engine import CpuEngine
+            self.eng = CpuEngine(spec, B=num_envs, n_jobs_cap=cap)
         for b in range(num_envs):
             self.eng.reset_env(b, self.scheds[b])
         self.completed_episode_stats: List[dict] = []

@@ -212,3 +212,57 @@ def test_bench_dp2_gloo_single_gpu_rehearsal(tmp_path):
     assert rec["n_gpus"] == 2
     assert rec["config"]["env_engine"] == "gpu_resident"
     assert rec["value"] > 0
+
+
+def test_ppo_on_cpu_engine_env(multi_model_files):
+    """The CpuEngine backend (vec_engine.CpuEngine — the parity-tested
+    mirror as a vectorised engine): EngineVectorEnv on device='cpu' runs
+    PPO end-to-end with the same trainer-facing surface as the GPU engine,
+    replacing subprocess env workers on CPU-only machines."""
+    from ddls_amd.cluster.vec_engine import CpuEngine
+    from ddls_amd.models.gnn import GNNPolicy
+    from ddls_amd.rl.engine_env import EngineVectorEnv
+    from ddls_amd.rl.ppo import PPOConfig, PPOTrainer
+
+    dev = torch.device("cpu")
+    torch.manual_seed(0)
+    policy = GNNPolicy(num_actions=17)
+    venv = EngineVectorEnv(
+        lambda: make_env(multi_model_files, "remove_and_repeat", 2, 3000, 15),
+        num_envs=8, device=dev, base_seed=11)
+    assert isinstance(venv.eng, CpuEngine)
+    cfg = PPOConfig(train_batch_size=8 * 8, sgd_minibatch_size=32,
+                    num_sgd_iter=3)
+    trainer = PPOTrainer(venv, policy, cfg, device=dev)
+    for _ in range(2):
+        stats = trainer.train(num_steps=8)
+        assert np.isfinite(stats["total_loss"])
+        assert np.isfinite(stats["mean_reward"])
+    assert trainer.total_env_steps == 2 * 8 * 8
+    venv.drain_episode_stats()
+    c, s = venv.jct_running_stats()
+    assert c > 0 and np.isfinite(s)
+
+
+def test_cpu_engine_episode_stats_roundtrip(multi_model_files):
+    """CpuEngine full episode: completed-episode stats drain through
+    EngineVectorEnv and carry the exact episode-stats vocabulary."""
+    from ddls_amd.models.gnn import GNNPolicy
+    from ddls_amd.rl.engine_env import EngineVectorEnv
+
+    torch.manual_seed(1)
+    policy = GNNPolicy(num_actions=17)
+    venv = EngineVectorEnv(
+        lambda: make_env(multi_model_files, "remove", 2, 2500, 15),
+        num_envs=4, device=torch.device("cpu"), base_seed=3)
+    total = 0
+    for _ in range(40):
+        venv.rollout(policy, steps=4)
+        stats = venv.drain_episode_stats()
+        total += len(stats)
+        for es in stats:
+            assert es["num_jobs_arrived"] > 0
+            assert 0.0 <= es["blocking_rate"] <= 1.0
+        if total >= 2:
+            break
+    assert total >= 1, "at least one episode should complete"
