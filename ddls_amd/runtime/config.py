@@ -136,17 +136,41 @@ def build_trainer_from_config(cfg: dict, device=None):
         env.reuse_jobs_generator = True
         return env
 
-    # fork env workers before any HIP context exists in this process
-    venv = SubprocVectorEnv(env_fn, num_envs=n_envs, num_workers=n_workers,
-                            base_seed=seed)
+    # env backend: "engine" = vectorised engine (GPU kernel on cuda,
+    # CpuEngine mirror on cpu), "subproc" = fork-based env workers,
+    # "auto" (default) = engine on cuda, subprocess on cpu (historical
+    # default; the engine supports 1-channel RAMP + the non-throughput
+    # rewards — unsupported configs raise with a clear message)
+    backend = loop_cfg.get("env_backend", "auto")
+    use_cuda = (device is not None and str(device) != "cpu"
+                and torch.cuda.is_available())
+    use_engine = (backend == "engine"
+                  or (backend == "auto" and use_cuda))
+    if use_engine:
+        from ..rl.engine_env import EngineVectorEnv
+        try:
+            venv = EngineVectorEnv(env_fn, num_envs=n_envs,
+                                   device=device if use_cuda
+                                   else torch.device("cpu"),
+                                   base_seed=seed)
+        except ValueError:
+            # engine-unsupported config (throughput reward, multi-channel,
+            # non-RAMP): explicit request surfaces the error, auto falls
+            # back to the subprocess path
+            if backend == "engine":
+                raise
+            use_engine = False
+    if not use_engine:
+        # fork env workers before any HIP context exists in this process
+        venv = SubprocVectorEnv(env_fn, num_envs=n_envs,
+                                num_workers=n_workers, base_seed=seed)
     num_actions = cfg.get("env_config", {}).get("max_partitions_per_op", 16) + 1
-    if loop_cfg.get("precompute_lookaheads", True):
+    if not use_engine and loop_cfg.get("precompute_lookaheads", True):
         from ..cluster.batched_lookahead import precompute_lookahead_memos
         scratch = build_env_from_config(cfg)
         scratch.reset(seed=seed)
         memo_l, memo_i = precompute_lookahead_memos(
-            scratch, device=device if (device is not None and str(device) != "cpu"
-                                       and torch.cuda.is_available()) else "cpu")
+            scratch, device=device if use_cuda else "cpu")
         del scratch
         venv.preload_memos(memo_l, memo_i)
     torch.manual_seed(seed)

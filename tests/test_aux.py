@@ -504,3 +504,38 @@ def test_fused_scratch_contract():
                       re.S).group(1)
     wnames = re.findall(r"\bW_[A-Z0-9_]+\b", re.sub(r"//[^\n]*", "", wbody))
     assert len(wnames) == len(_FusedCachedEngine._SLOTS) == 36
+
+
+def test_config_env_backend_engine(tmp_path):
+    """epoch_loop.env_backend: engine builds the trainer on the vectorised
+    engine (CpuEngine on CPU) instead of subprocess env workers."""
+    import os
+
+    import numpy as np
+    import torch
+    import yaml
+
+    from ddls_amd.cluster.vec_engine import CpuEngine
+    from ddls_amd.rl.engine_env import EngineVectorEnv
+    from ddls_amd.runtime.config import (build_trainer_from_config,
+                                         load_config)
+    from ddls_amd.workloads import generate_model, write_pipedream_txt
+
+    d = tmp_path / "jobs"
+    d.mkdir()
+    for name, (nn, sk, sc, seed) in {"m_a": (5, 0, 0.5, 21),
+                                     "m_b": (7, 1, 1.0, 22)}.items():
+        nodes, edges = generate_model(name, nn, sk, sc, seed)
+        write_pipedream_txt(str(d / f"{name}.txt"), nodes, edges)
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    cfg = load_config(os.path.join(root, "configs", "train_config.yaml"))
+    cfg["env_config"]["jobs_config"]["path_to_files"] = str(d)
+    cfg["env_config"]["jobs_config"]["replication_factor"] = 2
+    cfg["epoch_loop"]["num_envs"] = 4
+    cfg["epoch_loop"]["env_backend"] = "engine"
+    tr = build_trainer_from_config(cfg, device=torch.device("cpu"))
+    assert isinstance(tr.env, EngineVectorEnv)
+    assert isinstance(tr.env.eng, CpuEngine)
+    st = tr.train(num_steps=4)
+    assert np.isfinite(st["total_loss"])
