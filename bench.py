@@ -109,7 +109,9 @@ def main():
     policy = GNNPolicy(num_actions=17)
 
     env_fn = build_env_fn()
-    use_engine = use_cuda and not args.cpu_envs
+    # engine rollouts everywhere (GPU kernel on cuda, CpuEngine mirror on
+    # cpu); --cpu-envs selects the round-1 subprocess-worker path
+    use_engine = not args.cpu_envs
     jct_fn = None
     if use_engine:
         from ddls_amd.rl.engine_env import EngineVectorEnv
@@ -224,7 +226,8 @@ def main():
                 "parallelism": f"dp{world_size}",
                 "ramp": "4x4x2_32workers",
                 "envs_per_rank": args.envs_per_rank,
-                "env_engine": "gpu_resident" if use_engine else "cpu_subproc",
+                "env_engine": (("gpu_resident" if use_cuda else "cpu_engine")
+                               if use_engine else "cpu_subproc"),
                 "num_sgd_iter": args.num_sgd_iter,
                 "sgd_minibatch_size": args.sgd_minibatch_size,
                 "mean_simulated_jct": mean_jct,
