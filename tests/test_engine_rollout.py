@@ -266,3 +266,35 @@ def test_cpu_engine_episode_stats_roundtrip(multi_model_files):
         if total >= 2:
             break
     assert total >= 1, "at least one episode should complete"
+
+
+def test_bench_dp2_cpu_engine_rehearsal():
+    """World-size-2 bench on CPU (gloo + CpuEngine): the same distributed
+    code path (barriers, fused flat all-reduce, MAX-over-ranks timing,
+    rank-0 single JSON line) the driver runs at 8 GPUs — exercised HERE
+    without any GPU."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env.update({"DDLS_AMD_DIST_BACKEND": "gloo",
+                "MASTER_ADDR": "127.0.0.1"})
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+           "--master-port", "29741", "bench.py", "--gpus", "2",
+           "--steps", "1", "--warmup", "0", "--envs-per-rank", "8",
+           "--rollout-steps-per-env", "2", "--num-sgd-iter", "1",
+           "--sgd-minibatch-size", "16"]
+    out = subprocess.run(cmd, cwd=root, env=env, capture_output=True,
+                         text=True, timeout=600)
+    assert out.returncode == 0, out.stderr[-3000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    rec = json.loads(line)
+    assert rec["n_gpus"] == 2
+    assert rec["config"]["parallelism"] == "dp2"
+    assert rec["config"]["env_engine"] == "cpu_engine"
+    assert rec["value"] > 0
+    assert rec["config"]["mean_simulated_jct"] is not None
