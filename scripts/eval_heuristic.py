@@ -34,7 +34,15 @@ def main():
         for actor_name in cfg.get("actors", list(ACTORS)):
             seed_everything(seed)
             env = build_env_from_config(cfg)
-            actor = ACTORS[actor_name]()
+            if actor_name == "sip_ml":
+                # SiP-ML's proper static cap: at the env's cap (16) every
+                # job expands to unplaceable degrees and the actor
+                # degenerates to max_parallelism;
+                # 8 is argmin lookahead-JCT on both shipped workloads
+                actor = ACTORS[actor_name](
+                    max_partitions_per_op=cfg.get("sip_ml_cap", 8))
+            else:
+                actor = ACTORS[actor_name]()
             loop = EvalLoop(actor, env)
             r = loop.run(seed=seed)
             results[actor_name] = {
