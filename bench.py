@@ -89,7 +89,7 @@ def main():
     args = ap.parse_args()
 
     from ddls_amd.models.gnn import GNNPolicy
-    from ddls_amd.parallel import (get_rank, get_world_size,
+    from ddls_amd.parallel import (get_world_size,
                                    init_distributed_from_env, is_distributed)
     from ddls_amd.rl.ppo import PPOConfig, PPOTrainer
 

@@ -19,7 +19,6 @@ Mapping onto this stack:
 """
 from __future__ import annotations
 
-import time
 from dataclasses import dataclass
 from typing import Dict, List, Optional
 
@@ -29,7 +28,7 @@ import torch.nn.functional as F
 
 from ..parallel import all_reduce_gradients
 from .impala import ImpalaTrainer
-from .rollout import CompactObs, collate
+from .rollout import CompactObs
 
 
 @dataclass

@@ -16,7 +16,7 @@ from __future__ import annotations
 
 import sys
 import time
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 import numpy as np
 import torch

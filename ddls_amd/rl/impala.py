@@ -26,9 +26,8 @@ from typing import Dict, List, Optional
 
 import numpy as np
 import torch
-import torch.nn.functional as F
 
-from ..parallel import all_reduce_gradients, is_distributed
+from ..parallel import all_reduce_gradients
 from .rollout import CompactObs, collate
 
 

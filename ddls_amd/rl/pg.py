@@ -5,7 +5,6 @@ compute_advantages(use_gae=False, use_critic=False), zero bootstrap at
 fragment truncation).  Shares the rollout machinery with PPO/IMPALA."""
 from __future__ import annotations
 
-import time
 from dataclasses import dataclass
 from typing import Dict, Optional
 

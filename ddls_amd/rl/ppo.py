@@ -19,7 +19,7 @@ import torch
 import torch.nn.functional as F
 
 from ..models.gnn import GNNPolicy
-from ..parallel import all_reduce_gradients, get_rank, get_world_size, is_distributed
+from ..parallel import all_reduce_gradients, get_rank
 from .rollout import CompactObs, VectorEnv, collate
 
 

@@ -15,7 +15,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
-from ddls_amd.parallel import get_rank, init_distributed_from_env
+from ddls_amd.parallel import init_distributed_from_env
 from ddls_amd.runtime.config import (build_env_from_config,
                                      build_trainer_from_config, load_config)
 from ddls_amd.runtime.loops import EpochLoop, EvalLoop, Launcher, PolicyActor

@@ -4,7 +4,6 @@ Contract: CPU mirror == RampJobPartitioningEnvironment step-for-step
 (obs / reward / done, f64 exact), so the HIP kernel (tested separately on
 GPU against the mirror) is transitively equivalent to the reference env.
 """
-import math
 
 import numpy as np
 import pytest
