@@ -539,3 +539,40 @@ def test_config_env_backend_engine(tmp_path):
     assert isinstance(tr.env.eng, CpuEngine)
     st = tr.train(num_steps=4)
     assert np.isfinite(st["total_loss"])
+
+
+def test_train_cli_end_to_end(tmp_path):
+    """scripts/train.py CLI (README quick-start): tiny run with dotted
+    overrides — trains, evaluates, writes the RLlib-layout checkpoint."""
+    import os
+    import subprocess
+    import sys
+
+    from ddls_amd.workloads import generate_model, write_pipedream_txt
+
+    d = tmp_path / "jobs"
+    d.mkdir()
+    for name, (nn, sk, sc, seed) in {"m_a": (5, 0, 0.5, 31),
+                                     "m_b": (7, 1, 1.0, 32)}.items():
+        nodes, edges = generate_model(name, nn, sk, sc, seed)
+        write_pipedream_txt(str(d / f"{name}.txt"), nodes, edges)
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    run_dir = tmp_path / "runs"
+    cmd = [sys.executable, "scripts/train.py",
+           f"env_config.jobs_config.path_to_files={d}",
+           "env_config.jobs_config.replication_factor=2",
+           "epoch_loop.num_envs=2", "epoch_loop.num_env_workers=1",
+           "epoch_loop.precompute_lookaheads=false",
+           "algo.train_batch_size=16", "algo.sgd_minibatch_size=16",
+           "algo.num_sgd_iter=1", "algo.rollout_steps=8",
+           "num_epochs=2", "evaluation_interval=2",
+           f"experiment.path_to_save={run_dir}",
+           "experiment.name=cli_smoke",
+           "eval_config.max_steps=20"]
+    out = subprocess.run(cmd, cwd=root, capture_output=True, text=True,
+                         timeout=600)
+    assert out.returncode == 0, (out.stdout[-2000:], out.stderr[-2000:])
+    ckpts = list((run_dir / "cli_smoke" / "checkpoints").glob(
+        "checkpoint_*/checkpoint-*"))
+    assert ckpts, list((run_dir / "cli_smoke").rglob("*"))
