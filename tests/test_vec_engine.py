@@ -435,3 +435,22 @@ def test_mirror_throughput_rewards(multi_model_files, reward):
         multi_model_files, seed=17, steps=60, interarrival=25,
         replication=4, reward=reward)
     assert n > 0
+
+
+def test_cpu_engine_capacity_guard(multi_model_files):
+    """CpuEngine refuses a schedule longer than its job-log capacity with a
+    clear error instead of corrupting state."""
+    from ddls_amd.cluster.vec_engine import CpuEngine
+
+    env_c = make_env(multi_model_files)
+    env_c.reset(seed=1)
+    spec = compile_engine_spec(env_c)
+    gen = env_c.cluster.jobs_generator
+    sched = drain_episode_schedule(gen, spec, seed=2)
+    eng = CpuEngine(spec, B=1, n_jobs_cap=4)
+    if sched.n + 1 > 4:
+        with pytest.raises(ValueError, match="capacity"):
+            eng.reset_env(0, sched)
+    big = CpuEngine(spec, B=1, n_jobs_cap=sched.n + 8)
+    big.reset_env(0, sched)
+    assert int(big.T["obs_model"][0]) >= 0
