@@ -10,6 +10,8 @@ Sweep spec (YAML):
     method: grid | random
     num_trials: 8            # random only
     parallel: 2              # concurrent trials
+    base_overrides:          # applied to every trial
+      - num_epochs=100
     parameters:
       algo.lr: [1e-4, 2.785e-4, 1e-3]
       algo.entropy_coeff: [0.001, 0.003]
@@ -55,12 +57,21 @@ def main():
 
     trials = list(trials_from_spec(spec))
     parallel = spec.get("parallel", 1)
+    # divide CPU threads across concurrent trials: with OMP_NUM_THREADS
+    # unset every torch process spins ncpu OpenMP threads and parallel
+    # trials livelock on spin-waits (measured 2 trials on 8 cores: 244 s
+    # vs 4.4 s with the split)
+    child_env = dict(os.environ)
+    if "OMP_NUM_THREADS" not in child_env:
+        child_env["OMP_NUM_THREADS"] = str(
+            max(1, (os.cpu_count() or 1) // max(parallel, 1)))
     running, idx = [], 0
     while idx < len(trials) or running:
         while idx < len(trials) and len(running) < parallel:
             overrides = trials[idx]
             name = f"trial_{idx:03d}"
-            cli = [f"{k}={v}" for k, v in overrides.items()]
+            cli = [str(o) for o in spec.get("base_overrides", [])]
+            cli += [f"{k}={v}" for k, v in overrides.items()]
             cli += [f"experiment.name={name}",
                     f"experiment.path_to_save={args.out_dir}"]
             log = open(os.path.join(args.out_dir, f"{name}.log"), "w")
@@ -68,7 +79,7 @@ def main():
                 [sys.executable, os.path.join(os.path.dirname(__file__),
                                               "train.py"),
                  "--config", args.config] + cli,
-                stdout=log, stderr=subprocess.STDOUT)
+                stdout=log, stderr=subprocess.STDOUT, env=child_env)
             with open(os.path.join(args.out_dir, f"{name}.json"), "w") as f:
                 json.dump(overrides, f)
             running.append((name, p, log))

@@ -576,3 +576,43 @@ def test_train_cli_end_to_end(tmp_path):
     ckpts = list((run_dir / "cli_smoke" / "checkpoints").glob(
         "checkpoint_*/checkpoint-*"))
     assert ckpts, list((run_dir / "cli_smoke").rglob("*"))
+
+
+def test_run_sweep_cli(tmp_path):
+    """scripts/run_sweep.py (reference run_wandb_sweep.py replacement):
+    two-trial grid sweep over train.py subprocesses, summarised table."""
+    import os
+    import subprocess
+    import sys
+
+    import yaml
+
+    from ddls_amd.workloads import generate_model, write_pipedream_txt
+
+    d = tmp_path / "jobs"
+    d.mkdir()
+    nodes, edges = generate_model("m_a", 5, 0, 0.5, 41)
+    write_pipedream_txt(str(d / "m_a.txt"), nodes, edges)
+
+    spec = {"method": "grid",
+            "parallel": 2,
+            "parameters": {"algo.lr": [1e-4, 1e-3]},
+            "base_overrides": [
+                f"env_config.jobs_config.path_to_files={d}",
+                "env_config.jobs_config.replication_factor=2",
+                "epoch_loop.num_envs=2", "epoch_loop.num_env_workers=1",
+                "epoch_loop.precompute_lookaheads=false",
+                "algo.train_batch_size=16", "algo.sgd_minibatch_size=16",
+                "algo.num_sgd_iter=1", "algo.rollout_steps=8",
+                "num_epochs=1", "evaluation_interval=1",
+                "eval_config.max_steps=10"]}
+    spec_path = tmp_path / "sweep.yaml"
+    spec_path.write_text(yaml.safe_dump(spec))
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "scripts/run_sweep.py", str(spec_path),
+         "--out-dir", str(tmp_path / "sweep_out")],
+        cwd=root, capture_output=True, text=True, timeout=600)
+    assert out.returncode == 0, (out.stdout[-2000:], out.stderr[-2000:])
+    outs = list((tmp_path / "sweep_out").rglob("*"))
+    assert outs, "sweep should write trial outputs"
