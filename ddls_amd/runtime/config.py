@@ -34,9 +34,17 @@ def deep_merge(base: dict, override: dict) -> dict:
 
 def _parse_value(s: str) -> Any:
     try:
-        return yaml.safe_load(s)
+        v = yaml.safe_load(s)
     except yaml.YAMLError:
         return s
+    # YAML 1.1 leaves exponent-only floats ("1e-4") as strings; overrides
+    # like algo.lr=1e-4 must land as numbers
+    if isinstance(v, str):
+        try:
+            return float(v)
+        except ValueError:
+            return v
+    return v
 
 
 def apply_dotted_overrides(cfg: dict, overrides: List[str]) -> dict:
