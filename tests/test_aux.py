@@ -616,3 +616,41 @@ def test_run_sweep_cli(tmp_path):
     assert out.returncode == 0, (out.stdout[-2000:], out.stderr[-2000:])
     outs = list((tmp_path / "sweep_out").rglob("*"))
     assert outs, "sweep should write trial outputs"
+
+
+def test_plotting_renders_pngs(tmp_path):
+    """All five plotting entry points (reference plotting/plotting.py) render
+    to PNG with the installed matplotlib (Agg backend)."""
+    import matplotlib
+    matplotlib.use("Agg")
+
+    from ddls_amd.graphs import load_pipedream_graph
+    from ddls_amd.plotting import (plot_cluster_timeline,
+                                   plot_computation_graph,
+                                   plot_episode_stats_comparison,
+                                   plot_metric_distributions,
+                                   plot_training_curves)
+    from ddls_amd.workloads import generate_model, write_pipedream_txt
+
+    nodes, edges = generate_model("m_a", 6, 1, 0.5, 51)
+    path = tmp_path / "m_a.txt"
+    write_pipedream_txt(str(path), nodes, edges)
+    g = load_pipedream_graph(str(path), processor_type_profiled="A100")
+
+    figs = {
+        "graph": plot_computation_graph(g),
+        "curves": plot_training_curves(
+            {"mean_reward": [-3.0, -2.0, -1.0], "kl": [0.1, 0.05, 0.02]}),
+        "stats": plot_episode_stats_comparison(
+            {"a": {"blocking_rate": 0.1}, "b": {"blocking_rate": 0.4}},
+            metrics=("blocking_rate",)),
+        "dists": plot_metric_distributions(
+            {"a": [1.0, 2.0, 3.0], "b": [2.0, 2.5]}),
+        "timeline": plot_cluster_timeline(
+            {"num_jobs_running": [0, 1, 2, 1]},
+            metrics=("num_jobs_running",)),
+    }
+    for name, fig in figs.items():
+        out = tmp_path / f"{name}.png"
+        fig.savefig(str(out))
+        assert out.stat().st_size > 1000, name
