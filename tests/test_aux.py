@@ -654,3 +654,45 @@ def test_plotting_renders_pngs(tmp_path):
         out = tmp_path / f"{name}.png"
         fig.savefig(str(out))
         assert out.stat().st_size > 1000, name
+
+
+def test_eval_checkpoint_cli():
+    """scripts/eval_checkpoint.py (README quick-start) evaluates the
+    committed round-1 checkpoint deterministically."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    ckpt = os.path.join(root, "artifacts", "round1", "final_train_500",
+                        "checkpoint_000500", "checkpoint-500")
+    out = subprocess.run(
+        [sys.executable, "scripts/eval_checkpoint.py", ckpt,
+         "--episodes", "1"],
+        cwd=root, capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-2000:]
+    rec = json.loads([l for l in out.stdout.splitlines()
+                      if l.startswith("{")][-1])
+    # deterministic greedy eval of a committed artifact: pinned values
+    assert rec["episode_return"] == -48540.11973585138
+    assert rec["blocking_rate"] == 0.06593406593406594
+
+
+def test_run_sim_cli():
+    """scripts/run_sim.py (legacy dynamic cluster, reference run_sim.py):
+    20 jobs complete with the default placer/scheduler."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "scripts/run_sim.py", "--num-jobs", "10"],
+        cwd=root, capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-2000:]
+    rec = json.loads([l for l in out.stdout.splitlines()
+                      if l.startswith("{")][-1])
+    assert rec["num_jobs_arrived"] == 10
+    assert rec["num_jobs_completed"] + rec["num_jobs_blocked"] == 10
