@@ -696,3 +696,38 @@ def test_run_sim_cli():
                       if l.startswith("{")][-1])
     assert rec["num_jobs_arrived"] == 10
     assert rec["num_jobs_completed"] + rec["num_jobs_blocked"] == 10
+
+
+def test_eval_heuristic_cli(tmp_path):
+    """scripts/eval_heuristic.py over all seven baseline actors (tiny job
+    pool); SiP-ML must run at its proper cap (not degenerate to
+    max_parallelism)."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    from ddls_amd.workloads import generate_model, write_pipedream_txt
+
+    d = tmp_path / "jobs"
+    d.mkdir()
+    nodes, edges = generate_model("m_a", 6, 1, 0.8, 61)
+    write_pipedream_txt(str(d / "m_a.txt"), nodes, edges)
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "scripts/eval_heuristic.py",
+         f"env_config.jobs_config.path_to_files={d}",
+         "env_config.jobs_config.replication_factor=8"],
+        cwd=root, capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-2000:]
+    rows = {}
+    for line in out.stdout.splitlines():
+        if ": {" in line:
+            name, _, payload = line.partition(": ")
+            rows[name] = json.loads(payload)
+    for actor in ("random", "no_parallelism", "min_parallelism",
+                  "max_parallelism", "sip_ml", "acceptable_jct"):
+        assert actor in rows, (actor, list(rows))
+    assert rows["sip_ml"]["episode_return"] \
+        != rows["max_parallelism"]["episode_return"], \
+        "sip_ml must not degenerate to max_parallelism"
